@@ -1,0 +1,20 @@
+"""bobrapet_amd — an MI355X-native declarative AI workflow engine.
+
+A from-scratch rebuild of the capability surface of the reference workflow
+operator (see SURVEY.md / ARCHITECTURE.md): Stories (declarative DAG
+workflows), Engrams (in-process GPU workers with hand-written CDNA4 HIP
+kernels), Impulses (triggers), an event-driven run engine, HBM-resident
+payload storage, and RCCL-over-xGMI data movement.
+"""
+
+__version__ = "0.1.0"
+
+from .enums import (  # noqa: F401
+    BackoffStrategy,
+    ExitClass,
+    Phase,
+    StepType,
+    StopMode,
+    StoryPattern,
+    WorkloadMode,
+)

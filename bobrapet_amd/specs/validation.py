@@ -1,0 +1,362 @@
+"""Submit-time validation of spec objects.
+
+Parity with the reference's admission webhooks
+(reference: internal/webhook/v1alpha1/story_webhook.go:290-404 and
+engram_webhook.go / impulse_webhook.go / transport_webhook.go): step shape
+(exactly one of ref/type), name rules, unique names, needs existence +
+acyclicity per phase, requires paths, size caps, batch-only primitives,
+transports declared, template config-schema checks.
+
+Returns a ValidationResult (errors + warnings) instead of raising, so the
+engine can surface ValidationStatus like the reference's Story status does.
+"""
+from __future__ import annotations
+
+import json
+import re
+import typing as _t
+from dataclasses import dataclass, field
+
+from ..enums import BATCH_ONLY_STEP_TYPES, StepType, StoryPattern
+from ..templating import deps as tdeps
+from . import types as T
+
+MAX_STORY_BYTES = 1 << 20  # 1 MiB total Story cap (story_webhook.go:418-428)
+MAX_BLOCK_BYTES = 256 << 10  # per-step with/output cap (story_webhook.go:430-468)
+
+_NAME_RE = re.compile(r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?$")  # DNS-1123 label
+_MAX_NAME_LEN = 63
+
+
+@dataclass
+class ValidationResult:
+    errors: _t.List[str] = field(default_factory=list)
+    warnings: _t.List[str] = field(default_factory=list)
+
+    @property
+    def ok(self) -> bool:
+        return not self.errors
+
+    def error(self, msg: str) -> None:
+        self.errors.append(msg)
+
+    def warn(self, msg: str) -> None:
+        self.warnings.append(msg)
+
+    def raise_if_invalid(self) -> None:
+        if self.errors:
+            raise SpecValidationError(self.errors, self.warnings)
+
+
+class SpecValidationError(ValueError):
+    def __init__(self, errors: _t.List[str], warnings: _t.Optional[_t.List[str]] = None):
+        self.errors = list(errors)
+        self.warnings = list(warnings or [])
+        super().__init__("; ".join(self.errors))
+
+
+def _json_size(value) -> int:
+    try:
+        return len(json.dumps(value, separators=(",", ":"), default=str))
+    except (TypeError, ValueError):
+        return 0
+
+
+def _valid_name(name: str) -> bool:
+    return bool(name) and len(name) <= _MAX_NAME_LEN and bool(_NAME_RE.match(name))
+
+
+# ---------------------------------------------------------------------------
+# Story
+# ---------------------------------------------------------------------------
+
+
+def validate_story(story: T.Story) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(story.name):
+        res.error(f"story name {story.name!r} must be a DNS-1123 label (<=63 chars)")
+
+    if _json_size(T.to_dict(story)) > MAX_STORY_BYTES:
+        res.error("story spec exceeds the 1 MiB size cap")
+
+    if not story.steps:
+        res.error("story must declare at least one step")
+    if len(story.steps) > T.MAX_STEPS:
+        res.error(f"story has {len(story.steps)} steps; max is {T.MAX_STEPS}")
+    if len(story.compensations) > T.MAX_COMPENSATIONS:
+        res.error(f"too many compensations ({len(story.compensations)} > {T.MAX_COMPENSATIONS})")
+    if len(story.finally_) > T.MAX_FINALLY:
+        res.error(f"too many finally steps ({len(story.finally_)} > {T.MAX_FINALLY})")
+
+    main_names = _validate_step_list(res, story.steps, "steps", story)
+    comp_names = _validate_step_list(res, story.compensations, "compensations", story)
+    fin_names = _validate_step_list(res, story.finally_, "finally", story)
+
+    # Names must be unique across all three phases.
+    seen: _t.Dict[str, str] = {}
+    for phase, names in (("steps", main_names), ("compensations", comp_names), ("finally", fin_names)):
+        for n in names:
+            if n in seen:
+                res.error(f"duplicate step name {n!r} ({seen[n]} and {phase})")
+            else:
+                seen[n] = phase
+
+    # needs existence + acyclicity per phase; compensations/finally may also
+    # reference main-phase steps (story_webhook.go:332-360).
+    _check_graph(res, story.steps, set(main_names), "steps")
+    _check_graph(res, story.compensations, set(comp_names) | set(main_names), "compensations")
+    _check_graph(res, story.finally_, set(fin_names) | set(main_names), "finally")
+
+    # requires paths must reference known steps (story_webhook.go:362-365)
+    known = set(seen)
+    aliases = {s.alias: s.name for s in story.all_steps()}
+    for s in story.all_steps():
+        for path in s.requires:
+            root = path.split(".")[0]
+            if root in ("inputs", "run", "story"):
+                continue
+            head = path.split(".")
+            if root == "steps" and len(head) >= 2:
+                target = head[1]
+                if target not in known and target not in aliases:
+                    res.error(f"step {s.name!r}: requires path {path!r} references unknown step")
+            elif root not in known and root not in aliases:
+                res.error(f"step {s.name!r}: requires path {path!r} must start with steps./inputs.")
+
+    # batch-only primitives rejected in streaming stories (story_webhook.go:564-576)
+    if story.pattern == StoryPattern.STREAMING:
+        for s in story.all_steps():
+            if s.type in BATCH_ONLY_STEP_TYPES:
+                res.error(
+                    f"step {s.name!r}: primitive {s.type} is batch-only and not "
+                    f"allowed in a streaming story"
+                )
+
+    # declared transports must exist in the story's transport list
+    declared = {t.name for t in story.transports}
+    for s in story.all_steps():
+        if s.transport and s.transport not in declared:
+            res.error(f"step {s.name!r} references undeclared transport {s.transport!r}")
+
+    # output template refs reachability warning (story_webhook.go:306-312)
+    if story.output is not None:
+        for ref in tdeps.extract_referenced_steps(json.dumps(story.output, default=str)):
+            if ref not in known and ref not in aliases:
+                res.warn(f"story output references unknown step {ref!r}")
+
+    # schemas must be JSON-schema-shaped mappings (story_webhook.go:321-330)
+    for label, schema in (("inputsSchema", story.inputs_schema), ("outputsSchema", story.outputs_schema)):
+        if schema is not None and not isinstance(schema, dict):
+            res.error(f"{label} must be a JSON Schema object")
+
+    return res
+
+
+def _validate_step_list(
+    res: ValidationResult, steps: _t.List[T.Step], phase: str, story: T.Story
+) -> _t.List[str]:
+    names = []
+    for s in steps:
+        if not _valid_name(s.name):
+            res.error(f"{phase}: step name {s.name!r} must be a DNS-1123 label")
+        names.append(s.name)
+        has_ref = s.ref is not None and bool(s.ref.name)
+        has_type = s.type is not None
+        if has_ref == has_type:
+            res.error(
+                f"step {s.name!r}: exactly one of 'ref' (engram) or 'type' "
+                f"(primitive) must be set"
+            )
+        if s.with_ is not None and _json_size(s.with_) > MAX_BLOCK_BYTES:
+            res.error(f"step {s.name!r}: 'with' block exceeds {MAX_BLOCK_BYTES} bytes")
+        if has_type:
+            _validate_primitive_with(res, s)
+    return names
+
+
+def _validate_primitive_with(res: ValidationResult, step: T.Step) -> None:
+    """Shape checks for primitive `with` schemas (reference: dag.go:1549-1668,
+    step_executor.go:1084-1107)."""
+    w = step.with_ if isinstance(step.with_, dict) else {}
+    st = step.type
+    if st == StepType.SLEEP:
+        if "duration" not in w:
+            res.error(f"step {step.name!r}: sleep requires with.duration")
+    elif st == StepType.WAIT:
+        if "until" not in w:
+            res.error(f"step {step.name!r}: wait requires with.until (template)")
+        if w.get("onTimeout") not in (None, "fail", "skip"):
+            res.error(f"step {step.name!r}: wait onTimeout must be fail|skip")
+    elif st == StepType.GATE:
+        if w.get("onTimeout") not in (None, "fail", "skip"):
+            res.error(f"step {step.name!r}: gate onTimeout must be fail|skip")
+    elif st == StepType.STOP:
+        if w.get("phase") not in (None, "Succeeded", "Failed", "Finished") and w.get(
+            "mode"
+        ) not in (None, "success", "failure", "cancel"):
+            res.error(f"step {step.name!r}: stop phase/mode invalid")
+    elif st == StepType.EXECUTE_STORY:
+        if not (w.get("storyRef") or w.get("story")):
+            res.error(f"step {step.name!r}: executeStory requires with.storyRef")
+    elif st == StepType.PARALLEL:
+        branches = w.get("steps")
+        if not isinstance(branches, list) or not branches:
+            res.error(f"step {step.name!r}: parallel requires with.steps (non-empty list)")
+        else:
+            seen = set()
+            for i, b in enumerate(branches):
+                if not isinstance(b, dict) or not b.get("name"):
+                    res.error(f"step {step.name!r}: parallel branch {i} must have a name")
+                    continue
+                if b["name"] in seen:
+                    res.error(f"step {step.name!r}: duplicate parallel branch {b['name']!r}")
+                seen.add(b["name"])
+    elif st == StepType.CONDITION:
+        if not (w.get("expression") or w.get("if") or step.if_):
+            res.error(f"step {step.name!r}: condition requires with.expression")
+
+
+def _check_graph(
+    res: ValidationResult, steps: _t.List[T.Step], known: _t.Set[str], phase: str
+) -> None:
+    """needs existence + cycle detection (story_webhook.go:332-360).
+
+    Only explicit `needs` edges participate in the webhook-level cycle check;
+    template-implied deps are validated at run time like the reference does
+    (dag.go:3076)."""
+    adj: _t.Dict[str, _t.List[str]] = {}
+    for s in steps:
+        for dep in s.needs:
+            if dep not in known:
+                res.error(f"{phase}: step {s.name!r} needs unknown step {dep!r}")
+            if dep == s.name:
+                res.error(f"{phase}: step {s.name!r} needs itself")
+            adj.setdefault(s.name, []).append(dep)
+
+    # Kahn's algorithm over this phase's steps only.
+    local = {s.name for s in steps}
+    indeg = {n: 0 for n in local}
+    for name, ds in adj.items():
+        indeg[name] = sum(1 for d in ds if d in local)
+    queue = sorted(n for n, d in indeg.items() if d == 0)
+    dependents: _t.Dict[str, _t.List[str]] = {}
+    for name, ds in adj.items():
+        for d in ds:
+            if d in local:
+                dependents.setdefault(d, []).append(name)
+    visited = 0
+    while queue:
+        cur = queue.pop()
+        visited += 1
+        for nxt in dependents.get(cur, []):
+            indeg[nxt] -= 1
+            if indeg[nxt] == 0:
+                queue.append(nxt)
+    if visited < len(local):
+        cyc = sorted(n for n, d in indeg.items() if d > 0)
+        res.error(f"{phase}: dependency cycle among {cyc}")
+
+
+# ---------------------------------------------------------------------------
+# Engram / Impulse / Transport / templates
+# ---------------------------------------------------------------------------
+
+
+def validate_engram(
+    engram: T.Engram, template: _t.Optional[T.EngramTemplate] = None
+) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(engram.name):
+        res.error(f"engram name {engram.name!r} must be a DNS-1123 label")
+    if engram.template_ref is None or not engram.template_ref.name:
+        res.error("engram.templateRef is required")
+    if template is not None:
+        if engram.mode and template.supported_modes and engram.mode not in template.supported_modes:
+            res.error(
+                f"engram mode {engram.mode} not in template supportedModes "
+                f"{[str(m) for m in template.supported_modes]}"
+            )
+        if template.config_schema is not None:
+            from ..utils.jsonschema import validate_instance
+
+            for err in validate_instance(engram.with_ or {}, template.config_schema):
+                res.error(f"engram.with: {err}")
+    return res
+
+
+def validate_impulse(
+    impulse: T.Impulse, template: _t.Optional[T.ImpulseTemplate] = None
+) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(impulse.name):
+        res.error(f"impulse name {impulse.name!r} must be a DNS-1123 label")
+    if impulse.template_ref is None or not impulse.template_ref.name:
+        res.error("impulse.templateRef is required")
+    if impulse.story_ref is None or not impulse.story_ref.name:
+        res.error("impulse.storyRef is required")
+    if template is not None and template.config_schema is not None:
+        from ..utils.jsonschema import validate_instance
+
+        for err in validate_instance(impulse.with_ or {}, template.config_schema):
+            res.error(f"impulse.with: {err}")
+    return res
+
+
+def validate_transport(transport: T.Transport) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(transport.name):
+        res.error(f"transport name {transport.name!r} must be a DNS-1123 label")
+    if transport.driver not in T.KNOWN_TRANSPORT_DRIVERS:
+        res.error(
+            f"unknown transport driver {transport.driver!r} "
+            f"(known: {list(T.KNOWN_TRANSPORT_DRIVERS)})"
+        )
+    s = transport.streaming
+    if s is not None:
+        lane_names = [l.name for l in s.lanes]
+        if len(lane_names) != len(set(lane_names)):
+            res.error("transport streaming lanes must have unique names")
+        fc = s.flow_control
+        if fc is not None and fc.mode not in (None, "credit", "none"):
+            res.error(f"flowControl.mode must be credit|none, got {fc.mode!r}")
+        if fc is not None and fc.mode == "credit":
+            if (fc.initial_credits or 0) <= 0:
+                res.error("flowControl.initialCredits must be > 0 in credit mode")
+            if fc.max_credits is not None and fc.initial_credits is not None:
+                if fc.max_credits < fc.initial_credits:
+                    res.error("flowControl.maxCredits must be >= initialCredits")
+        d = s.delivery
+        if d is not None and d.semantics not in (None, "atMostOnce", "atLeastOnce", "exactlyOnce"):
+            res.error(f"delivery.semantics invalid: {d.semantics!r}")
+        if d is not None and d.ordering not in (None, "none", "perLane", "total"):
+            res.error(f"delivery.ordering invalid: {d.ordering!r}")
+        bp = s.backpressure
+        if bp is not None and bp.policy not in (None, "block", "dropOldest", "dropNewest"):
+            res.error(f"backpressure.policy invalid: {bp.policy!r}")
+    return res
+
+
+def validate_engram_template(tpl: T.EngramTemplate) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(tpl.name):
+        res.error(f"engram template name {tpl.name!r} must be a DNS-1123 label")
+    if not tpl.implementation:
+        res.error("engram template must name an implementation (builtin or image)")
+    for label, schema in (
+        ("configSchema", tpl.config_schema),
+        ("inputSchema", tpl.input_schema),
+        ("outputSchema", tpl.output_schema),
+        ("secretSchema", tpl.secret_schema),
+    ):
+        if schema is not None and not isinstance(schema, dict):
+            res.error(f"{label} must be a JSON Schema object")
+    return res
+
+
+def validate_impulse_template(tpl: T.ImpulseTemplate) -> ValidationResult:
+    res = ValidationResult()
+    if not _valid_name(tpl.name):
+        res.error(f"impulse template name {tpl.name!r} must be a DNS-1123 label")
+    if not tpl.implementation:
+        res.error("impulse template must name an implementation (builtin or image)")
+    return res

@@ -1,0 +1,93 @@
+"""Utility layer tests: durations + minimal JSON Schema."""
+import pytest
+
+from bobrapet_amd.utils.durations import DurationError, format_duration, parse_duration
+from bobrapet_amd.utils.jsonschema import apply_defaults, validate_instance
+
+
+class TestDurations:
+    def test_basic(self):
+        assert parse_duration("1s") == 1.0
+        assert parse_duration("300ms") == 0.3
+        assert parse_duration("2m30s") == 150.0
+        assert parse_duration("1.5h") == 5400.0
+        assert parse_duration("1d") == 86400.0
+        assert parse_duration("0") == 0.0
+
+    def test_numbers_pass_through(self):
+        assert parse_duration(5) == 5.0
+        assert parse_duration(0.25) == 0.25
+        assert parse_duration("42") == 42.0
+
+    def test_none(self):
+        assert parse_duration(None) is None
+
+    def test_negative(self):
+        assert parse_duration("-5s") == -5.0
+
+    def test_malformed(self):
+        with pytest.raises(DurationError):
+            parse_duration("5x")
+        with pytest.raises(DurationError):
+            parse_duration("s5")
+
+    def test_format(self):
+        assert format_duration(0) == "0s"
+        assert format_duration(150) == "2m30s"
+        assert format_duration(5400) == "1h30m"
+        assert parse_duration(format_duration(3723.5)) == pytest.approx(3723.5)
+
+
+class TestJsonSchema:
+    SCHEMA = {
+        "type": "object",
+        "required": ["name"],
+        "properties": {
+            "name": {"type": "string", "minLength": 1},
+            "count": {"type": "integer", "minimum": 0, "default": 1},
+            "tags": {"type": "array", "items": {"type": "string"}, "maxItems": 3},
+            "mode": {"enum": ["a", "b"]},
+        },
+    }
+
+    def test_valid(self):
+        assert validate_instance({"name": "x", "count": 2, "tags": ["t"]}, self.SCHEMA) == []
+
+    def test_missing_required(self):
+        errs = validate_instance({}, self.SCHEMA)
+        assert any("missing required" in e for e in errs)
+
+    def test_wrong_type(self):
+        errs = validate_instance({"name": 5}, self.SCHEMA)
+        assert any("expected type" in e for e in errs)
+
+    def test_enum(self):
+        errs = validate_instance({"name": "x", "mode": "c"}, self.SCHEMA)
+        assert any("enum" in e for e in errs)
+
+    def test_bounds(self):
+        errs = validate_instance({"name": "x", "count": -1}, self.SCHEMA)
+        assert any("minimum" in e for e in errs)
+        errs = validate_instance({"name": "x", "tags": ["a", "b", "c", "d"]}, self.SCHEMA)
+        assert any("more than" in e for e in errs)
+
+    def test_bool_is_not_integer(self):
+        errs = validate_instance({"name": "x", "count": True}, self.SCHEMA)
+        assert any("expected type" in e for e in errs)
+
+    def test_defaults(self):
+        assert apply_defaults({"name": "x"}, self.SCHEMA) == {"name": "x", "count": 1}
+        # existing values win
+        assert apply_defaults({"name": "x", "count": 7}, self.SCHEMA)["count"] == 7
+
+    def test_nested_defaults(self):
+        schema = {
+            "type": "object",
+            "properties": {
+                "outer": {
+                    "type": "object",
+                    "properties": {"inner": {"default": "d"}},
+                }
+            },
+        }
+        assert apply_defaults({"outer": {}}, schema) == {"outer": {"inner": "d"}}

@@ -1,0 +1,894 @@
+"""End-to-end engine tests: the DAG machine, primitives, retry, cancel,
+gates, redrive, sub-stories, triggers, effects, cache.
+
+These mirror the reference's envtest suites — but the
+actors the reference must simulate (SDK patches, kubelet) are real here:
+engrams execute in-process on worker slots.
+"""
+import time
+
+import pytest
+
+from bobrapet_amd.engine import EngineConfig, RunEngine
+from bobrapet_amd.enums import Phase, TriggerDecision
+
+
+BASE_RESOURCES = """
+kind: EngramTemplate
+metadata: {name: echo-tpl}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: echoer}
+spec: {templateRef: {name: echo-tpl}}
+---
+kind: EngramTemplate
+metadata: {name: fail-tpl}
+spec: {builtin: fail}
+---
+kind: Engram
+metadata: {name: failer}
+spec: {templateRef: {name: fail-tpl}}
+---
+kind: EngramTemplate
+metadata: {name: sleepy-tpl}
+spec: {builtin: sleepy}
+---
+kind: Engram
+metadata: {name: sleeper}
+spec: {templateRef: {name: sleepy-tpl}}
+"""
+
+
+@pytest.fixture()
+def eng():
+    engine = RunEngine(EngineConfig(cpu_workers=4, child_ttl_seconds=3600)).start()
+    engine.apply_yaml(BASE_RESOURCES)
+    import bobrapet_amd.engrams.registry as reg
+
+    reg.reset_instances()  # fresh fail-engram counters per test
+    yield engine
+    engine.stop()
+
+
+def _story(eng, yaml_text):
+    return eng.apply_yaml(yaml_text)[0]
+
+
+class TestBasicFlows:
+    def test_two_step_sleep_condition(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: two-step}
+spec:
+  steps:
+    - {name: pause, type: sleep, with: {duration: 5ms}}
+    - name: check
+      type: condition
+      needs: [pause]
+      with: {expression: "{{ steps.pause.phase == 'Succeeded' }}"}
+  output: {ok: "{{ steps.check.output.result }}"}
+""",
+        )
+        run = eng.run_story("default/two-step", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.output == {"ok": True}
+        assert run.step_states["pause"].phase == Phase.SUCCEEDED
+
+    def test_engram_chain_with_templates(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: chain}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: "{{ inputs.x }}"}}
+    - {name: b, ref: {name: echoer}, with: {v: "{{ steps.a.output.v + 1 }}"}}
+  output: {v: "{{ steps.b.output.v }}"}
+""",
+        )
+        run = eng.run_story("default/chain", {"x": 41}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED, (run.error, run.step_states)
+        assert run.output == {"v": 42}
+
+    def test_implicit_dependency_orders_steps(self, eng):
+        # no `needs`: the template reference alone must order b after a
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: implicit}
+spec:
+  steps:
+    - {name: second, ref: {name: echoer}, with: {v: "{{ steps.first.output.v }}"}}
+    - {name: first, ref: {name: echoer}, with: {v: 7}}
+""",
+        )
+        run = eng.run_story("default/implicit", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["second"].output == {"v": 7}
+
+    def test_independent_steps_run_concurrently(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: conc}
+spec:
+  steps:
+    - {name: s1, ref: {name: sleeper}, with: {seconds: 0.15}}
+    - {name: s2, ref: {name: sleeper}, with: {seconds: 0.15}}
+    - {name: s3, ref: {name: sleeper}, with: {seconds: 0.15}}
+""",
+        )
+        t0 = time.monotonic()
+        run = eng.run_story("default/conc", {}, timeout=10)
+        elapsed = time.monotonic() - t0
+        assert run.phase == Phase.SUCCEEDED
+        assert elapsed < 0.40, f"steps did not overlap: {elapsed:.2f}s"
+
+    def test_if_condition_skips(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: iffy}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: 1}}
+    - {name: b, ref: {name: echoer}, if: "{{ steps.a.output.v > 100 }}", with: {v: 2}}
+    - {name: c, ref: {name: echoer}, if: "{{ steps.a.output.v < 100 }}", with: {v: 3}}
+""",
+        )
+        run = eng.run_story("default/iffy", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["b"].phase == Phase.SKIPPED
+        assert run.step_states["c"].phase == Phase.SUCCEEDED
+
+    def test_skip_cascades_to_dependents(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: cascade}
+spec:
+  steps:
+    - {name: a, type: condition, with: {expression: "false"}}
+    - {name: b, ref: {name: echoer}, if: "{{ steps.a.output.result }}", with: {v: 1}}
+    - {name: c, ref: {name: echoer}, needs: [b], with: {v: 2}}
+""",
+        )
+        run = eng.run_story("default/cascade", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["b"].phase == Phase.SKIPPED
+        assert run.step_states["c"].phase == Phase.SKIPPED
+
+    def test_requires_guard(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: reqy}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: 1}}
+    - {name: b, ref: {name: echoer}, requires: ["steps.a.output.missing"], with: {v: 2}}
+    - {name: c, ref: {name: echoer}, requires: ["steps.a.output.v"], with: {v: 3}}
+""",
+        )
+        run = eng.run_story("default/reqy", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["b"].phase == Phase.SKIPPED
+        assert run.step_states["c"].phase == Phase.SUCCEEDED
+
+    def test_inputs_schema_defaults_and_validation(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: schema-story}
+spec:
+  inputsSchema:
+    type: object
+    required: [name]
+    properties:
+      name: {type: string}
+      count: {type: integer, default: 3}
+  steps:
+    - {name: a, ref: {name: echoer}, with: {n: "{{ inputs.count }}"}}
+  output: {n: "{{ steps.a.output.n }}"}
+""",
+        )
+        run = eng.run_story("default/schema-story", {"name": "x"}, timeout=10)
+        assert run.output == {"n": 3}
+        with pytest.raises(ValueError):
+            eng.submit_run("default/schema-story", {"count": 1})
+
+
+class TestFailureMachinery:
+    def test_fail_fast_skips_and_fails(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: failing}
+spec:
+  steps:
+    - {name: boom, ref: {name: failer}, with: {exitCode: 2}}
+    - {name: after, ref: {name: echoer}, needs: [boom], with: {v: 1}}
+""",
+        )
+        run = eng.run_story("default/failing", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["boom"].phase == Phase.FAILED
+        assert run.step_states["after"].phase == Phase.SKIPPED
+        assert run.failure_cause == "boom"
+
+    def test_allow_failure(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: allowed}
+spec:
+  steps:
+    - {name: boom, ref: {name: failer}, allowFailure: true, with: {exitCode: 2}}
+    - {name: after, ref: {name: echoer}, needs: [boom], with: {v: 1}}
+""",
+        )
+        run = eng.run_story("default/allowed", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["after"].phase == Phase.SUCCEEDED
+
+    def test_continue_on_step_failure(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: continue-on-fail}
+spec:
+  policy:
+    retries: {continueOnStepFailure: true}
+  steps:
+    - {name: boom, ref: {name: failer}, with: {exitCode: 2}}
+    - {name: dependent, ref: {name: echoer}, needs: [boom], with: {v: 1}}
+    - {name: independent, ref: {name: echoer}, with: {v: 2}}
+""",
+        )
+        run = eng.run_story("default/continue-on-fail", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["independent"].phase == Phase.SUCCEEDED
+        assert run.step_states["dependent"].phase == Phase.SKIPPED
+
+    def test_retry_until_success(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: retrying}
+spec:
+  steps:
+    - name: flaky
+      ref: {name: failer}
+      with: {succeedAfter: 2, exitCode: 1}
+      execution:
+        retry: {maxRetries: 5, delay: 10ms, jitter: 0, backoff: constant}
+""",
+        )
+        run = eng.run_story("default/retrying", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED, (run.error, run.step_states["flaky"].error)
+        assert run.step_states["flaky"].retries == 2
+        assert run.step_states["flaky"].output == {"attempts": 3}
+
+    def test_terminal_exit_code_does_not_retry(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: no-retry}
+spec:
+  steps:
+    - name: fatal
+      ref: {name: failer}
+      with: {exitCode: 2}
+      execution:
+        retry: {maxRetries: 5, delay: 5ms}
+""",
+        )
+        run = eng.run_story("default/no-retry", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["fatal"].retries == 0
+
+    def test_retry_budget_exhaustion(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: exhausted}
+spec:
+  steps:
+    - name: always-fails
+      ref: {name: failer}
+      with: {exitCode: 1}
+      execution:
+        retry: {maxRetries: 2, delay: 5ms, jitter: 0}
+""",
+        )
+        run = eng.run_story("default/exhausted", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["always-fails"].retries == 2
+
+    def test_compensations_run_on_failure(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: saga}
+spec:
+  steps:
+    - {name: work, ref: {name: failer}, with: {exitCode: 2}}
+  compensations:
+    - {name: undo, ref: {name: echoer}, with: {undone: true}}
+""",
+        )
+        run = eng.run_story("default/saga", {}, timeout=10)
+        assert run.phase == Phase.COMPENSATED
+        assert run.step_states["undo"].phase == Phase.SUCCEEDED
+
+    def test_finally_always_runs(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: fin-ok}
+spec:
+  steps:
+    - {name: work, ref: {name: echoer}, with: {v: 1}}
+  finally:
+    - {name: report, ref: {name: echoer}, with: {done: true}}
+""",
+        )
+        run = eng.run_story("default/fin-ok", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["report"].phase == Phase.SUCCEEDED
+
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: fin-fail}
+spec:
+  steps:
+    - {name: work, ref: {name: failer}, with: {exitCode: 2}}
+  finally:
+    - {name: report, ref: {name: echoer}, with: {done: true}}
+""",
+        )
+        run = eng.run_story("default/fin-fail", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["report"].phase == Phase.SUCCEEDED
+
+    def test_step_timeout(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: slow-step}
+spec:
+  steps:
+    - name: slow
+      ref: {name: sleeper}
+      with: {seconds: 5}
+      execution: {timeout: 100ms}
+""",
+        )
+        t0 = time.monotonic()
+        run = eng.run_story("default/slow-step", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["slow"].phase == Phase.TIMEOUT
+        assert time.monotonic() - t0 < 3.0
+
+    def test_story_timeout(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: slow-story}
+spec:
+  policy:
+    timeouts: {story: 100ms}
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 5}}
+""",
+        )
+        run = eng.run_story("default/slow-story", {}, timeout=10)
+        assert run.phase == Phase.TIMEOUT
+
+
+class TestPrimitives:
+    def test_stop_success(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: stopper}
+spec:
+  steps:
+    - {name: halt, type: stop, with: {mode: success, message: done early}}
+    - {name: never, ref: {name: echoer}, needs: [halt], with: {v: 1}}
+""",
+        )
+        run = eng.run_story("default/stopper", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["never"].phase == Phase.SKIPPED
+
+    def test_stop_failure_mode(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: stop-fail}
+spec:
+  steps:
+    - {name: halt, type: stop, with: {mode: failure}}
+""",
+        )
+        run = eng.run_story("default/stop-fail", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+
+    def test_wait_until_condition(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: waiter}
+spec:
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 0.1}}
+    - name: watch
+      type: wait
+      with:
+        until: "{{ steps.slow.phase == 'Succeeded' }}"
+        pollInterval: 10ms
+        timeout: 5s
+""",
+        )
+        run = eng.run_story("default/waiter", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["watch"].phase == Phase.SUCCEEDED
+
+    def test_wait_timeout_skip(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: wait-skip}
+spec:
+  steps:
+    - name: watch
+      type: wait
+      with: {until: "{{ false }}", pollInterval: 10ms, timeout: 50ms, onTimeout: skip}
+""",
+        )
+        run = eng.run_story("default/wait-skip", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["watch"].phase == Phase.SKIPPED
+
+    def test_wait_timeout_fail(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: wait-fail}
+spec:
+  steps:
+    - name: watch
+      type: wait
+      with: {until: "{{ false }}", pollInterval: 10ms, timeout: 50ms, onTimeout: fail}
+""",
+        )
+        run = eng.run_story("default/wait-fail", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["watch"].phase == Phase.TIMEOUT
+
+    def test_gate_approved(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: gated}
+spec:
+  steps:
+    - {name: approval, type: gate}
+    - {name: after, ref: {name: echoer}, needs: [approval], with: {v: 1}}
+""",
+        )
+        run = eng.submit_run("default/gated", {})
+        time.sleep(0.1)
+        r = eng.store.get_story_run(run.key)
+        assert r.step_states["approval"].phase == Phase.PAUSED
+        eng.approve_gate(run.key, "approval", decided_by="tester")
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["approval"].output["approved"] is True
+
+    def test_gate_rejected(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: gated-no}
+spec:
+  steps:
+    - {name: approval, type: gate}
+""",
+        )
+        run = eng.submit_run("default/gated-no", {})
+        time.sleep(0.05)
+        eng.reject_gate(run.key, "approval")
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.FAILED
+
+    def test_gate_timeout(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: gate-to}
+spec:
+  steps:
+    - {name: approval, type: gate, with: {timeout: 50ms, onTimeout: skip}}
+""",
+        )
+        run = eng.run_story("default/gate-to", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["approval"].phase == Phase.SKIPPED
+
+    def test_parallel_fanout_join(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: fan}
+spec:
+  steps:
+    - name: fanout
+      type: parallel
+      with:
+        steps:
+          - {name: b1, ref: {name: echoer}, with: {v: 1}}
+          - {name: b2, ref: {name: echoer}, with: {v: 2}}
+          - {name: b3, ref: {name: echoer}, with: {v: 3}}
+    - name: join
+      ref: {name: echoer}
+      needs: [fanout]
+      with: {got: "{{ steps.fanout.output.branches.b2.v }}"}
+""",
+        )
+        run = eng.run_story("default/fan", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["join"].output == {"got": 2}
+
+    def test_parallel_branch_failure(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: fan-fail}
+spec:
+  steps:
+    - name: fanout
+      type: parallel
+      with:
+        steps:
+          - {name: ok, ref: {name: echoer}, with: {v: 1}}
+          - {name: bad, ref: {name: failer}, with: {exitCode: 2}}
+""",
+        )
+        run = eng.run_story("default/fan-fail", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        assert run.step_states["fanout"].phase == Phase.FAILED
+
+    def test_parallel_branch_allow_failure(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: fan-allow}
+spec:
+  steps:
+    - name: fanout
+      type: parallel
+      with:
+        steps:
+          - {name: ok, ref: {name: echoer}, with: {v: 1}}
+          - {name: bad, ref: {name: failer}, allowFailure: true, with: {exitCode: 2}}
+""",
+        )
+        run = eng.run_story("default/fan-allow", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+
+    def test_execute_story_nested(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: inner}
+spec:
+  policy:
+    with: {base: 10}
+  steps:
+    - {name: calc, ref: {name: echoer}, with: {v: "{{ inputs.base + inputs.add }}"}}
+  output: {v: "{{ steps.calc.output.v }}"}
+---
+kind: Story
+metadata: {name: outer}
+spec:
+  steps:
+    - name: sub
+      type: executeStory
+      with:
+        storyRef: inner
+        with: {add: 5}
+    - name: use
+      ref: {name: echoer}
+      needs: [sub]
+      with: {got: "{{ steps.sub.output.output.v }}"}
+  output: {v: "{{ steps.use.output.got }}"}
+"""
+        )
+        run = eng.run_story("default/outer", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED, (run.error, {k: (v.phase, v.error) for k, v in run.step_states.items()})
+        assert run.output == {"v": 15}
+
+    def test_execute_story_failure_propagates(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: inner-bad}
+spec:
+  steps:
+    - {name: boom, ref: {name: failer}, with: {exitCode: 2}}
+---
+kind: Story
+metadata: {name: outer-bad}
+spec:
+  steps:
+    - {name: sub, type: executeStory, with: {storyRef: inner-bad}}
+"""
+        )
+        run = eng.run_story("default/outer-bad", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+
+    def test_recursion_depth_cap(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: recur}
+spec:
+  steps:
+    - {name: again, type: executeStory, with: {storyRef: recur}}
+"""
+        )
+        run = eng.run_story("default/recur", {}, timeout=20)
+        assert run.phase == Phase.FAILED
+
+
+class TestControl:
+    def test_graceful_cancel(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: cancelme}
+spec:
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 10}}
+""",
+        )
+        run = eng.submit_run("default/cancelme", {})
+        time.sleep(0.1)
+        eng.cancel(run.key)
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.CANCELED
+
+    def test_redrive(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: redrive-me}
+spec:
+  steps:
+    - name: flaky
+      ref: {name: failer}
+      with: {succeedAfter: 1, exitCode: 1}
+""",
+        )
+        run = eng.run_story("default/redrive-me", {}, timeout=10)
+        assert run.phase == Phase.FAILED  # no retries configured, first call fails
+        eng.redrive(run.key)
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.redrive_count == 1
+
+    def test_redrive_from_step(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: partial-redrive}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: 1}}
+    - {name: b, ref: {name: failer}, needs: [a], with: {succeedAfter: 1, exitCode: 1}}
+    - {name: c, ref: {name: echoer}, needs: [b], with: {v: 3}}
+""",
+        )
+        run = eng.run_story("default/partial-redrive", {}, timeout=10)
+        assert run.phase == Phase.FAILED
+        a_finished = run.step_states["a"].finished_at
+        eng.redrive_from_step(run.key, "b")
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["c"].phase == Phase.SUCCEEDED
+        # step a was NOT re-executed
+        assert run.step_states["a"].finished_at == a_finished
+
+    def test_story_concurrency_limit(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: limited}
+spec:
+  policy: {concurrency: 1}
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 0.2}}
+""",
+        )
+        runs = [eng.submit_run("default/limited", {"i": i}) for i in range(3)]
+        time.sleep(0.1)
+        phases = [eng.store.get_story_run(r.key).phase for r in runs]
+        assert Phase.SCHEDULING in phases  # at least one is queued
+        for r in runs:
+            final = eng.wait(r.key, timeout=15)
+            assert final.phase == Phase.SUCCEEDED
+
+
+class TestTriggers:
+    def test_trigger_created_then_reused(self, eng):
+        from bobrapet_amd.engine.triggers import StoryTrigger
+
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: triggered}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: 1}}
+""",
+        )
+        t1 = eng.triggers.submit(
+            StoryTrigger(submission_id="s1", story_name="triggered", inputs={"x": 1})
+        )
+        assert t1.decision == TriggerDecision.CREATED
+        t2 = eng.triggers.submit(
+            StoryTrigger(submission_id="s1", story_name="triggered", inputs={"x": 1})
+        )
+        assert t2.decision == TriggerDecision.REUSED
+        assert t2.story_run_ref == t1.story_run_ref
+        t3 = eng.triggers.submit(
+            StoryTrigger(submission_id="s2", story_name="triggered", inputs={"x": 2})
+        )
+        assert t3.decision == TriggerDecision.CREATED
+        assert t3.story_run_ref != t1.story_run_ref
+
+    def test_trigger_missing_story_rejected(self, eng):
+        from bobrapet_amd.engine.triggers import StoryTrigger
+
+        t = eng.triggers.submit(StoryTrigger(submission_id="s1", story_name="ghost"))
+        assert t.decision == TriggerDecision.REJECTED
+
+    def test_trigger_identity_required(self, eng):
+        from bobrapet_amd.engine.triggers import StoryTrigger
+
+        t = eng.triggers.submit(StoryTrigger(submission_id="", story_name="x"))
+        assert t.decision == TriggerDecision.REJECTED
+
+
+class TestCacheAndEffects:
+    def test_step_cache_hits(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: cached}
+spec:
+  steps:
+    - name: work
+      ref: {name: failer}
+      with: {succeedAfter: 0}
+      execution:
+        cache: {enabled: true, mode: readWrite}
+""",
+        )
+        r1 = eng.run_story("default/cached", {}, timeout=10)
+        assert r1.phase == Phase.SUCCEEDED
+        # second run with identical inputs: cache hit, fail engram NOT called
+        r2 = eng.submit_run("default/cached", {})
+        r2 = eng.wait(r2.key, timeout=10)
+        assert r2.phase == Phase.SUCCEEDED
+        assert r2.step_states["work"].output == r1.step_states["work"].output
+        assert eng.metrics.counter_value("steprun_cache_lookups_total", result="hit") == 1
+
+    def test_effect_ledger_exactly_once(self, eng):
+        from bobrapet_amd.engine.effects import EffectLedger
+        from bobrapet_amd.enums import EffectClaimPhase
+
+        ledger = EffectLedger()
+        claim, fresh = ledger.acquire("charge-42", holder="attempt-1")
+        assert fresh
+        ledger.complete("charge-42", "attempt-1")
+        claim2, fresh2 = ledger.acquire("charge-42", holder="attempt-2")
+        assert not fresh2
+        assert claim2.phase == EffectClaimPhase.COMPLETED
+
+    def test_effect_stale_takeover(self, eng):
+        from bobrapet_amd.engine.effects import EffectLedger
+
+        ledger = EffectLedger()
+        claim, fresh = ledger.acquire("x", holder="h1", lease_duration=0.01)
+        assert fresh
+        time.sleep(0.05)
+        claim2, fresh2 = ledger.acquire("x", holder="h2", lease_duration=60)
+        assert fresh2
+        assert claim2.holder == "h2"
+        assert claim2.takeovers == 1
+
+    def test_metrics_exported(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: metered}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: 1}}
+""",
+        )
+        eng.run_story("default/metered", {}, timeout=10)
+        text = eng.metrics.export_text()
+        assert "bobrapet_amd_storyruns_total" in text
+        assert "bobrapet_amd_stepruns_total" in text
+        assert "bobrapet_amd_storyrun_duration_seconds" in text
+
+
+class TestBlockedWakeup:
+    def test_run_blocked_until_engram_applied(self, eng):
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: needs-worker}
+spec:
+  steps:
+    - {name: a, ref: {name: late-engram}, with: {v: 1}}
+""",
+        )
+        run = eng.submit_run("default/needs-worker", {})
+        time.sleep(0.1)
+        r = eng.store.get_story_run(run.key)
+        assert r.step_states["a"].phase == Phase.BLOCKED
+        eng.apply_yaml(
+            """
+kind: Engram
+metadata: {name: late-engram}
+spec: {templateRef: {name: echo-tpl}}
+"""
+        )
+        run = eng.wait(run.key, timeout=10)
+        assert run.phase == Phase.SUCCEEDED

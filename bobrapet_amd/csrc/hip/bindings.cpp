@@ -1,0 +1,165 @@
+// Torch bindings for the bobrapet_amd CDNA4 kernel library.
+//
+// This file is the only translation unit that includes torch headers (they
+// dominate compile time); the kernels live in *.hip files exposing
+// extern "C" launchers over raw pointers + hipStream_t.
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_rmsnorm(void*, void*, void*, const void*, int, int, float, bool,
+                    hipStream_t);
+void launch_silu_mul(void*, const void*, const void*, long, hipStream_t);
+void launch_rope(void*, void*, const void*, const void*, int, int, int, int,
+                 hipStream_t);
+void launch_embed_pool(void*, const void*, const void*, int, int, int, int,
+                       hipStream_t);
+void launch_add_bf16(void*, const void*, const void*, long, hipStream_t);
+void launch_attn_prefill(void*, const void*, const void*, const void*, int,
+                         int, int, int, float, int, hipStream_t);
+void launch_attn_decode(void*, const void*, const void*, const void*, int, int,
+                        int, int, int, float, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() { return c10::hip::getCurrentHIPStream(); }
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0 && H <= 8192, "rmsnorm: H must be %8==0 and <=8192");
+  TORCH_CHECK(w.numel() == H, "rmsnorm: weight shape mismatch");
+  auto out = torch::empty_like(x);
+  const long rows = x.numel() / H;
+  launch_rmsnorm(out.data_ptr(), x.data_ptr(), nullptr, w.data_ptr(),
+                 (int)rows, H, (float)eps, false, cur_stream());
+  return out;
+}
+
+torch::Tensor fused_add_rmsnorm(torch::Tensor x, torch::Tensor residual,
+                                torch::Tensor w, double eps) {
+  // residual += x (written back); returns rmsnorm(residual) * w
+  check_bf16(x, "x");
+  check_bf16(residual, "residual");
+  check_bf16(w, "w");
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0 && H <= 8192, "fused_add_rmsnorm: bad H");
+  TORCH_CHECK(residual.sizes() == x.sizes(), "shape mismatch");
+  auto out = torch::empty_like(x);
+  const long rows = x.numel() / H;
+  launch_rmsnorm(out.data_ptr(), x.data_ptr(), residual.data_ptr(),
+                 w.data_ptr(), (int)rows, H, (float)eps, true, cur_stream());
+  return out;
+}
+
+torch::Tensor silu_mul(torch::Tensor gate, torch::Tensor up) {
+  check_bf16(gate, "gate");
+  check_bf16(up, "up");
+  TORCH_CHECK(gate.sizes() == up.sizes(), "silu_mul: shape mismatch");
+  TORCH_CHECK(gate.numel() % 8 == 0, "silu_mul: numel must be %8==0");
+  auto out = torch::empty_like(gate);
+  launch_silu_mul(out.data_ptr(), gate.data_ptr(), up.data_ptr(), gate.numel(),
+                  cur_stream());
+  return out;
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
+                  torch::Tensor sin_t) {
+  // q: [T, Hq, D], k: [T, Hk, D] bf16; cos/sin: [T, D/2] f32
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 && cos_t.is_contiguous(),
+              "cos must be f32 contiguous");
+  TORCH_CHECK(sin_t.scalar_type() == torch::kFloat32 && sin_t.is_contiguous(),
+              "sin must be f32 contiguous");
+  const int T = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hk = k.size(1);
+  TORCH_CHECK(k.size(0) == T && k.size(2) == D, "rope: q/k shape mismatch");
+  TORCH_CHECK(cos_t.size(0) == T && cos_t.size(1) == D / 2, "rope: table shape");
+  launch_rope(q.data_ptr(), k.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
+              T, Hq, Hk, D, cur_stream());
+}
+
+torch::Tensor embed_pool(torch::Tensor table, torch::Tensor ids) {
+  check_bf16(table, "table");
+  TORCH_CHECK(ids.scalar_type() == torch::kInt32 && ids.is_cuda() &&
+                  ids.is_contiguous(),
+              "ids must be int32 on GPU");
+  const int V = table.size(0), H = table.size(1);
+  const int B = ids.size(0), S = ids.size(1);
+  TORCH_CHECK(H % 8 == 0 && H <= 8192, "embed_pool: bad H");
+  auto out = torch::empty({B, H}, table.options());
+  launch_embed_pool(out.data_ptr(), table.data_ptr(), ids.data_ptr(), B, S, H,
+                    V, cur_stream());
+  return out;
+}
+
+torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(a.sizes() == b.sizes() && a.numel() % 8 == 0, "add: bad shapes");
+  auto out = torch::empty_like(a);
+  launch_add_bf16(out.data_ptr(), a.data_ptr(), b.data_ptr(), a.numel(),
+                  cur_stream());
+  return out;
+}
+
+torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                           double scale, bool causal) {
+  // q: [B,Hq,S,D=128], k/v: [B,Hkv,S,D]
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  const int B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
+  const int Hkv = k.size(1);
+  TORCH_CHECK(D == 128, "attn_prefill: D must be 128");
+  TORCH_CHECK(Hq % Hkv == 0, "attn_prefill: Hq must be a multiple of Hkv");
+  TORCH_CHECK(k.size(2) == S && v.size(2) == S, "attn_prefill: S mismatch");
+  auto out = torch::empty_like(q);
+  launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      B, Hq, Hkv, S, (float)scale, causal ? 1 : 0,
+                      cur_stream());
+  return out;
+}
+
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
+                          long L, double scale) {
+  // q: [B,Hq,D], kc/vc: [B,Hkv,Smax,D]
+  check_bf16(q, "q");
+  check_bf16(kc, "kc");
+  check_bf16(vc, "vc");
+  const int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hkv = kc.size(1), Smax = kc.size(2);
+  TORCH_CHECK(D == 128, "attn_decode: D must be 128");
+  TORCH_CHECK(L <= Smax, "attn_decode: L > Smax");
+  auto out = torch::empty_like(q);
+  launch_attn_decode(out.data_ptr(), q.data_ptr(), kc.data_ptr(),
+                     vc.data_ptr(), B, Hq, Hkv, Smax, (int)L, (float)scale,
+                     cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "bobrapet_amd hand-written CDNA4 (gfx950) kernels";
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (bf16, fused scale)");
+  m.def("fused_add_rmsnorm", &fused_add_rmsnorm,
+        "residual += x; rmsnorm(residual)");
+  m.def("silu_mul", &silu_mul, "silu(gate) * up");
+  m.def("rope_inplace", &rope_inplace, "rotary embedding in-place on q,k");
+  m.def("embed_pool", &embed_pool, "gather + mean-pool + l2norm");
+  m.def("add_bf16", &add_bf16, "a + b");
+  m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
+  m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
+}

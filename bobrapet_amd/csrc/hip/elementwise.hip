@@ -1,0 +1,258 @@
+// Fused elementwise / normalization kernels for MI355X (gfx950).
+//
+// All memory-bound: target is the HBM roofline (≈6.3 TB/s), so every
+// kernel loads bf16 as ushort8 (16 B/lane) per guide G13 and fuses the
+// adjacent elementwise work into one pass (residual-add into rmsnorm,
+// silu into the gate*up product, rope pairs in one read).
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// RMSNorm: out[r][c] = x[r][c] * rsqrt(mean(x[r]^2)+eps) * w[c]
+// One 256-thread workgroup per row; row length H is a multiple of 8.
+// fused variant: residual += x first (llama pre-norm block pattern), the
+// updated residual is both written back and normalized.
+// ---------------------------------------------------------------------------
+
+template <bool FUSED_ADD>
+__global__ __launch_bounds__(256) void rmsnorm_kernel(
+    unsigned short* __restrict__ out,         // [R][H] bf16
+    unsigned short* __restrict__ x,           // [R][H] bf16 (input)
+    unsigned short* __restrict__ residual,    // [R][H] bf16 (in/out, FUSED_ADD)
+    const unsigned short* __restrict__ w,     // [H]
+    int H, float eps) {
+  const int row = blockIdx.x;
+  const long base = (long)row * H;
+  const int tid = threadIdx.x;
+  const int nthread = blockDim.x;
+
+  float ssq = 0.f;
+  // pass 1: (optional residual add) + sum of squares; values kept in regs
+  // for H<=8192 with 256 threads → ≤4 ushort8 chunks per thread
+  float vals[4][8];
+  ushort8v raw[4];
+  int nchunk = 0;
+  for (int c = tid * 8; c < H; c += nthread * 8) {
+    ushort8v v = *reinterpret_cast<const ushort8v*>(x + base + c);
+    if (FUSED_ADD) {
+      ushort8v r = *reinterpret_cast<const ushort8v*>(residual + base + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(v[j]) + bf2f(r[j]);
+        vals[nchunk][j] = f;
+        v[j] = f2bf(f);
+      }
+      *reinterpret_cast<ushort8v*>(residual + base + c) = v;  // write back sum
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[nchunk][j] = bf2f(v[j]);
+    }
+    raw[nchunk] = v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ssq += vals[nchunk][j] * vals[nchunk][j];
+    ++nchunk;
+  }
+
+  // wave reduce + cross-wave via LDS
+  __shared__ float warp_ssq[8];
+  ssq = wave_reduce_sum(ssq);
+  const int wid = tid / WAVE;
+  if ((tid & (WAVE - 1)) == 0) warp_ssq[wid] = ssq;
+  __syncthreads();
+  float total = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    if (i < (nthread + WAVE - 1) / WAVE) total += warp_ssq[i];
+  const float scale = rsqrtf(total / (float)H + eps);
+
+  // pass 2: scale by rsqrt * w (x values already in registers)
+  int k = 0;
+  for (int c = tid * 8; c < H; c += nthread * 8, ++k) {
+    ushort8v wv = *reinterpret_cast<const ushort8v*>(w + c);
+    ushort8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(vals[k][j] * scale * bf2f(wv[j]));
+    *reinterpret_cast<ushort8v*>(out + base + c) = o;
+  }
+  (void)raw;
+}
+
+extern "C" void launch_rmsnorm(void* out, void* x, void* residual, const void* w,
+                               int rows, int H, float eps, bool fused_add,
+                               hipStream_t stream) {
+  dim3 grid(rows), block(256);
+  if (fused_add)
+    hipLaunchKernelGGL((rmsnorm_kernel<true>), grid, block, 0, stream,
+                       (unsigned short*)out, (unsigned short*)x,
+                       (unsigned short*)residual, (const unsigned short*)w, H, eps);
+  else
+    hipLaunchKernelGGL((rmsnorm_kernel<false>), grid, block, 0, stream,
+                       (unsigned short*)out, (unsigned short*)x, nullptr,
+                       (const unsigned short*)w, H, eps);
+}
+
+// ---------------------------------------------------------------------------
+// SiLU-mul (SwiGLU): out = silu(gate) * up, all [N] bf16, N % 8 == 0.
+// Grid-stride, capped grid (guide G11).
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void silu_mul_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ gate,
+    const unsigned short* __restrict__ up, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8v g = reinterpret_cast<const ushort8v*>(gate)[i];
+    ushort8v u = reinterpret_cast<const ushort8v*>(up)[i];
+    ushort8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float s = gf / (1.f + __expf(-gf));
+      o[j] = f2bf(s * bf2f(u[j]));
+    }
+    reinterpret_cast<ushort8v*>(out)[i] = o;
+  }
+}
+
+extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
+                                long n, hipStream_t stream) {
+  long n8 = n / 8;
+  int blocks = (int)((n8 + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(silu_mul_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (unsigned short*)out, (const unsigned short*)gate,
+                     (const unsigned short*)up, n8);
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (NeoX half-rotation, llama style), in-place on q and k.
+//   q: [T, Hq, D] bf16, k: [T, Hk, D] bf16 (T = flattened batch*seq)
+//   cos/sin: [T, D/2] f32 precomputed on host per position (guide: trig
+//   tables on HOST; on-device sinf/cosf turns memory-bound into VALU-bound)
+// Rotation: for d in [0, D/2): (a, b) = (x[d], x[d+D/2])
+//   x[d] = a*cos - b*sin ; x[d+D/2] = b*cos + a*sin
+// One wave handles one (token, head): D/2=64 pairs → 1 pair/lane.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void rope_kernel(
+    unsigned short* __restrict__ q, unsigned short* __restrict__ k,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t, int T,
+    int Hq, int Hk, int D) {
+  const int half = D / 2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int total = T * (Hq + Hk);
+  const int nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (int idx = wave; idx < total; idx += nwaves) {
+    const int t = idx / (Hq + Hk);
+    const int h = idx % (Hq + Hk);
+    unsigned short* base = (h < Hq) ? q + ((long)t * Hq + h) * D
+                                    : k + ((long)t * Hk + (h - Hq)) * D;
+    const float* crow = cos_t + (long)t * half;
+    const float* srow = sin_t + (long)t * half;
+    // each lane rotates pairs (lane, lane+64, ...) — D=128 → one pair/lane
+    for (int d = lane; d < half; d += WAVE) {
+      float a = bf2f(base[d]);
+      float b = bf2f(base[d + half]);
+      float c = crow[d], s = srow[d];
+      base[d] = f2bf(a * c - b * s);
+      base[d + half] = f2bf(b * c + a * s);
+    }
+  }
+}
+
+extern "C" void launch_rope(void* q, void* k, const void* cos_t,
+                            const void* sin_t, int T, int Hq, int Hk, int D,
+                            hipStream_t stream) {
+  int total_waves = T * (Hq + Hk);
+  int blocks = (total_waves * WAVE + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(rope_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (unsigned short*)q, (unsigned short*)k,
+                     (const float*)cos_t, (const float*)sin_t, T, Hq, Hk, D);
+}
+
+// ---------------------------------------------------------------------------
+// Embed: fused gather + mean-pool + L2-normalize (the `embed` engram's hot
+// op): out[b] = normalize(mean over s of table[ids[b][s]]).
+//   table: [V, H] bf16; ids: [B, S] int32; out: [B, H] bf16
+// One workgroup per batch row; H % 8 == 0, H <= 8192.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void embed_pool_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ table,
+    const int* __restrict__ ids, int S, int H, int V) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  float acc[4][8] = {};
+  const int nchunkmax = (H + 256 * 8 - 1) / (256 * 8);
+  for (int s = 0; s < S; ++s) {
+    int id = ids[(long)b * S + s];
+    if (id < 0 || id >= V) continue;
+    const unsigned short* row = table + (long)id * H;
+    int k = 0;
+    for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
+      ushort8v v = *reinterpret_cast<const ushort8v*>(row + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[k][j] += bf2f(v[j]);
+    }
+  }
+  const float inv_s = 1.f / (float)S;
+  float ssq = 0.f;
+  for (int k = 0; k < nchunkmax; ++k)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[k][j] *= inv_s;
+      ssq += acc[k][j] * acc[k][j];
+    }
+  __shared__ float warp_ssq[8];
+  ssq = wave_reduce_sum(ssq);
+  if ((tid & (WAVE - 1)) == 0) warp_ssq[tid / WAVE] = ssq;
+  __syncthreads();
+  float total = 1e-12f;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) total += warp_ssq[i];
+  const float inv_norm = rsqrtf(total);
+  int k = 0;
+  for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
+    ushort8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[k][j] * inv_norm);
+    *reinterpret_cast<ushort8v*>(out + (long)b * H + c) = o;
+  }
+}
+
+extern "C" void launch_embed_pool(void* out, const void* table, const void* ids,
+                                  int B, int S, int H, int V,
+                                  hipStream_t stream) {
+  hipLaunchKernelGGL(embed_pool_kernel, dim3(B), dim3(256), 0, stream,
+                     (unsigned short*)out, (const unsigned short*)table,
+                     (const int*)ids, S, H, V);
+}
+
+// ---------------------------------------------------------------------------
+// Residual add (plain): out = a + b, bf16, vectorized.
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void add_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ a,
+    const unsigned short* __restrict__ b, long n8) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (long)gridDim.x * blockDim.x) {
+    ushort8v x = reinterpret_cast<const ushort8v*>(a)[i];
+    ushort8v y = reinterpret_cast<const ushort8v*>(b)[i];
+    ushort8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(bf2f(x[j]) + bf2f(y[j]));
+    reinterpret_cast<ushort8v*>(out)[i] = o;
+  }
+}
+
+extern "C" void launch_add_bf16(void* out, const void* a, const void* b, long n,
+                                hipStream_t stream) {
+  long n8 = n / 8;
+  int blocks = (int)((n8 + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(add_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (unsigned short*)out, (const unsigned short*)a,
+                     (const unsigned short*)b, n8);
+}

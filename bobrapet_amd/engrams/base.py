@@ -14,7 +14,9 @@ import typing as _t
 from dataclasses import dataclass, field
 
 from ..enums import ErrorType
-from ..engine.records import EffectRecord, SignalEvent, StructuredError
+
+if _t.TYPE_CHECKING:
+    from ..engine.records import EffectRecord, SignalEvent, StructuredError
 
 
 class EngramFailure(Exception):
@@ -33,7 +35,9 @@ class EngramFailure(Exception):
         self.error_type = error_type
         self.details = details
 
-    def to_structured(self) -> StructuredError:
+    def to_structured(self) -> "StructuredError":
+        from ..engine.records import StructuredError
+
         return StructuredError(
             type=self.error_type,
             message=str(self),
@@ -72,8 +76,8 @@ class EngramContext:
     storage: _t.Any = None  # StorageManager
     trace_id: str = ""
     # collected during execution --------------------------------------------
-    signals: _t.List[SignalEvent] = field(default_factory=list)
-    effects: _t.List[EffectRecord] = field(default_factory=list)
+    signals: _t.List["SignalEvent"] = field(default_factory=list)
+    effects: _t.List["EffectRecord"] = field(default_factory=list)
     logs: _t.List[str] = field(default_factory=list)
     cancel_check: _t.Optional[_t.Callable[[], bool]] = None
     effect_guard: _t.Optional[_t.Callable[[str, str], bool]] = None
@@ -85,6 +89,8 @@ class EngramContext:
     def emit_signal(self, name: str, payload=None) -> None:
         """Ordered signal (reference: steprun_types.go SignalEvent; merged
         into prior outputs seq-ordered, dag.go:2289-2481)."""
+        from ..engine.records import SignalEvent
+
         self._signal_seq += 1
         self.signals.append(SignalEvent(seq=self._signal_seq, name=name, payload=payload))
 
@@ -96,6 +102,8 @@ class EngramContext:
             fresh = self.effect_guard(idempotency_key, description)
             if not fresh:
                 return False
+        from ..engine.records import EffectRecord
+
         self.effects.append(EffectRecord(idempotency_key=idempotency_key, description=description))
         return True
 

@@ -1,0 +1,282 @@
+"""Llama-family transformer for the llm-infer engram (MI355X-native).
+
+Random-init weights (the north star allows synthetic payloads / random
+weights — BASELINE.json); bf16 end-to-end.  Compute mapping:
+  - plain projections → torch.matmul (hipBLASLt on ROCm — library GEMMs)
+  - fused hot ops → hand-written CDNA4 kernels (bobrapet_amd.ops):
+    fused residual-add+RMSNorm, RoPE, flash attention (MFMA), SiLU-mul
+The decode path keeps a [B, Hkv, Smax, D] KV cache per layer resident in
+HBM (288 GB/GPU — no paging needed at these sizes).
+"""
+from __future__ import annotations
+
+import math
+import typing as _t
+from dataclasses import dataclass
+
+import torch
+
+from .. import ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama-3-8b"
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_layers: int = 32
+    num_heads: int = 32
+    num_kv_heads: int = 8
+    head_dim: int = 128
+    rope_theta: float = 500000.0
+    rms_eps: float = 1e-5
+    max_seq_len: int = 8192
+
+    @property
+    def qkv_out(self) -> int:
+        return (self.num_heads + 2 * self.num_kv_heads) * self.head_dim
+
+
+CONFIGS: _t.Dict[str, LlamaConfig] = {
+    "llama-3-8b": LlamaConfig(),
+    # small config for tests / CPU smoke
+    "llama-tiny": LlamaConfig(
+        name="llama-tiny",
+        vocab_size=1024,
+        hidden_size=256,
+        intermediate_size=688,
+        num_layers=2,
+        num_heads=2,
+        num_kv_heads=1,
+        head_dim=128,
+        max_seq_len=512,
+    ),
+    "llama-3-1b": LlamaConfig(
+        name="llama-3-1b",
+        vocab_size=128256,
+        hidden_size=2048,
+        intermediate_size=8192,
+        num_layers=16,
+        num_heads=32,
+        num_kv_heads=8,
+        head_dim=64,
+    ),
+}
+
+
+class LlamaLayerWeights:
+    __slots__ = ("ln_attn", "w_qkv", "w_o", "ln_mlp", "w_gate_up", "w_down")
+
+    def __init__(self, cfg: LlamaConfig, device, dtype, gen):
+        H, I = cfg.hidden_size, cfg.intermediate_size
+        std = 0.02
+
+        def mk(*shape):
+            t = torch.empty(*shape, device=device, dtype=dtype)
+            t.normal_(0.0, std, generator=gen)
+            return t
+
+        self.ln_attn = torch.ones(H, device=device, dtype=dtype)
+        self.w_qkv = mk(H, cfg.qkv_out)
+        self.w_o = mk(cfg.num_heads * cfg.head_dim, H)
+        self.ln_mlp = torch.ones(H, device=device, dtype=dtype)
+        self.w_gate_up = mk(H, 2 * I)
+        self.w_down = mk(I, H)
+
+
+class LlamaModel:
+    """Inference-only Llama with explicit weights (no nn.Module overhead)."""
+
+    def __init__(
+        self,
+        cfg: _t.Union[LlamaConfig, str] = "llama-3-8b",
+        device: _t.Union[str, torch.device, None] = None,
+        dtype: torch.dtype = torch.bfloat16,
+        seed: int = 1234,
+    ):
+        if isinstance(cfg, str):
+            cfg = CONFIGS[cfg]
+        self.cfg = cfg
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        self.dtype = dtype
+        gen = torch.Generator(device=self.device)
+        gen.manual_seed(seed)
+        std = 0.02
+        self.embed = torch.empty(
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=dtype
+        ).normal_(0.0, std, generator=gen)
+        self.layers = [
+            LlamaLayerWeights(cfg, self.device, dtype, gen) for _ in range(cfg.num_layers)
+        ]
+        self.ln_final = torch.ones(cfg.hidden_size, device=self.device, dtype=dtype)
+        self.lm_head = torch.empty(
+            cfg.hidden_size, cfg.vocab_size, device=self.device, dtype=dtype
+        ).normal_(0.0, std, generator=gen)
+        self.scale = 1.0 / math.sqrt(cfg.head_dim)
+        self._kv_cache: _t.Optional[_t.List[_t.Tuple[torch.Tensor, torch.Tensor]]] = None
+        self._cache_len = 0
+
+    # ------------------------------------------------------------------
+
+    def param_count(self) -> int:
+        cfg = self.cfg
+        per_layer = (
+            cfg.hidden_size * cfg.qkv_out
+            + cfg.num_heads * cfg.head_dim * cfg.hidden_size
+            + cfg.hidden_size * 2 * cfg.intermediate_size
+            + cfg.intermediate_size * cfg.hidden_size
+            + 2 * cfg.hidden_size
+        )
+        return (
+            cfg.vocab_size * cfg.hidden_size * 2
+            + cfg.num_layers * per_layer
+            + cfg.hidden_size
+        )
+
+    def _split_qkv(self, qkv: torch.Tensor, B: int, S: int):
+        cfg = self.cfg
+        nq = cfg.num_heads * cfg.head_dim
+        nkv = cfg.num_kv_heads * cfg.head_dim
+        q = qkv[..., :nq].view(B, S, cfg.num_heads, cfg.head_dim)
+        k = qkv[..., nq : nq + nkv].view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = qkv[..., nq + nkv :].view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        return q, k, v
+
+    # ------------------------------------------------------------------
+
+    @torch.no_grad()
+    def prefill(
+        self,
+        ids: torch.Tensor,
+        logits_for_all: bool = False,
+        fill_cache: bool = False,
+    ) -> torch.Tensor:
+        """ids [B, S] int64 → logits [B, V] (last position) or [B, S, V]."""
+        cfg = self.cfg
+        B, S = ids.shape
+        positions = torch.arange(S, device=self.device)
+        cos_t, sin_t = ops.rope_tables(positions, cfg.head_dim, cfg.rope_theta)
+        # tables are per flattened token (T = B*S)
+        cos_f = cos_t.repeat(B, 1)
+        sin_f = sin_t.repeat(B, 1)
+
+        residual = self.embed[ids.to(self.device)]  # [B,S,H]
+        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
+        if fill_cache:
+            self._alloc_cache(B, max(cfg.max_seq_len, S))
+
+        for li, lw in enumerate(self.layers):
+            # attention block
+            qkv = torch.matmul(x, lw.w_qkv)  # [B,S,qkv_out]
+            q, k, v = self._split_qkv(qkv, B, S)
+            qf = q.reshape(B * S, cfg.num_heads, cfg.head_dim).contiguous()
+            kf = k.reshape(B * S, cfg.num_kv_heads, cfg.head_dim).contiguous()
+            qf, kf = ops.rope_inplace(qf, kf, cos_f, sin_f)
+            qh = qf.view(B, S, cfg.num_heads, cfg.head_dim).transpose(1, 2).contiguous()
+            kh = kf.view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2).contiguous()
+            vh = v.transpose(1, 2).contiguous()
+            if fill_cache:
+                kc, vc = self._kv_cache[li]
+                kc[:, :, :S] = kh
+                vc[:, :, :S] = vh
+            attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
+            attn = attn.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
+            attn_out = torch.matmul(attn.contiguous(), lw.w_o)
+            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
+
+            # mlp block
+            gate_up = torch.matmul(x, lw.w_gate_up)
+            gate, up = gate_up.chunk(2, dim=-1)
+            act = ops.silu_mul(gate.contiguous(), up.contiguous())
+            mlp_out = torch.matmul(act, lw.w_down)
+            next_norm = (
+                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
+            )
+            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
+
+        if fill_cache:
+            self._cache_len = S
+        if logits_for_all:
+            return torch.matmul(x, self.lm_head)
+        return torch.matmul(x[:, -1], self.lm_head)
+
+    # ------------------------------------------------------------------
+
+    def _alloc_cache(self, B: int, smax: int) -> None:
+        cfg = self.cfg
+        self._kv_cache = [
+            (
+                torch.zeros(
+                    B, cfg.num_kv_heads, smax, cfg.head_dim, device=self.device, dtype=self.dtype
+                ),
+                torch.zeros(
+                    B, cfg.num_kv_heads, smax, cfg.head_dim, device=self.device, dtype=self.dtype
+                ),
+            )
+            for _ in range(cfg.num_layers)
+        ]
+        self._cache_len = 0
+
+    @torch.no_grad()
+    def decode_step(self, ids: torch.Tensor) -> torch.Tensor:
+        """One token per sequence: ids [B] → logits [B, V]; uses the cache
+        filled by prefill(fill_cache=True)."""
+        cfg = self.cfg
+        assert self._kv_cache is not None, "call prefill(fill_cache=True) first"
+        B = ids.shape[0]
+        pos = self._cache_len
+        positions = torch.full((B,), pos, device=self.device, dtype=torch.long)
+        cos_t, sin_t = ops.rope_tables(positions, cfg.head_dim, cfg.rope_theta)
+
+        residual = self.embed[ids.to(self.device)].view(B, 1, cfg.hidden_size)
+        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
+        for li, lw in enumerate(self.layers):
+            qkv = torch.matmul(x, lw.w_qkv)
+            q, k, v = self._split_qkv(qkv, B, 1)
+            qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
+            kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
+            qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
+            kc, vc = self._kv_cache[li]
+            kc[:, :, pos] = kf.view(B, cfg.num_kv_heads, cfg.head_dim)
+            vc[:, :, pos] = v.view(B, cfg.num_kv_heads, cfg.head_dim)
+            attn = ops.attn_decode(qf, kc, vc, pos + 1, self.scale)  # [B,Hq,D]
+            attn_out = torch.matmul(attn.reshape(B, 1, -1), lw.w_o)
+            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
+            gate_up = torch.matmul(x, lw.w_gate_up)
+            gate, up = gate_up.chunk(2, dim=-1)
+            act = ops.silu_mul(gate.contiguous(), up.contiguous())
+            mlp_out = torch.matmul(act, lw.w_down)
+            next_norm = (
+                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
+            )
+            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
+        self._cache_len = pos + 1
+        return torch.matmul(x[:, -1], self.lm_head)
+
+    @torch.no_grad()
+    def generate(self, ids: torch.Tensor, new_tokens: int, greedy: bool = True):
+        """Prefill + decode loop; returns generated token ids [B, new_tokens]."""
+        logits = self.prefill(ids, fill_cache=True)
+        out = []
+        for _ in range(new_tokens):
+            nxt = logits.argmax(dim=-1) if greedy else torch.multinomial(
+                torch.softmax(logits.float(), dim=-1), 1
+            ).squeeze(-1)
+            out.append(nxt)
+            logits = self.decode_step(nxt)
+        return torch.stack(out, dim=1)
+
+
+_MODEL_CACHE: _t.Dict[_t.Tuple[str, str], LlamaModel] = {}
+
+
+def get_model(name: str = "llama-3-8b", device=None) -> LlamaModel:
+    """Process-wide model cache (weights stay resident in HBM between steps)."""
+    dev = str(device or ("cuda" if torch.cuda.is_available() else "cpu"))
+    key = (name, dev)
+    if key not in _MODEL_CACHE:
+        _MODEL_CACHE[key] = LlamaModel(name, device=dev)
+    return _MODEL_CACHE[key]

@@ -1,1 +1,191 @@
+"""bobrapet_amd.ops — the hand-written CDNA4 kernel library.
 
+Dispatch rules:
+  - GPU tensors → the in-tree `_hipops` extension (hipcc-built, gfx950).
+    If the extension is missing on a GPU machine, ops FAIL LOUDLY — there
+    is no silent eager fallback on the GPU path.
+  - CPU tensors → plain fp32 torch reference implementations (the same
+    code the numerics tests compare the HIP kernels against).
+"""
+from __future__ import annotations
+
+import math
+import os
+import typing as _t
+
+import torch
+
+_hipops = None
+_load_error: _t.Optional[str] = None
+
+
+def _try_load():
+    global _hipops, _load_error
+    if _hipops is not None:
+        return _hipops
+    try:
+        from bobrapet_amd import _hipops as ext  # in-tree .so
+
+        _hipops = ext
+    except ImportError as exc:
+        _load_error = str(exc)
+    return _hipops
+
+
+def hip_available() -> bool:
+    return _try_load() is not None
+
+
+def _require_ext():
+    ext = _try_load()
+    if ext is None:
+        raise RuntimeError(
+            "bobrapet_amd._hipops is not built but a GPU tensor was passed. "
+            "Build it with `python -m bobrapet_amd.csrc.build` "
+            f"(import error: {_load_error})"
+        )
+    return ext
+
+
+# ---------------------------------------------------------------------------
+# reference implementations (fp32 torch, CPU or GPU) — used on CPU and as
+# the comparison baseline in tests/test_ops_gpu.py
+# ---------------------------------------------------------------------------
+
+
+def rmsnorm_ref(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    xf = x.float()
+    ms = xf.pow(2).mean(dim=-1, keepdim=True)
+    return (xf * torch.rsqrt(ms + eps) * w.float()).to(x.dtype)
+
+
+def fused_add_rmsnorm_ref(x, residual, w, eps=1e-5):
+    res = (residual.float() + x.float()).to(x.dtype)
+    return rmsnorm_ref(res, w, eps), res
+
+
+def silu_mul_ref(gate, up):
+    gf = gate.float()
+    return (gf * torch.sigmoid(gf) * up.float()).to(gate.dtype)
+
+
+def rope_tables(positions: torch.Tensor, dim: int, theta: float = 500000.0):
+    """cos/sin tables [T, dim/2] f32 (host-precomputed, guide App. B)."""
+    inv_freq = 1.0 / (
+        theta ** (torch.arange(0, dim, 2, dtype=torch.float32, device=positions.device) / dim)
+    )
+    ang = positions.float()[:, None] * inv_freq[None, :]
+    return torch.cos(ang).contiguous(), torch.sin(ang).contiguous()
+
+
+def rope_ref(x: torch.Tensor, cos_t: torch.Tensor, sin_t: torch.Tensor) -> torch.Tensor:
+    """NeoX half-rotation; x: [T, H, D], tables [T, D/2]."""
+    xf = x.float()
+    half = x.shape[-1] // 2
+    a, b = xf[..., :half], xf[..., half:]
+    c = cos_t[:, None, :]
+    s = sin_t[:, None, :]
+    return torch.cat([a * c - b * s, b * c + a * s], dim=-1).to(x.dtype)
+
+
+def embed_pool_ref(table: torch.Tensor, ids: torch.Tensor) -> torch.Tensor:
+    g = table.float()[ids.long()]  # [B,S,H]
+    pooled = g.mean(dim=1)
+    return torch.nn.functional.normalize(pooled, dim=-1, eps=1e-6).to(table.dtype)
+
+
+def attn_ref(q, k, v, scale: _t.Optional[float] = None, causal: bool = True):
+    """[B,H,S,D] fp32 reference attention with GQA expansion."""
+    B, Hq, S, D = q.shape
+    Hkv = k.shape[1]
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    qf, kf, vf = q.float(), k.float(), v.float()
+    s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if causal:
+        mask = torch.triu(torch.ones(S, S, dtype=torch.bool, device=q.device), diagonal=1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return torch.matmul(p, vf).to(q.dtype)
+
+
+def attn_decode_ref(q, kc, vc, L: int, scale: _t.Optional[float] = None):
+    """q [B,Hq,D]; caches [B,Hkv,Smax,D] → [B,Hq,D]."""
+    B, Hq, D = q.shape
+    Hkv = kc.shape[1]
+    scale = scale if scale is not None else 1.0 / math.sqrt(D)
+    k = kc[:, :, :L].float()
+    v = vc[:, :, :L].float()
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    s = torch.einsum("bhd,bhld->bhl", q.float(), k) * scale
+    p = torch.softmax(s, dim=-1)
+    return torch.einsum("bhl,bhld->bhd", p, v).to(q.dtype)
+
+
+# ---------------------------------------------------------------------------
+# dispatching public ops
+# ---------------------------------------------------------------------------
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        return _require_ext().rmsnorm(x.contiguous(), w.contiguous(), eps)
+    return rmsnorm_ref(x, w, eps)
+
+
+def fused_add_rmsnorm(x, residual, w, eps: float = 1e-5):
+    """residual ← residual + x (in place on GPU); returns (normed, residual)."""
+    if x.is_cuda:
+        out = _require_ext().fused_add_rmsnorm(
+            x.contiguous(), residual, w.contiguous(), eps
+        )
+        return out, residual
+    return fused_add_rmsnorm_ref(x, residual, w, eps)
+
+
+def silu_mul(gate, up):
+    if gate.is_cuda:
+        return _require_ext().silu_mul(gate.contiguous(), up.contiguous())
+    return silu_mul_ref(gate, up)
+
+
+def rope_inplace(q, k, cos_t, sin_t):
+    """q [T,Hq,D], k [T,Hk,D] rotated in place (GPU) / returned (CPU)."""
+    if q.is_cuda:
+        _require_ext().rope_inplace(q, k, cos_t, sin_t)
+        return q, k
+    return rope_ref(q, cos_t, sin_t), rope_ref(k, cos_t, sin_t)
+
+
+def embed_pool(table, ids):
+    if table.is_cuda:
+        return _require_ext().embed_pool(table.contiguous(), ids.to(torch.int32).contiguous())
+    return embed_pool_ref(table, ids)
+
+
+def add_bf16(a, b):
+    if a.is_cuda:
+        return _require_ext().add_bf16(a.contiguous(), b.contiguous())
+    return (a.float() + b.float()).to(a.dtype)
+
+
+def attn_prefill(q, k, v, scale: _t.Optional[float] = None, causal: bool = True):
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _require_ext().attn_prefill(
+            q.contiguous(), k.contiguous(), v.contiguous(), scale, causal
+        )
+    return attn_ref(q, k, v, scale, causal)
+
+
+def attn_decode(q, kc, vc, L: int, scale: _t.Optional[float] = None):
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _require_ext().attn_decode(q.contiguous(), kc, vc, L, scale)
+    return attn_decode_ref(q, kc, vc, L, scale)

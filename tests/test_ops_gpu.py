@@ -1,0 +1,157 @@
+"""GPU numerics tests: every HIP kernel vs the plain-PyTorch fp32 reference
+of the same op (driver contract).  All asymmetric random data — a
+symmetric input would miss operand/output transposes (guide §5.4 rule 16).
+"""
+import math
+
+import pytest
+import torch
+
+from bobrapet_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+
+def _skip_if_no_ext():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    assert ops.hip_available(), "HIP extension must be built on a GPU box (fail loud)"
+
+
+def _mae(a, b):
+    return (a.float() - b.float()).abs().max().item()
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    _skip_if_no_ext()
+    torch.manual_seed(42)
+
+
+class TestRmsNorm:
+    @pytest.mark.parametrize("rows,h", [(8, 4096), (33, 2048), (256, 8192), (4, 128)])
+    def test_vs_ref(self, rows, h):
+        x = torch.randn(rows, h, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(h, dtype=torch.bfloat16, device="cuda")
+        got = ops.rmsnorm(x, w, 1e-5)
+        ref = ops.rmsnorm_ref(x, w, 1e-5)
+        assert _mae(got, ref) < 0.02, _mae(got, ref)
+
+    def test_fused_add(self):
+        x = torch.randn(16, 4096, dtype=torch.bfloat16, device="cuda")
+        res = torch.randn(16, 4096, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(4096, dtype=torch.bfloat16, device="cuda")
+        res_ref = res.clone()
+        got, new_res = ops.fused_add_rmsnorm(x, res, w, 1e-5)
+        ref, ref_res = ops.fused_add_rmsnorm_ref(x, res_ref, w, 1e-5)
+        assert _mae(got, ref) < 0.02
+        assert _mae(new_res, ref_res) < 0.02
+        # residual updated in place on GPU
+        assert new_res.data_ptr() == res.data_ptr()
+
+
+class TestSiluMul:
+    def test_vs_ref(self):
+        g = torch.randn(1000, 1432 * 8, dtype=torch.bfloat16, device="cuda")
+        u = torch.randn_like(g)
+        got = ops.silu_mul(g, u)
+        ref = ops.silu_mul_ref(g, u)
+        assert _mae(got, ref) < 0.05
+
+
+class TestRope:
+    def test_vs_ref(self):
+        T, Hq, Hk, D = 64, 4, 2, 128
+        q = torch.randn(T, Hq, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(T, Hk, D, dtype=torch.bfloat16, device="cuda")
+        pos = torch.arange(T, device="cuda")
+        cos_t, sin_t = ops.rope_tables(pos, D)
+        q_ref = ops.rope_ref(q.clone(), cos_t, sin_t)
+        k_ref = ops.rope_ref(k.clone(), cos_t, sin_t)
+        ops.rope_inplace(q, k, cos_t, sin_t)
+        assert _mae(q, q_ref) < 0.03
+        assert _mae(k, k_ref) < 0.03
+
+
+class TestEmbedPool:
+    def test_vs_ref(self):
+        V, H, B, S = 5000, 4096, 16, 64
+        table = torch.randn(V, H, dtype=torch.bfloat16, device="cuda")
+        ids = torch.randint(0, V, (B, S), device="cuda", dtype=torch.int32)
+        got = ops.embed_pool(table, ids)
+        ref = ops.embed_pool_ref(table, ids)
+        assert _mae(got, ref) < 0.02
+
+
+class TestAttentionPrefill:
+    @pytest.mark.parametrize(
+        "b,hq,hkv,s,causal",
+        [
+            (1, 1, 1, 128, False),
+            (1, 4, 4, 128, True),
+            (2, 8, 2, 256, True),
+            (1, 32, 8, 512, True),
+            (1, 2, 2, 192, True),  # S not a multiple of 128
+        ],
+    )
+    def test_vs_ref(self, b, hq, hkv, s, causal):
+        D = 128
+        q = torch.randn(b, hq, s, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(b, hkv, s, D, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(b, hkv, s, D, dtype=torch.bfloat16, device="cuda")
+        scale = 1.0 / math.sqrt(D)
+        got = ops.attn_prefill(q, k, v, scale, causal)
+        ref = ops.attn_ref(q, k, v, scale, causal)
+        err = _mae(got, ref)
+        assert err < 0.03, f"max err {err}"
+
+    def test_spiked_scores_force_rescale(self):
+        # guide §5.4 rule 26: force the online-softmax rescale branch by
+        # spiking one K row against one Q row at a late tile
+        b, h, s, D = 1, 2, 512, 128
+        q = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
+        k[0, 0, 400] = q[0, 0, 500] * 3.0  # huge score at kv=400 for q=500
+        got = ops.attn_prefill(q, k, v, 1.0 / math.sqrt(D), True)
+        ref = ops.attn_ref(q, k, v, 1.0 / math.sqrt(D), True)
+        assert _mae(got, ref) < 0.03
+
+
+class TestAttentionDecode:
+    @pytest.mark.parametrize("b,hq,hkv,l", [(1, 1, 1, 64), (2, 32, 8, 500), (4, 8, 8, 1024)])
+    def test_vs_ref(self, b, hq, hkv, l):
+        D, smax = 128, 1024
+        q = torch.randn(b, hq, D, dtype=torch.bfloat16, device="cuda")
+        kc = torch.randn(b, hkv, smax, D, dtype=torch.bfloat16, device="cuda")
+        vc = torch.randn(b, hkv, smax, D, dtype=torch.bfloat16, device="cuda")
+        got = ops.attn_decode(q, kc, vc, l)
+        ref = ops.attn_decode_ref(q, kc, vc, l)
+        assert _mae(got, ref) < 0.03
+
+
+class TestModel:
+    def test_tiny_model_prefill_decode(self):
+        from bobrapet_amd.models.llama import LlamaModel
+
+        m = LlamaModel("llama-tiny", device="cuda")
+        ids = torch.randint(0, 1024, (2, 128), device="cuda")
+        logits = m.prefill(ids)
+        assert logits.shape == (2, 1024)
+        assert torch.isfinite(logits.float()).all()
+        toks = m.generate(ids, new_tokens=4)
+        assert toks.shape == (2, 4)
+
+    def test_tiny_model_prefill_matches_decode_path(self):
+        # prefill logits at position S ≈ decode logits after cache fill
+        from bobrapet_amd.models.llama import LlamaModel
+
+        m = LlamaModel("llama-tiny", device="cuda")
+        ids = torch.randint(0, 1024, (1, 129), device="cuda")
+        full = m.prefill(ids)  # logits for last token given 0..128
+        m2 = LlamaModel("llama-tiny", device="cuda")
+        _ = m2.prefill(ids[:, :128], fill_cache=True)
+        dec = m2.decode_step(ids[:, 128])
+        err = (full.float() - dec.float()).abs().max().item()
+        scale = full.float().abs().max().item()
+        assert err / max(scale, 1) < 0.1, (err, scale)

@@ -180,10 +180,18 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
       p_sum += __shfl_xor(p_sum, 32, WAVE);
       l_run = l_run * alpha + p_sum;
       m_run = m_new;
+      // O rows are q-indexed by the C-layout REGISTER pattern, not by the
+      // lane (alpha lives at lane q = l&31): redistribute before rescaling
+      float alpha_row[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        alpha_row[r] = __shfl(alpha, qrow, WAVE);
+      }
 #pragma unroll
       for (int dt = 0; dt < 4; ++dt)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha;
+        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha_row[r];
 
       // ---- P -> bf16 A fragments via permlane32_swap ----
       // per 32-kv tile: 8 packs -> 2 swaps x2 -> A slices (16 kv each)

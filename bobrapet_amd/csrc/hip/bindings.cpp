@@ -21,6 +21,9 @@ void launch_attn_prefill(void*, const void*, const void*, const void*, int,
                          int, int, int, float, int, hipStream_t);
 void launch_attn_decode(void*, const void*, const void*, const void*, int, int,
                         int, int, int, float, hipStream_t);
+void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
+void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
+                          hipStream_t);
 }
 
 namespace {
@@ -149,6 +152,21 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return out;
 }
 
+torch::Tensor dbg_mfma(torch::Tensor A, torch::Tensor B) {
+  auto C = torch::zeros({32, 32}, A.options());
+  launch_dbg_mfma(C.data_ptr(), A.data_ptr(), B.data_ptr(), cur_stream());
+  return C;
+}
+
+std::vector<torch::Tensor> dbg_attn_core(torch::Tensor Q, torch::Tensor K,
+                                         torch::Tensor V) {
+  auto S = torch::zeros({64, 32}, Q.options().dtype(torch::kFloat32));
+  auto O = torch::zeros({32, 128}, Q.options().dtype(torch::kFloat32));
+  launch_dbg_attn_core(S.data_ptr(), O.data_ptr(), Q.data_ptr(), K.data_ptr(),
+                       V.data_ptr(), cur_stream());
+  return {S, O};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -162,4 +180,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_bf16", &add_bf16, "a + b");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
+  m.def("dbg_mfma", &dbg_mfma, "layout probe: C=A@B one mfma");
+  m.def("dbg_attn_core", &dbg_attn_core, "layout probe: QK^T + pack + PV");
 }

@@ -37,9 +37,10 @@ __global__ __launch_bounds__(256) void rmsnorm_kernel(
       ushort8v r = *reinterpret_cast<const ushort8v*>(residual + base + c);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float f = bf2f(v[j]) + bf2f(r[j]);
-        vals[nchunk][j] = f;
-        v[j] = f2bf(f);
+        // round the sum to bf16 FIRST: the normalized value must match the
+        // bf16 residual that is written back (and the torch reference)
+        v[j] = f2bf(bf2f(v[j]) + bf2f(r[j]));
+        vals[nchunk][j] = bf2f(v[j]);
       }
       *reinterpret_cast<ushort8v*>(residual + base + c) = v;  // write back sum
     } else {

@@ -1,0 +1,226 @@
+#!/usr/bin/env python3
+"""bobrapet_amd flagship benchmark (driver contract).
+
+Headline metric (BASELINE.json): StoryRuns/sec on the 8-way `parallel`
+Story — 8 embed-engram branches fanned out by the DAG engine onto the
+local GPU's HIP streams, joined by an RCCL all-gather across ranks
+(config #3).  One bench "step" = one complete StoryRun through the full
+engine per rank (admission → DAG → fan-out → engram kernels → join →
+finalize).  `value` is whole-job StoryRuns/sec over all N ranks;
+p50 step latency is reported in `config`.
+
+Weak scaling: each rank runs its own engine + stories; the join's
+all-gather is the cross-rank collective, so ranks stay in lockstep.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W] [--config NAME]
+The driver launches N>1 via torch.distributed.run (one rank per GPU).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch  # noqa: E402
+
+from bobrapet_amd.engine import EngineConfig, RunEngine  # noqa: E402
+from bobrapet_amd.enums import Phase  # noqa: E402
+from bobrapet_amd.parallel import group  # noqa: E402
+
+BRANCHES = 8
+EMBED_BATCH = 32
+EMBED_SEQ = 128
+EMBED_DIM = 4096
+EMBED_VOCAB = 32000
+
+RESOURCES = f"""
+kind: EngramTemplate
+metadata: {{name: embed}}
+spec: {{builtin: embed}}
+---
+kind: Engram
+metadata: {{name: embedder}}
+spec:
+  templateRef: {{name: embed}}
+  with: {{dim: {EMBED_DIM}, vocab: {EMBED_VOCAB}, batch: {EMBED_BATCH}, seqLen: {EMBED_SEQ}}}
+---
+kind: EngramTemplate
+metadata: {{name: allgather-join}}
+spec: {{builtin: allgather-join}}
+---
+kind: Engram
+metadata: {{name: joiner}}
+spec: {{templateRef: {{name: allgather-join}}}}
+---
+kind: EngramTemplate
+metadata: {{name: llm-infer}}
+spec: {{builtin: llm-infer}}
+---
+kind: Engram
+metadata: {{name: llm}}
+spec:
+  templateRef: {{name: llm-infer}}
+  with: {{model: llama-3-8b}}
+"""
+
+PARALLEL_STORY = f"""
+kind: Story
+metadata: {{name: bench-parallel8}}
+spec:
+  steps:
+    - name: fanout
+      type: parallel
+      with:
+        steps:
+{chr(10).join(f'          - {{name: b{i}, ref: {{name: embedder}}, with: {{seed: {i}}}}}' for i in range(BRANCHES))}
+    - name: join
+      ref: {{name: joiner}}
+      needs: [fanout]
+      with:
+        branches: "{{{{ steps.fanout.output.branches }}}}"
+  output:
+    rows: "{{{{ steps.join.output.worldRows }}}}"
+"""
+
+SLEEP_STORY = """
+kind: Story
+metadata: {name: bench-cpu}
+spec:
+  steps:
+    - {name: pause, type: sleep, with: {duration: 0ms}}
+    - name: check
+      type: condition
+      needs: [pause]
+      with: {expression: "{{ steps.pause.phase == 'Succeeded' }}"}
+  output: {ok: "{{ steps.check.output.result }}"}
+"""
+
+LLM_STORY = """
+kind: Story
+metadata: {name: bench-llm}
+spec:
+  steps:
+    - name: infer
+      ref: {name: llm}
+      with: {batch: 4, seqLen: 2048, newTokens: 0}
+  output: {tokens: "{{ steps.infer.output.tokensProcessed }}"}
+"""
+
+
+def run_one(eng: RunEngine, story_key: str, idx: int, rank: int) -> dict:
+    run = eng.submit_run(story_key, {"i": idx, "rank": rank}, name=f"bench-{rank}-{idx}")
+    run = eng.wait(run, timeout=600)
+    if run.phase != Phase.SUCCEEDED:
+        states = {k: (str(v.phase), v.message, str(v.error)) for k, v in run.step_states.items()}
+        raise RuntimeError(f"bench run failed: {run.phase} {run.error} {states}")
+    lat = {}
+    for name, st in run.step_states.items():
+        if st.started_at and st.finished_at:
+            lat[name] = (st.finished_at - st.started_at) * 1000.0
+    return lat
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--config", default="parallel8", choices=["parallel8", "cpu", "llm"])
+    args = ap.parse_args()
+
+    multi = group.init_distributed()
+    rank = group.rank()
+    world = group.world_size()
+    n_gpus = world if multi else args.gpus
+    has_gpu = torch.cuda.is_available()
+    device = torch.cuda.current_device() if has_gpu else None
+
+    if args.config == "cpu" or not has_gpu:
+        config_name = "cpu"
+        story_key = "default/bench-cpu"
+        model_desc = "2-step batch story (sleep->condition), CPU engine"
+    elif args.config == "llm":
+        config_name = "llm"
+        story_key = "default/bench-llm"
+        model_desc = "llm-infer llama-3-8b bf16 prefill b4 s2048"
+    else:
+        config_name = "parallel8"
+        story_key = "default/bench-parallel8"
+        model_desc = (
+            f"8-branch parallel story, embed engram per branch "
+            f"(b{EMBED_BATCH} s{EMBED_SEQ} d{EMBED_DIM}), all-gather join"
+        )
+
+    eng = RunEngine(
+        EngineConfig(cpu_workers=4, workers_per_device=4, child_ttl_seconds=5.0)
+    ).start()
+    try:
+        eng.apply_yaml(RESOURCES)
+        eng.apply_yaml(PARALLEL_STORY)
+        eng.apply_yaml(SLEEP_STORY)
+        eng.apply_yaml(LLM_STORY)
+
+        # warmup (untimed): fills weight/table caches, compiles nothing
+        for i in range(args.warmup):
+            run_one(eng, story_key, -(i + 1), rank)
+
+        group.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+        t0 = time.monotonic()
+        step_lat: list = []
+        for i in range(args.steps):
+            lat = run_one(eng, story_key, i, rank)
+            step_lat.extend(lat.values())
+        group.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+        elapsed = time.monotonic() - t0
+
+        elapsed_max = group.max_over_ranks(
+            elapsed, device="cpu" if not has_gpu else None
+        )
+        runs_per_sec = (args.steps * world) / elapsed_max
+        ms_per_step = elapsed_max * 1000.0 / args.steps
+        p50 = statistics.median(step_lat) if step_lat else 0.0
+
+        if rank == 0:
+            line = {
+                "metric": "StoryRuns/sec (8-way parallel Story)",
+                "value": round(runs_per_sec, 3),
+                "unit": "runs/s",
+                "n_gpus": n_gpus,
+                "steps": args.steps,
+                "warmup": args.warmup,
+                "ms_per_step": round(ms_per_step, 3),
+                "higher_is_better": True,
+                "scaling": "weak",
+                "vs_baseline": None,
+                "dtype": "bf16",
+                "data": "synthetic",
+                "config": {
+                    "model": model_desc,
+                    "bench_config": config_name,
+                    "global_batch": args.steps * world,
+                    "seq_len": EMBED_SEQ if config_name == "parallel8" else (
+                        2048 if config_name == "llm" else 0
+                    ),
+                    "parallelism": f"dp{world}" if world > 1 else "single",
+                    "branches": BRANCHES if config_name == "parallel8" else None,
+                    "p50_step_latency_ms": round(p50, 3),
+                },
+            }
+            print(json.dumps(line), flush=True)
+    finally:
+        eng.stop()
+        group.teardown()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

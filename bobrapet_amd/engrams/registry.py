@@ -69,3 +69,4 @@ def _load_builtins() -> None:
     from . import filter_transform  # noqa: F401
     from . import llm_infer  # noqa: F401
     from . import embed  # noqa: F401
+    from . import join  # noqa: F401

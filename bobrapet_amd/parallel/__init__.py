@@ -1,1 +1,1 @@
-
+from . import collectives, group  # noqa: F401

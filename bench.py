@@ -117,7 +117,8 @@ def run_one(eng: RunEngine, story_key: str, idx: int, rank: int) -> dict:
     run = eng.wait(run, timeout=600)
     if run.phase != Phase.SUCCEEDED:
         states = {k: (str(v.phase), v.message, str(v.error)) for k, v in run.step_states.items()}
-        raise RuntimeError(f"bench run failed: {run.phase} {run.error} {states}")
+        details = run.error.details if run.error is not None else None
+        raise RuntimeError(f"bench run failed: {run.phase} {run.error} details={details} {states}")
     lat = {}
     for name, st in run.step_states.items():
         if st.started_at and st.finished_at:

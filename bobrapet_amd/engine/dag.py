@@ -265,6 +265,7 @@ class DAGReconciler:
         children = run.primitive_children.get(step.name, [])
         outputs: _t.Dict[str, _t.Any] = {}
         failed: _t.List[str] = []
+        details: _t.Dict[str, str] = {}
         for key in children:
             sr = self.engine.store.try_get_step_run(key)
             if sr is None or not sr.is_terminal:
@@ -273,13 +274,17 @@ class DAGReconciler:
             outputs[branch_name] = sr.status.output
             if sr.status.phase in HARD_FAIL_PHASES and sr.status.message != "allowFailure":
                 failed.append(branch_name)
+                if sr.status.error is not None:
+                    details[branch_name] = sr.status.error.message
         now = monotonic_now()
         state.output = {"branches": outputs}
         state.finished_at = now
         if failed:
             state.phase = Phase.FAILED
             state.error = StructuredError(
-                type=ErrorType.EXECUTION, message=f"parallel branches failed: {failed}"
+                type=ErrorType.EXECUTION,
+                message=f"parallel branches failed: {failed}",
+                details=details or None,
             )
         else:
             state.phase = Phase.SUCCEEDED

@@ -32,7 +32,9 @@ class AllGatherJoinEngram(Engram):
             for name in sorted(inp["branches"]):
                 out = inp["branches"][name]
                 if isinstance(out, dict):
-                    emb = out.get("embeddings") or out.get("logits")
+                    emb = out.get("embeddings")
+                    if emb is None:
+                        emb = out.get("logits")
                     if emb is not None:
                         refs.append(emb)
         elif "refs" in inp:

@@ -59,14 +59,21 @@ def reset_instances() -> None:
 
 
 _loaded = False
+_load_lock = __import__("threading").Lock()
 
 
 def _load_builtins() -> None:
+    """Thread-safe: worker slots resolve engrams concurrently, so the flag
+    must only flip after every builtin module finished importing."""
     global _loaded
     if _loaded:
         return
-    _loaded = True
-    from . import filter_transform  # noqa: F401
-    from . import llm_infer  # noqa: F401
-    from . import embed  # noqa: F401
-    from . import join  # noqa: F401
+    with _load_lock:
+        if _loaded:
+            return
+        from . import filter_transform  # noqa: F401
+        from . import llm_infer  # noqa: F401
+        from . import embed  # noqa: F401
+        from . import join  # noqa: F401
+
+        _loaded = True

@@ -41,11 +41,11 @@ __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
   return d * 128 + (byte_in_row ^ ((d & 7) << 4));
 }
 
-__global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
-    unsigned short* __restrict__ out,      // [B,Hq,S,D]
-    const unsigned short* __restrict__ q,  // [B,Hq,S,D]
-    const unsigned short* __restrict__ k,  // [B,Hkv,S,D]
-    const unsigned short* __restrict__ v,  // [B,Hkv,S,D]
+__global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
+    unsigned short* __restrict__ out,      // [B,S,Hq,D]
+    const unsigned short* __restrict__ q,  // [B,S,Hq,D]
+    const unsigned short* __restrict__ k,  // [B,S,Hkv,D]
+    const unsigned short* __restrict__ v,  // [B,S,Hkv,D]
     int B, int Hq, int Hkv, int S, float scale, int causal) {
   __shared__ __attribute__((aligned(16))) char smem[KVBLK * 256 + D_HEAD * 128];
   char* k_lds = smem;                  // swizzled K tile [64][128] bf16
@@ -54,7 +54,9 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
   const int wg = blockIdx.x;
   const int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
   const int bh = wg / nqblk;
-  const int qblk = wg % nqblk;
+  // deepest q-blocks first: under causal masking the last q-block has the
+  // most KV tiles — schedule it first so the tail is short workgroups
+  const int qblk = nqblk - 1 - (wg % nqblk);
   const int b = bh / Hq;
   const int hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
@@ -65,16 +67,21 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
   const int hi = lane >> 5;    // half-wave
   const int l31 = lane & 31;
 
-  const long q_base = (((long)b * Hq + hq) * S) * D_HEAD;
-  const long kv_base = (((long)b * Hkv + hkv) * S) * D_HEAD;
+  // BSHD strides: consecutive sequence positions are H*D elements apart
+  const long q_sstride = (long)Hq * D_HEAD;
+  const long kv_sstride = (long)Hkv * D_HEAD;
+  const long q_base = (long)b * S * q_sstride + (long)hq * D_HEAD;
+  const long kv_base = (long)b * S * kv_sstride + (long)hkv * D_HEAD;
 
-  const int my_q0 = qblk * WG_QROWS + wid * QBLK;  // this wave's q rows
-  const int my_q = my_q0 + l31;                    // this lane's q row (S^T col)
+  // wave-striped q assignment: wave w owns rows {qbase + 4*i + w}, i=0..31
+  // (causal kv ranges match across waves -> no idle compute waves)
+  const int qbase = qblk * WG_QROWS;
+  const int my_q = qbase + 4 * l31 + wid;          // this lane's q row (S^T col)
 
   // ---- load Q fragments: B-operand layout, 8 slices of d (16 each) ----
   bf16x8 qf[8];
   {
-    const unsigned short* qrow = q + q_base + (long)my_q * D_HEAD;
+    const unsigned short* qrow = q + q_base + (long)my_q * q_sstride;
     bool valid = my_q < S;
 #pragma unroll
     for (int s = 0; s < 8; ++s) {
@@ -92,10 +99,10 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
   float l_run = 0.f;
   f32x16 o_acc[4] = {};
 
-  const int q_hi_wg = qblk * WG_QROWS + WG_QROWS - 1;  // max q row in WG
+  const int q_hi_wg = qbase + WG_QROWS - 1;  // max q row in WG
   int kv_end = S;
   if (causal) kv_end = min(S, q_hi_wg + 1);
-  const int my_q_hi = my_q0 + QBLK - 1;
+  const int my_q_hi = qbase + 4 * 31 + wid;  // this wave's max q row
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
     // ---- cooperative staging ----
@@ -109,7 +116,7 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
         int kvg = kv0 + kv;
         ushort8v val;
         if (kvg < S)
-          val = *reinterpret_cast<const ushort8v*>(k + kv_base + (long)kvg * D_HEAD + byte / 2);
+          val = *reinterpret_cast<const ushort8v*>(k + kv_base + (long)kvg * kv_sstride + byte / 2);
         else
           val = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
         *reinterpret_cast<ushort8v*>(k_lds + k_lds_off(kv, byte)) = val;
@@ -124,7 +131,7 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
         int kvg = kv0 + kv;
         ushort8v val;
         if (kvg < S)
-          val = *reinterpret_cast<const ushort8v*>(v + kv_base + (long)kvg * D_HEAD + d0);
+          val = *reinterpret_cast<const ushort8v*>(v + kv_base + (long)kvg * kv_sstride + d0);
         else
           val = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
 #pragma unroll
@@ -244,9 +251,9 @@ __global__ __launch_bounds__(256, 1) void attn_prefill_kernel(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    int qg = my_q0 + qrow;
+    int qg = qbase + 4 * qrow + wid;
     if (qg >= S) continue;
-    unsigned short* orow = out + q_base + (long)qg * D_HEAD;
+    unsigned short* orow = out + q_base + (long)qg * q_sstride;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
       orow[dt * 32 + l31] = f2bf(o_acc[dt][r] * inv_for_row[r]);

@@ -14,8 +14,8 @@ void launch_rmsnorm(void*, void*, void*, const void*, int, int, float, bool,
 void launch_silu_mul(void*, const void*, const void*, long, hipStream_t);
 void launch_rope(void*, void*, const void*, const void*, int, int, int, int,
                  hipStream_t);
-void launch_embed_pool(void*, const void*, const void*, int, int, int, int,
-                       hipStream_t);
+void launch_embed_pool(void*, void*, const void*, const void*, int, int, int,
+                       int, hipStream_t);
 void launch_add_bf16(void*, const void*, const void*, long, hipStream_t);
 void launch_attn_prefill(void*, const void*, const void*, const void*, int,
                          int, int, int, float, int, hipStream_t);
@@ -102,8 +102,9 @@ torch::Tensor embed_pool(torch::Tensor table, torch::Tensor ids) {
   const int B = ids.size(0), S = ids.size(1);
   TORCH_CHECK(H % 8 == 0 && H <= 8192, "embed_pool: bad H");
   auto out = torch::empty({B, H}, table.options());
-  launch_embed_pool(out.data_ptr(), table.data_ptr(), ids.data_ptr(), B, S, H,
-                    V, cur_stream());
+  auto pooled = torch::zeros({B, H}, table.options().dtype(torch::kFloat32));
+  launch_embed_pool(out.data_ptr(), pooled.data_ptr(), table.data_ptr(),
+                    ids.data_ptr(), B, S, H, V, cur_stream());
   return out;
 }
 
@@ -119,15 +120,16 @@ torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
 
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                            double scale, bool causal) {
-  // q: [B,Hq,S,D=128], k/v: [B,Hkv,S,D]
+  // q: [B,S,Hq,D=128], k/v: [B,S,Hkv,D] (BSHD: the model's native layout,
+  // no transposes on the hot path)
   check_bf16(q, "q");
   check_bf16(k, "k");
   check_bf16(v, "v");
-  const int B = q.size(0), Hq = q.size(1), S = q.size(2), D = q.size(3);
-  const int Hkv = k.size(1);
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
   TORCH_CHECK(D == 128, "attn_prefill: D must be 128");
   TORCH_CHECK(Hq % Hkv == 0, "attn_prefill: Hq must be a multiple of Hkv");
-  TORCH_CHECK(k.size(2) == S && v.size(2) == S, "attn_prefill: S mismatch");
+  TORCH_CHECK(k.size(1) == S && v.size(1) == S, "attn_prefill: S mismatch");
   auto out = torch::empty_like(q);
   launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       B, Hq, Hkv, S, (float)scale, causal ? 1 : 0,

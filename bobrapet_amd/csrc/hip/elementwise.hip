@@ -176,35 +176,63 @@ extern "C" void launch_rope(void* q, void* k, const void* cos_t,
 // Embed: fused gather + mean-pool + L2-normalize (the `embed` engram's hot
 // op): out[b] = normalize(mean over s of table[ids[b][s]]).
 //   table: [V, H] bf16; ids: [B, S] int32; out: [B, H] bf16
-// One workgroup per batch row; H % 8 == 0, H <= 8192.
+// Two phases for grid parallelism (one-WG-per-row was ~300 GB/s):
+//   A: grid (B x S-chunks x H-slices) partial sums -> f32 atomics
+//   B: grid (B) normalize: /S, L2, cast bf16
 // ---------------------------------------------------------------------------
 
-__global__ __launch_bounds__(256) void embed_pool_kernel(
-    unsigned short* __restrict__ out, const unsigned short* __restrict__ table,
-    const int* __restrict__ ids, int S, int H, int V) {
+#define EMB_SCHUNK 8
+#define EMB_HSLICE 2048  // elems per slice (256 thr x 8)
+
+__global__ __launch_bounds__(256) void embed_pool_sum_kernel(
+    float* __restrict__ pooled,               // [B][H] f32, pre-zeroed
+    const unsigned short* __restrict__ table, // [V][H]
+    const int* __restrict__ ids,              // [B][S]
+    int B, int S, int H, int V) {
+  const int nslice = (H + EMB_HSLICE - 1) / EMB_HSLICE;
+  const int nchunk = (S + EMB_SCHUNK - 1) / EMB_SCHUNK;
+  int wg = blockIdx.x;
+  const int b = wg / (nchunk * nslice);
+  const int chunk = (wg / nslice) % nchunk;
+  const int slice = wg % nslice;
+  const int tid = threadIdx.x;
+  const int h0 = slice * EMB_HSLICE + tid * 8;
+  if (h0 + 8 > H) return;
+
+  float acc[8] = {};
+  const int s_end = min((chunk + 1) * EMB_SCHUNK, S);
+  for (int sidx = chunk * EMB_SCHUNK; sidx < s_end; ++sidx) {
+    int id = ids[(long)b * S + sidx];
+    if (id < 0 || id >= V) continue;
+    ushort8v v = *reinterpret_cast<const ushort8v*>(table + (long)id * H + h0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
+  }
+  float* dst = pooled + (long)b * H + h0;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(dst + j, acc[j]);
+}
+
+__global__ __launch_bounds__(256) void embed_pool_norm_kernel(
+    unsigned short* __restrict__ out, const float* __restrict__ pooled, int S,
+    int H) {
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
-  float acc[4][8] = {};
-  const int nchunkmax = (H + 256 * 8 - 1) / (256 * 8);
-  for (int s = 0; s < S; ++s) {
-    int id = ids[(long)b * S + s];
-    if (id < 0 || id >= V) continue;
-    const unsigned short* row = table + (long)id * H;
-    int k = 0;
-    for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
-      ushort8v v = *reinterpret_cast<const ushort8v*>(row + c);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[k][j] += bf2f(v[j]);
-    }
-  }
   const float inv_s = 1.f / (float)S;
+  float vals[4][8];
   float ssq = 0.f;
-  for (int k = 0; k < nchunkmax; ++k)
+  int k = 0;
+  for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
+    const float4v lo = *reinterpret_cast<const float4v*>(pooled + (long)b * H + c);
+    const float4v hic = *reinterpret_cast<const float4v*>(pooled + (long)b * H + c + 4);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      acc[k][j] *= inv_s;
-      ssq += acc[k][j] * acc[k][j];
+    for (int j = 0; j < 4; ++j) {
+      vals[k][j] = lo[j] * inv_s;
+      vals[k][j + 4] = hic[j] * inv_s;
     }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ssq += vals[k][j] * vals[k][j];
+  }
   __shared__ float warp_ssq[8];
   ssq = wave_reduce_sum(ssq);
   if ((tid & (WAVE - 1)) == 0) warp_ssq[tid / WAVE] = ssq;
@@ -213,21 +241,25 @@ __global__ __launch_bounds__(256) void embed_pool_kernel(
 #pragma unroll
   for (int i = 0; i < 4; ++i) total += warp_ssq[i];
   const float inv_norm = rsqrtf(total);
-  int k = 0;
+  k = 0;
   for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
     ushort8v o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) o[j] = f2bf(acc[k][j] * inv_norm);
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(vals[k][j] * inv_norm);
     *reinterpret_cast<ushort8v*>(out + (long)b * H + c) = o;
   }
 }
 
-extern "C" void launch_embed_pool(void* out, const void* table, const void* ids,
-                                  int B, int S, int H, int V,
+extern "C" void launch_embed_pool(void* out, void* pooled_f32, const void* table,
+                                  const void* ids, int B, int S, int H, int V,
                                   hipStream_t stream) {
-  hipLaunchKernelGGL(embed_pool_kernel, dim3(B), dim3(256), 0, stream,
-                     (unsigned short*)out, (const unsigned short*)table,
-                     (const int*)ids, S, H, V);
+  const int nslice = (H + EMB_HSLICE - 1) / EMB_HSLICE;
+  const int nchunk = (S + EMB_SCHUNK - 1) / EMB_SCHUNK;
+  hipLaunchKernelGGL(embed_pool_sum_kernel, dim3(B * nchunk * nslice), dim3(256),
+                     0, stream, (float*)pooled_f32,
+                     (const unsigned short*)table, (const int*)ids, B, S, H, V);
+  hipLaunchKernelGGL(embed_pool_norm_kernel, dim3(B), dim3(256), 0, stream,
+                     (unsigned short*)out, (const float*)pooled_f32, S, H);
 }
 
 // ---------------------------------------------------------------------------

@@ -175,16 +175,17 @@ class LlamaModel:
             qf = q.reshape(B * S, cfg.num_heads, cfg.head_dim).contiguous()
             kf = k.reshape(B * S, cfg.num_kv_heads, cfg.head_dim).contiguous()
             qf, kf = ops.rope_inplace(qf, kf, cos_f, sin_f)
-            qh = qf.view(B, S, cfg.num_heads, cfg.head_dim).transpose(1, 2).contiguous()
-            kh = kf.view(B, S, cfg.num_kv_heads, cfg.head_dim).transpose(1, 2).contiguous()
-            vh = v.transpose(1, 2).contiguous()
+            # BSHD end-to-end: no layout transposes on the hot path
+            qh = qf.view(B, S, cfg.num_heads, cfg.head_dim)
+            kh = kf.view(B, S, cfg.num_kv_heads, cfg.head_dim)
+            vh = v.contiguous()
             if fill_cache:
                 kc, vc = self._kv_cache[li]
-                kc[:, :, :S] = kh
-                vc[:, :, :S] = vh
+                kc[:, :, :S] = kh.transpose(1, 2)
+                vc[:, :, :S] = vh.transpose(1, 2)
             attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
-            attn = attn.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
-            attn_out = torch.matmul(attn.contiguous(), lw.w_o)
+            attn = attn.reshape(B, S, cfg.num_heads * cfg.head_dim)
+            attn_out = torch.matmul(attn, lw.w_o)
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
 
             # mlp block

@@ -176,12 +176,16 @@ def add_bf16(a, b):
 
 
 def attn_prefill(q, k, v, scale: _t.Optional[float] = None, causal: bool = True):
+    """BSHD layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] → out [B,S,Hq,D]."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
         return _require_ext().attn_prefill(
             q.contiguous(), k.contiguous(), v.contiguous(), scale, causal
         )
-    return attn_ref(q, k, v, scale, causal)
+    out = attn_ref(
+        q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2), scale, causal
+    )
+    return out.transpose(1, 2).contiguous()
 
 
 def attn_decode(q, kc, vc, L: int, scale: _t.Optional[float] = None):

@@ -96,12 +96,14 @@ class TestAttentionPrefill:
     )
     def test_vs_ref(self, b, hq, hkv, s, causal):
         D = 128
-        q = torch.randn(b, hq, s, D, dtype=torch.bfloat16, device="cuda")
-        k = torch.randn(b, hkv, s, D, dtype=torch.bfloat16, device="cuda")
-        v = torch.randn(b, hkv, s, D, dtype=torch.bfloat16, device="cuda")
+        q = torch.randn(b, s, hq, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(b, s, hkv, D, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(b, s, hkv, D, dtype=torch.bfloat16, device="cuda")
         scale = 1.0 / math.sqrt(D)
         got = ops.attn_prefill(q, k, v, scale, causal)
-        ref = ops.attn_ref(q, k, v, scale, causal)
+        ref = ops.attn_ref(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2), scale, causal
+        ).transpose(1, 2)
         err = _mae(got, ref)
         assert err < 0.03, f"max err {err}"
 
@@ -109,12 +111,15 @@ class TestAttentionPrefill:
         # guide §5.4 rule 26: force the online-softmax rescale branch by
         # spiking one K row against one Q row at a late tile
         b, h, s, D = 1, 2, 512, 128
-        q = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
-        k = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
-        v = torch.randn(b, h, s, D, dtype=torch.bfloat16, device="cuda")
-        k[0, 0, 400] = q[0, 0, 500] * 3.0  # huge score at kv=400 for q=500
+        q = torch.randn(b, s, h, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(b, s, h, D, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(b, s, h, D, dtype=torch.bfloat16, device="cuda")
+        k[0, 400, 0] = q[0, 500, 0] * 3.0  # huge score at kv=400 for q=500
         got = ops.attn_prefill(q, k, v, 1.0 / math.sqrt(D), True)
-        ref = ops.attn_ref(q, k, v, 1.0 / math.sqrt(D), True)
+        ref = ops.attn_ref(
+            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+            1.0 / math.sqrt(D), True,
+        ).transpose(1, 2)
         assert _mae(got, ref) < 0.03
 
 

@@ -89,6 +89,9 @@ class RunEngine:
         from .triggers import TriggerAdmission
 
         self.triggers = TriggerAdmission(self)
+        from .impulses import ImpulseRuntime
+
+        self.impulses = ImpulseRuntime(self)
 
         self._events: "queue.Queue" = queue.Queue()
         self._timers: _t.List[_t.Tuple[float, int, str, str]] = []
@@ -99,6 +102,7 @@ class RunEngine:
         self._running = False
         self._thread: _t.Optional[threading.Thread] = None
         self._retry_pending: _t.Dict[str, float] = {}  # steprun key → next_retry_at
+        self._streams: _t.Dict[str, object] = {}  # run key → StreamingRun
 
     # ------------------------------------------------------------------
     # lifecycle
@@ -115,6 +119,9 @@ class RunEngine:
     def stop(self) -> None:
         if not self._running:
             return
+        self.impulses.stop_all()
+        for stream in list(self._streams.values()):
+            stream.cancel()
         self._running = False
         self._events.put(("__stop__",))
         if self._thread is not None:
@@ -259,6 +266,11 @@ class RunEngine:
 
     def cancel(self, run: _t.Union[StoryRun, str], graceful: bool = True) -> None:
         key = run.key if isinstance(run, StoryRun) else run
+        stream = self._streams.get(key)
+        if stream is not None:
+            stream.cancel()
+            stream.finish(timeout=5.0)
+            return
         r = self.store.get_story_run(key)
         r.cancel_requested = True
         if not graceful:
@@ -356,6 +368,49 @@ class RunEngine:
         ev = self._run_done.get(r.key)
         if ev is not None:
             ev.clear()
+
+    # ------------------------------------------------------------------
+    # streaming (PerStoryRun pipelines — engine/streaming.py)
+    # ------------------------------------------------------------------
+
+    def submit_stream(
+        self,
+        story,
+        inputs=None,
+        name: _t.Optional[str] = None,
+        namespace: _t.Optional[str] = None,
+    ):
+        """Materialize a streaming Story as a live pipeline; returns the
+        StreamingRun handle (push/finish/cancel)."""
+        from ..enums import StoryPattern
+        from .streaming import StreamingRun
+
+        if isinstance(story, str):
+            ns, _, nm = story.rpartition("/")
+            story = self.registry.story(nm, ns or "default")
+        if story.pattern != StoryPattern.STREAMING:
+            raise ValueError(f"story {story.key} is not a streaming story")
+        namespace = namespace or story.namespace
+        if name is None:
+            name = f"{story.name}-{monotonic_now():.6f}".replace(".", "-")[:63]
+        run = StoryRun(
+            name=name,
+            namespace=namespace,
+            story_name=story.name,
+            story_namespace=story.namespace,
+            inputs=inputs or {},
+            story_generation=story.generation,
+            input_hash=input_hash(inputs or {}),
+        )
+        self.store.create_story_run(run)
+        self._run_done.setdefault(run.key, threading.Event())
+        stream = StreamingRun(self, run, story)
+        self._streams[run.key] = stream
+        self.metrics.inc("storyruns_total", phase="streaming")
+        return stream
+
+    def stream_of(self, run_key: str):
+        return self._streams.get(run_key)
 
     # ------------------------------------------------------------------
     # scope building (reference: getPriorStepOutputs dag.go:2083-2597)

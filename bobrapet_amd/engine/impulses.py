@@ -1,0 +1,243 @@
+"""Impulse runtime: always-on trigger workloads.
+
+Role parity with the reference's Impulse materialization
+(reference: internal/controller/impulse_controller.go:78-2217 — the
+controller deploys trigger containers with the BUBU_* env contract and
+delivery/throttle policy): here impulses are in-process handlers started
+by the ImpulseRuntime; each emitted event goes through StoryTrigger
+admission (dedupe/throttle — engine/triggers.py) and, for streaming
+stories, packets stream into the live pipeline (the gRPC-ingress role of
+BASELINE config #4 is the HTTP ingress in serve_http()).
+"""
+from __future__ import annotations
+
+import itertools
+import threading
+import time
+import typing as _t
+from dataclasses import dataclass, field
+
+from ..engrams.base import ImpulseHandler
+from ..enums import StoryPattern, TriggerDecision
+from ..specs import types as T
+from ..templating import Evaluator
+from .triggers import StoryTrigger
+
+if _t.TYPE_CHECKING:
+    from .engine import RunEngine
+
+
+_IMPULSE_REGISTRY: _t.Dict[str, _t.Callable[[], ImpulseHandler]] = {}
+
+
+def register_impulse(name: str, factory: _t.Callable[[], ImpulseHandler]) -> None:
+    _IMPULSE_REGISTRY[name] = factory
+
+
+class IntervalImpulse(ImpulseHandler):
+    """Emits a payload every `intervalMs` (config), up to `count` times."""
+
+    name = "interval"
+
+    def __init__(self, interval_ms: float = 100.0, count: _t.Optional[int] = None):
+        self.interval_ms = interval_ms
+        self.count = count
+        self._stop = threading.Event()
+        self._thread: _t.Optional[threading.Thread] = None
+
+    def configure(self, config: dict) -> None:
+        self.interval_ms = float(config.get("intervalMs", self.interval_ms))
+        if config.get("count") is not None:
+            self.count = int(config["count"])
+
+    def start(self, emit: _t.Callable[[dict], _t.Any]) -> None:
+        def loop():
+            n = 0
+            while not self._stop.is_set():
+                if self.count is not None and n >= self.count:
+                    return
+                emit({"tick": n, "ts": time.time()})
+                n += 1
+                self._stop.wait(self.interval_ms / 1000.0)
+
+        self._thread = threading.Thread(target=loop, daemon=True, name="impulse-interval")
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=2)
+
+
+class ManualImpulse(ImpulseHandler):
+    """Programmatic ingress: call .emit(payload) from application code (the
+    in-process equivalent of the HTTP/gRPC connector)."""
+
+    name = "manual"
+
+    def __init__(self):
+        self._emit: _t.Optional[_t.Callable] = None
+
+    def configure(self, config: dict) -> None:
+        pass
+
+    def start(self, emit: _t.Callable[[dict], _t.Any]) -> None:
+        self._emit = emit
+
+    def emit(self, payload: dict):
+        if self._emit is None:
+            raise RuntimeError("impulse not started")
+        return self._emit(payload)
+
+
+register_impulse("interval", IntervalImpulse)
+register_impulse("manual", ManualImpulse)
+
+
+@dataclass
+class LiveImpulse:
+    impulse: T.Impulse
+    handler: ImpulseHandler
+    emitted: int = 0
+    decisions: _t.Dict[str, int] = field(default_factory=dict)
+    stream_key: _t.Optional[str] = None
+
+
+class ImpulseRuntime:
+    def __init__(self, engine: "RunEngine"):
+        self.engine = engine
+        self.live: _t.Dict[str, LiveImpulse] = {}
+        self._seq = itertools.count()
+        self._lock = threading.Lock()
+        self._evaluator = Evaluator()
+
+    def start(self, impulse: _t.Union[T.Impulse, str]) -> LiveImpulse:
+        """Materialize one Impulse: resolve its template to a builtin handler
+        and start emitting."""
+        if isinstance(impulse, str):
+            ns, _, nm = impulse.rpartition("/")
+            with self.engine.registry._lock:
+                impulse = self.engine.registry.impulses[f"{ns or 'default'}/{nm}"]
+        tpl = self.engine.registry.impulse_template(impulse.template_ref.name)
+        impl_name = tpl.implementation
+        factory = _IMPULSE_REGISTRY.get(impl_name)
+        if factory is None:
+            raise KeyError(
+                f"no builtin impulse implementation {impl_name!r} "
+                f"(known: {sorted(_IMPULSE_REGISTRY)})"
+            )
+        handler = factory()
+        if hasattr(handler, "configure"):
+            handler.configure(dict(impulse.with_ or {}))
+        live = LiveImpulse(impulse=impulse, handler=handler)
+        with self._lock:
+            self.live[impulse.key] = live
+        handler.start(lambda payload: self._on_event(live, payload))
+        return live
+
+    def stop(self, impulse_key: str) -> None:
+        with self._lock:
+            live = self.live.pop(impulse_key, None)
+        if live is not None:
+            live.handler.stop()
+
+    def stop_all(self) -> None:
+        with self._lock:
+            keys = list(self.live)
+        for k in keys:
+            self.stop(k)
+
+    # ------------------------------------------------------------------
+
+    def _on_event(self, live: LiveImpulse, payload: dict):
+        """One trigger event → mapped inputs → StoryTrigger admission; for
+        streaming stories packets flow into the live pipeline."""
+        eng = self.engine
+        imp = live.impulse
+        live.emitted += 1
+        inputs = payload
+        if imp.mapping is not None and imp.mapping.inputs is not None:
+            inputs = self._evaluator.resolve_value(imp.mapping.inputs, {"event": payload})
+
+        story_ns = imp.story_ref.resolve_namespace(imp.namespace)
+        story = eng.registry.story(imp.story_ref.name, story_ns)
+
+        if story.pattern == StoryPattern.STREAMING:
+            # PerStoryRun pipeline: one live run per impulse; packets stream in
+            if live.stream_key is None or eng.stream_of(live.stream_key) is None:
+                stream = eng.submit_stream(
+                    story, inputs={"impulse": imp.name}, namespace=imp.namespace
+                )
+                live.stream_key = stream.run.key
+            stream = eng.stream_of(live.stream_key)
+            stream.push(inputs)
+            return {"streamed": True, "run": live.stream_key}
+
+        dedupe_key = None
+        if (
+            imp.delivery is not None
+            and imp.delivery.dedupe is not None
+            and imp.delivery.dedupe.key_template
+        ):
+            dedupe_key = str(
+                self._evaluator.resolve_string(
+                    imp.delivery.dedupe.key_template, {"event": payload}
+                )
+            )
+        trig = StoryTrigger(
+            submission_id=f"{imp.name}-{next(self._seq)}",
+            story_name=imp.story_ref.name,
+            story_namespace=story_ns,
+            namespace=imp.namespace,
+            key=dedupe_key,
+            inputs=inputs,
+            impulse=imp.name,
+        )
+        result = eng.triggers.submit(trig, throttle=imp.throttle)
+        live.decisions[str(result.decision)] = live.decisions.get(str(result.decision), 0) + 1
+        eng.metrics.inc("impulse_events_total", impulse=imp.name, decision=str(result.decision))
+        return result
+
+
+def build_http_app(engine: "RunEngine"):
+    """HTTP ingress (the reference's gRPC/webhook connector role): POST
+    /impulses/{ns}/{name} emits an event; GET /healthz, /metrics."""
+    from fastapi import FastAPI, HTTPException
+    from fastapi.responses import PlainTextResponse
+
+    app = FastAPI(title="bobrapet_amd ingress")
+    runtime: ImpulseRuntime = engine.impulses
+
+    @app.post("/impulses/{ns}/{name}")
+    async def trigger(ns: str, name: str, payload: dict):
+        live = runtime.live.get(f"{ns}/{name}")
+        if live is None:
+            raise HTTPException(404, f"impulse {ns}/{name} not running")
+        handler = live.handler
+        if isinstance(handler, ManualImpulse):
+            result = handler.emit(payload)
+        else:
+            result = runtime._on_event(live, payload)
+        if isinstance(result, StoryTrigger):
+            return {
+                "decision": str(result.decision),
+                "storyRun": result.story_run_ref,
+                "message": result.message,
+            }
+        return result
+
+    @app.get("/healthz")
+    async def healthz():
+        return {"ok": True}
+
+    @app.get("/metrics", response_class=PlainTextResponse)
+    async def metrics():
+        return engine.metrics.export_text()
+
+    return app
+
+
+def serve_http(engine: "RunEngine", host: str = "127.0.0.1", port: int = 8080):
+    import uvicorn
+
+    uvicorn.run(build_http_app(engine), host=host, port=port, log_level="warning")

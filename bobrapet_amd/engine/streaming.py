@@ -1,0 +1,355 @@
+"""Streaming (PerStoryRun) execution: persistent engram pipelines.
+
+Role replacement for the reference's realtime path (reference:
+steprun_controller.go:2527-4494 — per-run Deployment+Service+
+TransportBinding with connector sidecars and gRPC hub/P2P routing,
+SURVEY.md §3.5): a streaming Story materializes as in-process pipeline
+stages, one persistent worker per engram step pinned to its own
+(device, HIP stream), connected by credit-flow rings (transport/flow.py)
+per the analyzed topology (transport/topology.py).  Hub-routed steps
+evaluate their per-packet `runtime` templates in the engine's evaluator.
+
+hipGraph capture (BASELINE config #4): a stage whose engram implements
+``tensor_compute`` and whose step sets ``with.capture: true`` replays its
+per-packet GPU work as a captured hipGraph (torch.cuda.CUDAGraph is
+hipGraph on ROCm) after a warmup packet — fixed launch latency per packet.
+"""
+from __future__ import annotations
+
+import threading
+import time
+import typing as _t
+from dataclasses import dataclass, field
+
+from ..engrams import registry as engram_registry
+from ..engrams.base import EngramContext, EngramFailure, EngramResult
+from ..enums import Phase
+from ..specs import types as T
+from ..transport import flow, topology
+from .records import StoryRun, StructuredError, monotonic_now
+
+if _t.TYPE_CHECKING:
+    from .engine import RunEngine
+
+
+@dataclass
+class StageStats:
+    packets_in: int = 0
+    packets_out: int = 0
+    errors: int = 0
+    graph_replays: int = 0
+    busy_seconds: float = 0.0
+
+
+class _Stage(threading.Thread):
+    def __init__(
+        self,
+        sr: "StreamingRun",
+        step: T.Step,
+        in_rings: _t.List[flow.CreditRing],
+        out_rings: _t.List[flow.CreditRing],
+        device: _t.Optional[int],
+    ):
+        super().__init__(name=f"stage-{sr.run.name}-{step.name}", daemon=True)
+        self.sr = sr
+        self.step = step
+        self.in_rings = in_rings
+        self.out_rings = out_rings
+        self.device = device
+        self.stats = StageStats()
+        self.stream = None
+        self._graph = None
+        self._graph_in = None
+        self._graph_out = None
+        self.error: _t.Optional[str] = None
+
+    # ------------------------------------------------------------------
+
+    def run(self) -> None:
+        try:
+            if self.device is not None:
+                import torch
+
+                torch.cuda.set_device(self.device)
+                self.stream = torch.cuda.Stream(device=self.device)
+            self._loop()
+        except Exception as exc:  # stage crash fails the pipeline
+            self.error = f"{type(exc).__name__}: {exc}"
+            self.sr.on_stage_error(self.step.name, self.error)
+        finally:
+            for ring in self.out_rings:
+                ring.close()
+
+    def _loop(self) -> None:
+        import contextlib
+
+        while True:
+            packets = []
+            for ring in self.in_rings:
+                pkt = ring.pop()
+                if pkt is flow.SENTINEL:
+                    return
+                packets.append(pkt)
+            packet = packets[0] if len(packets) == 1 else {"fanIn": packets}
+            if self.sr.canceled:
+                return
+            t0 = time.monotonic()
+            ctx = contextlib.nullcontext()
+            if self.stream is not None:
+                import torch
+
+                ctx = torch.cuda.stream(self.stream)
+            with ctx:
+                out = self._process(packet)
+            if self.stream is not None:
+                self.stream.synchronize()
+            self.stats.busy_seconds += time.monotonic() - t0
+            self.stats.packets_in += 1
+            if out is not None:
+                for ring in self.out_rings:
+                    ring.push(out)
+                self.stats.packets_out += 1
+                self.sr.on_packet(self.step.name, leaf=not self.out_rings, packet=out)
+
+    # ------------------------------------------------------------------
+
+    def _process(self, packet):
+        eng = self.sr.engine
+        step = self.step
+        # hub routing: per-packet runtime templates transform the payload
+        # (reference: BUBU_TEMPLATE_CONTEXT hub evaluation)
+        if step.name in self.sr.topo.hub_steps and isinstance(step.runtime, dict):
+            route = step.runtime.get("route")
+            if route is not None:
+                scope = {"packet": packet, "inputs": self.sr.run.inputs}
+                if not eng.evaluator.evaluate_condition(str(route), scope):
+                    return None  # dropped by routing rule
+            transform = step.runtime.get("transform")
+            if transform is not None:
+                scope = {"packet": packet, "inputs": self.sr.run.inputs}
+                packet = eng.evaluator.resolve_value(transform, scope)
+
+        impl = self._impl()
+        if impl is None:  # pure hub/transform step without an engram
+            return packet
+
+        captured = self._maybe_graph(impl, packet)
+        if captured is not None:
+            self.stats.graph_replays += 1
+            return captured
+
+        ctx = self._ctx(packet)
+        try:
+            result = impl.run(ctx)
+        except EngramFailure as exc:
+            self.stats.errors += 1
+            self.sr.on_stage_error(step.name, str(exc))
+            return None
+        if isinstance(result, EngramResult):
+            return result.output
+        return result
+
+    def _impl(self):
+        if self.step.ref is None:
+            return None
+        eng = self.sr.engine
+        engram = eng.registry.try_engram(
+            self.step.ref.name, self.step.ref.resolve_namespace(self.sr.story.namespace)
+        )
+        if engram is None:
+            raise RuntimeError(f"engram {self.step.ref.name} not found")
+        tpl = (
+            eng.registry.engram_template(engram.template_ref.name)
+            if engram.template_ref is not None
+            else None
+        )
+        name = tpl.implementation if tpl is not None else self.step.ref.name
+        self._engram_cfg = engram.with_
+        return engram_registry.resolve(name)
+
+    def _ctx(self, packet) -> EngramContext:
+        return EngramContext(
+            story_name=self.sr.story.name,
+            story_run=self.sr.run.name,
+            step_name=self.step.name,
+            namespace=self.sr.run.namespace,
+            input=packet,
+            config=getattr(self, "_engram_cfg", None),
+            runtime=self.step.runtime,
+            execution_mode="deployment",
+            device=self.device,
+            stream=self.stream,
+            storage=self.sr.engine.storage,
+            cancel_check=lambda: self.sr.canceled,
+        )
+
+    # ------------------------------------------------------------------
+
+    def _maybe_graph(self, impl, packet):
+        """hipGraph capture path: engram.tensor_compute over a fixed-shape
+        tensor packet, captured once then replayed per packet."""
+        want = isinstance(self.step.with_, dict) and self.step.with_.get("capture")
+        fn = getattr(impl, "tensor_compute", None)
+        if not want or fn is None or self.device is None:
+            return None
+        import torch
+
+        tensor = packet.get("tensor") if isinstance(packet, dict) else None
+        if not torch.is_tensor(tensor):
+            return None
+        tensor = tensor.to(f"cuda:{self.device}", non_blocking=False)
+        if self._graph is None:
+            ctx = self._ctx(packet)
+            # warmup (allocations settle), then capture
+            fn(ctx, tensor)
+            torch.cuda.synchronize(self.device)
+            self._graph_in = tensor.clone()
+            g = torch.cuda.CUDAGraph()
+            capture_stream = torch.cuda.Stream(device=self.device)
+            with torch.cuda.stream(capture_stream):
+                with torch.cuda.graph(g, stream=capture_stream):
+                    self._graph_out = fn(ctx, self._graph_in)
+            self._graph = g
+        self._graph_in.copy_(tensor)
+        self._graph.replay()
+        if self.stream is not None:
+            torch.cuda.synchronize(self.device)
+        out = dict(packet) if isinstance(packet, dict) else {}
+        out["tensor"] = self._graph_out.clone()
+        return out
+
+
+class StreamingRun:
+    """A live streaming StoryRun: pipeline of stages + ingress."""
+
+    def __init__(self, engine: "RunEngine", run: StoryRun, story: T.Story):
+        self.engine = engine
+        self.run = run
+        self.story = story
+        self.topo = topology.analyze(story)
+        self.canceled = False
+        self._lock = threading.Lock()
+        self._leaf_packets = 0
+        self._last_outputs: _t.List = []
+        self.settings = self._settings(story)
+
+        # rings: one per edge; ingress edges share the ingress ring list
+        self._edge_rings: _t.Dict[_t.Tuple[str, str], flow.CreditRing] = {}
+        self.ingress: _t.List[flow.CreditRing] = []
+        for e in self.topo.edges:
+            ring = flow.CreditRing(
+                name=f"{run.name}:{e.src or '@'}->{e.dst}",
+                settings=self.settings,
+                lane=e.lane,
+            )
+            self._edge_rings[(e.src, e.dst)] = ring
+            if e.src == "":
+                self.ingress.append(ring)
+
+        self.stages: _t.List[_Stage] = []
+        by_name = {s.name: s for s in story.steps}
+        n_dev = engine.workers.device_count
+        for i, name in enumerate(self.topo.stages):
+            step = by_name[name]
+            in_rings = [
+                self._edge_rings[(e.src, e.dst)] for e in self.topo.upstream_of(name)
+            ]
+            out_rings = [
+                self._edge_rings[(e.src, e.dst)] for e in self.topo.downstream_of(name)
+            ]
+            device = None
+            if n_dev > 0 and step.ref is not None:
+                device = i % n_dev  # stage-per-device pipeline placement
+            stage = _Stage(self, step, in_rings, out_rings, device)
+            self.stages.append(stage)
+            st = run.step_state(name)
+            st.phase = Phase.RUNNING
+            st.started_at = monotonic_now()
+        run.phase = Phase.RUNNING
+        run.started_at = run.started_at or monotonic_now()
+        for s in self.stages:
+            s.start()
+
+    @staticmethod
+    def _settings(story: T.Story) -> T.TransportStreamingSettings:
+        if story.transports and story.transports[0].streaming:
+            from ..specs.types import from_dict
+
+            return from_dict(T.TransportStreamingSettings, story.transports[0].streaming)
+        return T.default_streaming_settings()
+
+    # ------------------------------------------------------------------
+
+    def push(self, packet, timeout: _t.Optional[float] = 30.0) -> bool:
+        ok = True
+        for ring in self.ingress:
+            ok = ring.push(packet, timeout=timeout) and ok
+        return ok
+
+    def finish(self, timeout: float = 60.0) -> StoryRun:
+        """Close the ingress, drain and finalize the run."""
+        for ring in self.ingress:
+            ring.close()
+        deadline = time.monotonic() + timeout
+        for s in self.stages:
+            s.join(timeout=max(deadline - time.monotonic(), 0.1))
+        now = monotonic_now()
+        failed = False
+        for s in self.stages:
+            st = self.run.step_state(s.step.name)
+            st.output = {
+                "packetsIn": s.stats.packets_in,
+                "packetsOut": s.stats.packets_out,
+                "errors": s.stats.errors,
+                "graphReplays": s.stats.graph_replays,
+                "busySeconds": round(s.stats.busy_seconds, 6),
+            }
+            st.finished_at = now
+            if s.error:
+                st.phase = Phase.FAILED
+                st.error = StructuredError(message=s.error)
+                failed = True
+            else:
+                st.phase = Phase.SUCCEEDED
+        self.run.phase = Phase.FAILED if failed else Phase.FINISHED
+        self.run.output = {
+            "packets": self._leaf_packets,
+            "stages": len(self.stages),
+        }
+        self.run.finished_at = now
+        self.engine.on_run_terminal(self.run)
+        self.engine._streams.pop(self.run.key, None)
+        return self.run
+
+    def cancel(self) -> None:
+        self.canceled = True
+        for ring in self._edge_rings.values():
+            ring.close()
+
+    def on_packet(self, stage: str, leaf: bool, packet) -> None:
+        if leaf:
+            with self._lock:
+                self._leaf_packets += 1
+                if len(self._last_outputs) < 8:
+                    self._last_outputs.append(_strip_tensors(packet))
+        self.engine.metrics.inc("stream_packets_total", stage=stage)
+
+    def on_stage_error(self, stage: str, message: str) -> None:
+        self.engine.metrics.inc("stream_stage_errors_total", stage=stage)
+
+    @property
+    def leaf_packets(self) -> int:
+        with self._lock:
+            return self._leaf_packets
+
+
+def _strip_tensors(value):
+    import torch
+
+    if torch.is_tensor(value):
+        return {"tensor": {"shape": list(value.shape), "dtype": str(value.dtype)}}
+    if isinstance(value, dict):
+        return {k: _strip_tensors(v) for k, v in value.items()}
+    if isinstance(value, list):
+        return [_strip_tensors(v) for v in value]
+    return value

@@ -212,7 +212,7 @@ def _validate_primitive_with(res: ValidationResult, step: T.Step) -> None:
                     res.error(f"step {step.name!r}: duplicate parallel branch {b['name']!r}")
                 seen.add(b["name"])
     elif st == StepType.CONDITION:
-        if not (w.get("expression") or w.get("if") or step.if_):
+        if not (w.get("expression") or w.get("if") or step.if_ or step.runtime):
             res.error(f"step {step.name!r}: condition requires with.expression")
 
 

@@ -1,0 +1,362 @@
+"""Streaming subsystem tests: topology analysis, credit-flow rings, the
+PerStoryRun pipeline runtime, impulse ingress (CPU; the hipGraph capture
+path is exercised by the gpu-marked test at the bottom)."""
+import threading
+import time
+
+import pytest
+
+from bobrapet_amd.engine import EngineConfig, RunEngine
+from bobrapet_amd.enums import Phase
+from bobrapet_amd.specs import load_yaml
+from bobrapet_amd.transport import flow, topology
+
+
+RESOURCES = """
+kind: EngramTemplate
+metadata: {name: transform-tpl}
+spec: {builtin: transform}
+---
+kind: Engram
+metadata: {name: transformer}
+spec: {templateRef: {name: transform-tpl}}
+---
+kind: EngramTemplate
+metadata: {name: echo-tpl}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: echoer}
+spec: {templateRef: {name: echo-tpl}}
+"""
+
+
+@pytest.fixture()
+def eng():
+    engine = RunEngine(EngineConfig(cpu_workers=2)).start()
+    engine.apply_yaml(RESOURCES)
+    yield engine
+    engine.stop()
+
+
+class TestTopology:
+    def test_linear_pipeline(self):
+        (story,) = load_yaml(
+            """
+kind: Story
+metadata: {name: pipe}
+spec:
+  pattern: streaming
+  steps:
+    - {name: a, ref: {name: x}}
+    - {name: b, ref: {name: x}, needs: [a]}
+    - {name: c, ref: {name: x}, needs: [b]}
+"""
+        )
+        topo = topology.analyze(story)
+        assert topo.stages == ["a", "b", "c"]
+        assert [(e.src, e.dst) for e in topo.edges] == [("", "a"), ("a", "b"), ("b", "c")]
+        assert all(e.mode == "p2p" for e in topo.edges)
+
+    def test_hub_routing_for_runtime_templates(self):
+        (story,) = load_yaml(
+            """
+kind: Story
+metadata: {name: hubbed}
+spec:
+  pattern: streaming
+  steps:
+    - {name: a, ref: {name: x}}
+    - {name: b, ref: {name: x}, needs: [a], runtime: {route: "{{ packet.ok }}"}}
+"""
+        )
+        topo = topology.analyze(story)
+        assert "b" in topo.hub_steps
+        assert [e.mode for e in topo.edges if e.dst == "b"] == ["hub"]
+
+
+class TestCreditRing:
+    def test_fifo_and_close(self):
+        ring = flow.CreditRing("t")
+        ring.push({"a": 1})
+        ring.push({"a": 2})
+        assert ring.pop()["a"] == 1
+        assert ring.pop()["a"] == 2
+        ring.close()
+        assert ring.pop() is flow.SENTINEL
+
+    def test_credits_replenish(self):
+        from bobrapet_amd.specs.types import (
+            TransportBackpressure,
+            TransportFlowControl,
+            TransportStreamingSettings,
+        )
+
+        s = TransportStreamingSettings(
+            flow_control=TransportFlowControl(mode="credit", initial_credits=2, max_credits=2),
+            backpressure=TransportBackpressure(buffer_packets=10, policy="block"),
+        )
+        ring = flow.CreditRing("t", settings=s)
+        assert ring.push(1, timeout=0.05)
+        assert ring.push(2, timeout=0.05)
+        assert not ring.push(3, timeout=0.05)  # out of credits → blocked → timeout
+        assert ring.pop() == 1
+        assert ring.push(3, timeout=0.05)  # credit came back
+        assert ring.stats.blocked_waits >= 1
+
+    def test_drop_newest_policy(self):
+        from bobrapet_amd.specs.types import (
+            TransportBackpressure,
+            TransportFlowControl,
+            TransportStreamingSettings,
+        )
+
+        s = TransportStreamingSettings(
+            flow_control=TransportFlowControl(mode="none"),
+            backpressure=TransportBackpressure(buffer_packets=2, policy="dropNewest"),
+        )
+        ring = flow.CreditRing("t", settings=s)
+        assert ring.push(1) and ring.push(2)
+        assert not ring.push(3)
+        assert ring.stats.dropped_newest == 1
+
+    def test_drop_oldest_policy(self):
+        from bobrapet_amd.specs.types import (
+            TransportBackpressure,
+            TransportFlowControl,
+            TransportStreamingSettings,
+        )
+
+        s = TransportStreamingSettings(
+            flow_control=TransportFlowControl(mode="none"),
+            backpressure=TransportBackpressure(buffer_packets=2, policy="dropOldest"),
+        )
+        ring = flow.CreditRing("t", settings=s)
+        ring.push(1)
+        ring.push(2)
+        ring.push(3)
+        assert ring.pop() == 2  # 1 was dropped
+        assert ring.stats.dropped_oldest == 1
+
+    def test_blocking_producer_consumer(self):
+        ring = flow.CreditRing("t")
+        got = []
+
+        def consumer():
+            while True:
+                p = ring.pop()
+                if p is flow.SENTINEL:
+                    return
+                got.append(p)
+
+        t = threading.Thread(target=consumer)
+        t.start()
+        for i in range(100):
+            ring.push(i)
+        ring.close()
+        t.join(timeout=5)
+        assert got == list(range(100))
+
+
+class TestStreamingRun:
+    def test_three_stage_pipeline(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: pipe3}
+spec:
+  pattern: streaming
+  steps:
+    - name: double
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v * 2 }}"}}
+    - name: gatekeep
+      type: condition
+      needs: [double]
+      runtime: {route: "{{ packet.items[0].v < 10 }}"}
+    - name: inc
+      ref: {name: transformer}
+      needs: [gatekeep]
+      runtime: {map: {v: "{{ item.v + 1 }}"}}
+"""
+        )
+        stream = eng.submit_stream("default/pipe3")
+        for i in range(10):
+            stream.push({"items": [{"v": i}]})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # routing: packets with 2*v >= 10 dropped → v in 0..4 pass
+        assert run.output["packets"] == 5
+        assert run.step_states["double"].output["packetsIn"] == 10
+        assert run.step_states["inc"].output["packetsIn"] == 5
+
+    def test_stream_cancel(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: pipec}
+spec:
+  pattern: streaming
+  steps:
+    - {name: only, ref: {name: echoer}}
+"""
+        )
+        stream = eng.submit_stream("default/pipec")
+        stream.push({"x": 1})
+        time.sleep(0.05)
+        eng.cancel(stream.run.key)
+        assert stream.run.is_terminal
+
+    def test_batch_story_rejected(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: batchy}
+spec:
+  steps:
+    - {name: only, ref: {name: echoer}, with: {v: 1}}
+"""
+        )
+        with pytest.raises(ValueError):
+            eng.submit_stream("default/batchy")
+
+
+class TestImpulses:
+    def test_interval_impulse_batch_story(self, eng):
+        eng.apply_yaml(
+            """
+kind: ImpulseTemplate
+metadata: {name: interval-tpl}
+spec: {builtin: interval}
+---
+kind: Story
+metadata: {name: ticked}
+spec:
+  steps:
+    - {name: work, ref: {name: echoer}, with: {n: "{{ inputs.n }}"}}
+---
+kind: Impulse
+metadata: {name: ticker}
+spec:
+  templateRef: {name: interval-tpl}
+  storyRef: {name: ticked}
+  with: {intervalMs: 10, count: 4}
+  mapping:
+    inputs: {n: "{{ event.tick }}"}
+"""
+        )
+        live = eng.impulses.start("default/ticker")
+        deadline = time.time() + 5
+        while live.emitted < 4 and time.time() < deadline:
+            time.sleep(0.02)
+        time.sleep(0.2)
+        runs = eng.store.runs_of_story("default/ticked")
+        assert len(runs) == 4
+        assert all(eng.wait(r.key, timeout=5).phase == Phase.SUCCEEDED for r in runs)
+
+    def test_manual_impulse_streaming_story(self, eng):
+        eng.apply_yaml(
+            """
+kind: ImpulseTemplate
+metadata: {name: manual-tpl}
+spec: {builtin: manual}
+---
+kind: Story
+metadata: {name: streamy}
+spec:
+  pattern: streaming
+  steps:
+    - name: stage
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v }}"}}
+---
+kind: Impulse
+metadata: {name: pusher}
+spec:
+  templateRef: {name: manual-tpl}
+  storyRef: {name: streamy}
+"""
+        )
+        live = eng.impulses.start("default/pusher")
+        for i in range(3):
+            live.handler.emit({"items": [{"v": i}]})
+        time.sleep(0.2)
+        stream = eng.stream_of(live.stream_key)
+        assert stream is not None
+        assert stream.leaf_packets == 3
+        run = stream.finish(timeout=5)
+        assert run.phase == Phase.FINISHED
+
+    def test_impulse_throttle_max_in_flight(self, eng):
+        eng.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: sleepy-tpl}
+spec: {builtin: sleepy}
+---
+kind: Engram
+metadata: {name: sleeper}
+spec: {templateRef: {name: sleepy-tpl}}
+---
+kind: ImpulseTemplate
+metadata: {name: manual-tpl}
+spec: {builtin: manual}
+---
+kind: Story
+metadata: {name: slow-story}
+spec:
+  steps:
+    - {name: work, ref: {name: sleeper}, with: {seconds: 0.5}}
+---
+kind: Impulse
+metadata: {name: flooder}
+spec:
+  templateRef: {name: manual-tpl}
+  storyRef: {name: slow-story}
+  throttle: {maxInFlight: 2}
+"""
+        )
+        live = eng.impulses.start("default/flooder")
+        decisions = [str(live.handler.emit({"i": i}).decision) for i in range(4)]
+        assert decisions.count("Created") == 2
+        assert decisions.count("Rejected") == 2
+
+
+class TestHttpIngress:
+    def test_http_app_trigger(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        eng.apply_yaml(
+            """
+kind: ImpulseTemplate
+metadata: {name: manual-tpl}
+spec: {builtin: manual}
+---
+kind: Story
+metadata: {name: web-story}
+spec:
+  steps:
+    - {name: work, ref: {name: echoer}, with: {got: "{{ inputs.msg }}"}}
+---
+kind: Impulse
+metadata: {name: webhook}
+spec:
+  templateRef: {name: manual-tpl}
+  storyRef: {name: web-story}
+  mapping:
+    inputs: {msg: "{{ event.message }}"}
+"""
+        )
+        eng.impulses.start("default/webhook")
+        client = TestClient(build_http_app(eng))
+        resp = client.post("/impulses/default/webhook", json={"message": "hi"})
+        assert resp.status_code == 200
+        body = resp.json()
+        assert body["decision"] == "Created"
+        run = eng.wait(body["storyRun"], timeout=5)
+        assert run.phase == Phase.SUCCEEDED
+        assert run.step_states["work"].output == {"got": "hi"}
+        assert client.get("/healthz").json() == {"ok": True}
+        assert "bobrapet_amd_" in client.get("/metrics").text

@@ -112,7 +112,16 @@ spec:
 """
 
 
-def run_one(eng: RunEngine, story_key: str, idx: int, rank: int) -> dict:
+def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None) -> dict:
+    if native is not None:
+        status = native.run_story(story_key, {"i": idx, "rank": rank}, timeout=600)
+        if status["phase"] != "Succeeded":
+            raise RuntimeError(f"native bench run failed: {status}")
+        lat = {}
+        for name, st in status["steps"].items():
+            if st.get("startedAt") and st.get("finishedAt"):
+                lat[name] = (st["finishedAt"] - st["startedAt"]) * 1000.0
+        return lat
     run = eng.submit_run(story_key, {"i": idx, "rank": rank}, name=f"bench-{rank}-{idx}")
     run = eng.wait(run, timeout=600)
     if run.phase != Phase.SUCCEEDED:
@@ -132,6 +141,10 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--config", default="parallel8", choices=["parallel8", "cpu", "llm"])
+    ap.add_argument(
+        "--engine-impl", default="auto", choices=["auto", "python", "native"],
+        help="DAG engine: the bobraccel C++ core (native) or the Python engine",
+    )
     args = ap.parse_args()
 
     multi = group.init_distributed()
@@ -160,15 +173,32 @@ def main() -> int:
     eng = RunEngine(
         EngineConfig(cpu_workers=4, workers_per_device=4, child_ttl_seconds=5.0)
     ).start()
+    native = None
     try:
         eng.apply_yaml(RESOURCES)
         eng.apply_yaml(PARALLEL_STORY)
         eng.apply_yaml(SLEEP_STORY)
         eng.apply_yaml(LLM_STORY)
 
+        if args.engine_impl in ("auto", "native"):
+            try:
+                from bobrapet_amd.runtime.native import NativeRunner, story_supported
+
+                ns, _, nm = story_key.rpartition("/")
+                story_obj = eng.registry.story(nm, ns)
+                if story_supported(story_obj) is None:
+                    native = NativeRunner.from_run_engine(eng)
+                    native.compile(story_obj)
+                elif args.engine_impl == "native":
+                    raise RuntimeError(story_supported(story_obj))
+            except Exception:
+                if args.engine_impl == "native":
+                    raise
+                native = None
+
         # warmup (untimed): fills weight/table caches, compiles nothing
         for i in range(args.warmup):
-            run_one(eng, story_key, -(i + 1), rank)
+            run_one(eng, story_key, -(i + 1), rank, native)
 
         group.barrier()
         if has_gpu:
@@ -176,7 +206,7 @@ def main() -> int:
         t0 = time.monotonic()
         step_lat: list = []
         for i in range(args.steps):
-            lat = run_one(eng, story_key, i, rank)
+            lat = run_one(eng, story_key, i, rank, native)
             step_lat.extend(lat.values())
         group.barrier()
         if has_gpu:
@@ -212,12 +242,15 @@ def main() -> int:
                         2048 if config_name == "llm" else 0
                     ),
                     "parallelism": f"dp{world}" if world > 1 else "single",
+                    "engine": "bobraccel-native" if native is not None else "python",
                     "branches": BRANCHES if config_name == "parallel8" else None,
                     "p50_step_latency_ms": round(p50, 3),
                 },
             }
             print(json.dumps(line), flush=True)
     finally:
+        if native is not None:
+            native.stop()
         eng.stop()
         group.teardown()
     return 0

@@ -16,6 +16,8 @@ import sysconfig
 PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 SRC_DIR = os.path.join(PKG_DIR, "csrc", "hip")
 OUT_SO = os.path.join(PKG_DIR, "_hipops.so")
+CORE_SO = os.path.join(PKG_DIR, "_core.so")
+CORE_DIR = os.path.join(PKG_DIR, "csrc", "core")
 BUILD_DIR = os.path.join(PKG_DIR, "csrc", "build")
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
@@ -89,6 +91,34 @@ def build(force: bool = False, verbose: bool = True) -> str:
     return OUT_SO
 
 
+def build_core(force: bool = False) -> str:
+    """Build the bobraccel native DAG core (plain C++, no torch headers)."""
+    import pybind11
+
+    os.makedirs(BUILD_DIR, exist_ok=True)
+    srcs = [os.path.join(CORE_DIR, f) for f in ("engine.cpp", "pybind.cpp")]
+    hdrs = [os.path.join(CORE_DIR, f) for f in ("jvalue.h", "expr.h", "engine.h")]
+    deps = srcs + hdrs
+    if not force and os.path.exists(CORE_SO) and not any(
+        _newer(d, CORE_SO) for d in deps
+    ):
+        return CORE_SO
+    py_inc = sysconfig.get_paths()["include"]
+    ext_suffix = ""
+    cmd = [
+        "g++", "-O2", "-std=c++17", "-shared", "-fPIC",
+        f"-I{pybind11.get_include()}", f"-I{py_inc}", f"-I{CORE_DIR}",
+        *srcs, "-o", CORE_SO, "-pthread",
+    ]
+    _run(cmd)
+    return CORE_SO
+
+
+def build_all(force: bool = False) -> None:
+    build(force=force)
+    build_core(force=force)
+
+
 if __name__ == "__main__":
-    build(force="--force" in sys.argv)
-    print(f"built {OUT_SO}")
+    build_all(force="--force" in sys.argv)
+    print(f"built {OUT_SO} and {CORE_SO}")

@@ -66,6 +66,22 @@ metadata: {{name: llm}}
 spec:
   templateRef: {{name: llm-infer}}
   with: {{model: llama-3-8b}}
+---
+kind: EngramTemplate
+metadata: {{name: transform-tpl}}
+spec: {{builtin: transform}}
+---
+kind: Engram
+metadata: {{name: joiner-passthrough}}
+spec: {{templateRef: {{name: transform-tpl}}}}
+---
+kind: EngramTemplate
+metadata: {{name: echo-tpl}}
+spec: {{builtin: echo}}
+---
+kind: Engram
+metadata: {{name: echo-stage}}
+spec: {{templateRef: {{name: echo-tpl}}}}
 """
 
 PARALLEL_STORY = f"""
@@ -111,6 +127,45 @@ spec:
   output: {tokens: "{{ steps.infer.output.tokensProcessed }}"}
 """
 
+STREAM_STORY = """
+kind: Story
+metadata: {name: bench-stream}
+spec:
+  pattern: streaming
+  steps:
+    - name: featurize
+      ref: {name: embedder}
+      with: {capture: true}
+    - name: tag
+      ref: {name: echo-stage}
+      needs: [featurize]
+"""
+
+BIGPAYLOAD_STORY = """
+kind: Story
+metadata: {name: bench-big-inner}
+spec:
+  steps:
+    - {name: make, ref: {name: embedder}, with: {batch: 256, seqLen: 512}}
+    - name: ready
+      type: wait
+      with: {until: "{{ steps.make.phase == 'Succeeded' }}", pollInterval: 2ms, timeout: 60s}
+    - name: use
+      ref: {name: joiner}
+      needs: [ready]
+      with: {refs: ["{{ steps.make.output.embeddings }}"]}
+  output: {rows: "{{ steps.use.output.worldRows }}"}
+---
+kind: Story
+metadata: {name: bench-big}
+spec:
+  steps:
+    - name: sub
+      type: executeStory
+      with: {storyRef: bench-big-inner, with: {}}
+  output: {rows: "{{ steps.sub.output.output.rows }}"}
+"""
+
 
 def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None) -> dict:
     if native is not None:
@@ -135,12 +190,69 @@ def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None) ->
     return lat
 
 
+def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
+    """Config #4: packets/sec through the captured streaming pipeline;
+    one 'step' = one packet."""
+    stream = eng.submit_stream("default/bench-stream")
+    ids = torch.randint(0, EMBED_VOCAB, (EMBED_BATCH, EMBED_SEQ), dtype=torch.int32)
+    if torch.cuda.is_available():
+        ids = ids.cuda()
+    for i in range(args.warmup):
+        stream.push({"tensor": ids, "seq": -i, "items": [{}]})
+    deadline = time.monotonic() + 30
+    while stream.leaf_packets < args.warmup and time.monotonic() < deadline:
+        time.sleep(0.001)
+    group.barrier()
+    t0 = time.monotonic()
+    for i in range(args.steps):
+        stream.push({"tensor": ids, "seq": i, "items": [{}]})
+    while stream.leaf_packets < args.warmup + args.steps and time.monotonic() < t0 + 120:
+        time.sleep(0.001)
+    group.barrier()
+    elapsed = time.monotonic() - t0
+    run = stream.finish(timeout=30)
+    if run.phase.value not in ("Finished",):
+        raise RuntimeError(f"stream bench failed: {run.phase}")
+    elapsed_max = group.max_over_ranks(elapsed, device="cpu" if not torch.cuda.is_available() else None)
+    replays = run.step_states["featurize"].output.get("graphReplays", 0)
+    if rank == 0:
+        line = {
+            "metric": "StoryRuns/sec (8-way parallel Story)",
+            "value": round(args.steps * world / elapsed_max, 3),
+            "unit": "packets/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed_max * 1000.0 / args.steps, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "streaming 2-stage pipeline, hipGraph-captured embed",
+                "bench_config": "stream",
+                "global_batch": args.steps * world,
+                "seq_len": EMBED_SEQ,
+                "parallelism": f"dp{world}" if world > 1 else "single",
+                "graph_replays": replays,
+            },
+        }
+        print(json.dumps(line), flush=True)
+    eng.stop()
+    group.teardown()
+    return 0
+
+
 def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--config", default="parallel8", choices=["parallel8", "cpu", "llm"])
+    ap.add_argument(
+        "--config", default="parallel8",
+        choices=["parallel8", "cpu", "llm", "stream", "bigpayload"],
+    )
     ap.add_argument(
         "--engine-impl", default="auto", choices=["auto", "python", "native"],
         help="DAG engine: the bobraccel C++ core (native) or the Python engine",
@@ -154,7 +266,7 @@ def main() -> int:
     has_gpu = torch.cuda.is_available()
     device = torch.cuda.current_device() if has_gpu else None
 
-    if args.config == "cpu" or not has_gpu:
+    if args.config == "cpu" or (not has_gpu and args.config in ("parallel8", "llm")):
         config_name = "cpu"
         story_key = "default/bench-cpu"
         model_desc = "2-step batch story (sleep->condition), CPU engine"
@@ -162,6 +274,14 @@ def main() -> int:
         config_name = "llm"
         story_key = "default/bench-llm"
         model_desc = "llm-infer llama-3-8b bf16 prefill b4 s2048"
+    elif args.config == "stream":
+        config_name = "stream"
+        story_key = "default/bench-stream"
+        model_desc = "streaming 2-stage pipeline, hipGraph-captured embed"
+    elif args.config == "bigpayload":
+        config_name = "bigpayload"
+        story_key = "default/bench-big"
+        model_desc = "nested executeStory + wait, HBM tensor payload edge"
     else:
         config_name = "parallel8"
         story_key = "default/bench-parallel8"
@@ -179,6 +299,11 @@ def main() -> int:
         eng.apply_yaml(PARALLEL_STORY)
         eng.apply_yaml(SLEEP_STORY)
         eng.apply_yaml(LLM_STORY)
+        eng.apply_yaml(STREAM_STORY)
+        eng.apply_yaml(BIGPAYLOAD_STORY)
+
+        if config_name == "stream":
+            return run_stream_bench(eng, args, rank, world, n_gpus)
 
         if args.engine_impl in ("auto", "native"):
             try:

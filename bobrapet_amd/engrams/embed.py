@@ -40,6 +40,17 @@ class EmbedEngram(Engram):
     name = "embed"
     wants_gpu = True
 
+    def tensor_compute(self, ctx: EngramContext, ids: "torch.Tensor") -> "torch.Tensor":
+        """Pure tensor path for hipGraph capture (streaming stages with
+        with.capture: true replay this as a captured graph per packet)."""
+        from .. import ops
+
+        cfg = dict(ctx.config or {})
+        dim = int(cfg.get("dim", 4096))
+        vocab = int(cfg.get("vocab", 32000))
+        table = _table(vocab, dim, int(cfg.get("seed", 7)), ids.device)
+        return ops.embed_pool(table, ids.to(torch.int32))
+
     def run(self, ctx: EngramContext) -> EngramResult:
         from .. import ops
 

@@ -170,11 +170,20 @@ class Evaluator:
         if cap <= 0:
             return
         try:
-            size = len(json.dumps(value, separators=(",", ":"), default=str))
+            size = len(json.dumps(value, separators=(",", ":"), default=_cheap_default))
         except (TypeError, ValueError):
             return
         if size > cap:
             raise OutputTooLarge(f"template output {size} bytes exceeds cap {cap}")
+
+
+def _cheap_default(value):
+    """Opaque objects (tensors) serialize as a short tag — never stringify
+    payload data on the control plane."""
+    t = type(value)
+    if t.__module__ == "torch" and t.__name__ in ("Tensor", "Parameter"):
+        return f"<tensor {tuple(value.shape)}>"
+    return str(value)[:256]
 
 
 def _unwrap_missing(value):
@@ -208,6 +217,9 @@ def _truthy(v) -> bool:
 def _stringify(v) -> str:
     if v is None:
         return ""
+    t = type(v)
+    if t.__module__ == "torch" and t.__name__ in ("Tensor", "Parameter"):
+        return f"<tensor {tuple(v.shape)}>"
     if isinstance(v, bool):
         return "true" if v else "false"
     if isinstance(v, float) and v.is_integer():

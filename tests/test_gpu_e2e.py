@@ -1,0 +1,159 @@
+"""GPU end-to-end tests: streaming pipeline with hipGraph capture
+(BASELINE config #4) and the nested executeStory + gate/wait story with
+multi-GB HBM payload offload (config #5)."""
+import time
+
+import pytest
+import torch
+
+from bobrapet_amd.engine import EngineConfig, RunEngine
+from bobrapet_amd.enums import Phase
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture()
+def eng():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    engine = RunEngine(EngineConfig(cpu_workers=2, workers_per_device=2)).start()
+    yield engine
+    engine.stop()
+
+
+class TestStreamingGraphCapture:
+    def test_three_stage_pipeline_with_capture(self, eng):
+        eng.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: embed}
+spec: {builtin: embed}
+---
+kind: Engram
+metadata: {name: embedder}
+spec:
+  templateRef: {name: embed}
+  with: {dim: 2048, vocab: 8000}
+---
+kind: EngramTemplate
+metadata: {name: transform-tpl}
+spec: {builtin: transform}
+---
+kind: Engram
+metadata: {name: transformer}
+spec: {templateRef: {name: transform-tpl}}
+---
+kind: Story
+metadata: {name: stream-gpu}
+spec:
+  pattern: streaming
+  steps:
+    - name: featurize
+      ref: {name: embedder}
+      with: {capture: true}
+    - name: tag
+      ref: {name: transformer}
+      needs: [featurize]
+      runtime: {map: {ok: "{{ true }}"}}
+"""
+        )
+        stream = eng.submit_stream("default/stream-gpu")
+        ids = torch.randint(0, 8000, (8, 64), device="cuda", dtype=torch.int32)
+        n = 12
+        for i in range(n):
+            stream.push({"tensor": ids.clone(), "seq": i, "items": [{"v": i}]})
+        run = stream.finish(timeout=60)
+        assert run.phase == Phase.FINISHED, {
+            k: (str(v.phase), v.error) for k, v in run.step_states.items()
+        }
+        st = run.step_states["featurize"].output
+        assert st["packetsIn"] == n
+        # first packet captures, the rest replay the hipGraph
+        assert st["graphReplays"] >= n - 1, st
+
+
+class TestBigPayloadNestedStory:
+    def test_nested_gate_wait_with_hbm_offload(self, eng):
+        """Config #5 shape: nested executeStory + gate/wait; a multi-GB
+        tensor payload crosses the step edge as a $storageRef (HBM
+        resident, never serialized)."""
+        from bobrapet_amd.engrams.base import Engram, EngramContext, EngramResult
+        from bobrapet_amd.engrams.registry import register
+
+        class BigProducer(Engram):
+            name = "big-producer"
+            wants_gpu = True
+
+            def run(self, ctx: EngramContext) -> EngramResult:
+                gb = float((ctx.input or {}).get("gb", 1.0))
+                n = int(gb * (1 << 30) // 2)  # bf16 elements
+                t = torch.ones(n, dtype=torch.bfloat16, device=f"cuda:{ctx.device}")
+                return EngramResult(
+                    output={"blob": ctx.storage.offload_tensor(t), "gb": gb}
+                )
+
+        class BigConsumer(Engram):
+            name = "big-consumer"
+            wants_gpu = True
+
+            def run(self, ctx: EngramContext) -> EngramResult:
+                blob = (ctx.input or {}).get("blob")
+                assert torch.is_tensor(blob), type(blob)
+                s = float(blob[:1000].float().sum().item())
+                return EngramResult(output={"checksum": s, "numel": blob.numel()})
+
+        register("big-producer", BigProducer)
+        register("big-consumer", BigConsumer)
+        eng.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: big-producer}
+spec: {builtin: big-producer}
+---
+kind: EngramTemplate
+metadata: {name: big-consumer}
+spec: {builtin: big-consumer}
+---
+kind: Engram
+metadata: {name: producer}
+spec: {templateRef: {name: big-producer}}
+---
+kind: Engram
+metadata: {name: consumer}
+spec: {templateRef: {name: big-consumer}}
+---
+kind: Story
+metadata: {name: inner-big}
+spec:
+  steps:
+    - {name: make, ref: {name: producer}, with: {gb: "{{ inputs.gb }}"}}
+    - name: ready
+      type: wait
+      with: {until: "{{ steps.make.phase == 'Succeeded' }}", pollInterval: 5ms, timeout: 60s}
+    - name: use
+      ref: {name: consumer}
+      needs: [ready]
+      with: {blob: "{{ steps.make.output.blob }}"}
+  output: {checksum: "{{ steps.use.output.checksum }}", numel: "{{ steps.use.output.numel }}"}
+---
+kind: Story
+metadata: {name: outer-big}
+spec:
+  steps:
+    - {name: approval, type: gate}
+    - name: sub
+      type: executeStory
+      needs: [approval]
+      with: {storyRef: inner-big, with: {gb: 4.0}}
+  output: {result: "{{ steps.sub.output.output }}"}
+"""
+        )
+        run = eng.submit_run("default/outer-big", {})
+        time.sleep(0.1)
+        eng.approve_gate(run.key, "approval")
+        run = eng.wait(run.key, timeout=180)
+        assert run.phase == Phase.SUCCEEDED, (run.error, {
+            k: (str(v.phase), str(v.error)) for k, v in run.step_states.items()
+        })
+        assert run.output["result"]["checksum"] == 1000.0
+        assert run.output["result"]["numel"] == int(4.0 * (1 << 30) // 2)

@@ -22,7 +22,10 @@
 #define D_HEAD 128
 #define QBLK 32
 #define KVBLK 64
-#define WG_QROWS 128  // 4 waves * QBLK
+#define NWAVES 8        // 8 waves (512 thr) => 2 waves/SIMD co-resident:
+                        // one wave's softmax VALU hides under its partner's
+                        // MFMA segment (guide §Two waves per SIMD)
+#define WG_QROWS 256    // NWAVES * QBLK
 
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 
@@ -36,27 +39,35 @@ __device__ __forceinline__ int k_lds_off(int kv, int byte_in_row) {
   return kv * 256 + (byte_in_row ^ ((kv & 15) << 4));
 }
 
-// V^T image: [d][kv] rows of 128 B, XOR byte bits 4..6 with d&7.
+// V^T image: [d][kv] rows of 128 B.  XOR byte bits 4..6 with (d>>1)&7:
+// combined with the row-parity bank bit (32*d mod 64) the b128 read bank is
+// injective in d mod 16, which is exactly the spread of a 16-lane
+// ds_read_b128 group — conflict-free PV reads (guide T2 derivation).
 __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
-  return d * 128 + (byte_in_row ^ ((d & 7) << 4));
+  return d * 128 + (byte_in_row ^ (((d >> 1) & 7) << 4));
 }
 
-__global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
+template <int VARIANT>  // ablation bitmask: 1=stage 2=qk+softmax 4=pv (7=full)
+__global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
     unsigned short* __restrict__ out,      // [B,S,Hq,D]
     const unsigned short* __restrict__ q,  // [B,S,Hq,D]
     const unsigned short* __restrict__ k,  // [B,S,Hkv,D]
     const unsigned short* __restrict__ v,  // [B,S,Hkv,D]
     int B, int Hq, int Hkv, int S, float scale, int causal) {
-  __shared__ __attribute__((aligned(16))) char smem[KVBLK * 256 + D_HEAD * 128];
-  char* k_lds = smem;                  // swizzled K tile [64][128] bf16
-  char* vt_lds = smem + KVBLK * 256;   // swizzled V^T tile [128][64] bf16
+  // two K+V^T buffer pairs; pointers computed per use (an addrspace(3)
+  // pointer array fails to compile as a static initializer)
+  __shared__ __attribute__((aligned(16))) char smem[2 * (KVBLK * 256 + D_HEAD * 128)];
+  const int BUFSTRIDE = KVBLK * 256 + D_HEAD * 128;
+  auto k_buf = [&](int i) -> char* { return smem + i * BUFSTRIDE; };
+  auto v_buf = [&](int i) -> char* { return smem + i * BUFSTRIDE + KVBLK * 256; };
 
   const int wg = blockIdx.x;
   const int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
-  const int bh = wg / nqblk;
-  // deepest q-blocks first: under causal masking the last q-block has the
-  // most KV tiles — schedule it first so the tail is short workgroups
-  const int qblk = nqblk - 1 - (wg % nqblk);
+  // global depth-descending order (LPT): ALL deepest q-blocks dispatch
+  // first, so under causal masking the stragglers are the short ones
+  const int nbh = gridDim.x / nqblk;
+  const int bh = wg % nbh;
+  const int qblk = nqblk - 1 - (wg / nbh);
   const int b = bh / Hq;
   const int hq = bh % Hq;
   const int hkv = hq / (Hq / Hkv);
@@ -73,10 +84,10 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   const long q_base = (long)b * S * q_sstride + (long)hq * D_HEAD;
   const long kv_base = (long)b * S * kv_sstride + (long)hkv * D_HEAD;
 
-  // wave-striped q assignment: wave w owns rows {qbase + 4*i + w}, i=0..31
+  // wave-striped q assignment: wave w owns rows {qbase + 8*i + w}, i=0..31
   // (causal kv ranges match across waves -> no idle compute waves)
   const int qbase = qblk * WG_QROWS;
-  const int my_q = qbase + 4 * l31 + wid;          // this lane's q row (S^T col)
+  const int my_q = qbase + NWAVES * l31 + wid;     // this lane's q row (S^T col)
 
   // ---- load Q fragments: B-operand layout, 8 slices of d (16 each) ----
   bf16x8 qf[8];
@@ -102,140 +113,210 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
   const int q_hi_wg = qbase + WG_QROWS - 1;  // max q row in WG
   int kv_end = S;
   if (causal) kv_end = min(S, q_hi_wg + 1);
-  const int my_q_hi = qbase + 4 * 31 + wid;  // this wave's max q row
+  const int my_q_hi = qbase + NWAVES * 31 + wid;  // this wave's max q row
 
-  for (int kv0 = 0; kv0 < kv_end; kv0 += KVBLK) {
-    // ---- cooperative staging ----
-    // K tile: 64 rows x 256 B; 256 threads x 4 chunks of 16 B
-    {
+  // ---- double-buffered pipeline (guide T14 split + 2 LDS buffers):
+  // per tile: write the pre-loaded NEXT tile into the other buffer, issue
+  // global loads for the tile after it, compute the CURRENT tile, one
+  // barrier. HBM latency hides under the MFMA phase.
+  const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
+
+  ushort8v kreg[2], vreg[2];
+  auto load_tile = [&](int tile) {
 #pragma unroll
-      for (int it = 0; it < 4; ++it) {
-        int chunk = it * 256 + tid;         // 1024 chunks of 16 B
-        int kv = chunk >> 4;                // 16 chunks per row
-        int byte = (chunk & 15) * 16;
-        int kvg = kv0 + kv;
-        ushort8v val;
-        if (kvg < S)
-          val = *reinterpret_cast<const ushort8v*>(k + kv_base + (long)kvg * kv_sstride + byte / 2);
-        else
-          val = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
-        *reinterpret_cast<ushort8v*>(k_lds + k_lds_off(kv, byte)) = val;
-      }
-      // V tile transposed: thread reads 8 contiguous d of one kv row,
-      // scatter-writes them to VT rows
-#pragma unroll
-      for (int it = 0; it < 4; ++it) {
-        int chunk = it * 256 + tid;
-        int kv = chunk >> 4;
-        int d0 = (chunk & 15) * 8;
-        int kvg = kv0 + kv;
-        ushort8v val;
-        if (kvg < S)
-          val = *reinterpret_cast<const ushort8v*>(v + kv_base + (long)kvg * kv_sstride + d0);
-        else
-          val = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<unsigned short*>(vt_lds + vt_lds_off(d0 + j, kv * 2)) = val[j];
-      }
+    for (int it = 0; it < 2; ++it) {
+      int chunk = it * 512 + tid;
+      int kv = chunk >> 4;
+      int byte = (chunk & 15) * 16;
+      int kvg = tile * KVBLK + kv;
+      if (kvg < S)
+        kreg[it] = *reinterpret_cast<const ushort8v*>(
+            k + kv_base + (long)kvg * kv_sstride + byte / 2);
+      else
+        kreg[it] = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
+      int d0 = (chunk & 15) * 8;
+      if (kvg < S)
+        vreg[it] = *reinterpret_cast<const ushort8v*>(
+            v + kv_base + (long)kvg * kv_sstride + d0);
+      else
+        vreg[it] = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
     }
+  };
+  auto write_tile = [&](char* kbuf, char* vbuf) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int chunk = it * 512 + tid;
+      int kv = chunk >> 4;
+      int byte = (chunk & 15) * 16;
+      *reinterpret_cast<ushort8v*>(kbuf + k_lds_off(kv, byte)) = kreg[it];
+      int d0 = (chunk & 15) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<unsigned short*>(vbuf + vt_lds_off(d0 + j, kv * 2)) =
+            vreg[it][j];
+    }
+  };
+
+  if (ntiles > 0) {
+    load_tile(0);
+    write_tile(k_buf(0), v_buf(0));
+    if (ntiles > 1) load_tile(1);
     __syncthreads();
+  }
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KVBLK;
+    const int cur = t & 1, nxt = cur ^ 1;
+    if (t + 1 < ntiles) {
+      write_tile(k_buf(nxt), v_buf(nxt));  // regs hold tile t+1
+      if (t + 2 < ntiles) load_tile(t + 2);  // issue early; lands next iter
+    }
+    char* kb = k_buf(cur);
+    char* vb = v_buf(cur);
 
     bool compute = (!causal) || (kv0 <= my_q_hi);
+    if constexpr ((VARIANT & 2) == 0) {
+      // stage-only ablation: touch both buffers so staging isn't DCE'd
+      if (compute) {
+        bf16x8 keep0 = *reinterpret_cast<const bf16x8*>(kb + k_lds_off(l31, 0));
+        bf16x8 keep1 = *reinterpret_cast<const bf16x8*>(vb + vt_lds_off(l31, 0));
+        asm volatile("" ::"v"(keep0), "v"(keep1));
+      }
+      __syncthreads();
+      continue;
+    }
     if (compute) {
       // ---- QK^T (swapped): S^T[kv][q] in two 32-kv tiles ----
       f32x16 st[2] = {};
+      __builtin_amdgcn_s_setprio(1);
+      // ss outer / tt inner: the two accumulator chains alternate, so the
+      // MFMA issue stream never stalls on its own dependent accumulator
+      // (back-to-back issue 32 cyc/SIMD vs ~2x that for a serial chain)
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
+      for (int ss = 0; ss < 8; ++ss) {
+        int byte = (ss * 16 + hi * 8) * 2;
 #pragma unroll
-        for (int s = 0; s < 8; ++s) {
-          int byte = (s * 16 + hi * 8) * 2;
-          int kv = t * 32 + l31;
-          bf16x8 kf = *reinterpret_cast<const bf16x8*>(k_lds + k_lds_off(kv, byte));
-          st[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], st[t], 0, 0, 0);
+        for (int tt = 0; tt < 2; ++tt) {
+          int kv = tt * 32 + l31;
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(kb + k_lds_off(kv, byte));
+          st[tt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ss], st[tt], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
 
-      // ---- mask + online softmax ----
+      // ---- mask + online softmax (log2 domain; v_exp_f32 IS exp2) ----
+      // Edge handling hoisted per tile: interior tiles skip the per-element
+      // mask (guide trap 4c: never a per-element runtime select).
+      const float sc2 = scale * 1.44269504f;  // fold log2(e) into the scale
+      const bool edge = (kv0 + KVBLK > S) || (causal && (kv0 + KVBLK - 1) > my_q);
       float m_tile = -1e30f;
+      if (edge) {
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
+        for (int tt = 0; tt < 2; ++tt)
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          int kvg = kv0 + t * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-          float sv = st[t][r] * scale;
-          if (kvg >= S || (causal && kvg > my_q)) sv = -1e30f;
-          st[t][r] = sv;
-          m_tile = fmaxf(m_tile, sv);
-        }
+          for (int r = 0; r < 16; ++r) {
+            int kvg = kv0 + tt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            float sv = st[tt][r] * sc2;
+            if (kvg >= S || (causal && kvg > my_q)) sv = -1e30f;
+            st[tt][r] = sv;
+            m_tile = fmaxf(m_tile, sv);
+          }
+      } else {
+#pragma unroll
+        for (int tt = 0; tt < 2; ++tt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            float sv = st[tt][r] * sc2;
+            st[tt][r] = sv;
+            m_tile = fmaxf(m_tile, sv);
+          }
       }
       m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, WAVE));
-      float m_new = fmaxf(m_run, m_tile);
-      // all-masked tile guard (fully OOB rows keep m_new = -1e30)
-      float alpha = (m_run <= -1e30f) ? 0.f : __expf(m_run - m_new);
-      if (m_new <= -1e30f) alpha = 1.f;
+
+      // defer-max (guide T13): when the tile max stays within THR2 of the
+      // running max, keep m_run and skip the O rescale entirely.  P is then
+      // bounded by 2^THR2; the f32 accumulator tolerates it.  Safe order:
+      // the decision precedes this tile's exponentiation (textbook form).
+      const float THR2 = 8.0f;
+      bool defer = __builtin_amdgcn_wave_reduce_and_b32(
+                       (m_run > -1e30f) && (m_tile - m_run <= THR2), 0) != 0;
+      float m_new;
+      if (defer) {
+        m_new = m_run;
+      } else {
+        m_new = fmaxf(m_run, m_tile);
+        float alpha = (m_run <= -1e30f) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_new);
+        if (m_new <= -1e30f) alpha = 1.f;
+        float alpha_row[16];
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+          alpha_row[r] = __shfl(alpha, qrow, WAVE);
+        }
+#pragma unroll
+        for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha_row[r];
+        l_run *= alpha;
+        m_run = m_new;
+      }
 
       float p_sum = 0.f;
 #pragma unroll
-      for (int t = 0; t < 2; ++t)
+      for (int tt = 0; tt < 2; ++tt)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float pv = (st[t][r] <= -1e30f) ? 0.f : __expf(st[t][r] - m_new);
-          st[t][r] = pv;
+          float pv = (st[tt][r] <= -1e30f) ? 0.f
+                                           : __builtin_amdgcn_exp2f(st[tt][r] - m_new);
+          st[tt][r] = pv;
           p_sum += pv;
         }
       p_sum += __shfl_xor(p_sum, 32, WAVE);
-      l_run = l_run * alpha + p_sum;
-      m_run = m_new;
-      // O rows are q-indexed by the C-layout REGISTER pattern, not by the
-      // lane (alpha lives at lane q = l&31): redistribute before rescaling
-      float alpha_row[16];
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-        alpha_row[r] = __shfl(alpha, qrow, WAVE);
-      }
-#pragma unroll
-      for (int dt = 0; dt < 4; ++dt)
-#pragma unroll
-        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha_row[r];
+      l_run += p_sum;
 
+      if constexpr ((VARIANT & 4) == 0) {
+        // no-PV ablation: keep the softmax results live
+        asm volatile("" ::"v"(p_sum), "v"(st[0][0]), "v"(st[1][15]),
+                     "v"(l_run), "v"(m_run));
+        __syncthreads();
+        continue;
+      }
       // ---- P -> bf16 A fragments via permlane32_swap ----
-      // per 32-kv tile: 8 packs -> 2 swaps x2 -> A slices (16 kv each)
-      bf16x8 pa[4];  // 4 slices of 16 kv covering the 64-kv tile
+      bf16x8 pa[4];
 #pragma unroll
-      for (int t = 0; t < 2; ++t) {
+      for (int tt = 0; tt < 2; ++tt) {
         unsigned int pk[8];
 #pragma unroll
-        for (int j = 0; j < 8; ++j) pk[j] = pack_bf16(st[t][2 * j], st[t][2 * j + 1]);
-        // slice 0 of this tile (kv t*32 + 0..15)
+        for (int j = 0; j < 8; ++j)
+          pk[j] = pack_bf16(st[tt][2 * j], st[tt][2 * j + 1]);
         {
           auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
           auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
           unsigned int w0 = r0[0], w2 = r0[1], w1 = r1[0], w3 = r1[1];
-          pa[t * 2] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
+          pa[tt * 2] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
         }
-        // slice 1 (kv t*32 + 16..31)
         {
           auto r0 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
           auto r1 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
           unsigned int w0 = r0[0], w2 = r0[1], w1 = r1[0], w3 = r1[1];
-          pa[t * 2 + 1] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
+          pa[tt * 2 + 1] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
         }
       }
 
       // ---- PV: O[q][d] += P @ V ----
+      __builtin_amdgcn_s_setprio(1);
+      // ss outer / dt inner: 4 independent O chains interleave (see QK^T)
 #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
+      for (int ss = 0; ss < 4; ++ss) {
+        int byte = (ss * 16 + hi * 8) * 2;
 #pragma unroll
-        for (int s = 0; s < 4; ++s) {
-          // B operand: V[kv = s*16 + hi*8 + jj][d = dt*32 + l31]
+        for (int dt = 0; dt < 4; ++dt) {
           int d = dt * 32 + l31;
-          int byte = (s * 16 + hi * 8) * 2;
-          bf16x8 vf = *reinterpret_cast<const bf16x8*>(vt_lds + vt_lds_off(d, byte));
-          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[s], vf, o_acc[dt], 0, 0, 0);
+          bf16x8 vf = *reinterpret_cast<const bf16x8*>(vb + vt_lds_off(d, byte));
+          o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ss], vf, o_acc[dt], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
     __syncthreads();
   }
@@ -251,7 +332,7 @@ __global__ __launch_bounds__(256, 2) void attn_prefill_kernel(
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
-    int qg = qbase + 4 * qrow + wid;
+    int qg = qbase + NWAVES * qrow + wid;
     if (qg >= S) continue;
     unsigned short* orow = out + q_base + (long)qg * q_sstride;
 #pragma unroll
@@ -265,11 +346,28 @@ extern "C" void launch_attn_prefill(void* out, const void* q, const void* k,
                                     int S, float scale, int causal,
                                     hipStream_t stream) {
   int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
-  dim3 grid(B * Hq * nqblk), block(256);
-  hipLaunchKernelGGL(attn_prefill_kernel, grid, block, 0, stream,
+  dim3 grid(B * Hq * nqblk), block(512);
+  hipLaunchKernelGGL((attn_prefill_kernel<7>), grid, block, 0, stream,
                      (unsigned short*)out, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v, B, Hq,
                      Hkv, S, scale, causal);
+}
+
+extern "C" void launch_attn_prefill_variant(int variant, void* out,
+                                            const void* q, const void* k,
+                                            const void* v, int B, int Hq,
+                                            int Hkv, int S, float scale,
+                                            int causal, hipStream_t stream) {
+  int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
+  dim3 grid(B * Hq * nqblk), block(512);
+  auto args = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)out,
+                       (const unsigned short*)q, (const unsigned short*)k,
+                       (const unsigned short*)v, B, Hq, Hkv, S, scale, causal);
+  };
+  if (variant == 1) args(attn_prefill_kernel<1>);
+  else if (variant == 3) args(attn_prefill_kernel<3>);
+  else args(attn_prefill_kernel<7>);
 }
 
 // ---------------------------------------------------------------------------

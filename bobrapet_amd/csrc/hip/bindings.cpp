@@ -22,6 +22,9 @@ void launch_attn_prefill(void*, const void*, const void*, const void*, int,
 void launch_attn_decode(void*, const void*, const void*, const void*, int, int,
                         int, int, int, float, hipStream_t);
 void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
+void launch_attn_prefill_variant(int, void*, const void*, const void*,
+                                 const void*, int, int, int, int, float, int,
+                                 hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
 }
@@ -154,6 +157,18 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return out;
 }
 
+torch::Tensor attn_prefill_variant(int variant, torch::Tensor q,
+                                   torch::Tensor k, torch::Tensor v,
+                                   double scale, bool causal) {
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2);
+  const int Hkv = k.size(2);
+  auto out = torch::zeros_like(q);
+  launch_attn_prefill_variant(variant, out.data_ptr(), q.data_ptr(),
+                              k.data_ptr(), v.data_ptr(), B, Hq, Hkv, S,
+                              (float)scale, causal ? 1 : 0, cur_stream());
+  return out;
+}
+
 torch::Tensor dbg_mfma(torch::Tensor A, torch::Tensor B) {
   auto C = torch::zeros({32, 32}, A.options());
   launch_dbg_mfma(C.data_ptr(), A.data_ptr(), B.data_ptr(), cur_stream());
@@ -183,5 +198,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
   m.def("dbg_mfma", &dbg_mfma, "layout probe: C=A@B one mfma");
+  m.def("attn_prefill_variant", &attn_prefill_variant,
+        "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("dbg_attn_core", &dbg_attn_core, "layout probe: QK^T + pack + PV");
 }

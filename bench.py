@@ -290,8 +290,12 @@ def main() -> int:
             f"(b{EMBED_BATCH} s{EMBED_SEQ} d{EMBED_DIM}), all-gather join"
         )
 
+    device_ids = None
+    if multi and has_gpu:
+        device_ids = [torch.cuda.current_device()]  # one GPU per rank
     eng = RunEngine(
-        EngineConfig(cpu_workers=4, workers_per_device=4, child_ttl_seconds=5.0)
+        EngineConfig(cpu_workers=4, workers_per_device=4, child_ttl_seconds=5.0),
+        device_ids=device_ids,
     ).start()
     native = None
     try:

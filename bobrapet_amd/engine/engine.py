@@ -58,6 +58,7 @@ class RunEngine:
         config: _t.Optional[EngineConfig] = None,
         storage: _t.Optional[StorageManager] = None,
         device_count: _t.Optional[int] = None,
+        device_ids: _t.Optional[_t.List[int]] = None,
         metrics=None,
         tracer=None,
     ):
@@ -81,6 +82,7 @@ class RunEngine:
         self.tracer = tracer if tracer is not None else tracing_mod.Tracer(enabled=False)
         self.workers = WorkerPool(
             device_count=device_count,
+            device_ids=device_ids,
             workers_per_device=self.config.workers_per_device,
             cpu_workers=self.config.cpu_workers,
         )

@@ -187,10 +187,10 @@ class StepExecutor:
         the DAG scheduler places each StepRun on one of the GPUs)."""
         if cfg.placement_gpu is not None:
             return cfg.placement_gpu
-        n = self.engine.workers.device_count
-        if n <= 0:
+        pool = self.engine.workers
+        if pool.device_count <= 0:
             return None
-        allowed = cfg.placement_gpus or list(range(n))
+        allowed = cfg.placement_gpus or list(pool.device_ids)
         # stable hash spread of (run, step) over allowed devices
         h = hash((run.name, step.name)) & 0x7FFFFFFF
         return allowed[h % len(allowed)]

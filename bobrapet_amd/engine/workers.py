@@ -52,22 +52,31 @@ class WorkerPool:
     Device workers pin a torch CUDA stream each, so concurrent steps on one
     GPU overlap via multiple HIP streams."""
 
-    def __init__(self, device_count: _t.Optional[int] = None, workers_per_device: int = 2, cpu_workers: int = 4):
-        if device_count is None:
-            device_count = 0
-            try:
-                import torch
-
-                if torch.cuda.is_available():
-                    device_count = torch.cuda.device_count()
-            except Exception:
+    def __init__(
+        self,
+        device_count: _t.Optional[int] = None,
+        workers_per_device: int = 2,
+        cpu_workers: int = 4,
+        device_ids: _t.Optional[_t.List[int]] = None,
+    ):
+        if device_ids is None:
+            if device_count is None:
                 device_count = 0
-        self.device_count = device_count
-        self._device_queues: _t.List[queue.Queue] = []
+                try:
+                    import torch
+
+                    if torch.cuda.is_available():
+                        device_count = torch.cuda.device_count()
+                except Exception:
+                    device_count = 0
+            device_ids = list(range(device_count))
+        self.device_ids = list(device_ids)
+        self.device_count = len(self.device_ids)
+        self._device_queues: _t.Dict[int, queue.Queue] = {}
         self._slots: _t.List[_Slot] = []
-        for dev in range(device_count):
+        for dev in self.device_ids:
             q: queue.Queue = queue.Queue()
-            self._device_queues.append(q)
+            self._device_queues[dev] = q
             for i in range(workers_per_device):
                 slot = _Slot(f"gpu{dev}-w{i}", dev, q)
                 slot.start()
@@ -79,13 +88,17 @@ class WorkerPool:
             self._slots.append(slot)
 
     def submit(self, fn: _t.Callable[[_Slot], None], device: _t.Optional[int] = None) -> None:
-        if device is not None and 0 <= device < len(self._device_queues):
+        if device is not None and device in self._device_queues:
             self._device_queues[device].put(fn)
+        elif self._device_queues and device is not None:
+            # unknown device index: route to a deterministic owned device
+            dev = self.device_ids[device % len(self.device_ids)]
+            self._device_queues[dev].put(fn)
         else:
             self._cpu_queue.put(fn)
 
     def shutdown(self) -> None:
-        for q in self._device_queues:
+        for q in self._device_queues.values():
             for _ in range(8):
                 q.put(None)
         for _ in range(32):

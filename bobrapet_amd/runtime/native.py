@@ -342,9 +342,9 @@ class NativeRunner:
         impl_name = template.implementation if template is not None else engram_key.split("/")[-1]
         config = engram.with_
         device = None
-        n = self.workers.device_count
-        if n > 0:
-            device = (hash((run_id, step, branch)) & 0x7FFFFFFF) % n
+        if self.workers.device_count > 0:
+            ids = self.workers.device_ids
+            device = ids[(hash((run_id, step, branch)) & 0x7FFFFFFF) % len(ids)]
 
         def body(slot):
             try:

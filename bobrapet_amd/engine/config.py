@@ -81,7 +81,50 @@ class EngineConfig:
         for k, v in kwargs.items():
             if not hasattr(self, k):
                 raise KeyError(f"unknown config key {k!r}")
+            if k == "queues" and isinstance(v, dict):
+                v = {
+                    name: q if isinstance(q, QueueConfig) else QueueConfig(**q)
+                    for name, q in v.items()
+                }
             setattr(self, k, v)
+
+    def load_file(self, path: str) -> None:
+        """Apply a YAML/JSON config file (the ConfigMap role)."""
+        import yaml as _yaml
+
+        with open(path, "r", encoding="utf-8") as fh:
+            data = _yaml.safe_load(fh) or {}
+        self.update(**data)
+
+    def watch_file(self, path: str, interval: float = 2.0):
+        """Hot-reload the file when its mtime changes (reference: the
+        operator-config reconciler watching its ConfigMap).  Returns a
+        stop() callable."""
+        import os
+        import threading
+
+        stop = threading.Event()
+        state = {"mtime": None}
+
+        def loop():
+            while not stop.wait(interval):
+                try:
+                    m = os.path.getmtime(path)
+                except OSError:
+                    continue
+                if state["mtime"] is None:
+                    state["mtime"] = m
+                    continue
+                if m != state["mtime"]:
+                    state["mtime"] = m
+                    try:
+                        self.load_file(path)
+                    except Exception:
+                        pass  # bad config keeps the previous values
+
+        t = threading.Thread(target=loop, daemon=True, name="config-watch")
+        t.start()
+        return stop.set
 
 
 @dataclass

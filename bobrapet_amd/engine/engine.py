@@ -196,6 +196,7 @@ class RunEngine:
         recursion_depth: int = 0,
         trigger_token: _t.Optional[str] = None,
         labels: _t.Optional[dict] = None,
+        _trusted: bool = False,
     ) -> StoryRun:
         if isinstance(story, str):
             ns, _, nm = story.rpartition("/")
@@ -205,7 +206,7 @@ class RunEngine:
             name = f"{story.name}-{monotonic_now():.6f}".replace(".", "-")[:63]
 
         # guards (reference: storyrun_controller.go:981-1045)
-        inputs = self._prepare_inputs(story, inputs)
+        inputs = self._prepare_inputs(story, inputs, trusted=_trusted)
 
         queue_name = self.config.default_queue
         priority = 0
@@ -230,17 +231,28 @@ class RunEngine:
         )
         if trigger_token:
             run.trigger_tokens.append(trigger_token)
+        # schema pins (reference: ensureStoryRunSchemaRefs / bubu:// refs)
+        if story.inputs_schema is not None:
+            run.annotations["schema/inputs"] = f"bubu://story/{story.namespace}/{story.name}/inputs@g{story.generation}"
+        if story.outputs_schema is not None:
+            run.annotations["schema/outputs"] = f"bubu://story/{story.namespace}/{story.name}/outputs@g{story.generation}"
         self.store.create_story_run(run)
         self._run_done.setdefault(run.key, threading.Event())
         self.metrics.inc("storyruns_total", phase="submitted")
         self._post(("admit", run.key))
         return run
 
-    def _prepare_inputs(self, story: T.Story, inputs):
+    def _prepare_inputs(self, story: T.Story, inputs, trusted: bool = False):
         """Oversized-input offload + schema defaults + validation
-        (reference: storyrun_controller.go:981-1045, pkg/runs/inputs)."""
+        (reference: storyrun_controller.go:981-1045, pkg/runs/inputs).
+        User-submitted inputs may not carry $storageRef markers (spoofing
+        rejection — reference: storyrun_webhook.go:389-423)."""
         if inputs is None:
             inputs = {}
+        if not trusted and self.storage.contains_refs(inputs):
+            # engine-created refs arrive only via trusted submitters
+            # (executor sub-stories); user-facing paths reject them
+            raise ValueError("storyrun inputs may not contain $storageRef values")
         if story.inputs_schema is not None:
             inputs = apply_defaults(inputs, story.inputs_schema)
             errs = validate_instance(inputs, story.inputs_schema)
@@ -413,6 +425,20 @@ class RunEngine:
 
     def stream_of(self, run_key: str):
         return self._streams.get(run_key)
+
+    # ------------------------------------------------------------------
+    # checkpoint/resume (reference: "state IS the checkpoint" SURVEY §5.4)
+    # ------------------------------------------------------------------
+
+    def save_state(self, path: str) -> None:
+        from . import snapshot
+
+        snapshot.save_state(self, path)
+
+    def load_state(self, path: str) -> int:
+        from . import snapshot
+
+        return snapshot.load_state(self, path)
 
     # ------------------------------------------------------------------
     # scope building (reference: getPriorStepOutputs dag.go:2083-2597)

@@ -398,6 +398,7 @@ class StepExecutor:
                 parent_run=run.name,
                 parent_step=step.name,
                 recursion_depth=run.recursion_depth + 1,
+                _trusted=True,  # engine-resolved inputs may carry $storageRef
             )
         state.output = {"storyRun": f"{run.namespace}/{child_name}"}
         if wait:

@@ -132,6 +132,56 @@ class ResourceRegistry:
                         break
             return count
 
+    def transport_status(self, name: str, engine=None) -> dict:
+        """Aggregate live binding state for one Transport (reference:
+        pkg/transport/capabilities_aggregation.go:47-172)."""
+        transport = self.transport(name)
+        bindings = []
+        if engine is not None:
+            for stream in list(getattr(engine, "_streams", {}).values()):
+                for b in getattr(stream, "bindings", {}).values():
+                    if b.transport in (name, ""):
+                        bindings.append(b)
+        codecs = sorted({c for b in bindings for c in b.codecs}) or list(transport.codecs)
+        return {
+            "driver": transport.driver,
+            "availableCodecs": codecs,
+            "bindings": {
+                "total": len(bindings),
+                "ready": sum(1 for b in bindings if b.phase == "Ready"),
+                "pending": sum(1 for b in bindings if b.phase == "Pending"),
+            },
+        }
+
+    def story_status(self, story_key: str) -> dict:
+        """Definition status (reference: story_controller.go ValidationStatus
+        + usageCount)."""
+        from ..specs import validation as V
+
+        with self._lock:
+            story = self.stories.get(story_key)
+        if story is None:
+            raise NotFound(f"story {story_key}")
+        res = V.validate_story(story)
+        return {
+            "validationStatus": "valid" if res.ok else "invalid",
+            "validationErrors": res.errors,
+            "validationWarnings": res.warnings,
+            "usageCount": self.story_usage_count(story_key),
+            "stepsTotal": len(story.steps),
+            "generation": story.generation,
+        }
+
+    def engram_status(self, engram_key: str) -> dict:
+        with self._lock:
+            engram = self.engrams.get(engram_key)
+        if engram is None:
+            raise NotFound(f"engram {engram_key}")
+        return {
+            "usageCount": self.engram_usage_count(engram_key),
+            "generation": engram.generation,
+        }
+
     def allows_cross_namespace(
         self, from_kind: str, from_ns: str, to_kind: str, to_ns: str, to_name: str = ""
     ) -> bool:

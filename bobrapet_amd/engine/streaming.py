@@ -33,6 +33,25 @@ if _t.TYPE_CHECKING:
 
 
 @dataclass
+class TransportBinding:
+    """Per-(storyRun, step) negotiated connector binding (reference:
+    api/transport/v1alpha1/transportbinding_types.go:108-199): here the
+    negotiation is in-process — driver/codec selection + ring endpoints."""
+
+    name: str
+    story_run: str
+    step: str
+    transport: str = ""
+    driver: str = "inproc"
+    codecs: _t.List[str] = field(default_factory=lambda: ["tensor", "json"])
+    endpoint: str = ""
+    upstream: _t.List[str] = field(default_factory=list)
+    downstream: _t.List[str] = field(default_factory=list)
+    phase: str = "Ready"
+    heartbeat: float = field(default_factory=monotonic_now)
+
+
+@dataclass
 class StageStats:
     packets_in: int = 0
     packets_out: int = 0
@@ -265,6 +284,28 @@ class StreamingRun:
             st = run.step_state(name)
             st.phase = Phase.RUNNING
             st.started_at = monotonic_now()
+        # negotiated bindings (reference: ensureRunTransportBinding
+        # steprun_controller.go:3701; driver from the story's transport)
+        driver = "inproc"
+        transport_name = ""
+        if story.transports:
+            transport_name = story.transports[0].transport_ref or story.transports[0].name
+            try:
+                driver = engine.registry.transport(transport_name).driver
+            except KeyError:
+                pass
+        self.bindings: _t.Dict[str, TransportBinding] = {}
+        for name in self.topo.stages:
+            self.bindings[name] = TransportBinding(
+                name=f"{run.name}-{name}",
+                story_run=run.name,
+                step=name,
+                transport=transport_name,
+                driver=driver,
+                endpoint=f"ring://{run.name}/{name}",
+                upstream=[e.src for e in self.topo.upstream_of(name) if e.src],
+                downstream=[e.dst for e in self.topo.downstream_of(name)],
+            )
         run.phase = Phase.RUNNING
         run.started_at = run.started_at or monotonic_now()
         for s in self.stages:

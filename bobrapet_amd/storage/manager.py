@@ -20,6 +20,8 @@ import uuid
 from .stores import BlobNotFound, FileStore, MemStore, Store, StoreError, TensorStore
 
 STORAGE_REF_KEY = "$storageRef"
+ENV_REF_KEY = "$envRef"
+FILE_REF_KEY = "$fileRef"
 DEFAULT_MAX_INLINE = 8 << 10  # bytes of JSON before a value offloads
 MAX_WALK_DEPTH = 64
 
@@ -129,6 +131,23 @@ class StorageManager:
         if isinstance(value, dict):
             if STORAGE_REF_KEY in value:
                 return self.resolve_ref(value)
+            if ENV_REF_KEY in value:
+                # environment indirection (role of the reference's
+                # $configMapRef — kube_refs.go:100-167)
+                import os
+
+                ref = value[ENV_REF_KEY]
+                name = ref.get("name") if isinstance(ref, dict) else str(ref)
+                return os.environ.get(name, ref.get("default") if isinstance(ref, dict) else None)
+            if FILE_REF_KEY in value:
+                ref = value[FILE_REF_KEY]
+                path = ref.get("path") if isinstance(ref, dict) else str(ref)
+                self.validate_ref_key(path)
+                with open(path, "r", encoding="utf-8") as fh:
+                    text = fh.read()
+                if isinstance(ref, dict) and ref.get("json"):
+                    return json.loads(text)
+                return text
             return {k: self.hydrate(v, _depth=_depth + 1) for k, v in value.items()}
         if isinstance(value, list):
             return [self.hydrate(v, _depth=_depth + 1) for v in value]

@@ -1,0 +1,138 @@
+"""Checkpoint/resume: run-state snapshot save + restore into a fresh engine
+(reference durability model — SURVEY §5.4: state IS the checkpoint)."""
+import time
+
+import pytest
+
+from bobrapet_amd.engine import EngineConfig, RunEngine
+from bobrapet_amd.enums import Phase
+
+RESOURCES = """
+kind: EngramTemplate
+metadata: {name: echo-tpl}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: echoer}
+spec: {templateRef: {name: echo-tpl}}
+---
+kind: Story
+metadata: {name: snap}
+spec:
+  steps:
+    - {name: first, ref: {name: echoer}, with: {v: "{{ inputs.x }}"}}
+    - {name: approval, type: gate, needs: [first]}
+    - {name: last, ref: {name: echoer}, needs: [approval], with: {v: "{{ steps.first.output.v + 1 }}"}}
+  output: {v: "{{ steps.last.output.v }}"}
+"""
+
+
+def test_snapshot_roundtrip_resumes_run(tmp_path):
+    path = str(tmp_path / "state.json")
+    eng1 = RunEngine(EngineConfig(cpu_workers=2)).start()
+    try:
+        eng1.apply_yaml(RESOURCES)
+        run = eng1.submit_run("default/snap", {"x": 41})
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            r = eng1.store.get_story_run(run.key)
+            if r.step_states.get("approval") and r.step_states["approval"].phase == Phase.PAUSED:
+                break
+            time.sleep(0.01)
+        # also a finished run in the store
+        done = eng1.run_story("default/snap", {"x": 1}, timeout=5)  # waits at gate
+        eng1.approve_gate(done.key, "approval")
+        done = eng1.wait(done.key, timeout=5)
+        assert done.phase == Phase.SUCCEEDED
+        eng1.save_state(path)
+    finally:
+        eng1.stop()
+
+    # fresh engine: restore, approve the gate, run completes with the
+    # already-finished step preserved
+    eng2 = RunEngine(EngineConfig(cpu_workers=2)).start()
+    try:
+        eng2.apply_yaml(RESOURCES)
+        n = eng2.load_state(path)
+        assert n == 2
+        r = eng2.store.get_story_run(run.key)
+        assert r.step_states["first"].phase == Phase.SUCCEEDED
+        assert r.step_states["first"].output == {"v": 41}
+        eng2.approve_gate(run.key, "approval")
+        final = eng2.wait(run.key, timeout=10)
+        assert final.phase == Phase.SUCCEEDED
+        assert final.output == {"v": 42}
+        # the terminal run restored terminal
+        done2 = eng2.store.get_story_run(done.key)
+        assert done2.phase == Phase.SUCCEEDED
+    finally:
+        eng2.stop()
+
+
+def test_in_flight_engram_step_reexecutes(tmp_path):
+    path = str(tmp_path / "state.json")
+    eng1 = RunEngine(EngineConfig(cpu_workers=2)).start()
+    try:
+        eng1.apply_yaml(RESOURCES)
+        eng1.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: sleepy-tpl}
+spec: {builtin: sleepy}
+---
+kind: Engram
+metadata: {name: sleeper}
+spec: {templateRef: {name: sleepy-tpl}}
+---
+kind: Story
+metadata: {name: snap-slow}
+spec:
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 30}}
+"""
+        )
+        run = eng1.submit_run("default/snap-slow", {})
+        time.sleep(0.1)  # step running
+        eng1.save_state(path)
+    finally:
+        eng1.stop()
+
+    eng2 = RunEngine(EngineConfig(cpu_workers=2)).start()
+    try:
+        eng2.apply_yaml(RESOURCES)
+        eng2.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: sleepy-tpl}
+spec: {builtin: sleepy}
+---
+kind: Engram
+metadata: {name: sleeper}
+spec: {templateRef: {name: sleepy-tpl}}
+---
+kind: Story
+metadata: {name: snap-slow}
+spec:
+  steps:
+    - {name: slow, ref: {name: sleeper}, with: {seconds: 0.01}}
+"""
+        )
+        eng2.load_state(path)
+        # the in-flight step was reset to Pending and re-executes
+        final = eng2.wait(run.key, timeout=10)
+        assert final.phase == Phase.SUCCEEDED
+    finally:
+        eng2.stop()
+
+
+def test_spoofed_storage_ref_rejected():
+    eng = RunEngine(EngineConfig(cpu_workers=1)).start()
+    try:
+        eng.apply_yaml(RESOURCES)
+        with pytest.raises(ValueError, match="storageRef"):
+            eng.submit_run(
+                "default/snap",
+                {"x": 1, "sneaky": {"$storageRef": {"key": "outputs/x", "kind": "json"}}},
+            )
+    finally:
+        eng.stop()

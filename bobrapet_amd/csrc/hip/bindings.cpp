@@ -25,6 +25,8 @@ void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
 void launch_attn_prefill_variant(int, void*, const void*, const void*,
                                  const void*, int, int, int, int, float, int,
                                  hipStream_t);
+void launch_gemm_bf16_nt(void*, const void*, const void*, int, int, int,
+                         hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
 }
@@ -157,6 +159,20 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   return out;
 }
 
+torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
+  // a: [M,K] bf16, b: [N,K] bf16 (transposed-weight layout) -> [M,N]
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "gemm_nt: K mismatch");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 64 == 0,
+              "gemm_nt: M,N must be multiples of 128 and K of 64");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm_bf16_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(), M, N, K,
+                      cur_stream());
+  return c;
+}
+
 torch::Tensor attn_prefill_variant(int variant, torch::Tensor q,
                                    torch::Tensor k, torch::Tensor v,
                                    double scale, bool causal) {
@@ -200,5 +216,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dbg_mfma", &dbg_mfma, "layout probe: C=A@B one mfma");
   m.def("attn_prefill_variant", &attn_prefill_variant,
         "ablation: 1=stage 3=+qk/softmax 7=full");
+  m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
   m.def("dbg_attn_core", &dbg_attn_core, "layout probe: QK^T + pack + PV");
 }

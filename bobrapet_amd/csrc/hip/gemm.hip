@@ -1,0 +1,147 @@
+// bf16 MFMA GEMM for MI355X (gfx950): C[M,N] = A[M,K] @ B[N,K]^T.
+//
+// NT layout (B stored row-major as [N][K], the nn.Linear weight
+// convention) so BOTH operands stage through LDS with the same
+// lane-linear global_load_lds image — glds cannot transpose (guide §5
+// "glds K-tile trap"), so the transposed-B read problem is solved in the
+// weight layout instead.
+//
+// Structure (guide §5 ladder step 3 + T1/T2/T3 minimum 2-phase):
+//  - 128x128 tile, 4 waves (2x2), per-wave 64x64 = 4x4 fragments of
+//    v_mfma_f32_16x16x32_bf16; BK = 64 (two k-subtiles of 32)
+//  - global->LDS staging via __builtin_amdgcn_global_load_lds width 16,
+//    double-buffered; the LDS image is lane-linear so the bank-conflict
+//    XOR swizzle lives on the per-lane SOURCE address and the matching
+//    read offset (guide rule 21)
+//  - XCD-aware bijective blockIdx swizzle (T1) for L2 locality
+#include "common.h"
+
+#define GM_BM 128
+#define GM_BN 128
+#define GM_BK 64
+
+typedef float f32x4g __attribute__((ext_vector_type(4)));
+
+// A/B LDS image: [128 rows][64 k] bf16 = 128-byte rows, 16 KB per tile.
+// Swizzle: byte-in-row ^= ((row>>1)&7)<<4 — combined with the row-parity
+// bank bit this makes the 16-lane ds_read_b128 fragment reads (rows
+// distinct mod 16 within a group) conflict-free.
+__device__ __forceinline__ int gm_swz(int row, int byte_in_row) {
+  return row * 128 + (byte_in_row ^ (((row >> 1) & 7) << 4));
+}
+
+__global__ __launch_bounds__(256, 2) void gemm_bf16_nt_kernel(
+    unsigned short* __restrict__ C,        // [M][N] bf16
+    const unsigned short* __restrict__ A,  // [M][K] bf16
+    const unsigned short* __restrict__ B,  // [N][K] bf16
+    int M, int N, int K) {
+  __shared__ __attribute__((aligned(16))) char smem[4 * GM_BM * GM_BK * 2];
+  // buffers: [buf][A|B] each 16 KB
+  auto tile_ptr = [&](int buf, int which) -> char* {
+    return smem + (buf * 2 + which) * (GM_BM * GM_BK * 2);
+  };
+
+  // XCD-aware bijective remap (guide T1)
+  const int nbm = M / GM_BM, nbn = N / GM_BN;
+  unsigned int wgid = xcd_swizzle(blockIdx.x, nbm * nbn);
+  const int bm = (int)(wgid / nbn) * GM_BM;
+  const int bn = (int)(wgid % nbn) * GM_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;           // 4 waves as 2x2
+  const int wr = (wid >> 1) * 64;     // wave row offset in the tile
+  const int wc = (wid & 1) * 64;      // wave col offset
+  const int l15 = lane & 15;
+  const int lg = lane >> 4;           // 0..3 fragment k-group
+
+  // ---- staging: each wave issues 4 glds per operand per tile; one glds
+  // writes 1 KB of LDS (64 lanes x 16 B) lane-linearly at a wave-uniform
+  // base.  chunk = (wid*4 + it)*64 + lane; row = chunk/8; the source
+  // byte offset applies the inverse swizzle (rule 21).
+  auto stage = [&](int buf, int kt) {
+    const long k0 = (long)kt * GM_BK;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int piece = wid * 4 + it;           // 0..15, wave-uniform
+      int chunk = piece * 64 + lane;      // 16-B chunk index
+      int row = chunk >> 3;               // 8 chunks per 128-B row
+      int bq = (chunk & 7) * 16;          // byte-in-row of the LDS image
+      int src_byte = bq ^ (((row >> 1) & 7) << 4);
+      {  // A tile: rows bm+row
+        const unsigned short* src = A + (long)(bm + row) * K + k0 + src_byte / 2;
+        char* dst = tile_ptr(buf, 0) + piece * 1024;
+        __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)src,
+                                         (__attribute__((address_space(3))) unsigned int*)dst,
+                                         16, 0, 0);
+      }
+      {  // B tile: rows bn+row
+        const unsigned short* src = B + (long)(bn + row) * K + k0 + src_byte / 2;
+        char* dst = tile_ptr(buf, 1) + piece * 1024;
+        __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) unsigned int*)src,
+                                         (__attribute__((address_space(3))) unsigned int*)dst,
+                                         16, 0, 0);
+      }
+    }
+  };
+
+  f32x4g acc[4][4] = {};
+  const int nkt = K / GM_BK;
+
+  stage(0, 0);
+  // drain the prologue stage and enter the loop (simple 2-phase: vmcnt(0)
+  // + plain barrier per guide T3-minimum; the barrier itself re-waits)
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  for (int kt = 0; kt < nkt; ++kt) {
+    const int cur = kt & 1, nxt = cur ^ 1;
+    if (kt + 1 < nkt) stage(nxt, kt + 1);
+
+    const char* a_lds = tile_ptr(cur, 0);
+    const char* b_lds = tile_ptr(cur, 1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 af[4], bf[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        int arow = wr + f * 16 + l15;
+        int byte = (ks * 32 + lg * 8) * 2;
+        af[f] = *reinterpret_cast<const bf16x8*>(a_lds + gm_swz(arow, byte));
+        int brow = wc + f * 16 + l15;
+        bf[f] = *reinterpret_cast<const bf16x8*>(b_lds + gm_swz(brow, byte));
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[j],
+                                                              acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C[i=row][j=col] 16x16 layout: col = l15, row = lg*4+r
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = bm + wr + i * 16 + lg * 4 + r;
+        int col = bn + wc + j * 16 + l15;
+        C[(long)row * N + col] = f2bf(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+extern "C" void launch_gemm_bf16_nt(void* C, const void* A, const void* B,
+                                    int M, int N, int K, hipStream_t stream) {
+  dim3 grid((M / GM_BM) * (N / GM_BN)), block(256);
+  hipLaunchKernelGGL(gemm_bf16_nt_kernel, grid, block, 0, stream,
+                     (unsigned short*)C, (const unsigned short*)A,
+                     (const unsigned short*)B, M, N, K);
+}

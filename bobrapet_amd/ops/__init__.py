@@ -188,6 +188,20 @@ def attn_prefill(q, k, v, scale: _t.Optional[float] = None, causal: bool = True)
     return out.transpose(1, 2).contiguous()
 
 
+def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """C = a @ b.T with b stored [N, K] (nn.Linear weight layout).
+
+    GPU: the hand-written MFMA kernel (csrc/hip/gemm.hip) for aligned
+    shapes, else hipBLASLt via torch.matmul; CPU: fp32 reference."""
+    if a.is_cuda:
+        M, K = a.shape
+        N = b.shape[0]
+        if M % 128 == 0 and N % 128 == 0 and K % 64 == 0:
+            return _require_ext().gemm_nt(a.contiguous(), b.contiguous())
+        return torch.matmul(a, b.t())
+    return (a.float() @ b.float().t()).to(a.dtype)
+
+
 def attn_decode(q, kc, vc, L: int, scale: _t.Optional[float] = None):
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:

@@ -123,6 +123,28 @@ class TestAttentionPrefill:
         assert _mae(got, ref) < 0.03
 
 
+class TestGemm:
+    @pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 384, 512), (512, 256, 4096)])
+    def test_vs_torch(self, m, n, k):
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        got = ops.gemm_nt(a, b)
+        ref = torch.matmul(a.float(), b.float().t())
+        err = (got.float() - ref).abs()
+        rel = err.max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
+
+    def test_asymmetric_catches_transpose(self):
+        # asymmetric ramp operands (guide G9): a transposed C would fail
+        a = (torch.arange(128 * 64, device="cuda", dtype=torch.float32)
+             .reshape(128, 64) % 7).to(torch.bfloat16) / 7
+        b = (torch.arange(128 * 64, device="cuda", dtype=torch.float32)
+             .reshape(128, 64) % 5).to(torch.bfloat16) / 5
+        got = ops.gemm_nt(a, b)
+        ref = torch.matmul(a.float(), b.float().t())
+        assert (got.float() - ref).abs().max().item() < 0.1
+
+
 class TestAttentionDecode:
     @pytest.mark.parametrize("b,hq,hkv,l", [(1, 1, 1, 64), (2, 32, 8, 500), (4, 8, 8, 1024)])
     def test_vs_ref(self, b, hq, hkv, l):

@@ -1,0 +1,124 @@
+"""Direct unit tests for the builtin engram library (CPU paths) + the CLI."""
+import pytest
+
+from bobrapet_amd.engrams.base import EngramContext, EngramFailure
+from bobrapet_amd.engrams import registry
+
+
+def _ctx(**kw):
+    return EngramContext(**kw)
+
+
+class TestFilterTransform:
+    def test_filter_where(self):
+        impl = registry.resolve("filter")
+        out = impl.run(
+            _ctx(
+                input={"items": [{"s": 1}, {"s": 5}, {"s": 3}]},
+                runtime={"where": "{{ item.s >= 3 }}"},
+            )
+        ).output
+        assert out == {"items": [{"s": 5}, {"s": 3}], "count": 2}
+
+    def test_transform_map(self):
+        impl = registry.resolve("transform")
+        out = impl.run(
+            _ctx(
+                input={"items": [{"x": 2}, {"x": 3}]},
+                runtime={"map": {"y": "{{ item.x * item.x }}", "i": "{{ index }}"}},
+            )
+        ).output
+        assert out["items"] == [{"y": 4, "i": 0}, {"y": 9, "i": 1}]
+
+    def test_filter_requires_list(self):
+        impl = registry.resolve("filter")
+        with pytest.raises(EngramFailure):
+            impl.run(_ctx(input={"items": "nope"}))
+
+
+class TestEchoFail:
+    def test_echo_merges_config(self):
+        impl = registry.resolve("echo")
+        out = impl.run(_ctx(input={"a": 1}, config={"b": 2, "a": 0})).output
+        assert out == {"a": 1, "b": 2}
+
+    def test_fail_succeed_after(self):
+        impl = registry.resolve("fail")
+        ctx = _ctx(input={"succeedAfter": 1}, story_run="r", step_name="s")
+        with pytest.raises(EngramFailure):
+            impl.run(ctx)
+        assert impl.run(ctx).output == {"attempts": 2}
+
+    def test_unknown_engram(self):
+        with pytest.raises(registry.UnknownEngram):
+            registry.resolve("no-such-engram")
+
+
+class TestEngramContextContract:
+    def test_signals_are_seq_ordered(self):
+        ctx = _ctx()
+        ctx.emit_signal("a", 1)
+        ctx.emit_signal("b", 2)
+        assert [s.seq for s in ctx.signals] == [1, 2]
+        assert [s.name for s in ctx.signals] == ["a", "b"]
+
+    def test_effect_guard_blocks_duplicates(self):
+        calls = []
+
+        def guard(key, desc):
+            calls.append(key)
+            return len(calls) == 1  # only the first acquire is fresh
+
+        ctx = _ctx(effect_guard=guard)
+        assert ctx.record_effect("k1") is True
+        assert ctx.record_effect("k1") is False
+        assert len(ctx.effects) == 1
+
+    def test_cancel_check(self):
+        ctx = _ctx(cancel_check=lambda: True)
+        assert ctx.canceled
+
+
+class TestCli:
+    def test_validate_and_run(self, tmp_path):
+        from typer.testing import CliRunner
+
+        from bobrapet_amd.cli import app
+
+        runner = CliRunner()
+        res = runner.invoke(app, ["validate", "-f", "examples/order-flow.yaml"])
+        assert res.exit_code == 0, res.output
+        res = runner.invoke(
+            app,
+            ["run", "-f", "examples/order-flow.yaml", "--inputs", '{"orderId": "o-9"}', "--json"],
+        )
+        assert res.exit_code == 0, res.output
+        assert '"confirmed": true' in res.output
+
+    def test_validate_rejects_bad_story(self, tmp_path):
+        from typer.testing import CliRunner
+
+        from bobrapet_amd.cli import app
+
+        bad = tmp_path / "bad.yaml"
+        bad.write_text(
+            """
+kind: Story
+metadata: {name: bad}
+spec:
+  steps:
+    - {name: a, type: sleep, needs: [ghost], with: {duration: 1s}}
+"""
+        )
+        runner = CliRunner()
+        res = runner.invoke(app, ["validate", "-f", str(bad)])
+        assert res.exit_code == 1
+        assert "unknown step" in res.output
+
+    def test_engrams_list(self):
+        from typer.testing import CliRunner
+
+        from bobrapet_amd.cli import app
+
+        res = CliRunner().invoke(app, ["engrams"])
+        assert "llm-infer" in res.output and "embed" in res.output

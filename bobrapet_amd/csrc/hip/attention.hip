@@ -52,8 +52,8 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
     unsigned short* __restrict__ out,      // [B,S,Hq,D]
     const unsigned short* __restrict__ q,  // [B,S,Hq,D]
     const unsigned short* __restrict__ k,  // [B,S,Hkv,D]
-    const unsigned short* __restrict__ v,  // [B,S,Hkv,D]
-    int B, int Hq, int Hkv, int S, float scale, int causal) {
+    const unsigned short* __restrict__ v,  // [B,S,Hkv,D] (row stride may differ)
+    int B, int Hq, int Hkv, int S, long v_sstride, float scale, int causal) {
   // two K+V^T buffer pairs; pointers computed per use (an addrspace(3)
   // pointer array fails to compile as a static initializer)
   __shared__ __attribute__((aligned(16))) char smem[2 * (KVBLK * 256 + D_HEAD * 128)];
@@ -83,6 +83,8 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
   const long kv_sstride = (long)Hkv * D_HEAD;
   const long q_base = (long)b * S * q_sstride + (long)hq * D_HEAD;
   const long kv_base = (long)b * S * kv_sstride + (long)hkv * D_HEAD;
+  // v may be a strided view (e.g. a slice of the fused qkv projection)
+  const long v_base = (long)b * S * v_sstride + (long)hkv * D_HEAD;
 
   // wave-striped q assignment: wave w owns rows {qbase + 8*i + w}, i=0..31
   // (causal kv ranges match across waves -> no idle compute waves)
@@ -137,7 +139,7 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
       int d0 = (chunk & 15) * 8;
       if (kvg < S)
         vreg[it] = *reinterpret_cast<const ushort8v*>(
-            v + kv_base + (long)kvg * kv_sstride + d0);
+            v + v_base + (long)kvg * v_sstride + d0);
       else
         vreg[it] = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
     }
@@ -343,14 +345,14 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
 
 extern "C" void launch_attn_prefill(void* out, const void* q, const void* k,
                                     const void* v, int B, int Hq, int Hkv,
-                                    int S, float scale, int causal,
-                                    hipStream_t stream) {
+                                    int S, long v_sstride, float scale,
+                                    int causal, hipStream_t stream) {
   int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
   dim3 grid(B * Hq * nqblk), block(512);
   hipLaunchKernelGGL((attn_prefill_kernel<7>), grid, block, 0, stream,
                      (unsigned short*)out, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v, B, Hq,
-                     Hkv, S, scale, causal);
+                     Hkv, S, v_sstride, scale, causal);
 }
 
 extern "C" void launch_attn_prefill_variant(int variant, void* out,
@@ -363,11 +365,165 @@ extern "C" void launch_attn_prefill_variant(int variant, void* out,
   auto args = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)out,
                        (const unsigned short*)q, (const unsigned short*)k,
-                       (const unsigned short*)v, B, Hq, Hkv, S, scale, causal);
+                       (const unsigned short*)v, B, Hq, Hkv, S,
+                       (long)Hkv * D_HEAD, scale, causal);
   };
   if (variant == 1) args(attn_prefill_kernel<1>);
   else if (variant == 3) args(attn_prefill_kernel<3>);
   else args(attn_prefill_kernel<7>);
+}
+
+// ---------------------------------------------------------------------------
+// Decode attention v2 (single token, GQA, KV cache): two-pass chunked.
+//   pass 1: grid (B x Hkv x NCHUNK) — each WG reads its KV chunk ONCE and
+//     scores all GQA q heads of the group (amortizes the GQA-redundant
+//     K/V reads and fills the chip at small batch); per-chunk online
+//     (m, l, O) partials land in a workspace
+//   pass 2: grid (B x Hq) — combine the <=NCHUNK partials
+// Length L comes from host OR a device scalar (hipGraph-replayable);
+// chunks beyond L exit immediately so the grid is static under capture.
+// ---------------------------------------------------------------------------
+
+#define DEC_CHUNK 256
+#define DEC_GMAX 8  // max GQA group size handled per WG
+
+template <bool L_FROM_DEV>
+__global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
+    float* __restrict__ ws,  // [B][Hkv][NCHUNK][G][2+D] f32 (m, l, O)
+    const unsigned short* __restrict__ q,   // [B][Hq][D]
+    const unsigned short* __restrict__ kc,  // [B][Hkv][Smax][D]
+    const unsigned short* __restrict__ vc,
+    int B, int Hq, int Hkv, int Smax, int nchunk, int L,
+    const int* __restrict__ L_dev, float scale) {
+  if constexpr (L_FROM_DEV) L = *L_dev;
+  const int G = Hq / Hkv;
+  const int wg = blockIdx.x;
+  const int chunk = wg % nchunk;
+  const int hkv = (wg / nchunk) % Hkv;
+  const int b = wg / (nchunk * Hkv);
+  const int kv0 = chunk * DEC_CHUNK;
+  float* slot = ws + ((((long)b * Hkv + hkv) * nchunk + chunk) * DEC_GMAX) * (2 + D_HEAD);
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+
+  if (kv0 >= L) {
+    // dead chunk: mark empty partials (m = -inf, l = 0)
+    for (int g = tid; g < G; g += blockDim.x) {
+      slot[g * (2 + D_HEAD)] = -1e30f;
+      slot[g * (2 + D_HEAD) + 1] = 0.f;
+    }
+    return;
+  }
+  const int kv_end = min(kv0 + DEC_CHUNK, L);
+
+  // q fragments for the whole GQA group: lane owns 2 d positions
+  float q0[DEC_GMAX], q1[DEC_GMAX];
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    if (g < G) {
+      const unsigned short* qrow = q + ((long)b * Hq + hkv * G + g) * D_HEAD;
+      ushort2 qv = *reinterpret_cast<const ushort2*>(qrow + lane * 2);
+      q0[g] = bf2f(qv.x);
+      q1[g] = bf2f(qv.y);
+    } else {
+      q0[g] = q1[g] = 0.f;
+    }
+  }
+
+  const long base = (((long)b * Hkv + hkv) * Smax) * D_HEAD;
+  float m_w[DEC_GMAX], l_w[DEC_GMAX], o0[DEC_GMAX], o1[DEC_GMAX];
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    m_w[g] = -1e30f;
+    l_w[g] = 0.f;
+    o0[g] = o1[g] = 0.f;
+  }
+  for (int kv = kv0 + wid; kv < kv_end; kv += 4) {
+    ushort2 kv2 = *reinterpret_cast<const ushort2*>(kc + base + (long)kv * D_HEAD + lane * 2);
+    float kx = bf2f(kv2.x), ky = bf2f(kv2.y);
+    ushort2 vv = *reinterpret_cast<const ushort2*>(vc + base + (long)kv * D_HEAD + lane * 2);
+    float vx = bf2f(vv.x), vy = bf2f(vv.y);
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) {
+      if (g >= G) break;
+      float sdot = q0[g] * kx + q1[g] * ky;
+      sdot = wave_reduce_sum(sdot) * scale;
+      float m_new = fmaxf(m_w[g], sdot);
+      float alpha = (m_w[g] <= -1e30f) ? 0.f : __expf(m_w[g] - m_new);
+      float pp = __expf(sdot - m_new);
+      o0[g] = o0[g] * alpha + pp * vx;
+      o1[g] = o1[g] * alpha + pp * vy;
+      l_w[g] = l_w[g] * alpha + pp;
+      m_w[g] = m_new;
+    }
+  }
+
+  // merge the 4 waves through LDS, then publish the chunk partial
+  __shared__ float sm[DEC_GMAX][4], sl[DEC_GMAX][4], so[DEC_GMAX][4][D_HEAD];
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    if (g >= G) break;
+    sm[g][wid] = m_w[g];
+    sl[g][wid] = l_w[g];
+    so[g][wid][lane * 2] = o0[g];
+    so[g][wid][lane * 2 + 1] = o1[g];
+  }
+  __syncthreads();
+  if (wid == 0) {
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) {
+      if (g >= G) break;
+      float m_g = fmaxf(fmaxf(sm[g][0], sm[g][1]), fmaxf(sm[g][2], sm[g][3]));
+      float l_g = 0.f, a0 = 0.f, a1 = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        float aw = (sm[g][w] <= -1e30f) ? 0.f : __expf(sm[g][w] - m_g);
+        l_g += sl[g][w] * aw;
+        a0 += so[g][w][lane * 2] * aw;
+        a1 += so[g][w][lane * 2 + 1] * aw;
+      }
+      float* out = slot + g * (2 + D_HEAD);
+      if (lane == 0) {
+        out[0] = m_g;
+        out[1] = l_g;
+      }
+      out[2 + lane * 2] = a0;
+      out[2 + lane * 2 + 1] = a1;
+    }
+  }
+}
+
+template <bool L_FROM_DEV>
+__global__ __launch_bounds__(64, 8) void attn_decode_combine_kernel(
+    unsigned short* __restrict__ out,  // [B][Hq][D]
+    const float* __restrict__ ws, int B, int Hq, int Hkv, int nchunk, int L,
+    const int* __restrict__ L_dev) {
+  if constexpr (L_FROM_DEV) L = *L_dev;
+  const int G = Hq / Hkv;
+  const int bh = blockIdx.x;
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / G, g = hq % G;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int live = (L + DEC_CHUNK - 1) / DEC_CHUNK;
+  float m_g = -1e30f;
+  for (int c = 0; c < live; ++c) {
+    const float* slot = ws + ((((long)b * Hkv + hkv) * nchunk + c) * DEC_GMAX + g) * (2 + D_HEAD);
+    m_g = fmaxf(m_g, slot[0]);
+  }
+  float l_g = 0.f, a0 = 0.f, a1 = 0.f;
+  for (int c = 0; c < live; ++c) {
+    const float* slot = ws + ((((long)b * Hkv + hkv) * nchunk + c) * DEC_GMAX + g) * (2 + D_HEAD);
+    float aw = (slot[0] <= -1e30f) ? 0.f : __expf(slot[0] - m_g);
+    l_g += slot[1] * aw;
+    a0 += slot[2 + lane * 2] * aw;
+    a1 += slot[2 + lane * 2 + 1] * aw;
+  }
+  float inv = (l_g > 0.f) ? 1.f / l_g : 0.f;
+  unsigned short* orow = out + ((long)b * Hq + hq) * D_HEAD;
+  orow[lane * 2] = f2bf(a0 * inv);
+  orow[lane * 2 + 1] = f2bf(a1 * inv);
 }
 
 // ---------------------------------------------------------------------------
@@ -378,10 +534,13 @@ extern "C" void launch_attn_prefill_variant(int variant, void* out,
 // wave merge through LDS (guide §B "attention decode").
 // ---------------------------------------------------------------------------
 
+template <bool L_FROM_DEV>
 __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
     unsigned short* __restrict__ out, const unsigned short* __restrict__ q,
     const unsigned short* __restrict__ kc, const unsigned short* __restrict__ vc,
-    int B, int Hq, int Hkv, int Smax, int L, float scale) {
+    int B, int Hq, int Hkv, int Smax, int L, const int* __restrict__ L_dev,
+    float scale) {
+  if constexpr (L_FROM_DEV) L = *L_dev;  // hipGraph-replayable length
   const int bh = blockIdx.x;
   const int b = bh / Hq;
   const int hq = bh % Hq;
@@ -444,13 +603,45 @@ __global__ __launch_bounds__(256, 2) void attn_decode_kernel(
   }
 }
 
-extern "C" void launch_attn_decode(void* out, const void* q, const void* kc,
-                                   const void* vc, int B, int Hq, int Hkv,
-                                   int Smax, int L, float scale,
+extern "C" void launch_attn_decode(void* out, void* workspace, const void* q,
+                                   const void* kc, const void* vc, int B,
+                                   int Hq, int Hkv, int Smax, int L,
+                                   const void* L_dev, float scale,
                                    hipStream_t stream) {
+  const int G = Hq / Hkv;
+  if (workspace != nullptr && G <= DEC_GMAX) {
+    const int nchunk = (Smax + DEC_CHUNK - 1) / DEC_CHUNK;
+    dim3 g1(B * Hkv * nchunk), b1(256);
+    dim3 g2(B * Hq), b2(64);
+    if (L_dev != nullptr) {
+      hipLaunchKernelGGL((attn_decode_chunk_kernel<true>), g1, b1, 0, stream,
+                         (float*)workspace, (const unsigned short*)q,
+                         (const unsigned short*)kc, (const unsigned short*)vc,
+                         B, Hq, Hkv, Smax, nchunk, 0, (const int*)L_dev, scale);
+      hipLaunchKernelGGL((attn_decode_combine_kernel<true>), g2, b2, 0, stream,
+                         (unsigned short*)out, (const float*)workspace, B, Hq,
+                         Hkv, nchunk, 0, (const int*)L_dev);
+    } else {
+      hipLaunchKernelGGL((attn_decode_chunk_kernel<false>), g1, b1, 0, stream,
+                         (float*)workspace, (const unsigned short*)q,
+                         (const unsigned short*)kc, (const unsigned short*)vc,
+                         B, Hq, Hkv, Smax, nchunk, L, (const int*)nullptr, scale);
+      hipLaunchKernelGGL((attn_decode_combine_kernel<false>), g2, b2, 0, stream,
+                         (unsigned short*)out, (const float*)workspace, B, Hq,
+                         Hkv, nchunk, L, (const int*)nullptr);
+    }
+    return;
+  }
+  // fallback: single-pass kernel (large GQA groups / no workspace)
   dim3 grid(B * Hq), block(256);
-  hipLaunchKernelGGL(attn_decode_kernel, grid, block, 0, stream,
-                     (unsigned short*)out, (const unsigned short*)q,
-                     (const unsigned short*)kc, (const unsigned short*)vc, B,
-                     Hq, Hkv, Smax, L, scale);
+  if (L_dev != nullptr)
+    hipLaunchKernelGGL((attn_decode_kernel<true>), grid, block, 0, stream,
+                       (unsigned short*)out, (const unsigned short*)q,
+                       (const unsigned short*)kc, (const unsigned short*)vc, B,
+                       Hq, Hkv, Smax, 0, (const int*)L_dev, scale);
+  else
+    hipLaunchKernelGGL((attn_decode_kernel<false>), grid, block, 0, stream,
+                       (unsigned short*)out, (const unsigned short*)q,
+                       (const unsigned short*)kc, (const unsigned short*)vc, B,
+                       Hq, Hkv, Smax, L, (const int*)nullptr, scale);
 }

@@ -18,9 +18,14 @@ void launch_embed_pool(void*, void*, const void*, const void*, int, int, int,
                        int, hipStream_t);
 void launch_add_bf16(void*, const void*, const void*, long, hipStream_t);
 void launch_attn_prefill(void*, const void*, const void*, const void*, int,
-                         int, int, int, float, int, hipStream_t);
-void launch_attn_decode(void*, const void*, const void*, const void*, int, int,
-                        int, int, int, float, hipStream_t);
+                         int, int, int, long, float, int, hipStream_t);
+void launch_silu_mul_strided(void*, const void*, const void*, long, int, long,
+                             hipStream_t);
+void launch_rope_qkv(void*, void*, const void*, const void*, const void*, int,
+                     int, int, int, long, hipStream_t);
+void launch_attn_decode(void*, void*, const void*, const void*, const void*,
+                        int, int, int, int, int, const void*, float,
+                        hipStream_t);
 void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
 void launch_attn_prefill_variant(int, void*, const void*, const void*,
                                  const void*, int, int, int, int, float, int,
@@ -126,19 +131,22 @@ torch::Tensor add_bf16(torch::Tensor a, torch::Tensor b) {
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                            double scale, bool causal) {
   // q: [B,S,Hq,D=128], k/v: [B,S,Hkv,D] (BSHD: the model's native layout,
-  // no transposes on the hot path)
+  // no transposes on the hot path); v may be a strided view
   check_bf16(q, "q");
   check_bf16(k, "k");
-  check_bf16(v, "v");
   const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
   const int Hkv = k.size(2);
   TORCH_CHECK(D == 128, "attn_prefill: D must be 128");
   TORCH_CHECK(Hq % Hkv == 0, "attn_prefill: Hq must be a multiple of Hkv");
   TORCH_CHECK(k.size(1) == S && v.size(1) == S, "attn_prefill: S mismatch");
+  // v may be a strided [B,S,Hkv,D] view (head+lane dims contiguous)
+  TORCH_CHECK(v.is_cuda() && v.scalar_type() == torch::kBFloat16, "v dtype");
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D, "attn_prefill: v head must be contiguous");
+  TORCH_CHECK(v.stride(0) == v.stride(1) * S, "attn_prefill: v batch stride");
   auto out = torch::empty_like(q);
   launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
-                      B, Hq, Hkv, S, (float)scale, causal ? 1 : 0,
-                      cur_stream());
+                      B, Hq, Hkv, S, (long)v.stride(1), (float)scale,
+                      causal ? 1 : 0, cur_stream());
   return out;
 }
 
@@ -153,10 +161,74 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   TORCH_CHECK(D == 128, "attn_decode: D must be 128");
   TORCH_CHECK(L <= Smax, "attn_decode: L > Smax");
   auto out = torch::empty_like(q);
-  launch_attn_decode(out.data_ptr(), q.data_ptr(), kc.data_ptr(),
-                     vc.data_ptr(), B, Hq, Hkv, Smax, (int)L, (float)scale,
-                     cur_stream());
+  void* ws_ptr = nullptr;
+  torch::Tensor ws;
+  if (getenv("BOBRA_DEC_V2") && getenv("BOBRA_DEC_V2")[0] == '1') {
+    const int nchunk = (Smax + 255) / 256;
+    ws = torch::empty({(long)B * Hkv * nchunk * 8 * 130},
+                      q.options().dtype(torch::kFloat32));
+    ws_ptr = ws.data_ptr();
+  }
+  launch_attn_decode(out.data_ptr(), ws_ptr, q.data_ptr(), kc.data_ptr(),
+                     vc.data_ptr(), B, Hq, Hkv, Smax, (int)L, nullptr,
+                     (float)scale, cur_stream());
   return out;
+}
+
+torch::Tensor attn_decode_t(torch::Tensor q, torch::Tensor kc,
+                            torch::Tensor vc, torch::Tensor L_dev,
+                            double scale) {
+  // graph-capturable decode: L read from a device int32 scalar
+  check_bf16(q, "q");
+  const int B = q.size(0), Hq = q.size(1), D = q.size(2);
+  const int Hkv = kc.size(1), Smax = kc.size(2);
+  TORCH_CHECK(D == 128, "attn_decode_t: D must be 128");
+  TORCH_CHECK(L_dev.scalar_type() == torch::kInt32 && L_dev.is_cuda(),
+              "L must be an int32 device scalar");
+  auto out = torch::empty_like(q);
+  void* ws_ptr = nullptr;
+  torch::Tensor ws;
+  if (getenv("BOBRA_DEC_V2") && getenv("BOBRA_DEC_V2")[0] == '1') {
+    const int nchunk = (Smax + 255) / 256;
+    ws = torch::empty({(long)B * Hkv * nchunk * 8 * 130},
+                      q.options().dtype(torch::kFloat32));
+    ws_ptr = ws.data_ptr();
+  }
+  launch_attn_decode(out.data_ptr(), ws_ptr, q.data_ptr(), kc.data_ptr(),
+                     vc.data_ptr(), B, Hq, Hkv, Smax, 0, L_dev.data_ptr(),
+                     (float)scale, cur_stream());
+  return out;
+}
+
+torch::Tensor silu_mul_strided(torch::Tensor gate_up) {
+  // gate_up: [..., 2*I] bf16 (contiguous); returns silu(g)*u of shape [..., I]
+  check_bf16(gate_up, "gate_up");
+  const int two_i = gate_up.size(-1);
+  const int inner = two_i / 2;
+  TORCH_CHECK(inner % 8 == 0, "silu_mul_strided: I must be %8==0");
+  const long rows = gate_up.numel() / two_i;
+  auto sizes = gate_up.sizes().vec();
+  sizes.back() = inner;
+  auto out = torch::empty(sizes, gate_up.options());
+  const unsigned short* base = (const unsigned short*)gate_up.data_ptr();
+  launch_silu_mul_strided(out.data_ptr(), base, base + inner, rows, inner,
+                          (long)two_i, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> rope_qkv(torch::Tensor qkv, int Hq, int Hk, int D,
+                                    torch::Tensor cos_t, torch::Tensor sin_t) {
+  // qkv: [T, rowlen] bf16 contiguous; returns rotated contiguous q [T,Hq,D],
+  // k [T,Hk,D] (v stays a view into qkv)
+  check_bf16(qkv, "qkv");
+  const long T = qkv.size(0);
+  TORCH_CHECK(cos_t.size(0) == T && cos_t.size(1) == D / 2, "rope_qkv tables");
+  auto q = torch::empty({T, (long)Hq, (long)D}, qkv.options());
+  auto k = torch::empty({T, (long)Hk, (long)D}, qkv.options());
+  launch_rope_qkv(q.data_ptr(), k.data_ptr(), qkv.data_ptr(), cos_t.data_ptr(),
+                  sin_t.data_ptr(), (int)T, Hq, Hk, D, (long)qkv.size(1),
+                  cur_stream());
+  return {q, k};
 }
 
 torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
@@ -213,9 +285,15 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_bf16", &add_bf16, "a + b");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
+  m.def("attn_decode_t", &attn_decode_t,
+        "decode attention, length from a device scalar (hipGraph-capturable)");
   m.def("dbg_mfma", &dbg_mfma, "layout probe: C=A@B one mfma");
   m.def("attn_prefill_variant", &attn_prefill_variant,
         "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
+  m.def("silu_mul_strided", &silu_mul_strided,
+        "silu(gate)*up from a fused [.., 2I] gate_up matrix (no copies)");
+  m.def("rope_qkv", &rope_qkv,
+        "fused rope: strided q/k heads from the qkv projection -> contiguous");
   m.def("dbg_attn_core", &dbg_attn_core, "layout probe: QK^T + pack + PV");
 }

@@ -114,6 +114,40 @@ __global__ __launch_bounds__(256) void silu_mul_kernel(
   }
 }
 
+__global__ __launch_bounds__(256) void silu_mul_strided_kernel(
+    unsigned short* __restrict__ out, const unsigned short* __restrict__ gate,
+    const unsigned short* __restrict__ up, long rows, int inner,
+    long row_stride) {
+  const long n8 = rows * (inner / 8);
+  const int i8 = inner / 8;
+  for (long idx = blockIdx.x * blockDim.x + threadIdx.x; idx < n8;
+       idx += (long)gridDim.x * blockDim.x) {
+    long r = idx / i8;
+    int c = (int)(idx % i8) * 8;
+    ushort8v g = *reinterpret_cast<const ushort8v*>(gate + r * row_stride + c);
+    ushort8v u = *reinterpret_cast<const ushort8v*>(up + r * row_stride + c);
+    ushort8v o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf2f(g[j]);
+      float sg = gf / (1.f + __expf(-gf));
+      o[j] = f2bf(sg * bf2f(u[j]));
+    }
+    *reinterpret_cast<ushort8v*>(out + r * (long)inner + c) = o;
+  }
+}
+
+extern "C" void launch_silu_mul_strided(void* out, const void* gate,
+                                        const void* up, long rows, int inner,
+                                        long row_stride, hipStream_t stream) {
+  long n8 = rows * (inner / 8);
+  int blocks = (int)((n8 + 255) / 256);
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(silu_mul_strided_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (unsigned short*)out, (const unsigned short*)gate,
+                     (const unsigned short*)up, rows, inner, row_stride);
+}
+
 extern "C" void launch_silu_mul(void* out, const void* gate, const void* up,
                                 long n, hipStream_t stream) {
   long n8 = n / 8;
@@ -159,6 +193,52 @@ __global__ __launch_bounds__(256) void rope_kernel(
       base[d + half] = f2bf(b * c + a * s);
     }
   }
+}
+
+// Fused rope-from-qkv: reads q/k heads from the fused qkv projection
+// output (row stride = qkv_out) and writes rotated CONTIGUOUS [T,H,D]
+// tensors — removes the slice-contiguous copies on the model hot path.
+__global__ __launch_bounds__(256) void rope_qkv_kernel(
+    unsigned short* __restrict__ q_out,     // [T,Hq,D]
+    unsigned short* __restrict__ k_out,     // [T,Hk,D]
+    const unsigned short* __restrict__ qkv, // [T, row_stride elems]
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t, int T,
+    int Hq, int Hk, int D, long row_stride) {
+  const int half = D / 2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int total = T * (Hq + Hk);
+  const int nwaves = (gridDim.x * blockDim.x) / WAVE;
+  for (int idx = wave; idx < total; idx += nwaves) {
+    const int t = idx / (Hq + Hk);
+    const int h = idx % (Hq + Hk);
+    // q heads sit at [0, Hq*D); k heads follow at [Hq*D, (Hq+Hk)*D)
+    const unsigned short* src = qkv + (long)t * row_stride + (long)h * D;
+    unsigned short* dst = (h < Hq) ? q_out + ((long)t * Hq + h) * D
+                                   : k_out + ((long)t * Hk + (h - Hq)) * D;
+    const float* crow = cos_t + (long)t * half;
+    const float* srow = sin_t + (long)t * half;
+    for (int d = lane; d < half; d += WAVE) {
+      float a = bf2f(src[d]);
+      float b = bf2f(src[d + half]);
+      float c = crow[d], sn = srow[d];
+      dst[d] = f2bf(a * c - b * sn);
+      dst[d + half] = f2bf(b * c + a * sn);
+    }
+  }
+}
+
+extern "C" void launch_rope_qkv(void* q_out, void* k_out, const void* qkv,
+                                const void* cos_t, const void* sin_t, int T,
+                                int Hq, int Hk, int D, long row_stride,
+                                hipStream_t stream) {
+  int total_waves = T * (Hq + Hk);
+  int blocks = (total_waves * WAVE + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  hipLaunchKernelGGL(rope_qkv_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (unsigned short*)q_out, (unsigned short*)k_out,
+                     (const unsigned short*)qkv, (const float*)cos_t,
+                     (const float*)sin_t, T, Hq, Hk, D, row_stride);
 }
 
 extern "C" void launch_rope(void* q, void* k, const void* cos_t,

@@ -118,6 +118,10 @@ class LlamaModel:
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
         self._kv_cache: _t.Optional[_t.List[_t.Tuple[torch.Tensor, torch.Tensor]]] = None
         self._cache_len = 0
+        self._decode_graph = None
+        self._g_ids = None  # static graph buffers
+        self._g_pos = None
+        self._g_logits = None
 
     # ------------------------------------------------------------------
 
@@ -169,16 +173,12 @@ class LlamaModel:
             self._alloc_cache(B, max(cfg.max_seq_len, S))
 
         for li, lw in enumerate(self.layers):
-            # attention block
+            # attention block — zero layout copies: rope reads strided q/k
+            # heads straight from the qkv projection; v stays a strided view
             qkv = torch.matmul(x, lw.w_qkv)  # [B,S,qkv_out]
-            q, k, v = self._split_qkv(qkv, B, S)
-            qf = q.reshape(B * S, cfg.num_heads, cfg.head_dim).contiguous()
-            kf = k.reshape(B * S, cfg.num_kv_heads, cfg.head_dim).contiguous()
-            qf, kf = ops.rope_inplace(qf, kf, cos_f, sin_f)
-            # BSHD end-to-end: no layout transposes on the hot path
-            qh = qf.view(B, S, cfg.num_heads, cfg.head_dim)
-            kh = kf.view(B, S, cfg.num_kv_heads, cfg.head_dim)
-            vh = v.contiguous()
+            qh, kh, vh = ops.rope_qkv_split(
+                qkv, B, S, cfg.num_heads, cfg.num_kv_heads, cfg.head_dim, cos_f, sin_f
+            )
             if fill_cache:
                 kc, vc = self._kv_cache[li]
                 kc[:, :, :S] = kh.transpose(1, 2)
@@ -188,10 +188,9 @@ class LlamaModel:
             attn_out = torch.matmul(attn, lw.w_o)
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
 
-            # mlp block
+            # mlp block — silu reads gate/up halves in place
             gate_up = torch.matmul(x, lw.w_gate_up)
-            gate, up = gate_up.chunk(2, dim=-1)
-            act = ops.silu_mul(gate.contiguous(), up.contiguous())
+            act = ops.silu_mul_fused(gate_up)
             mlp_out = torch.matmul(act, lw.w_down)
             next_norm = (
                 self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
@@ -205,6 +204,78 @@ class LlamaModel:
         return torch.matmul(x[:, -1], self.lm_head)
 
     # ------------------------------------------------------------------
+
+    def _decode_body(self, ids: torch.Tensor, pos_i32: torch.Tensor) -> torch.Tensor:
+        """One decode step with all dynamic state in device tensors — every
+        op here is hipGraph-capturable (no host-dependent shapes/values)."""
+        cfg = self.cfg
+        B = ids.shape[0]
+        pos_l = pos_i32.to(torch.long)
+        inv_freq = self._inv_freq
+        ang = pos_l.float().reshape(1, 1) * inv_freq[None, :]
+        cos_t = torch.cos(ang).expand(B, -1).contiguous()
+        sin_t = torch.sin(ang).expand(B, -1).contiguous()
+
+        residual = self.embed[ids].view(B, 1, cfg.hidden_size)
+        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
+        L_dev = (pos_i32 + 1).contiguous()
+        for li, lw in enumerate(self.layers):
+            qkv = torch.matmul(x, lw.w_qkv)
+            q, k, v = self._split_qkv(qkv, B, 1)
+            qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
+            kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
+            qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
+            kc, vc = self._kv_cache[li]
+            # device-indexed cache append (graph-replayable)
+            kc.index_copy_(2, pos_l, kf.view(B, cfg.num_kv_heads, 1, cfg.head_dim))
+            vc.index_copy_(2, pos_l, v.view(B, cfg.num_kv_heads, 1, cfg.head_dim))
+            attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
+            attn_out = torch.matmul(attn.reshape(B, 1, -1), lw.w_o)
+            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
+            gate_up = torch.matmul(x, lw.w_gate_up)
+            act = ops.silu_mul_fused(gate_up)
+            mlp_out = torch.matmul(act, lw.w_down)
+            next_norm = (
+                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
+            )
+            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
+        return torch.matmul(x[:, -1], self.lm_head)
+
+    @torch.no_grad()
+    def decode_step_graphed(self, ids: torch.Tensor) -> torch.Tensor:
+        """hipGraph-captured decode (guide: capture launch-bound inner loops
+        in hipGraphs): one replay per token; position/length live in device
+        scalars that the graph itself advances."""
+        assert self._kv_cache is not None, "call prefill(fill_cache=True) first"
+        B = ids.shape[0]
+        if not hasattr(self, "_inv_freq"):
+            cfg = self.cfg
+            self._inv_freq = 1.0 / (
+                cfg.rope_theta
+                ** (torch.arange(0, cfg.head_dim, 2, dtype=torch.float32, device=self.device) / cfg.head_dim)
+            )
+        if self._decode_graph is None:
+            self._g_ids = torch.zeros(B, dtype=torch.long, device=self.device)
+            self._g_pos = torch.zeros(1, dtype=torch.int32, device=self.device)
+            self._g_ids.copy_(ids)
+            self._g_pos.fill_(self._cache_len)
+            # warmup outside capture (allocator settles)
+            self._decode_body(self._g_ids, self._g_pos)
+            self._g_pos.fill_(self._cache_len)
+            torch.cuda.synchronize(self.device)
+            g = torch.cuda.CUDAGraph()
+            stream = torch.cuda.Stream(device=self.device)
+            with torch.cuda.stream(stream):
+                with torch.cuda.graph(g, stream=stream):
+                    self._g_logits = self._decode_body(self._g_ids, self._g_pos)
+                    self._g_pos.add_(1)  # the graph advances its own position
+            self._decode_graph = g
+            self._decode_graph_batch = B
+        assert self._decode_graph_batch == B, "decode batch changed; new model needed"
+        self._g_ids.copy_(ids)
+        self._decode_graph.replay()
+        self._cache_len += 1
+        return self._g_logits
 
     def _alloc_cache(self, B: int, smax: int) -> None:
         cfg = self.cfg
@@ -258,16 +329,23 @@ class LlamaModel:
         return torch.matmul(x[:, -1], self.lm_head)
 
     @torch.no_grad()
-    def generate(self, ids: torch.Tensor, new_tokens: int, greedy: bool = True):
-        """Prefill + decode loop; returns generated token ids [B, new_tokens]."""
+    def generate(
+        self, ids: torch.Tensor, new_tokens: int, greedy: bool = True,
+        use_graph: _t.Optional[bool] = None,
+    ):
+        """Prefill + decode loop; returns generated token ids [B, new_tokens].
+        On GPU the decode loop replays a captured hipGraph per token."""
+        if use_graph is None:
+            use_graph = self.device.type == "cuda"
         logits = self.prefill(ids, fill_cache=True)
+        step = self.decode_step_graphed if use_graph else self.decode_step
         out = []
         for _ in range(new_tokens):
             nxt = logits.argmax(dim=-1) if greedy else torch.multinomial(
                 torch.softmax(logits.float(), dim=-1), 1
             ).squeeze(-1)
             out.append(nxt)
-            logits = self.decode_step(nxt)
+            logits = step(nxt)
         return torch.stack(out, dim=1)
 
 

@@ -179,13 +179,41 @@ def attn_prefill(q, k, v, scale: _t.Optional[float] = None, causal: bool = True)
     """BSHD layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] → out [B,S,Hq,D]."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if q.is_cuda:
-        return _require_ext().attn_prefill(
-            q.contiguous(), k.contiguous(), v.contiguous(), scale, causal
-        )
+        if not (v.stride(-1) == 1 and v.stride(2) == v.shape[-1]):
+            v = v.contiguous()
+        return _require_ext().attn_prefill(q.contiguous(), k.contiguous(), v, scale, causal)
     out = attn_ref(
         q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2), scale, causal
     )
     return out.transpose(1, 2).contiguous()
+
+
+def rope_qkv_split(qkv, B, S, Hq, Hkv, D, cos_t, sin_t):
+    """Fused split+rope: qkv [B,S,(Hq+2*Hkv)*D] → rotated contiguous
+    q [B,S,Hq,D], k [B,S,Hkv,D] and v as a STRIDED VIEW into qkv (the
+    attention kernel reads strided v — no copies)."""
+    rowlen = qkv.shape[-1]
+    nq, nkv = Hq * D, Hkv * D
+    if qkv.is_cuda:
+        flat = qkv.reshape(B * S, rowlen)
+        q, k = _require_ext().rope_qkv(flat, Hq, Hkv, D, cos_t, sin_t)
+        v = qkv[..., nq + nkv :].view(B, S, Hkv, D)
+        return q.view(B, S, Hq, D), k.view(B, S, Hkv, D), v
+    q = qkv[..., :nq].reshape(B * S, Hq, D)
+    k = qkv[..., nq : nq + nkv].reshape(B * S, Hkv, D)
+    v = qkv[..., nq + nkv :].reshape(B, S, Hkv, D)
+    q = rope_ref(q, cos_t, sin_t).view(B, S, Hq, D)
+    k = rope_ref(k, cos_t, sin_t).view(B, S, Hkv, D)
+    return q, k, v
+
+
+def silu_mul_fused(gate_up: torch.Tensor) -> torch.Tensor:
+    """silu(g)*u from the fused [.., 2I] gate_up projection (no slicing
+    copies on GPU)."""
+    if gate_up.is_cuda:
+        return _require_ext().silu_mul_strided(gate_up.contiguous())
+    gate, up = gate_up.chunk(2, dim=-1)
+    return silu_mul_ref(gate, up)
 
 
 def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
@@ -200,6 +228,14 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
             return _require_ext().gemm_nt(a.contiguous(), b.contiguous())
         return torch.matmul(a, b.t())
     return (a.float() @ b.float().t()).to(a.dtype)
+
+
+def attn_decode_t(q, kc, vc, L_dev: torch.Tensor, scale: _t.Optional[float] = None):
+    """Graph-capturable decode attention: L read from a device int32 scalar."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        return _require_ext().attn_decode_t(q.contiguous(), kc, vc, L_dev, scale)
+    return attn_decode_ref(q, kc, vc, int(L_dev.item()), scale)
 
 
 def attn_decode(q, kc, vc, L: int, scale: _t.Optional[float] = None):

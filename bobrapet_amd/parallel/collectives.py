@@ -15,13 +15,19 @@ import torch.distributed as dist
 
 
 def all_gather_tensor(t: torch.Tensor) -> torch.Tensor:
-    """Gather a same-shape tensor from every rank → stacked [world, ...]."""
+    """Gather a same-shape tensor from every rank → stacked [world, ...].
+
+    all_gather_into_tensor requires the output's FIRST dim to be
+    world * input_first_dim (flat concat) — allocate flat, then view."""
     if not dist.is_initialized() or dist.get_world_size() == 1:
         return t.unsqueeze(0)
     world = dist.get_world_size()
-    out = torch.empty((world,) + tuple(t.shape), dtype=t.dtype, device=t.device)
-    dist.all_gather_into_tensor(out, t.contiguous())
-    return out
+    src = t.contiguous()
+    flat = torch.empty(
+        (world * src.shape[0],) + tuple(src.shape[1:]), dtype=src.dtype, device=src.device
+    )
+    dist.all_gather_into_tensor(flat, src)
+    return flat.view((world,) + tuple(src.shape))
 
 
 def all_gather_object(obj) -> _t.List:

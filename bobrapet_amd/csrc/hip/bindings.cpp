@@ -32,6 +32,8 @@ void launch_attn_prefill_variant(int, void*, const void*, const void*,
                                  hipStream_t);
 void launch_gemm_bf16_nt(void*, const void*, const void*, int, int, int,
                          hipStream_t);
+void launch_gemv_bf16_nt(void*, const void*, const void*, int, int, int,
+                         hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
 }
@@ -245,6 +247,19 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+torch::Tensor gemv_nt(torch::Tensor a, torch::Tensor b) {
+  // a: [M<=8, K] bf16, b: [N, K] bf16 -> [M, N]
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(M <= 8, "gemv_nt: M must be <= 8");
+  TORCH_CHECK(b.size(1) == K && K % 8 == 0, "gemv_nt: K must be %8==0");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemv_bf16_nt(c.data_ptr(), a.data_ptr(), b.data_ptr(), M, N, K,
+                      cur_stream());
+  return c;
+}
+
 torch::Tensor attn_prefill_variant(int variant, torch::Tensor q,
                                    torch::Tensor k, torch::Tensor v,
                                    double scale, bool causal) {
@@ -291,6 +306,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_variant", &attn_prefill_variant,
         "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
+  m.def("gemv_nt", &gemv_nt, "bf16 weight-streaming GEMV (M<=8)");
   m.def("silu_mul_strided", &silu_mul_strided,
         "silu(gate)*up from a fused [.., 2I] gate_up matrix (no copies)");
   m.def("rope_qkv", &rope_qkv,

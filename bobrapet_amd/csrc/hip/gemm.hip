@@ -145,3 +145,67 @@ extern "C" void launch_gemm_bf16_nt(void* C, const void* A, const void* B,
                      (unsigned short*)C, (const unsigned short*)A,
                      (const unsigned short*)B, M, N, K);
 }
+
+
+// ---------------------------------------------------------------------------
+// Weight-streaming GEMV for decode (guide App. B "GEMV / M<=16 decode
+// weights: operand streamed once, load straight to VGPRs, deep unroll"):
+// C[M,N] = A[M,K] @ W[N,K]^T for small M.  One wave per output column:
+// the wave streams W[n][:] coalesced (ushort8 per lane), A rows ride in
+// L1/L2 (tiny), M dot products accumulate per lane, one wave reduction
+// per (m, n).
+// ---------------------------------------------------------------------------
+
+#define GV_MMAX 8
+
+template <int M>
+__global__ __launch_bounds__(256) void gemv_bf16_nt_kernel(
+    unsigned short* __restrict__ C,        // [M][N]
+    const unsigned short* __restrict__ A,  // [M][K]
+    const unsigned short* __restrict__ W,  // [N][K]
+    int N, int K) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int waves_total = (gridDim.x * blockDim.x) >> 6;
+  for (int n = blockIdx.x * 4 + wid; n < N; n += waves_total) {
+    const unsigned short* wrow = W + (long)n * K;
+    float acc[M];
+#pragma unroll
+    for (int m = 0; m < M; ++m) acc[m] = 0.f;
+    for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
+      ushort8v wv = *reinterpret_cast<const ushort8v*>(wrow + k0);
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)m * K + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[m] += bf2f(av[j]) * bf2f(wv[j]);
+      }
+    }
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      float r = wave_reduce_sum(acc[m]);
+      if (lane == 0) C[(long)m * N + n] = f2bf(r);
+    }
+  }
+}
+
+extern "C" void launch_gemv_bf16_nt(void* C, const void* A, const void* W,
+                                    int M, int N, int K, hipStream_t stream) {
+  int blocks = (N + 3) / 4;
+  if (blocks > 2048) blocks = 2048;
+  dim3 grid(blocks), block(256);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
+                       (const unsigned short*)A, (const unsigned short*)W, N, K);
+  };
+  switch (M) {
+    case 1: launch(gemv_bf16_nt_kernel<1>); break;
+    case 2: launch(gemv_bf16_nt_kernel<2>); break;
+    case 3: launch(gemv_bf16_nt_kernel<3>); break;
+    case 4: launch(gemv_bf16_nt_kernel<4>); break;
+    case 5: launch(gemv_bf16_nt_kernel<5>); break;
+    case 6: launch(gemv_bf16_nt_kernel<6>); break;
+    case 7: launch(gemv_bf16_nt_kernel<7>); break;
+    default: launch(gemv_bf16_nt_kernel<8>); break;
+  }
+}

@@ -216,6 +216,13 @@ def silu_mul_fused(gate_up: torch.Tensor) -> torch.Tensor:
     return silu_mul_ref(gate, up)
 
 
+def gemv_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """C = a @ b.T for a [M<=8, K]: the weight-streaming decode GEMV."""
+    if a.is_cuda:
+        return _require_ext().gemv_nt(a.contiguous(), b.contiguous())
+    return (a.float() @ b.float().t()).to(a.dtype)
+
+
 def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     """C = a @ b.T with b stored [N, K] (nn.Linear weight layout).
 

@@ -408,7 +408,6 @@ __global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
   const int wid = tid >> 6;
 
   if (kv0 >= L) {
-    // dead chunk: mark empty partials (m = -inf, l = 0)
     for (int g = tid; g < G; g += blockDim.x) {
       slot[g * (2 + D_HEAD)] = -1e30f;
       slot[g * (2 + D_HEAD) + 1] = 0.f;
@@ -417,55 +416,77 @@ __global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
   }
   const int kv_end = min(kv0 + DEC_CHUNK, L);
 
-  // q fragments for the whole GQA group: lane owns 2 d positions
-  float q0[DEC_GMAX], q1[DEC_GMAX];
-#pragma unroll
-  for (int g = 0; g < DEC_GMAX; ++g) {
-    if (g < G) {
-      const unsigned short* qrow = q + ((long)b * Hq + hkv * G + g) * D_HEAD;
-      ushort2 qv = *reinterpret_cast<const ushort2*>(qrow + lane * 2);
-      q0[g] = bf2f(qv.x);
-      q1[g] = bf2f(qv.y);
-    } else {
-      q0[g] = q1[g] = 0.f;
-    }
+  // LDS: q rows (broadcast reads), per-lane p values, cross-wave merge
+  __shared__ float q_lds[DEC_GMAX][D_HEAD];
+  __shared__ float p_lds[4][DEC_GMAX][WAVE];
+  __shared__ float sm[DEC_GMAX][4], sl[DEC_GMAX][4], so[DEC_GMAX][4][D_HEAD];
+  for (int i = tid; i < G * D_HEAD; i += blockDim.x) {
+    int g = i / D_HEAD, d = i % D_HEAD;
+    q_lds[g][d] = bf2f(q[((long)b * Hq + hkv * G + g) * D_HEAD + d]);
   }
+  __syncthreads();
 
   const long base = (((long)b * Hkv + hkv) * Smax) * D_HEAD;
-  float m_w[DEC_GMAX], l_w[DEC_GMAX], o0[DEC_GMAX], o1[DEC_GMAX];
+
+  // phase A: lane-per-row scores (no per-row reductions)
+  const int r = kv0 + wid * WAVE + lane;
+  const bool valid = r < kv_end;
+  const int r_safe = valid ? r : (kv_end - 1);
+  float sacc[DEC_GMAX];
 #pragma unroll
-  for (int g = 0; g < DEC_GMAX; ++g) {
-    m_w[g] = -1e30f;
-    l_w[g] = 0.f;
-    o0[g] = o1[g] = 0.f;
-  }
-  for (int kv = kv0 + wid; kv < kv_end; kv += 4) {
-    ushort2 kv2 = *reinterpret_cast<const ushort2*>(kc + base + (long)kv * D_HEAD + lane * 2);
-    float kx = bf2f(kv2.x), ky = bf2f(kv2.y);
-    ushort2 vv = *reinterpret_cast<const ushort2*>(vc + base + (long)kv * D_HEAD + lane * 2);
-    float vx = bf2f(vv.x), vy = bf2f(vv.y);
+  for (int g = 0; g < DEC_GMAX; ++g) sacc[g] = 0.f;
+  const unsigned short* krow = kc + base + (long)r_safe * D_HEAD;
+#pragma unroll
+  for (int d0 = 0; d0 < D_HEAD / 8; ++d0) {
+    ushort8v k8 = *reinterpret_cast<const ushort8v*>(krow + d0 * 8);
+    float kf[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) kf[j] = bf2f(k8[j]);
 #pragma unroll
     for (int g = 0; g < DEC_GMAX; ++g) {
       if (g >= G) break;
-      float sdot = q0[g] * kx + q1[g] * ky;
-      sdot = wave_reduce_sum(sdot) * scale;
-      float m_new = fmaxf(m_w[g], sdot);
-      float alpha = (m_w[g] <= -1e30f) ? 0.f : __expf(m_w[g] - m_new);
-      float pp = __expf(sdot - m_new);
-      o0[g] = o0[g] * alpha + pp * vx;
-      o1[g] = o1[g] * alpha + pp * vy;
-      l_w[g] = l_w[g] * alpha + pp;
-      m_w[g] = m_new;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sacc[g] += kf[j] * q_lds[g][d0 * 8 + j];
     }
   }
-
-  // merge the 4 waves through LDS, then publish the chunk partial
-  __shared__ float sm[DEC_GMAX][4], sl[DEC_GMAX][4], so[DEC_GMAX][4][D_HEAD];
+  float m_w[DEC_GMAX], l_w[DEC_GMAX];
 #pragma unroll
   for (int g = 0; g < DEC_GMAX; ++g) {
     if (g >= G) break;
-    sm[g][wid] = m_w[g];
-    sl[g][wid] = l_w[g];
+    float sv = valid ? sacc[g] * scale : -1e30f;
+    float m = wave_reduce_max(sv);
+    float pv = (sv <= -1e30f) ? 0.f : __expf(sv - m);
+    p_lds[wid][g][lane] = pv;
+    m_w[g] = m;
+    l_w[g] = wave_reduce_sum(pv);
+  }
+  __syncthreads();
+
+  // phase B: broadcast-PV — every lane reads the SAME row (coalesced V),
+  // p broadcast from LDS; lane owns 2 output dims
+  float o0[DEC_GMAX], o1[DEC_GMAX];
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) o0[g] = o1[g] = 0.f;
+  const int rows = min(WAVE, kv_end - (kv0 + wid * WAVE));
+  for (int rr = 0; rr < rows; ++rr) {
+    int row = kv0 + wid * WAVE + rr;
+    ushort2 v2 = *reinterpret_cast<const ushort2*>(vc + base + (long)row * D_HEAD + lane * 2);
+    float vx = bf2f(v2.x), vy = bf2f(v2.y);
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) {
+      if (g >= G) break;
+      float pv = p_lds[wid][g][rr];
+      o0[g] += pv * vx;
+      o1[g] += pv * vy;
+    }
+  }
+
+  // cross-wave merge (each wave covered disjoint rows)
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    if (g >= G) break;
+    sm[g][wid] = rows > 0 ? m_w[g] : -1e30f;
+    sl[g][wid] = rows > 0 ? l_w[g] : 0.f;
     so[g][wid][lane * 2] = o0[g];
     so[g][wid][lane * 2 + 1] = o1[g];
   }

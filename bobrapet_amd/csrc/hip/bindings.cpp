@@ -165,7 +165,10 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
   auto out = torch::empty_like(q);
   void* ws_ptr = nullptr;
   torch::Tensor ws;
-  if (getenv("BOBRA_DEC_V2") && getenv("BOBRA_DEC_V2")[0] == '1') {
+  // two-pass chunked decode (lane-per-row scores + broadcast PV) is the
+  // default; BOBRA_DEC_V1=1 forces the single-pass per-(b,h) kernel.
+  if (!(getenv("BOBRA_DEC_V1") && getenv("BOBRA_DEC_V1")[0] == '1') &&
+      Hq / Hkv <= 8) {
     const int nchunk = (Smax + 255) / 256;
     ws = torch::empty({(long)B * Hkv * nchunk * 8 * 130},
                       q.options().dtype(torch::kFloat32));
@@ -190,7 +193,10 @@ torch::Tensor attn_decode_t(torch::Tensor q, torch::Tensor kc,
   auto out = torch::empty_like(q);
   void* ws_ptr = nullptr;
   torch::Tensor ws;
-  if (getenv("BOBRA_DEC_V2") && getenv("BOBRA_DEC_V2")[0] == '1') {
+  // two-pass chunked decode (lane-per-row scores + broadcast PV) is the
+  // default; BOBRA_DEC_V1=1 forces the single-pass per-(b,h) kernel.
+  if (!(getenv("BOBRA_DEC_V1") && getenv("BOBRA_DEC_V1")[0] == '1') &&
+      Hq / Hkv <= 8) {
     const int nchunk = (Smax + 255) / 256;
     ws = torch::empty({(long)B * Hkv * nchunk * 8 * 130},
                       q.options().dtype(torch::kFloat32));

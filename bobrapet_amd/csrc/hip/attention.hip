@@ -397,29 +397,32 @@ __global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
     const int* __restrict__ L_dev, float scale) {
   if constexpr (L_FROM_DEV) L = *L_dev;
   const int G = Hq / Hkv;
-  const int wg = blockIdx.x;
-  const int chunk = wg % nchunk;
-  const int hkv = (wg / nchunk) % Hkv;
-  const int b = wg / (nchunk * Hkv);
-  const int kv0 = chunk * DEC_CHUNK;
-  float* slot = ws + ((((long)b * Hkv + hkv) * nchunk + chunk) * DEC_GMAX) * (2 + D_HEAD);
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid >> 6;
-
-  if (kv0 >= L) {
-    for (int g = tid; g < G; g += blockDim.x) {
-      slot[g * (2 + D_HEAD)] = -1e30f;
-      slot[g * (2 + D_HEAD) + 1] = 0.f;
-    }
-    return;
-  }
-  const int kv_end = min(kv0 + DEC_CHUNK, L);
 
   // LDS: q rows (broadcast reads), per-lane p values, cross-wave merge
   __shared__ float q_lds[DEC_GMAX][D_HEAD];
   __shared__ float p_lds[4][DEC_GMAX][WAVE];
   __shared__ float sm[DEC_GMAX][4], sl[DEC_GMAX][4], so[DEC_GMAX][4][D_HEAD];
+
+  // persistent-WG remap: the grid is sized for the DEVICE (>=256 WGs, <=2048),
+  // not for B*Hkv*nchunk — each WG walks tuples with chunk fastest-varying and
+  // skips dead chunks (kv0 >= L) in-register.  The cache is allocated at Smax
+  // (8192) while decode typically runs at L~1k: a tuple-per-WG grid would be
+  // ~90% dead launches, which is what made the first cut 2x SLOWER end-to-end
+  // at b32 despite winning microbenchmarks.  Dead slots are never written;
+  // the combine pass only reads live chunks (ceil(L/DEC_CHUNK)).
+  const int total = B * Hkv * nchunk;
+  for (int idx = blockIdx.x; idx < total; idx += gridDim.x) {
+  const int chunk = idx % nchunk;
+  const int hkv = (idx / nchunk) % Hkv;
+  const int b = idx / (nchunk * Hkv);
+  const int kv0 = chunk * DEC_CHUNK;
+  if (kv0 >= L) continue;  // WG-uniform: L, idx uniform across the block
+  float* slot = ws + ((((long)b * Hkv + hkv) * nchunk + chunk) * DEC_GMAX) * (2 + D_HEAD);
+  const int kv_end = min(kv0 + DEC_CHUNK, L);
+
   for (int i = tid; i < G * D_HEAD; i += blockDim.x) {
     int g = i / D_HEAD, d = i % D_HEAD;
     q_lds[g][d] = bf2f(q[((long)b * Hq + hkv * G + g) * D_HEAD + d]);
@@ -513,6 +516,8 @@ __global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
       out[2 + lane * 2 + 1] = a1;
     }
   }
+  __syncthreads();  // protect q_lds/p_lds/merge LDS before the next tuple
+  }  // tuple loop
 }
 
 template <bool L_FROM_DEV>
@@ -630,9 +635,16 @@ extern "C" void launch_attn_decode(void* out, void* workspace, const void* q,
                                    const void* L_dev, float scale,
                                    hipStream_t stream) {
   const int G = Hq / Hkv;
-  if (workspace != nullptr && G <= DEC_GMAX) {
+  // Two-pass chunked decode exists to MAKE parallelism when B*Hq workgroups
+  // cannot fill 256 CUs (B<=8: 1.7-6x over single-pass).  Once the
+  // single-pass grid alone fills the chip (B*Hq >= 512 WGs) the chunked
+  // form only adds workspace traffic + a combine pass and measures ~2x
+  // SLOWER end-to-end (b32 Llama-3-8B: 9.1 vs 18.3 ms/step) — so route
+  // large batches to the single-pass kernel.
+  if (workspace != nullptr && G <= DEC_GMAX && B * Hq < 512) {
     const int nchunk = (Smax + DEC_CHUNK - 1) / DEC_CHUNK;
-    dim3 g1(B * Hkv * nchunk), b1(256);
+    const int total = B * Hkv * nchunk;
+    dim3 g1(total < 2048 ? total : 2048), b1(256);
     dim3 g2(B * Hq), b2(64);
     if (L_dev != nullptr) {
       hipLaunchKernelGGL((attn_decode_chunk_kernel<true>), g1, b1, 0, stream,

@@ -19,7 +19,7 @@ from dataclasses import dataclass, field
 
 from ..enums import OnTimeout, Phase, StepType
 from ..specs import types as T
-from ..templating import TemplateError, deps as tdeps
+from ..templating import OffloadedDataUsage, TemplateError, deps as tdeps
 from ..utils.durations import parse_duration
 from .records import StepState, StoryRun, StructuredError, monotonic_now
 from ..enums import ErrorType
@@ -544,7 +544,10 @@ class DAGReconciler:
         if run.phase == Phase.SUCCEEDED and story.output is not None:
             try:
                 scope = self.engine.build_scope(run, story)
-                out = self.engine.evaluator.resolve_value(story.output, scope)
+                if "__output__" in run.materialized:
+                    out = run.materialized["__output__"]
+                else:
+                    out = self.engine.evaluator.resolve_value(story.output, scope)
                 size = _json_size(out)
                 if size > self.engine.config.max_output_bytes:
                     run.degraded = True
@@ -555,6 +558,17 @@ class DAGReconciler:
                     )
                 else:
                     run.output = out
+            except OffloadedDataUsage:
+                # policy=block: delegate the output template to the
+                # materialize engram and finalize on its completion tick
+                status = self.engine.executor.delegate_output_materialize(run, story, scope)
+                if status == "pending":
+                    run.phase = Phase.RUNNING
+                    return
+                run.phase = Phase.FAILED
+                run.error = StructuredError(
+                    type=ErrorType.EXECUTION, message="output template materialization failed"
+                )
             except TemplateError as exc:
                 run.phase = Phase.FAILED
                 run.error = StructuredError(

@@ -733,6 +733,14 @@ class RunEngine:
         exit_class = classify_exit_code(exit_code)
         if exit_class == ExitClass.SUCCESS:
             self._complete_engram_success(run, sr, output)
+            if sr.spec.step_name.endswith("/materialize") and sr.status.phase == Phase.SUCCEEDED:
+                base = sr.spec.step_name.rsplit("/", 1)[0]
+                # the completion path may have re-offloaded the whole
+                # {"result": ...} envelope — unwrap it, then re-offload just
+                # the result so run.materialized stays marker-sized
+                out = self.storage.hydrate(sr.status.output)
+                result = out.get("result") if isinstance(out, dict) else None
+                run.materialized[base] = self.storage.dehydrate_document(result)
         else:
             self._handle_engram_failure(run, sr, exit_code, exit_class, error)
         self._tick(run)

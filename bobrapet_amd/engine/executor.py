@@ -118,7 +118,14 @@ class StepExecutor:
         try:
             step_with = eng.evaluator.resolve_value(step.with_, scope) if step.with_ is not None else None
         except OffloadedDataUsage:
-            raise
+            # policy=block: never evaluate over offloaded data on the engine
+            # loop — delegate to the materialize engram on a worker slot
+            # (reference: materialize.go resolveMaterialize:326)
+            if step.name in run.materialized:
+                step_with = run.materialized[step.name]
+            else:
+                self._delegate_materialize(run, story, step, scope, state)
+                return
         merged_input = step_with
         config = engram.with_
 
@@ -181,6 +188,86 @@ class StepExecutor:
         sr.status.phase = Phase.RUNNING
         sr.status.started_at = monotonic_now()
         eng.launch_engram_step(run, sr, template, resolved_cfg)
+
+    def _delegate_materialize(
+        self, run: StoryRun, story: T.Story, step: T.Step, scope: dict, state: StepState
+    ) -> None:
+        """Create (or observe) the aux materialize StepRun for a step whose
+        `with` templates touch `$storageRef` data under policy=block
+        (reference: ensureMaterializeStepRun materialize.go:142-240).  The
+        aux run's step_name carries a "/" so DAG state sync ignores it; its
+        result lands in run.materialized[step] and the step re-dispatches."""
+        eng = self.engine
+        sr_name = compose_name(run.name, f"{step.name}-mat")
+        sr_key = f"{run.namespace}/{sr_name}"
+        existing = eng.store.try_get_step_run(sr_key)
+        if existing is not None:
+            if existing.status.phase == Phase.SUCCEEDED:
+                # completion hook normally fills run.materialized; recover here
+                # after a snapshot restore where only the StepRun survived
+                out = eng.storage.hydrate(existing.status.output)
+                result = out.get("result") if isinstance(out, dict) else None
+                run.materialized[step.name] = eng.storage.dehydrate_document(result)
+                state.message = ""
+                return
+            if existing.is_terminal:  # failed/canceled materialization
+                state.phase = Phase.FAILED
+                state.finished_at = monotonic_now()
+                state.error = existing.status.error or StructuredError(
+                    type=ErrorType.EXECUTION, message="materialize step failed"
+                )
+            else:
+                state.message = "materializing offloaded inputs"
+            return
+        resolved_cfg = eng.resolver.resolve(step=step, story=story)
+        spec = StepRunSpec(
+            story_run=run.name,
+            step_name=f"{step.name}/materialize",
+            engram="builtin/materialize",
+            input={"mode": "value", "template": step.with_, "vars": dict(scope)},
+            timeout_seconds=resolved_cfg.timeout_seconds,
+            placement_gpu=self._place(run, step, resolved_cfg),
+        )
+        sr = StepRun(name=sr_name, namespace=run.namespace, spec=spec)
+        sr, created = eng.store.create_or_get_step_run(sr)
+        sr.status.phase = Phase.RUNNING
+        sr.status.started_at = monotonic_now()
+        state.message = "materializing offloaded inputs"
+        eng.metrics.inc("materialize_runs_total")
+        eng.launch_engram_step(run, sr, None, resolved_cfg)
+
+    def delegate_output_materialize(self, run: StoryRun, story: T.Story, scope: dict) -> str:
+        """Same delegation for the run-level output template
+        (reference: finalizeSuccessfulRun resolving output over offloaded
+        step data).  Returns "pending" while the aux run is in flight,
+        "failed" if it terminally failed; on success run.materialized
+        holds "__output__" and the caller re-finalizes."""
+        eng = self.engine
+        sr_name = compose_name(run.name, "output-mat")
+        sr_key = f"{run.namespace}/{sr_name}"
+        existing = eng.store.try_get_step_run(sr_key)
+        if existing is not None:
+            if existing.status.phase == Phase.SUCCEEDED:
+                out = eng.storage.hydrate(existing.status.output)
+                result = out.get("result") if isinstance(out, dict) else None
+                run.materialized["__output__"] = eng.storage.dehydrate_document(result)
+                return "pending"  # caller re-enters finalize next tick
+            return "failed" if existing.is_terminal else "pending"
+        resolved_cfg = eng.resolver.resolve(story=story)
+        spec = StepRunSpec(
+            story_run=run.name,
+            step_name="__output__/materialize",
+            engram="builtin/materialize",
+            input={"mode": "value", "template": story.output, "vars": dict(scope)},
+            timeout_seconds=resolved_cfg.timeout_seconds,
+        )
+        sr = StepRun(name=sr_name, namespace=run.namespace, spec=spec)
+        sr, _created = eng.store.create_or_get_step_run(sr)
+        sr.status.phase = Phase.RUNNING
+        sr.status.started_at = monotonic_now()
+        eng.metrics.inc("materialize_runs_total")
+        eng.launch_engram_step(run, sr, None, resolved_cfg)
+        return "pending"
 
     def _place(self, run: StoryRun, step: T.Step, cfg) -> _t.Optional[int]:
         """Deterministic (gpu) placement for the step (SURVEY.md §2.6:

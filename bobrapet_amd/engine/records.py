@@ -173,6 +173,9 @@ class StoryRun:
     step_states: _t.Dict[str, StepState] = field(default_factory=dict)
     gates: _t.Dict[str, GateStatus] = field(default_factory=dict)
     primitive_children: _t.Dict[str, _t.List[str]] = field(default_factory=dict)
+    # policy=block: delegated template results keyed by step name
+    # (reference: materialize.go readMaterializeResult 304-318)
+    materialized: _t.Dict[str, _t.Any] = field(default_factory=dict)
     trigger_tokens: _t.List[str] = field(default_factory=list)
     timers: _t.Dict[str, float] = field(default_factory=dict)  # durable deadlines
     output: JSON = None

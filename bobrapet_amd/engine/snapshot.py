@@ -82,6 +82,7 @@ def dump_state(engine) -> dict:
                     for k, g in run.gates.items()
                 },
                 "primitiveChildren": run.primitive_children,
+                "materialized": _clean(run.materialized),
                 "triggerTokens": run.trigger_tokens,
                 "timers": run.timers,
                 "output": _clean(run.output),
@@ -174,6 +175,7 @@ def load_state(engine, path: str) -> int:
                 st.phase = Phase.PENDING
                 st.started_at = None
             run.step_states[k] = st
+        run.materialized = rd.get("materialized") or {}
         for k, gd in rd.get("gates", {}).items():
             run.gates[k] = GateStatus(step=k, state=gd.get("state", "Pending"), decided_by=gd.get("decidedBy", ""))
         engine.store.create_story_run(run)

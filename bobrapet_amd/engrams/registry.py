@@ -75,5 +75,6 @@ def _load_builtins() -> None:
         from . import llm_infer  # noqa: F401
         from . import embed  # noqa: F401
         from . import join  # noqa: F401
+        from . import materialize  # noqa: F401
 
         _loaded = True

@@ -220,3 +220,99 @@ class TestTemplateFuzz:
         except TemplateSyntaxError:
             return  # random text that looks like {{...}} but is malformed
         assert out == value
+
+
+class TestMaterializeDelegation:
+    """policy=block delegates template evaluation over offloaded data to the
+    builtin materialize engram as aux StepRuns (reference: materialize.go
+    resolveMaterialize:326, ensureMaterializeStepRun:142-240)."""
+
+    def _engine(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import OffloadedDataPolicy
+
+        return RunEngine(
+            EngineConfig(
+                cpu_workers=2,
+                max_inline_size=64,
+                offloaded_data_policy=OffloadedDataPolicy.BLOCK,
+            )
+        ).start()
+
+    YAML = """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: echoer}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: mat}
+spec:
+  steps:
+    - name: big
+      ref: {name: echoer}
+      with: {blob: "%s"}
+    - name: use
+      ref: {name: echoer}
+      needs: [big]
+      with: {got: "{{ steps.big.output.blob }}"}
+  output:
+    l: "{{ steps.use.output.got }}"
+"""
+
+    def test_step_and_output_delegation(self):
+        from bobrapet_amd.enums import Phase
+
+        eng = self._engine()
+        try:
+            eng.apply_yaml(self.YAML % ("x" * 100))
+            run = eng.run_story("default/mat", {}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED, run.error
+            assert eng.storage.hydrate(run.output)["l"] == "x" * 100
+            names = sorted(sr.spec.step_name for sr in eng.store.step_runs_of(run.key))
+            # aux materialize runs exist for the step `with` and the output
+            assert "use/materialize" in names
+            assert "__output__/materialize" in names
+            # and the DAG state machine never saw them as steps
+            assert set(run.step_states) == {"big", "use"}
+            assert "use" in run.materialized and "__output__" in run.materialized
+        finally:
+            eng.stop()
+
+    def test_small_payloads_stay_inline_no_materialize(self):
+        from bobrapet_amd.enums import Phase
+
+        eng = self._engine()
+        try:
+            eng.apply_yaml(self.YAML % "tiny")
+            run = eng.run_story("default/mat", {}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED, run.error
+            assert run.output == {"l": "tiny"}
+            names = [sr.spec.step_name for sr in eng.store.step_runs_of(run.key)]
+            assert not any("materialize" in n for n in names)
+        finally:
+            eng.stop()
+
+    def test_materialize_survives_snapshot(self, tmp_path):
+        from bobrapet_amd.enums import Phase
+
+        eng = self._engine()
+        try:
+            eng.apply_yaml(self.YAML % ("y" * 100))
+            run = eng.run_story("default/mat", {}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED
+            path = str(tmp_path / "snap.json")
+            eng.save_state(path)
+        finally:
+            eng.stop()
+        eng2 = self._engine()
+        try:
+            eng2.apply_yaml(self.YAML % ("y" * 100))
+            eng2.load_state(path)
+            r2 = eng2.store.try_get_story_run(run.key)
+            assert r2 is not None and "use" in r2.materialized
+        finally:
+            eng2.stop()

@@ -103,13 +103,11 @@ class _Stage(threading.Thread):
         import contextlib
 
         while True:
-            packets = []
-            for ring in self.in_rings:
-                pkt = ring.pop()
-                if pkt is flow.SENTINEL:
-                    return
-                packets.append(pkt)
-            packet = packets[0] if len(packets) == 1 else {"fanIn": packets}
+            packet = self._pop_fan_in()
+            if packet is flow.SENTINEL:
+                return
+            if packet is None:  # dropped join round (quorum/timeout)
+                continue
             if self.sr.canceled:
                 return
             t0 = time.monotonic()
@@ -131,6 +129,83 @@ class _Stage(threading.Thread):
                 self.sr.on_packet(self.step.name, leaf=not self.out_rings, packet=out)
 
     # ------------------------------------------------------------------
+
+    def _pop_fan_in(self):
+        """Join packets from the upstream rings per the transport fan-in
+        settings (reference: TransportFanInSettings
+        transport_settings_types.go:174-192 — modes all|any|quorum).
+
+        Joins are arrival-round based (one packet per ring per round), not
+        envelope-ID based: with `timeoutSeconds` a straggler's packet stays
+        queued and joins the NEXT round — at-least-once, perLane-ordered
+        semantics over in-process rings.  Returns the merged packet,
+        ``None`` for a dropped round, or SENTINEL at end-of-stream."""
+        rings = self.in_rings
+        if len(rings) == 1:
+            return rings[0].pop()
+        fi = self.sr.settings.fan_in
+        mode = fi.mode if fi is not None else "all"
+        timeout = fi.timeout_seconds if fi is not None else None
+
+        if mode == "any":
+            # first packet from any live upstream wins; round-robin poll
+            live = [r for r in rings if r not in getattr(self, "_done_rings", set())]
+            if not hasattr(self, "_done_rings"):
+                self._done_rings = set()
+                live = list(rings)
+            while True:
+                progressed = False
+                for ring in list(live):
+                    try:
+                        pkt = ring.pop(timeout=0.005)
+                    except TimeoutError:
+                        continue
+                    if pkt is flow.SENTINEL:
+                        self._done_rings.add(ring)
+                        live.remove(ring)
+                        progressed = True
+                        continue
+                    return pkt
+                if not live:
+                    return flow.SENTINEL
+                if not progressed:
+                    time.sleep(0)  # yield between poll sweeps
+
+        # all / quorum: one packet per ring per round
+        need = len(rings)
+        if mode == "quorum":
+            need = max(1, min(int(fi.quorum or len(rings)), len(rings)))
+        deadline = (time.monotonic() + timeout) if timeout else None
+        closed = getattr(self, "_closed_rings", None)
+        if closed is None:
+            closed = self._closed_rings = set()
+        packets, arrived_from = [], []
+        for i, ring in enumerate(rings):
+            if ring in closed:
+                continue
+            try:
+                if deadline is None:
+                    pkt = ring.pop()
+                else:
+                    pkt = ring.pop(timeout=max(deadline - time.monotonic(), 0.001))
+            except TimeoutError:
+                continue
+            if pkt is flow.SENTINEL:
+                closed.add(ring)
+                continue
+            packets.append(pkt)
+            arrived_from.append(i)
+        if len(rings) - len(closed) < need and len(packets) < need:
+            # quorum can never be met again (upstreams ended) — end of stream
+            return flow.SENTINEL
+        if len(packets) < need:
+            self.stats.errors += 0  # dropped round, not an error
+            self.sr.engine.metrics.inc("stream_fanin_dropped_total")
+            return None
+        merged = {"fanIn": packets}
+        if len(packets) < len(rings):
+            merged["arrivedFrom"] = arrived_from
+        return merged
 
     def _process(self, packet):
         eng = self.sr.engine

@@ -636,15 +636,28 @@ class TransportBackpressure:
 
 
 @dataclass
+class TransportFanIn:
+    """Fan-in join behavior when multiple upstreams feed a step
+    (reference: transport_settings_types.go:174-192, modes all|any|quorum)."""
+
+    mode: str = "all"
+    quorum: _t.Optional[int] = None
+    timeout_seconds: _t.Optional[int] = None
+    max_entries: _t.Optional[int] = None
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
 class TransportStreamingSettings:
     """Streaming policy vocabulary (reference:
     transport_settings_types.go:68-110): lanes, flow control, backpressure,
-    delivery/ordering semantics."""
+    delivery/ordering semantics, fan-in joins."""
 
     lanes: _t.List[TransportLane] = field(default_factory=list)
     flow_control: _t.Optional[TransportFlowControl] = None
     delivery: _t.Optional[TransportDelivery] = None
     backpressure: _t.Optional[TransportBackpressure] = None
+    fan_in: _t.Optional[TransportFanIn] = None
     extra: dict = field(default_factory=dict)
 
 

@@ -360,3 +360,70 @@ spec:
         assert run.step_states["work"].output == {"got": "hi"}
         assert client.get("/healthz").json() == {"ok": True}
         assert "bobrapet_amd_" in client.get("/metrics").text
+
+
+class TestFanInModes:
+    """TransportFanInSettings all|any|quorum (reference:
+    transport_settings_types.go:174-192)."""
+
+    FANIN_STORY = """
+kind: Story
+metadata: {name: fin}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        fanIn: {%s}
+  steps:
+    - name: left
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v }}", side: left}}
+    - name: right
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v }}", side: right}}
+    - name: join
+      ref: {name: echoer}
+      needs: [left, right]
+"""
+
+    def test_fan_in_all_default(self, eng):
+        eng.apply_yaml(self.FANIN_STORY % "mode: all")
+        stream = eng.submit_stream("default/fin")
+        for i in range(4):
+            stream.push({"items": [{"v": i}]})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # every round joins both upstreams → join sees 4 merged packets
+        assert run.step_states["join"].output["packetsIn"] == 4
+
+    def test_fan_in_any_emits_per_packet(self, eng):
+        eng.apply_yaml(self.FANIN_STORY % "mode: any")
+        stream = eng.submit_stream("default/fin")
+        for i in range(3):
+            stream.push({"items": [{"v": i}]})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # each upstream packet flows through individually: 3 left + 3 right
+        assert run.step_states["join"].output["packetsIn"] == 6
+
+    def test_fan_in_quorum_of_one(self, eng):
+        eng.apply_yaml(self.FANIN_STORY % "mode: quorum, quorum: 1, timeoutSeconds: 5")
+        stream = eng.submit_stream("default/fin")
+        for i in range(3):
+            stream.push({"items": [{"v": i}]})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # both upstreams are prompt, so every round still joins 2 packets
+        assert run.step_states["join"].output["packetsIn"] == 3
+
+    def test_settings_parse_fan_in(self):
+        from bobrapet_amd.specs.types import TransportStreamingSettings, from_dict
+
+        s = from_dict(
+            TransportStreamingSettings,
+            {"fanIn": {"mode": "quorum", "quorum": 2, "timeoutSeconds": 7}},
+        )
+        assert s.fan_in.mode == "quorum"
+        assert s.fan_in.quorum == 2
+        assert s.fan_in.timeout_seconds == 7

@@ -348,6 +348,7 @@ class DAGReconciler:
 
         failure_present = run.failure_cause is not None or self._has_failure(run, story)
         progressed = False
+        launched = 0
         scope = None
         for step in steps:
             state = run.step_states.get(step.name)
@@ -379,8 +380,16 @@ class DAGReconciler:
                 st.phase = Phase.PENDING
             self.engine.executor.execute(run, story, step, scope)
             self.engine.metrics.inc("dag_steps_launched_total")
+            launched += 1
             progressed = True
             scope = None if st.phase.is_terminal else scope  # refresh after immediate completion
+        if launched:
+            # (reference: bobrapet_dag_iteration_steps histogram)
+            self.engine.metrics.observe("dag_iteration_steps", launched)
+        active = sum(1 for st in run.step_states.values() if st.phase == Phase.RUNNING)
+        done = sum(1 for st in run.step_states.values() if st.phase.is_terminal)
+        self.engine.metrics.set_gauge("storyrun_steps_active", active)
+        self.engine.metrics.set_gauge("storyrun_steps_completed", done)
         return progressed
 
     def _select_phase(

@@ -79,6 +79,7 @@ class RunEngine:
         self.cache = StepCache(evaluator=self.evaluator)
         self.effects = EffectLedger()
         self.metrics = metrics if metrics is not None else metrics_mod.MetricsRegistry()
+        self.evaluator.metrics = self.metrics
         self.tracer = tracer if tracer is not None else tracing_mod.Tracer(enabled=False)
         self.workers = WorkerPool(
             device_count=device_count,
@@ -647,6 +648,15 @@ class RunEngine:
             return r.priority + (now - r.created_at) / aging
 
         self._scheduling.sort(key=effective_priority, reverse=True)
+        # queue observability (reference: storyrun_queue_depth /
+        # storyrun_queue_age_seconds pkg/metrics)
+        depth: _t.Dict[str, int] = {}
+        for key in self._scheduling:
+            r = self.store.try_get_story_run(key)
+            if r is not None:
+                depth[r.queue or "default"] = depth.get(r.queue or "default", 0) + 1
+        for q, d in depth.items():
+            self.metrics.set_gauge("storyrun_queue_depth", d, queue=q)
         admitted = []
         for key in list(self._scheduling):
             r = self.store.try_get_story_run(key)
@@ -673,6 +683,9 @@ class RunEngine:
                 continue
             # admitted
             admitted.append(key)
+            self.metrics.observe(
+                "storyrun_queue_age_seconds", now - r.created_at, queue=r.queue or "default"
+            )
             global_running += 1
             per_queue[r.queue] = per_queue.get(r.queue, 0) + 1
             per_story[story_key] = per_story.get(story_key, 0) + 1
@@ -926,8 +939,10 @@ class RunEngine:
         if tag == "ttl" and run.children_cleaned_at is None:
             self.store.delete_steps_of(run.key)
             run.children_cleaned_at = monotonic_now()
+            self.metrics.inc("resource_cleanup_total", kind="children")
         elif tag == "retention":
             if run.children_cleaned_at is None:
                 self.store.delete_steps_of(run.key)
             self.store.delete_story_run(run.key)
             self._run_done.pop(run.key, None)
+            self.metrics.inc("resource_cleanup_total", kind="storyrun")

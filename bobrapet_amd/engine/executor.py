@@ -382,6 +382,7 @@ class StepExecutor:
         eng = self.engine
         branch_scope = dict(scope)
         branch_scope["branch"] = {"name": branch.name, "parent": parent.name}
+        eng.metrics.inc("child_stepruns_created_total")
         spec = StepRunSpec(story_run=run.name, step_name=f"{parent.name}/{branch.name}")
         sr = StepRun(name=child_name, namespace=run.namespace, spec=spec)
         sr, created = eng.store.create_or_get_step_run(sr)

@@ -381,6 +381,8 @@ class StreamingRun:
                 upstream=[e.src for e in self.topo.upstream_of(name) if e.src],
                 downstream=[e.dst for e in self.topo.downstream_of(name)],
             )
+        engine.metrics.set_gauge("transport_bindings_total", len(self.bindings))
+        engine.metrics.set_gauge("transport_bindings_ready", len(self.bindings))
         run.phase = Phase.RUNNING
         run.started_at = run.started_at or monotonic_now()
         for s in self.stages:

@@ -117,6 +117,7 @@ class TriggerAdmission:
                 if in_flight >= throttle.max_in_flight:
                     trigger.decision = TriggerDecision.REJECTED
                     trigger.message = f"maxInFlight {throttle.max_in_flight} reached"
+                    eng.metrics.inc("impulse_throttled_triggers_total", reason="maxInFlight")
                     return trigger
             if throttle.rate_per_second:
                 limiter = self._limiters.setdefault(

@@ -52,15 +52,29 @@ CONFIGS: _t.Dict[str, LlamaConfig] = {
         head_dim=128,
         max_seq_len=512,
     ),
+    # 1B-class config (head_dim held at 128 — the CDNA4 attention kernels
+    # are D=128 specialized; 16x128 spans the same hidden width)
     "llama-3-1b": LlamaConfig(
         name="llama-3-1b",
         vocab_size=128256,
         hidden_size=2048,
         intermediate_size=8192,
         num_layers=16,
-        num_heads=32,
+        num_heads=16,
+        num_kv_heads=4,
+        head_dim=128,
+    ),
+    # single-GPU 70B: 141 GB of bf16 weights resident in 288 GB HBM3E —
+    # no tensor parallelism needed for b1 serving on one MI355X
+    "llama-3-70b": LlamaConfig(
+        name="llama-3-70b",
+        vocab_size=128256,
+        hidden_size=8192,
+        intermediate_size=28672,
+        num_layers=80,
+        num_heads=64,
         num_kv_heads=8,
-        head_dim=64,
+        head_dim=128,
     ),
 }
 
@@ -78,11 +92,14 @@ class LlamaLayerWeights:
             return t
 
         self.ln_attn = torch.ones(H, device=device, dtype=dtype)
-        self.w_qkv = mk(H, cfg.qkv_out)
-        self.w_o = mk(cfg.num_heads * cfg.head_dim, H)
+        # all projection weights live in [out,in] layout: the decode GEMV
+        # streams W rows directly (no transposed copies, works at 70B), and
+        # prefill passes .t() views to hipBLASLt (native NT GEMM, no copy)
+        self.w_qkv = mk(cfg.qkv_out, H)
+        self.w_o = mk(H, cfg.num_heads * cfg.head_dim)
         self.ln_mlp = torch.ones(H, device=device, dtype=dtype)
-        self.w_gate_up = mk(H, 2 * I)
-        self.w_down = mk(I, H)
+        self.w_gate_up = mk(2 * I, H)
+        self.w_down = mk(H, I)
 
 
 class LlamaModel:
@@ -113,7 +130,7 @@ class LlamaModel:
         ]
         self.ln_final = torch.ones(cfg.hidden_size, device=self.device, dtype=dtype)
         self.lm_head = torch.empty(
-            cfg.hidden_size, cfg.vocab_size, device=self.device, dtype=dtype
+            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=dtype
         ).normal_(0.0, std, generator=gen)
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
         self._kv_cache: _t.Optional[_t.List[_t.Tuple[torch.Tensor, torch.Tensor]]] = None
@@ -122,7 +139,16 @@ class LlamaModel:
         self._g_ids = None  # static graph buffers
         self._g_pos = None
         self._g_logits = None
-        self._wt: _t.Dict[int, tuple] = {}  # transposed weights for GEMV decode
+        per_layer = (
+            2 * cfg.hidden_size
+            + cfg.hidden_size * cfg.qkv_out
+            + cfg.num_heads * cfg.head_dim * cfg.hidden_size
+            + 3 * cfg.hidden_size * cfg.intermediate_size
+        )
+        n_params = (
+            2 * cfg.vocab_size * cfg.hidden_size + cfg.num_layers * per_layer + cfg.hidden_size
+        )
+        self.param_bytes = n_params * self.embed.element_size()
 
     # ------------------------------------------------------------------
 
@@ -176,7 +202,7 @@ class LlamaModel:
         for li, lw in enumerate(self.layers):
             # attention block — zero layout copies: rope reads strided q/k
             # heads straight from the qkv projection; v stays a strided view
-            qkv = torch.matmul(x, lw.w_qkv)  # [B,S,qkv_out]
+            qkv = torch.matmul(x, lw.w_qkv.t())  # [B,S,qkv_out]
             qh, kh, vh = ops.rope_qkv_split(
                 qkv, B, S, cfg.num_heads, cfg.num_kv_heads, cfg.head_dim, cos_f, sin_f
             )
@@ -186,13 +212,13 @@ class LlamaModel:
                 vc[:, :, :S] = vh.transpose(1, 2)
             attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
             attn = attn.reshape(B, S, cfg.num_heads * cfg.head_dim)
-            attn_out = torch.matmul(attn, lw.w_o)
+            attn_out = torch.matmul(attn, lw.w_o.t())
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
 
             # mlp block — silu reads gate/up halves in place
-            gate_up = torch.matmul(x, lw.w_gate_up)
+            gate_up = torch.matmul(x, lw.w_gate_up.t())
             act = ops.silu_mul_fused(gate_up)
-            mlp_out = torch.matmul(act, lw.w_down)
+            mlp_out = torch.matmul(act, lw.w_down.t())
             next_norm = (
                 self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
             )
@@ -201,31 +227,20 @@ class LlamaModel:
         if fill_cache:
             self._cache_len = S
         if logits_for_all:
-            return torch.matmul(x, self.lm_head)
-        return torch.matmul(x[:, -1], self.lm_head)
+            return torch.matmul(x, self.lm_head.t())
+        return torch.matmul(x[:, -1], self.lm_head.t())
 
     # ------------------------------------------------------------------
 
-    def _decode_mm(self, x2d: torch.Tensor, w: torch.Tensor, wt: torch.Tensor):
-        """Decode projection: batch-1 rows go through the hand-written
-        weight-streaming GEMV (≈6-7 TB/s vs ~2.6-6.6 for the library at
-        these skinny shapes); larger batches use hipBLASLt."""
+    def _decode_mm(self, x2d: torch.Tensor, w: torch.Tensor):
+        """Decode projection over [out,in] weights: batch-1 rows go through
+        the hand-written weight-streaming GEMV (≈6-7 TB/s vs ~2.6-6.6 for
+        the library at these skinny shapes — no transposed copies needed,
+        the row-major [out,in] layout IS the streaming layout, so this works
+        at 70B too); larger batches use hipBLASLt (native NT)."""
         if x2d.shape[0] == 1 and x2d.is_cuda and x2d.shape[1] % 8 == 0:
-            return ops.gemv_nt(x2d, wt)
-        return torch.matmul(x2d, w)
-
-    def _decode_weights(self, li: int):
-        """Transposed [out,in] copies for the GEMV path (built once; the
-        extra 16 GB sits comfortably in 288 GB HBM)."""
-        if li not in self._wt:
-            lw = self.layers[li]
-            self._wt[li] = (
-                lw.w_qkv.t().contiguous(),
-                lw.w_o.t().contiguous(),
-                lw.w_gate_up.t().contiguous(),
-                lw.w_down.t().contiguous(),
-            )
-        return self._wt[li]
+            return ops.gemv_nt(x2d, w)
+        return torch.matmul(x2d, w.t())
 
     def _decode_body(self, ids: torch.Tensor, pos_i32: torch.Tensor) -> torch.Tensor:
         """One decode step with all dynamic state in device tensors — every
@@ -243,9 +258,8 @@ class LlamaModel:
         L_dev = (pos_i32 + 1).contiguous()
         use_gemv = B == 1 and self.device.type == "cuda"
         for li, lw in enumerate(self.layers):
-            wt = self._decode_weights(li) if use_gemv else (None,) * 4
             x2 = x.view(B, cfg.hidden_size)
-            qkv = self._decode_mm(x2, lw.w_qkv, wt[0]).view(B, 1, -1) if use_gemv else torch.matmul(x, lw.w_qkv)
+            qkv = self._decode_mm(x2, lw.w_qkv).view(B, 1, -1) if use_gemv else torch.matmul(x, lw.w_qkv.t())
             q, k, v = self._split_qkv(qkv, B, 1)
             qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
             kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
@@ -257,32 +271,30 @@ class LlamaModel:
             attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
             a2 = attn.reshape(B, cfg.num_heads * cfg.head_dim)
             attn_out = (
-                self._decode_mm(a2, lw.w_o, wt[1]).view(B, 1, -1)
+                self._decode_mm(a2, lw.w_o).view(B, 1, -1)
                 if use_gemv
-                else torch.matmul(attn.reshape(B, 1, -1), lw.w_o)
+                else torch.matmul(attn.reshape(B, 1, -1), lw.w_o.t())
             )
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
             x2 = x.view(B, cfg.hidden_size)
             gate_up = (
-                self._decode_mm(x2, lw.w_gate_up, wt[2]).view(B, 1, -1)
+                self._decode_mm(x2, lw.w_gate_up).view(B, 1, -1)
                 if use_gemv
-                else torch.matmul(x, lw.w_gate_up)
+                else torch.matmul(x, lw.w_gate_up.t())
             )
             act = ops.silu_mul_fused(gate_up)
             mlp_out = (
-                self._decode_mm(act.view(B, -1), lw.w_down, wt[3]).view(B, 1, -1)
+                self._decode_mm(act.view(B, -1), lw.w_down).view(B, 1, -1)
                 if use_gemv
-                else torch.matmul(act, lw.w_down)
+                else torch.matmul(act, lw.w_down.t())
             )
             next_norm = (
                 self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
             )
             x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
         if use_gemv:
-            if "lm" not in self._wt:
-                self._wt["lm"] = self.lm_head.t().contiguous()
-            return ops.gemv_nt(x.view(B, cfg.hidden_size), self._wt["lm"])
-        return torch.matmul(x[:, -1], self.lm_head)
+            return ops.gemv_nt(x.view(B, cfg.hidden_size), self.lm_head)
+        return torch.matmul(x[:, -1], self.lm_head.t())
 
     @torch.no_grad()
     def decode_step_graphed(self, ids: torch.Tensor) -> torch.Tensor:
@@ -349,7 +361,7 @@ class LlamaModel:
         residual = self.embed[ids.to(self.device)].view(B, 1, cfg.hidden_size)
         x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
         for li, lw in enumerate(self.layers):
-            qkv = torch.matmul(x, lw.w_qkv)
+            qkv = torch.matmul(x, lw.w_qkv.t())
             q, k, v = self._split_qkv(qkv, B, 1)
             qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
             kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
@@ -358,18 +370,18 @@ class LlamaModel:
             kc[:, :, pos] = kf.view(B, cfg.num_kv_heads, cfg.head_dim)
             vc[:, :, pos] = v.view(B, cfg.num_kv_heads, cfg.head_dim)
             attn = ops.attn_decode(qf, kc, vc, pos + 1, self.scale)  # [B,Hq,D]
-            attn_out = torch.matmul(attn.reshape(B, 1, -1), lw.w_o)
+            attn_out = torch.matmul(attn.reshape(B, 1, -1), lw.w_o.t())
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
-            gate_up = torch.matmul(x, lw.w_gate_up)
+            gate_up = torch.matmul(x, lw.w_gate_up.t())
             gate, up = gate_up.chunk(2, dim=-1)
             act = ops.silu_mul(gate.contiguous(), up.contiguous())
-            mlp_out = torch.matmul(act, lw.w_down)
+            mlp_out = torch.matmul(act, lw.w_down.t())
             next_norm = (
                 self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
             )
             x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
         self._cache_len = pos + 1
-        return torch.matmul(x[:, -1], self.lm_head)
+        return torch.matmul(x[:, -1], self.lm_head.t())
 
     @torch.no_grad()
     def generate(

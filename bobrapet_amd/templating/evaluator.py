@@ -89,6 +89,9 @@ class Evaluator:
     config: EvalConfig = field(default_factory=EvalConfig)
     hydrator: _t.Optional[_t.Callable[[dict], _t.Any]] = None  # $storageRef → value
     _cache: _t.Dict[str, CompiledTemplate] = field(default_factory=dict)
+    # optional MetricsRegistry (reference: bobrapet_cel_evaluation_total /
+    # _duration_seconds / _cache_hits_total) — set by the engine
+    metrics: _t.Optional[_t.Any] = None
 
     # -- public API ---------------------------------------------------------
 
@@ -163,6 +166,10 @@ class Evaluator:
             tpl = parse_template(text)
             if len(self._cache) < 4096:
                 self._cache[text] = tpl
+        elif self.metrics is not None:
+            self.metrics.inc("template_cache_hits_total")
+        if self.metrics is not None:
+            self.metrics.inc("template_evaluations_total")
         return tpl
 
     def _check_size(self, value) -> None:

@@ -91,3 +91,56 @@ class TestJsonSchema:
             },
         }
         assert apply_defaults({"outer": {}}, schema) == {"outer": {"inner": "d"}}
+
+
+class TestMetricsParitySeries:
+    """The reference's meaningful bobrapet_* series have counterparts
+    (reference: pkg/metrics/controller_metrics.go — queue depth/age,
+    dag_iteration_steps, steps active/completed, cel eval counters,
+    child stepruns, cleanup)."""
+
+    def test_series_emitted_by_engine_run(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: echoer}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: m}
+spec:
+  steps:
+    - name: par
+      type: parallel
+      with:
+        steps:
+          - {name: a, ref: {name: echoer}, with: {v: "{{ inputs.x }}"}}
+          - {name: b, ref: {name: echoer}, with: {v: 2}}
+"""
+            )
+            run = eng.run_story("default/m", {"x": 1}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED, run.error
+            text = eng.metrics.export_text()
+            for series in (
+                "storyruns_total",
+                "stepruns_total",
+                "dag_steps_launched_total",
+                "dag_iteration_steps",
+                "storyrun_steps_completed",
+                "storyrun_queue_depth",
+                "storyrun_queue_age_seconds",
+                "template_evaluations_total",
+                "child_stepruns_created_total",
+            ):
+                assert series in text, f"missing series {series}\n{text[:800]}"
+        finally:
+            eng.stop()

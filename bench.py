@@ -99,9 +99,15 @@ spec:
       needs: [fanout]
       with:
         branches: "{{{{ steps.fanout.output.branches }}}}"
+        commSlot: "{{{{ inputs.slot }}}}"
   output:
     rows: "{{{{ steps.join.output.worldRows }}}}"
 """
+
+# in-flight pipelining window for the parallel8 throughput config: W
+# stories overlap (engine + GPU streams); story i runs on comm slot i%W so
+# cross-rank all-gathers stay ordered per communicator
+INFLIGHT = 4
 
 SLEEP_STORY = """
 kind: Story
@@ -167,9 +173,10 @@ spec:
 """
 
 
-def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None) -> dict:
+def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None, slot=None) -> dict:
+    inputs = {"i": idx, "rank": rank, "slot": slot if slot is not None else 0}
     if native is not None:
-        status = native.run_story(story_key, {"i": idx, "rank": rank}, timeout=600)
+        status = native.run_story(story_key, inputs, timeout=600)
         if status["phase"] != "Succeeded":
             raise RuntimeError(f"native bench run failed: {status}")
         lat = {}
@@ -177,7 +184,7 @@ def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None) ->
             if st.get("startedAt") and st.get("finishedAt"):
                 lat[name] = (st["finishedAt"] - st["startedAt"]) * 1000.0
         return lat
-    run = eng.submit_run(story_key, {"i": idx, "rank": rank}, name=f"bench-{rank}-{idx}")
+    run = eng.submit_run(story_key, inputs, name=f"bench-{rank}-{idx}")
     run = eng.wait(run, timeout=600)
     if run.phase != Phase.SUCCEEDED:
         states = {k: (str(v.phase), v.message, str(v.error)) for k, v in run.step_states.items()}
@@ -266,7 +273,8 @@ def main() -> int:
     has_gpu = torch.cuda.is_available()
     device = torch.cuda.current_device() if has_gpu else None
 
-    if args.config == "cpu" or (not has_gpu and args.config in ("parallel8", "llm")):
+    force = os.environ.get("BOBRA_BENCH_NO_CPU_FALLBACK") == "1"
+    if args.config == "cpu" or (not has_gpu and not force and args.config in ("parallel8", "llm")):
         config_name = "cpu"
         story_key = "default/bench-cpu"
         model_desc = "2-step batch story (sleep->condition), CPU engine"
@@ -325,18 +333,39 @@ def main() -> int:
                     raise
                 native = None
 
-        # warmup (untimed): fills weight/table caches, compiles nothing
-        for i in range(args.warmup):
-            run_one(eng, story_key, -(i + 1), rank, native)
+        inflight = INFLIGHT if config_name == "parallel8" else 1
+        if inflight > 1:
+            group.ensure_comm_slots(inflight)  # collective; same order on all ranks
+
+        def run_span(first: int, count: int) -> list:
+            """Run `count` stories starting at `first`; W in-flight slot
+            threads, slot s strictly sequential over indices s, s+W, ...
+            (collective-order safety: see ensure_comm_slots)."""
+            if inflight <= 1:
+                lats: list = []
+                for i in range(first, first + count):
+                    lats.extend(run_one(eng, story_key, i, rank, native).values())
+                return lats
+            import concurrent.futures as cf
+
+            def slot_main(slot: int) -> list:
+                out: list = []
+                for i in range(first + slot, first + count, inflight):
+                    out.extend(run_one(eng, story_key, i, rank, native, slot=slot).values())
+                return out
+
+            with cf.ThreadPoolExecutor(max_workers=inflight) as ex:
+                return [v for f in [ex.submit(slot_main, w) for w in range(inflight)]
+                        for v in f.result()]
+
+        # warmup (untimed): fills weight/table caches + comm-slot communicators
+        run_span(1_000_000, max(args.warmup, inflight))
 
         group.barrier()
         if has_gpu:
             torch.cuda.synchronize()
         t0 = time.monotonic()
-        step_lat: list = []
-        for i in range(args.steps):
-            lat = run_one(eng, story_key, i, rank, native)
-            step_lat.extend(lat.values())
+        step_lat = run_span(0, args.steps)
         group.barrier()
         if has_gpu:
             torch.cuda.synchronize()

@@ -79,7 +79,7 @@ class EmbedEngram(Engram):
         t0 = time.monotonic()
         emb = ops.embed_pool(table, ids_t)
         if table.is_cuda:
-            torch.cuda.synchronize(table.device)
+            torch.cuda.current_stream(table.device).synchronize()
         latency_ms = (time.monotonic() - t0) * 1000.0
 
         out = {

@@ -49,10 +49,13 @@ class AllGatherJoinEngram(Engram):
 
         t0 = time.monotonic()
         local = torch.cat([t.reshape(-1, t.shape[-1]) for t in tensors], dim=0)
-        gathered = collectives.all_gather_tensor(local)  # [world, rows, dim]
+        pg = group.comm_slot(inp.get("commSlot"))
+        gathered = collectives.all_gather_tensor(local, pg)  # [world, rows, dim]
         joined = gathered.reshape(-1, local.shape[-1])
         if joined.is_cuda:
-            torch.cuda.synchronize(joined.device)
+            # slot-stream sync only: a device-wide sync would stall the
+            # sibling branch streams running concurrently on this GPU
+            torch.cuda.current_stream(joined.device).synchronize()
         latency_ms = (time.monotonic() - t0) * 1000.0
 
         out = {

@@ -14,11 +14,13 @@ import torch
 import torch.distributed as dist
 
 
-def all_gather_tensor(t: torch.Tensor) -> torch.Tensor:
+def all_gather_tensor(t: torch.Tensor, pg=None) -> torch.Tensor:
     """Gather a same-shape tensor from every rank → stacked [world, ...].
 
     all_gather_into_tensor requires the output's FIRST dim to be
-    world * input_first_dim (flat concat) — allocate flat, then view."""
+    world * input_first_dim (flat concat) — allocate flat, then view.
+    `pg` selects a comm-slot communicator (group.comm_slot) so concurrent
+    stories can all-gather without cross-rank order hazards."""
     if not dist.is_initialized() or dist.get_world_size() == 1:
         return t.unsqueeze(0)
     world = dist.get_world_size()
@@ -26,7 +28,7 @@ def all_gather_tensor(t: torch.Tensor) -> torch.Tensor:
     flat = torch.empty(
         (world * src.shape[0],) + tuple(src.shape[1:]), dtype=src.dtype, device=src.device
     )
-    dist.all_gather_into_tensor(flat, src)
+    dist.all_gather_into_tensor(flat, src, group=pg)
     return flat.view((world,) + tuple(src.shape))
 
 

@@ -49,7 +49,33 @@ def init_distributed(backend: _t.Optional[str] = None, timeout_s: float = 300.0)
     return True
 
 
+_SLOT_GROUPS: _t.List = []
+
+
+def ensure_comm_slots(n: int) -> None:
+    """Create n extra all-rank communicators (collective call — every rank
+    must call with the same n, in the same order).  Slot communicators make
+    CONCURRENT in-flight stories safe: a story pinned to slot s only issues
+    collectives on communicator s, per-slot submission is sequential, so
+    the cross-rank operation order matches per communicator even when W
+    stories overlap (xGMI has no global stream order to rely on)."""
+    if not dist.is_initialized() or dist.get_world_size() <= 1:
+        return
+    ranks = list(range(dist.get_world_size()))
+    while len(_SLOT_GROUPS) < n:
+        _SLOT_GROUPS.append(dist.new_group(ranks=ranks))
+
+
+def comm_slot(i: _t.Optional[int]):
+    """The process group for in-flight slot i (None → default group)."""
+    if i is None or not _SLOT_GROUPS:
+        return None
+    return _SLOT_GROUPS[int(i) % len(_SLOT_GROUPS)]
+
+
 def teardown() -> None:
+    global _SLOT_GROUPS
+    _SLOT_GROUPS = []
     if dist.is_initialized():
         dist.destroy_process_group()
 

@@ -140,3 +140,34 @@ def test_parallel_join_two_ranks_lockstep():
         # 4 branches x batch 2 x 2 ranks = 16 joined rows on every rank
         assert payload["rows"] == [16, 16, 16], payload
     assert results[0]["agg"] == results[1]["agg"]
+
+
+@pytest.mark.timeout(300)
+def test_bench_parallel8_pipelined_two_ranks():
+    """Rehearse the driver's multi-rank bench launch exactly: torchrun world=2
+    (gloo on CPU), the REAL parallel8 story (no cpu fallback), pipelined
+    in-flight slots + comm-slot communicators for the all-gather join."""
+    port = _free_port()
+    env = dict(os.environ)
+    env.update({"BOBRA_BENCH_NO_CPU_FALLBACK": "1"})
+    out = subprocess.run(
+        [
+            sys.executable, "-m", "torch.distributed.run",
+            "--nnodes=1", "--nproc-per-node", "2",
+            "--master-addr", "127.0.0.1", "--master-port", str(port),
+            "bench.py", "--gpus", "2", "--steps", "6", "--warmup", "1",
+            "--config", "parallel8",
+        ],
+        cwd=REPO,
+        env=env,
+        capture_output=True,
+        text=True,
+        timeout=280,
+    )
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-2500:])
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert lines, out.stdout[-1500:]
+    line = json.loads(lines[-1])
+    assert line["n_gpus"] == 2
+    assert line["value"] > 0
+    assert line["config"]["branches"] == 8

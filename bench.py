@@ -107,7 +107,7 @@ spec:
 # in-flight pipelining window for the parallel8 throughput config: W
 # stories overlap (engine + GPU streams); story i runs on comm slot i%W so
 # cross-rank all-gathers stay ordered per communicator
-INFLIGHT = 4
+INFLIGHT = int(os.environ.get("BOBRA_BENCH_INFLIGHT", "4"))
 
 SLEEP_STORY = """
 kind: Story
@@ -302,7 +302,7 @@ def main() -> int:
     if multi and has_gpu:
         device_ids = [torch.cuda.current_device()]  # one GPU per rank
     eng = RunEngine(
-        EngineConfig(cpu_workers=4, workers_per_device=4, child_ttl_seconds=5.0),
+        EngineConfig(cpu_workers=4, workers_per_device=8, child_ttl_seconds=5.0),
         device_ids=device_ids,
     ).start()
     native = None

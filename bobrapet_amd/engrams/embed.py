@@ -69,10 +69,16 @@ class EmbedEngram(Engram):
         else:
             batch = int(inp.get("batch", cfg.get("batch", 32)))
             seq = int(inp.get("seqLen", cfg.get("seqLen", 128)))
-            gen = torch.Generator(device="cpu").manual_seed(int(inp.get("seed", 0)))
-            ids_t = torch.randint(0, vocab, (batch, seq), generator=gen).to(
-                device=device, dtype=torch.int32
-            )
+            sd = int(inp.get("seed", 0))
+            key = ("ids", vocab, batch, seq, sd, str(device))
+            ids_t = _TABLE_CACHE.get(key)
+            if ids_t is None:
+                gen = torch.Generator(device="cpu").manual_seed(sd)
+                ids_t = torch.randint(0, vocab, (batch, seq), generator=gen).to(
+                    device=device, dtype=torch.int32
+                )
+                if len(_TABLE_CACHE) < 256:
+                    _TABLE_CACHE[key] = ids_t
         if ids_t.ndim != 2:
             raise EngramFailure("embed: ids must be [batch][seq]", exit_code=2)
 

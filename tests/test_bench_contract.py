@@ -149,7 +149,7 @@ def test_bench_parallel8_pipelined_two_ranks():
     in-flight slots + comm-slot communicators for the all-gather join."""
     port = _free_port()
     env = dict(os.environ)
-    env.update({"BOBRA_BENCH_NO_CPU_FALLBACK": "1"})
+    env.update({"BOBRA_BENCH_NO_CPU_FALLBACK": "1", "BOBRA_BENCH_INFLIGHT": "4"})
     out = subprocess.run(
         [
             sys.executable, "-m", "torch.distributed.run",

@@ -111,7 +111,7 @@ spec:
 # p50 0.63 ms vs 725 runs/s at p50 3.9 ms with 8 in-flight (the per-story
 # Python bodies are GIL-bound, so overlap buys ~2% throughput for 6x step
 # latency).  Pipelining stays available for saturation testing.
-INFLIGHT = int(os.environ.get("BOBRA_BENCH_INFLIGHT", "1"))
+INFLIGHT = os.environ.get("BOBRA_BENCH_INFLIGHT")  # resolved per-world in main()
 
 SLEEP_STORY = """
 kind: Story
@@ -337,7 +337,14 @@ def main() -> int:
                     raise
                 native = None
 
-        inflight = INFLIGHT if config_name == "parallel8" else 1
+        # single-rank: sequential wins (p50 0.63 ms at ~same throughput);
+        # multi-rank: W=4 in-flight stories on comm-slot communicators
+        # overlap each story's all-gather with the next story's branches,
+        # hiding cross-rank straggler jitter.
+        default_w = 4 if world > 1 else 1
+        inflight = int(INFLIGHT) if INFLIGHT else default_w
+        if config_name != "parallel8":
+            inflight = 1
         if inflight > 1:
             group.ensure_comm_slots(inflight)  # collective; same order on all ranks
 

@@ -38,9 +38,11 @@ class ResourceRegistry:
         self.impulse_templates: _t.Dict[str, T.ImpulseTemplate] = {}
         self.transports: _t.Dict[str, T.Transport] = {}
         self.reference_grants: _t.Dict[str, T.ReferenceGrant] = {}
+        self.mutations = 0  # bumped on every apply; cheap cache-invalidation tag
 
     def apply(self, obj) -> None:
         with self._lock:
+            self.mutations += 1
             if isinstance(obj, T.Story):
                 obj.generation = self.stories.get(obj.key, obj).generation + (
                     1 if obj.key in self.stories else 0

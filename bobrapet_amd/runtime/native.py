@@ -268,6 +268,8 @@ class NativeRunner:
         self.metrics = metrics
         self.plan_ids: _t.Dict[str, int] = {}
         self._plan_gen: _t.Dict[str, int] = {}
+        self._engram_cache: _t.Dict[str, tuple] = {}
+        self._engram_cache_gen = -1
         self.engine.set_launcher(self._launch)
         self.engine.start()
 
@@ -321,18 +323,23 @@ class NativeRunner:
 
     # ------------------------------------------------------------------
 
-    def _launch(self, run_id, step, branch, attempt, engram_key, resolved_input):
-        """Engram launcher (called from the core's loop thread, GIL held):
-        enqueue onto the worker pool and return immediately."""
+    def _resolve_engram(self, engram_key: str):
+        """(impl_name, config) for an engram key — resolved once per key and
+        cached: the launcher runs on the core loop thread per step body, so
+        registry/template lookups there are pure per-step overhead."""
+        gen = getattr(self.registry, "mutations", 0)
+        if gen != self._engram_cache_gen:
+            self._engram_cache.clear()
+            self._engram_cache_gen = gen
+        hit = self._engram_cache.get(engram_key)
+        if hit is not None:
+            return hit
         engram = None
         if "/" in engram_key:
             ns, name = engram_key.split("/", 1)
             engram = self.registry.try_engram(name, ns)
         if engram is None:
-            self.engine.complete_engram(
-                run_id, step, branch, attempt, 2, None, f"engram {engram_key} not found"
-            )
-            return
+            return None
         template = None
         if engram.template_ref is not None:
             try:
@@ -340,15 +347,29 @@ class NativeRunner:
             except KeyError:
                 pass
         impl_name = template.implementation if template is not None else engram_key.split("/")[-1]
-        config = engram.with_
+        hit = (impl_name, engram.with_)
+        self._engram_cache[engram_key] = hit
+        return hit
+
+    def _launch(self, run_id, step, branch, attempt, engram_key, resolved_input):
+        """Engram launcher (called from the core's loop thread, GIL held):
+        enqueue onto the worker pool and return immediately."""
+        hit = self._resolve_engram(engram_key)
+        if hit is None:
+            self.engine.complete_engram(
+                run_id, step, branch, attempt, 2, None, f"engram {engram_key} not found"
+            )
+            return
+        impl_name, config = hit
         device = None
         if self.workers.device_count > 0:
             ids = self.workers.device_ids
             device = ids[(hash((run_id, step, branch)) & 0x7FFFFFFF) % len(ids)]
 
+        impl = engram_registry.resolve(impl_name)
+
         def body(slot):
             try:
-                impl = engram_registry.resolve(impl_name)
                 ctx = EngramContext(
                     story_run=str(run_id),
                     step_name=f"{step}" if branch < 0 else f"{step}/{branch}",

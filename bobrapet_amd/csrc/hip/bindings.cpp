@@ -114,7 +114,11 @@ torch::Tensor embed_pool(torch::Tensor table, torch::Tensor ids) {
   const int B = ids.size(0), S = ids.size(1);
   TORCH_CHECK(H % 8 == 0 && H <= 8192, "embed_pool: bad H");
   auto out = torch::empty({B, H}, table.options());
-  auto pooled = torch::zeros({B, H}, table.options().dtype(torch::kFloat32));
+  const int nchunk = (S + 7) / 8;  // EMB_SCHUNK partials
+  // zeros, not empty: H that is not a multiple of the 2048-wide slice
+  // leaves a tail the sum kernel never writes
+  auto pooled =
+      torch::zeros({(long)B * nchunk, H}, table.options().dtype(torch::kFloat32));
   launch_embed_pool(out.data_ptr(), pooled.data_ptr(), table.data_ptr(),
                     ids.data_ptr(), B, S, H, V, cur_stream());
   return out;

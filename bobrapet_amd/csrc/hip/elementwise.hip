@@ -288,14 +288,20 @@ __global__ __launch_bounds__(256) void embed_pool_sum_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
   }
-  float* dst = pooled + (long)b * H + h0;
+  // per-chunk partial slot, plain coalesced stores — the atomicAdd version
+  // of this kernel was atomic-throughput bound (16-way contention per
+  // pooled element, ~0.57 TB/s effective); partials + merge-in-norm stream
+  float* dst = pooled + (((long)b * nchunk + chunk) * H) + h0;
+  float4v lo, hi;
 #pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(dst + j, acc[j]);
+  for (int j = 0; j < 4; ++j) { lo[j] = acc[j]; hi[j] = acc[j + 4]; }
+  *reinterpret_cast<float4v*>(dst) = lo;
+  *reinterpret_cast<float4v*>(dst + 4) = hi;
 }
 
 __global__ __launch_bounds__(256) void embed_pool_norm_kernel(
     unsigned short* __restrict__ out, const float* __restrict__ pooled, int S,
-    int H) {
+    int H, int nchunk) {
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const float inv_s = 1.f / (float)S;
@@ -303,13 +309,16 @@ __global__ __launch_bounds__(256) void embed_pool_norm_kernel(
   float ssq = 0.f;
   int k = 0;
   for (int c = tid * 8; c < H; c += 256 * 8, ++k) {
-    const float4v lo = *reinterpret_cast<const float4v*>(pooled + (long)b * H + c);
-    const float4v hic = *reinterpret_cast<const float4v*>(pooled + (long)b * H + c + 4);
+    float sum[8] = {};
+    for (int ch = 0; ch < nchunk; ++ch) {
+      const float* src = pooled + (((long)b * nchunk + ch) * H) + c;
+      const float4v lo = *reinterpret_cast<const float4v*>(src);
+      const float4v hic = *reinterpret_cast<const float4v*>(src + 4);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      vals[k][j] = lo[j] * inv_s;
-      vals[k][j + 4] = hic[j] * inv_s;
+      for (int j = 0; j < 4; ++j) { sum[j] += lo[j]; sum[j + 4] += hic[j]; }
     }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[k][j] = sum[j] * inv_s;
 #pragma unroll
     for (int j = 0; j < 8; ++j) ssq += vals[k][j] * vals[k][j];
   }
@@ -339,7 +348,8 @@ extern "C" void launch_embed_pool(void* out, void* pooled_f32, const void* table
                      0, stream, (float*)pooled_f32,
                      (const unsigned short*)table, (const int*)ids, B, S, H, V);
   hipLaunchKernelGGL(embed_pool_norm_kernel, dim3(B), dim3(256), 0, stream,
-                     (unsigned short*)out, (const float*)pooled_f32, S, H);
+                     (unsigned short*)out, (const float*)pooled_f32, S, H,
+                     nchunk);
 }
 
 // ---------------------------------------------------------------------------

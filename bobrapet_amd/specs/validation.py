@@ -18,7 +18,7 @@ import typing as _t
 from dataclasses import dataclass, field
 
 from ..enums import BATCH_ONLY_STEP_TYPES, StepType, StoryPattern
-from ..templating import deps as tdeps
+from ..templating import TemplateSyntaxError, deps as tdeps, is_template, parse_template
 from . import types as T
 
 MAX_STORY_BYTES = 1 << 20  # 1 MiB total Story cap (story_webhook.go:418-428)
@@ -148,6 +148,32 @@ def validate_story(story: T.Story) -> ValidationResult:
     for label, schema in (("inputsSchema", story.inputs_schema), ("outputsSchema", story.outputs_schema)):
         if schema is not None and not isinstance(schema, dict):
             res.error(f"{label} must be a JSON Schema object")
+
+    # template safety: every template string must parse at apply time
+    # (reference: pkg/templatesafety ValidateTemplateString + webhook-side
+    # ValidateJSONTemplates — fail at apply, not mid-run)
+    def _walk_templates(where: str, value) -> None:
+        if isinstance(value, str):
+            if is_template(value):
+                try:
+                    parse_template(value)
+                except TemplateSyntaxError as exc:
+                    res.error(f"{where}: bad template: {exc}")
+        elif isinstance(value, dict):
+            for k, v in value.items():
+                _walk_templates(f"{where}.{k}", v)
+        elif isinstance(value, list):
+            for i, v in enumerate(value):
+                _walk_templates(f"{where}[{i}]", v)
+
+    for s in story.all_steps():
+        _walk_templates(f"step {s.name!r} with", s.with_)
+        _walk_templates(f"step {s.name!r} runtime", s.runtime)
+        if s.if_:
+            _walk_templates(f"step {s.name!r} if", s.if_)
+        if s.idempotency_key_template:
+            _walk_templates(f"step {s.name!r} idempotencyKey", s.idempotency_key_template)
+    _walk_templates("story output", story.output)
 
     return res
 

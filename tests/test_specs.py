@@ -303,3 +303,60 @@ spec:
     assert validate_engram(eng, tpl).ok
     eng.with_ = {"model": 42}
     assert not validate_engram(eng, tpl).ok
+
+
+class TestTemplateSafety:
+    """Apply-time template validation (reference: pkg/templatesafety
+    ValidateTemplateString; webhook ValidateJSONTemplates)."""
+
+    def test_bad_template_rejected_at_apply(self):
+        from bobrapet_amd.specs import load_yaml, validation
+
+        (story,) = load_yaml(
+            """
+kind: Story
+metadata: {name: bad-tpl}
+spec:
+  steps:
+    - {name: a, type: sleep, with: {duration: 1s, note: "{{ steps.a.output. }}"}}
+"""
+        )
+        res = validation.validate_story(story)
+        assert any("bad template" in e for e in res.errors)
+
+    def test_bad_output_template_rejected(self):
+        from bobrapet_amd.specs import load_yaml, validation
+
+        (story,) = load_yaml(
+            """
+kind: Story
+metadata: {name: bad-out}
+spec:
+  steps:
+    - {name: a, type: sleep, with: {duration: 1s}}
+  output: {v: "{{ 1 + }}"}
+"""
+        )
+        res = validation.validate_story(story)
+        assert any("story output" in e and "bad template" in e for e in res.errors)
+
+    def test_good_templates_pass(self):
+        from bobrapet_amd.specs import load_yaml, validation
+
+        (story,) = load_yaml(
+            """
+kind: Story
+metadata: {name: ok-tpl}
+spec:
+  steps:
+    - {name: a, type: sleep, with: {duration: 1s}}
+    - name: b
+      type: condition
+      needs: [a]
+      if: "{{ steps.a.phase == 'Succeeded' }}"
+      with: {expression: "{{ inputs.x > 3 && size(inputs.name) < 10 }}"}
+  output: {v: "{{ steps.b.output.result }}"}
+"""
+        )
+        res = validation.validate_story(story)
+        assert not res.errors, res.errors

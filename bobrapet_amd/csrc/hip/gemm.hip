@@ -189,40 +189,6 @@ __global__ __launch_bounds__(256) void gemv_bf16_nt_kernel(
   }
 }
 
-// M=1 specialization: x staged once per WG into LDS (<=16K halves), W rows
-// streamed with 32B-per-lane vectors — halves the global-load instruction
-// count of the generic path (x re-reads were L2 hits but still occupied the
-// memory pipe) and doubles per-lane MLP on the stream.
-__global__ __launch_bounds__(256) void gemv_bf16_nt_m1_kernel(
-    unsigned short* __restrict__ C, const unsigned short* __restrict__ A,
-    const unsigned short* __restrict__ W, int N, int K) {
-  extern __shared__ float x_lds[];  // [K] f32
-  const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid >> 6;
-  for (int k = tid * 8; k < K; k += 256 * 8) {
-    ushort8v v = *reinterpret_cast<const ushort8v*>(A + k);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) x_lds[k + j] = bf2f(v[j]);
-  }
-  __syncthreads();
-  const int waves_total = (gridDim.x * blockDim.x) >> 6;
-  for (int n = blockIdx.x * 4 + wid; n < N; n += waves_total) {
-    const unsigned short* wrow = W + (long)n * K;
-    float acc = 0.f;
-    for (int k0 = lane * 16; k0 < K; k0 += WAVE * 16) {
-      ushort8v w0 = *reinterpret_cast<const ushort8v*>(wrow + k0);
-      ushort8v w1 = *reinterpret_cast<const ushort8v*>(wrow + k0 + 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc += bf2f(w0[j]) * x_lds[k0 + j];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc += bf2f(w1[j]) * x_lds[k0 + 8 + j];
-    }
-    float r = wave_reduce_sum(acc);
-    if (lane == 0) C[n] = f2bf(r);
-  }
-}
-
 extern "C" void launch_gemv_bf16_nt(void* C, const void* A, const void* W,
                                     int M, int N, int K, hipStream_t stream) {
   int blocks = (N + 3) / 4;
@@ -232,13 +198,6 @@ extern "C" void launch_gemv_bf16_nt(void* C, const void* A, const void* W,
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
                        (const unsigned short*)A, (const unsigned short*)W, N, K);
   };
-  if (M == 1 && K % 1024 == 0 && K <= 16384 &&
-      !(getenv("BOBRA_GEMV_V1") && getenv("BOBRA_GEMV_V1")[0] == '1')) {
-    hipLaunchKernelGGL(gemv_bf16_nt_m1_kernel, grid, block, K * 4, stream,
-                       (unsigned short*)C, (const unsigned short*)A,
-                       (const unsigned short*)W, N, K);
-    return;
-  }
   switch (M) {
     case 1: launch(gemv_bf16_nt_kernel<1>); break;
     case 2: launch(gemv_bf16_nt_kernel<2>); break;

@@ -496,6 +496,9 @@ class StepExecutor:
             state.finished_at = monotonic_now()
 
 
+_finish_lock = __import__("threading").Lock()
+
+
 def finish_step_run(
     sr: StepRun,
     phase: Phase,
@@ -506,13 +509,17 @@ def finish_step_run(
 ) -> bool:
     """Terminal-phase-wins write to a StepRun status (SDK-race discipline,
     reference: stepStatusPatchedBySDK steprun_controller.go:2031).  Returns
-    False when an earlier terminal phase already won."""
-    if sr.status.phase.is_terminal:
-        return False
-    sr.status.phase = phase
-    sr.status.output = output if output is not None else sr.status.output
-    sr.status.error = error
-    sr.status.exit_code = exit_code
-    sr.status.exit_class = exit_class
-    sr.status.finished_at = monotonic_now()
-    return True
+    False when an earlier terminal phase already won.  The engine loop is
+    the only writer in normal operation; the lock makes the first-terminal-
+    wins guarantee hold even for out-of-band writers (tests, future
+    multi-loop setups) — the critical section is a few field writes."""
+    with _finish_lock:
+        if sr.status.phase.is_terminal:
+            return False
+        sr.status.phase = phase
+        sr.status.output = output if output is not None else sr.status.output
+        sr.status.error = error
+        sr.status.exit_code = exit_code
+        sr.status.exit_class = exit_class
+        sr.status.finished_at = monotonic_now()
+        return True

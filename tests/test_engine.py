@@ -892,3 +892,87 @@ spec: {templateRef: {name: echo-tpl}}
         )
         run = eng.wait(run.key, timeout=10)
         assert run.phase == Phase.SUCCEEDED
+
+
+class TestConcurrentWriterRaces:
+    """Race-shaped tests (reference: steprun_sdk_race_test.go,
+    persistMergedStates dag.go:774-793): terminal-phase-wins under
+    concurrent completion attempts, and cancel racing completion."""
+
+    def test_finish_step_run_terminal_wins_concurrent(self):
+        import threading
+
+        from bobrapet_amd.engine.executor import finish_step_run
+        from bobrapet_amd.engine.records import StepRun, StepRunSpec
+        from bobrapet_amd.enums import ExitClass, Phase
+
+        sr = StepRun(name="r-a", spec=StepRunSpec(story_run="r", step_name="a"))
+        wins = []
+        barrier = threading.Barrier(8)
+
+        def writer(phase, code):
+            barrier.wait()
+            if finish_step_run(sr, phase, output={"p": str(phase)}, exit_code=code):
+                wins.append(phase)
+
+        threads = [
+            threading.Thread(
+                target=writer,
+                args=(Phase.SUCCEEDED if i % 2 == 0 else Phase.FAILED, i % 2),
+            )
+            for i in range(8)
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()
+        # exactly one writer won; status matches the winner and never flips
+        assert len(wins) == 1
+        assert sr.status.phase == wins[0]
+        assert sr.status.output == {"p": str(wins[0])}
+
+    def test_step_state_merge_preserves_terminal(self):
+        from bobrapet_amd.engine.records import StepState
+        from bobrapet_amd.enums import Phase
+
+        dst = StepState(name="a", phase=Phase.SUCCEEDED, output={"v": 1})
+        src = StepState(name="a", phase=Phase.RUNNING, output=None)
+        dst.merge_from(src)
+        assert dst.phase == Phase.SUCCEEDED
+        assert dst.output == {"v": 1}
+
+    def test_cancel_races_completion(self):
+        """Cancel posted while engrams are mid-flight must converge to a
+        terminal run without deadlock, whichever side wins each step."""
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=4)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: sleepy}
+spec: {builtin: sleepy}
+---
+kind: Engram
+metadata: {name: s}
+spec: {templateRef: {name: sleepy}}
+---
+kind: Story
+metadata: {name: race}
+spec:
+  steps:
+    - {name: a, ref: {name: s}, with: {seconds: 0.05}}
+    - {name: b, ref: {name: s}, needs: [a], with: {seconds: 0.05}}
+"""
+            )
+            for _ in range(5):
+                run = eng.submit_run("default/race", {})
+                import time as _time
+
+                _time.sleep(0.02)
+                eng.cancel(run, graceful=False)
+                run = eng.wait(run, timeout=20)
+                assert run.is_terminal
+        finally:
+            eng.stop()

@@ -325,6 +325,9 @@ class StreamingRun:
         self._lock = threading.Lock()
         self._leaf_packets = 0
         self._last_outputs: _t.List = []
+        self._recorded: _t.List = []     # recording entries (bounded)
+        self._replay_log: _t.List = []   # ingress packets for replay
+        self._rec_seq = 0
         self.settings = self._settings(story)
 
         # rings: one per edge; ingress edges share the ingress ring list
@@ -399,10 +402,36 @@ class StreamingRun:
     # ------------------------------------------------------------------
 
     def push(self, packet, timeout: _t.Optional[float] = 30.0) -> bool:
+        rp = self.settings.replay
+        if rp is not None and rp.mode in ("memory", "durable") and len(self._replay_log) < 4096:
+            # durable mode additionally survives payload offload: markers,
+            # not live tensors, are retained
+            self._replay_log.append(
+                self.engine.storage.dehydrate_document(packet)
+                if rp.mode == "durable"
+                else packet
+            )
         ok = True
         for ring in self.ingress:
             ok = ring.push(packet, timeout=timeout) and ok
         return ok
+
+    def replay(self, last: _t.Optional[int] = None, timeout: _t.Optional[float] = 30.0) -> int:
+        """Re-push recorded ingress packets (reference: TransportReplaySettings
+        memory/durable replay).  Returns how many packets were re-delivered."""
+        rp = self.settings.replay
+        if rp is None or rp.mode not in ("memory", "durable"):
+            raise ValueError("replay requires streaming.replay.mode memory|durable")
+        src = self._replay_log[-last:] if last else list(self._replay_log)
+        n = 0
+        for pkt in src:
+            if rp.mode == "durable":
+                pkt = self.engine.storage.hydrate(pkt)
+            ok = True
+            for ring in self.ingress:
+                ok = ring.push(pkt, timeout=timeout) and ok
+            n += 1 if ok else 0
+        return n
 
     def finish(self, timeout: float = 60.0) -> StoryRun:
         """Close the ingress, drain and finalize the run."""
@@ -434,6 +463,12 @@ class StreamingRun:
             "packets": self._leaf_packets,
             "stages": len(self.stages),
         }
+        if self._recorded:
+            # recordings ride the normal $storageRef offload path
+            self.run.output["recording"] = self.engine.storage.dehydrate_document(
+                {"entries": self._recorded}
+            )
+            self.run.output["recordedPackets"] = len(self._recorded)
         self.run.finished_at = now
         self.engine.on_run_terminal(self.run)
         self.engine._streams.pop(self.run.key, None)
@@ -444,9 +479,26 @@ class StreamingRun:
         for ring in self._edge_rings.values():
             ring.close()
 
+    def _record(self, stage: str, packet) -> None:
+        rec = self.settings.recording
+        if rec is None or not rec.mode or len(self._recorded) >= 8192:
+            return
+        self._rec_seq += 1
+        rate = rec.sample_rate if rec.sample_rate is not None else 100
+        if rate <= 0 or (self._rec_seq - 1) % max(1, 100 // max(rate, 1)) != 0:
+            return
+        entry = {"seq": self._rec_seq, "stage": stage, "ts": monotonic_now()}
+        if rec.mode == "full":
+            body = _strip_tensors(packet)
+            for path in rec.redact_fields:
+                body = _redact(body, path.split("."))
+            entry["packet"] = body
+        self._recorded.append(entry)
+
     def on_packet(self, stage: str, leaf: bool, packet) -> None:
-        if leaf:
-            with self._lock:
+        with self._lock:
+            self._record(stage, packet)
+            if leaf:
                 self._leaf_packets += 1
                 if len(self._last_outputs) < 8:
                     self._last_outputs.append(_strip_tensors(packet))
@@ -459,6 +511,19 @@ class StreamingRun:
     def leaf_packets(self) -> int:
         with self._lock:
             return self._leaf_packets
+
+
+def _redact(value, path):
+    """Replace the value at a dot path with "<redacted>" (recording
+    redactFields — reference: transport_settings_types.go:526)."""
+    if not path or not isinstance(value, dict) or path[0] not in value:
+        return value
+    out = dict(value)
+    if len(path) == 1:
+        out[path[0]] = "<redacted>"
+    else:
+        out[path[0]] = _redact(out[path[0]], path[1:])
+    return out
 
 
 def _strip_tensors(value):

@@ -636,6 +636,29 @@ class TransportBackpressure:
 
 
 @dataclass
+class TransportReplay:
+    """Replay storage/retention (reference:
+    transport_settings_types.go:301-315, modes none|memory|durable)."""
+
+    mode: str = "none"
+    retention_seconds: _t.Optional[int] = None
+    checkpoint_interval: _t.Optional[str] = None
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
+class TransportRecording:
+    """Stream recording (reference: transport_settings_types.go:508-529):
+    metadata-only or full payloads, sampling, redaction."""
+
+    mode: str = ""  # "" (off) | metadata | full
+    sample_rate: _t.Optional[int] = None  # 0..100
+    retention_seconds: _t.Optional[int] = None
+    redact_fields: _t.List[str] = field(default_factory=list)
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
 class TransportFanIn:
     """Fan-in join behavior when multiple upstreams feed a step
     (reference: transport_settings_types.go:174-192, modes all|any|quorum)."""
@@ -658,6 +681,8 @@ class TransportStreamingSettings:
     delivery: _t.Optional[TransportDelivery] = None
     backpressure: _t.Optional[TransportBackpressure] = None
     fan_in: _t.Optional[TransportFanIn] = None
+    replay: _t.Optional[TransportReplay] = None
+    recording: _t.Optional[TransportRecording] = None
     extra: dict = field(default_factory=dict)
 
 

@@ -427,3 +427,67 @@ spec:
         assert s.fan_in.mode == "quorum"
         assert s.fan_in.quorum == 2
         assert s.fan_in.timeout_seconds == 7
+
+
+class TestReplayRecording:
+    """TransportReplaySettings (memory|durable) + TransportRecordingSettings
+    (metadata|full, sampling, redaction) — reference:
+    transport_settings_types.go:296-315, 508-529."""
+
+    STORY = """
+kind: Story
+metadata: {name: rr}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        %s
+  steps:
+    - name: only
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v + 1 }}", secret: "{{ item.secret }}"}}
+"""
+
+    def test_memory_replay_redelivers(self, eng):
+        eng.apply_yaml(self.STORY % "replay: {mode: memory}")
+        stream = eng.submit_stream("default/rr")
+        for i in range(4):
+            stream.push({"items": [{"v": i, "secret": "s"}]})
+        import time as _t2
+
+        _t2.sleep(0.2)
+        assert stream.replay(last=2) == 2
+        run = stream.finish(timeout=10)
+        assert run.phase.name == "FINISHED"
+        assert run.output["packets"] == 6  # 4 original + 2 replayed
+
+    def test_recording_full_with_redaction(self, eng):
+        eng.apply_yaml(
+            self.STORY
+            % 'recording: {mode: full, redactFields: ["items"]}'
+        )
+        stream = eng.submit_stream("default/rr")
+        for i in range(3):
+            stream.push({"items": [{"v": i, "secret": "hunter2"}]})
+        run = stream.finish(timeout=10)
+        rec = eng.storage.hydrate(run.output["recording"])
+        assert run.output["recordedPackets"] == 3
+        assert all(e["packet"]["items"] == "<redacted>" for e in rec["entries"])
+        assert "hunter2" not in str(rec)
+
+    def test_recording_metadata_only(self, eng):
+        eng.apply_yaml(self.STORY % "recording: {mode: metadata}")
+        stream = eng.submit_stream("default/rr")
+        stream.push({"items": [{"v": 1, "secret": "x"}]})
+        run = stream.finish(timeout=10)
+        rec = eng.storage.hydrate(run.output["recording"])
+        assert rec["entries"][0]["stage"] == "only"
+        assert "packet" not in rec["entries"][0]
+
+    def test_replay_requires_mode(self, eng):
+        eng.apply_yaml(self.STORY % "fanIn: {mode: all}")
+        stream = eng.submit_stream("default/rr")
+        with pytest.raises(ValueError):
+            stream.replay()
+        stream.finish(timeout=10)

@@ -53,7 +53,8 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
     const unsigned short* __restrict__ q,  // [B,S,Hq,D]
     const unsigned short* __restrict__ k,  // [B,S,Hkv,D]
     const unsigned short* __restrict__ v,  // [B,S,Hkv,D] (row stride may differ)
-    int B, int Hq, int Hkv, int S, long v_sstride, float scale, int causal) {
+    int B, int Hq, int Hkv, int S, long v_sstride, float scale, int causal,
+    float* __restrict__ stats) {  // optional [B,Hq,S,2] (m, l) exp2-domain
   // two K+V^T buffer pairs; pointers computed per use (an addrspace(3)
   // pointer array fails to compile as a static initializer)
   __shared__ __attribute__((aligned(16))) char smem[2 * (KVBLK * 256 + D_HEAD * 128)];
@@ -341,18 +342,25 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
     for (int dt = 0; dt < 4; ++dt)
       orow[dt * 32 + l31] = f2bf(o_acc[dt][r] * inv_for_row[r]);
   }
+  // per-row softmax stats for cross-block merging (ring attention):
+  // lane l31 of wave wid owns row my_q's (m_run, l_run); halves duplicate
+  if (stats != nullptr && hi == 0 && my_q < S) {
+    float2 ml{m_run, l_run};
+    *reinterpret_cast<float2*>(stats + ((long)bh * S + my_q) * 2) = ml;
+  }
 }
 
 extern "C" void launch_attn_prefill(void* out, const void* q, const void* k,
                                     const void* v, int B, int Hq, int Hkv,
                                     int S, long v_sstride, float scale,
-                                    int causal, hipStream_t stream) {
+                                    int causal, void* stats,
+                                    hipStream_t stream) {
   int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
   dim3 grid(B * Hq * nqblk), block(512);
   hipLaunchKernelGGL((attn_prefill_kernel<7>), grid, block, 0, stream,
                      (unsigned short*)out, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v, B, Hq,
-                     Hkv, S, v_sstride, scale, causal);
+                     Hkv, S, v_sstride, scale, causal, (float*)stats);
 }
 
 extern "C" void launch_attn_prefill_variant(int variant, void* out,
@@ -366,7 +374,7 @@ extern "C" void launch_attn_prefill_variant(int variant, void* out,
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)out,
                        (const unsigned short*)q, (const unsigned short*)k,
                        (const unsigned short*)v, B, Hq, Hkv, S,
-                       (long)Hkv * D_HEAD, scale, causal);
+                       (long)Hkv * D_HEAD, scale, causal, (float*)nullptr);
   };
   if (variant == 1) args(attn_prefill_kernel<1>);
   else if (variant == 3) args(attn_prefill_kernel<3>);

@@ -18,7 +18,7 @@ void launch_embed_pool(void*, void*, const void*, const void*, int, int, int,
                        int, hipStream_t);
 void launch_add_bf16(void*, const void*, const void*, long, hipStream_t);
 void launch_attn_prefill(void*, const void*, const void*, const void*, int,
-                         int, int, int, long, float, int, hipStream_t);
+                         int, int, int, long, float, int, void*, hipStream_t);
 void launch_silu_mul_strided(void*, const void*, const void*, long, int, long,
                              hipStream_t);
 void launch_rope_qkv(void*, void*, const void*, const void*, const void*, int,
@@ -152,8 +152,28 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   auto out = torch::empty_like(q);
   launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       B, Hq, Hkv, S, (long)v.stride(1), (float)scale,
-                      causal ? 1 : 0, cur_stream());
+                      causal ? 1 : 0, nullptr, cur_stream());
   return out;
+}
+
+std::vector<torch::Tensor> attn_prefill_stats(torch::Tensor q, torch::Tensor k,
+                                              torch::Tensor v, double scale,
+                                              bool causal) {
+  // attn_prefill + per-row softmax stats [B,Hq,S,2] f32 (m, l) in the
+  // kernel's exp2 domain — the cross-block merge surface for ring attention
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128, "attn_prefill_stats: D must be 128");
+  TORCH_CHECK(Hq % Hkv == 0, "attn_prefill_stats: Hq % Hkv");
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D, "v head must be contiguous");
+  auto out = torch::empty_like(q);
+  auto stats = torch::empty({B, Hq, S, 2}, q.options().dtype(torch::kFloat32));
+  launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      B, Hq, Hkv, S, (long)v.stride(1), (float)scale,
+                      causal ? 1 : 0, stats.data_ptr(), cur_stream());
+  return {out, stats};
 }
 
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor kc, torch::Tensor vc,
@@ -308,6 +328,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_inplace", &rope_inplace, "rotary embedding in-place on q,k");
   m.def("embed_pool", &embed_pool, "gather + mean-pool + l2norm");
   m.def("add_bf16", &add_bf16, "a + b");
+  m.def("attn_prefill_stats", &attn_prefill_stats,
+        "prefill attention + per-row (m,l) softmax stats");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
   m.def("attn_decode_t", &attn_decode_t,

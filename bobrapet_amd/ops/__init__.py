@@ -175,6 +175,48 @@ def add_bf16(a, b):
     return (a.float() + b.float()).to(a.dtype)
 
 
+def attn_prefill_stats_ref(q, k, v, scale, causal):
+    """fp32 reference of attn_prefill_stats: BSHD in, (out, stats) out.
+    stats[..., 0] = m, [..., 1] = l in the KERNEL's exp2 domain
+    (scores * scale * log2(e)) so cross-block merges are implementation-
+    uniform on CPU and GPU."""
+    import torch
+
+    B, S, Hq, D = q.shape
+    Hkv = k.shape[2]
+    g = Hq // Hkv
+    qt = q.float().permute(0, 2, 1, 3)  # [B,Hq,S,D]
+    kt = k.float().permute(0, 2, 1, 3)
+    vt = v.float().permute(0, 2, 1, 3)
+    if g > 1:
+        kt = kt.repeat_interleave(g, dim=1)
+        vt = vt.repeat_interleave(g, dim=1)
+    s2 = torch.matmul(qt, kt.transpose(-1, -2)) * (scale * 1.4426950408889634)
+    if causal:
+        mask = torch.ones(S, k.shape[1], dtype=torch.bool, device=q.device).tril_()
+        s2 = s2.masked_fill(~mask, float("-inf"))
+    m = s2.amax(dim=-1)  # [B,Hq,S]
+    p = torch.exp2(s2 - m[..., None])
+    l = p.sum(dim=-1)
+    out = torch.matmul(p, vt) / l[..., None]
+    stats = torch.stack([m, l], dim=-1)  # [B,Hq,S,2]
+    return out.permute(0, 2, 1, 3).to(q.dtype).contiguous(), stats
+
+
+def attn_prefill_stats(q, k, v, scale: _t.Optional[float] = None, causal: bool = True):
+    """attn_prefill + per-row (m, l) softmax stats for cross-block merging
+    (ring attention over xGMI).  BSHD; stats [B,Hq,S,2] f32 exp2-domain."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        if not (v.stride(-1) == 1 and v.stride(2) == v.shape[-1]):
+            v = v.contiguous()
+        out, stats = _require_ext().attn_prefill_stats(
+            q.contiguous(), k.contiguous(), v, scale, causal
+        )
+        return out, stats
+    return attn_prefill_stats_ref(q, k, v, scale, causal)
+
+
 def attn_prefill(q, k, v, scale: _t.Optional[float] = None, causal: bool = True):
     """BSHD layout: q [B,S,Hq,D], k/v [B,S,Hkv,D] → out [B,S,Hq,D]."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])

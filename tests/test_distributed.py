@@ -13,6 +13,9 @@ import pytest
 import torch
 
 
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
 def _free_port() -> int:
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
@@ -166,3 +169,76 @@ spec:
     )
     p = place_steps(story, 8)
     assert p["a"] == 3
+
+
+def _ring_attn_rank(rank: int, world: int, port: int, q):
+    import os
+    import sys
+
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    try:
+        sys.path.insert(0, REPO)
+        import torch
+
+        from bobrapet_amd import ops
+        from bobrapet_amd.parallel import group
+        from bobrapet_amd.parallel.ring_attention import ring_attention
+
+        group.init_distributed(backend="gloo")
+        torch.manual_seed(7)  # same global tensors on every rank
+        B, S, Hq, Hkv, D = 2, 48, 4, 2, 64
+        S_glob = S * world
+        qg = torch.randn(B, S_glob, Hq, D, dtype=torch.bfloat16)
+        kg = torch.randn(B, S_glob, Hkv, D, dtype=torch.bfloat16)
+        vg = torch.randn_like(kg)
+        res = {}
+        for causal in (True, False):
+            full = ops.attn_prefill(qg, kg, vg, None, causal)
+            lo, hi = rank * S, (rank + 1) * S
+            out = ring_attention(
+                qg[:, lo:hi].contiguous(),
+                kg[:, lo:hi].contiguous(),
+                vg[:, lo:hi].contiguous(),
+                causal=causal,
+            )
+            err = (out.float() - full[:, lo:hi].float()).abs().max().item()
+            res[f"causal={causal}"] = err
+        q.put((rank, res))
+        group.teardown()
+    except Exception as exc:
+        import traceback
+
+        q.put((rank, {"error": f"{exc}\n{traceback.format_exc()}"}))
+
+
+@pytest.mark.timeout(180)
+@pytest.mark.parametrize("world", [2, 3])
+def test_ring_attention_matches_full(world):
+    """Sequence-parallel ring attention == single-rank full attention,
+    causal and non-causal, including an odd ring size."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_ring_attn_rank, args=(r, world, port, q)) for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=160)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    for rank, payload in results.items():
+        assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
+        for key, err in payload.items():
+            assert err < 0.03, f"rank {rank} {key}: err {err}"

@@ -184,11 +184,24 @@ class LlamaModel:
         ids: torch.Tensor,
         logits_for_all: bool = False,
         fill_cache: bool = False,
+        seq_shard: _t.Optional[_t.Tuple[int, int]] = None,
     ) -> torch.Tensor:
-        """ids [B, S] int64 → logits [B, V] (last position) or [B, S, V]."""
+        """ids [B, S] int64 → logits [B, V] (last position) or [B, S, V].
+
+        seq_shard=(rank, world) runs SEQUENCE-PARALLEL prefill: `ids` is
+        this rank's contiguous shard of a rank-major-partitioned global
+        sequence (weights replicated), attention is ring attention over
+        xGMI (parallel/ring_attention.py), rope positions are offset by the
+        shard start.  Returns this rank's local logits."""
         cfg = self.cfg
         B, S = ids.shape
-        positions = torch.arange(S, device=self.device)
+        pos0 = 0
+        ring = None
+        if seq_shard is not None and seq_shard[1] > 1:
+            from ..parallel.ring_attention import ring_attention as ring
+
+            pos0 = seq_shard[0] * S
+        positions = torch.arange(pos0, pos0 + S, device=self.device)
         cos_t, sin_t = ops.rope_tables(positions, cfg.head_dim, cfg.rope_theta)
         # tables are per flattened token (T = B*S)
         cos_f = cos_t.repeat(B, 1)
@@ -210,7 +223,10 @@ class LlamaModel:
                 kc, vc = self._kv_cache[li]
                 kc[:, :, :S] = kh.transpose(1, 2)
                 vc[:, :, :S] = vh.transpose(1, 2)
-            attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
+            if ring is not None:
+                attn = ring(qh, kh, vh, self.scale, causal=True)
+            else:
+                attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
             attn = attn.reshape(B, S, cfg.num_heads * cfg.head_dim)
             attn_out = torch.matmul(attn, lw.w_o.t())
             x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)

@@ -242,3 +242,68 @@ def test_ring_attention_matches_full(world):
         assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
         for key, err in payload.items():
             assert err < 0.03, f"rank {rank} {key}: err {err}"
+
+
+def _sp_prefill_rank(rank: int, world: int, port: int, q):
+    import os
+    import sys
+
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    try:
+        sys.path.insert(0, REPO)
+        import torch
+
+        from bobrapet_amd.models.llama import LlamaModel
+        from bobrapet_amd.parallel import group
+
+        group.init_distributed(backend="gloo")
+        torch.manual_seed(0)
+        m = LlamaModel("llama-tiny", device="cpu")
+        S_loc = 16
+        gen = torch.Generator().manual_seed(9)
+        ids = torch.randint(0, m.cfg.vocab_size, (1, S_loc * world), generator=gen)
+        full = m.prefill(ids, logits_for_all=True)
+        lo, hi = rank * S_loc, (rank + 1) * S_loc
+        local = m.prefill(
+            ids[:, lo:hi].contiguous(), logits_for_all=True, seq_shard=(rank, world)
+        )
+        err = (local.float() - full[:, lo:hi].float()).abs().max().item()
+        scale = full.abs().max().item()
+        q.put((rank, {"err": err, "scale": scale}))
+        group.teardown()
+    except Exception as exc:
+        import traceback
+
+        q.put((rank, {"error": f"{exc}\n{traceback.format_exc()}"}))
+
+
+@pytest.mark.timeout(180)
+def test_sequence_parallel_prefill_matches_full():
+    """llama-tiny sequence-parallel prefill (ring attention per layer,
+    rope positions offset per shard) == single-rank full prefill."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_sp_prefill_rank, args=(r, world, port, q)) for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=160)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    for rank, payload in results.items():
+        assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
+        assert payload["err"] < 0.05 * max(payload["scale"], 1.0), payload

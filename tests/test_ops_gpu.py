@@ -219,11 +219,6 @@ class TestAttnPrefillStats:
         q = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
         k = torch.randn(B, 2 * S, Hq, D, device="cuda", dtype=torch.bfloat16)
         v = torch.randn_like(k)
-        full = ops.attn_prefill(q, k[:, :S], v[:, :S], None, False)  # warm
-        fullref = ops.attn_prefill_stats_ref(
-            q.float(), k.float(), v.float(), D ** -0.5, False
-        )  # unused shape check
-        del full, fullref
         o1, s1 = ops.attn_prefill_stats(q, k[:, :S].contiguous(), v[:, :S].contiguous(), None, False)
         o2, s2 = ops.attn_prefill_stats(q, k[:, S:].contiguous(), v[:, S:].contiguous(), None, False)
         m = torch.maximum(s1[..., 0], s2[..., 0])
@@ -231,4 +226,5 @@ class TestAttnPrefillStats:
         wb = (s2[..., 1] * torch.exp2(s2[..., 0] - m)).permute(0, 2, 1)[..., None]
         merged = (o1.float() * wa + o2.float() * wb) / (wa + wb)
         ref = ops.attn_prefill_stats_ref(q, k, v, D ** -0.5, False)[0]
-        assert (merged - ref.float()).abs().max().item() < 0.03
+        err = (merged - ref.float()).abs().max().item()
+        assert err < 0.03, f"merge err {err}"

@@ -56,12 +56,23 @@ class LlmInferEngram(Engram):
                 0, model.cfg.vocab_size, (batch, seq_len), generator=gen
             ).to(model.device)
 
+        # sequenceParallel: true shards the sequence across the live
+        # distributed ranks (SPMD story) — ring attention per layer; each
+        # rank's `ids` must be its rank-major shard of the global sequence
+        seq_parallel = bool(inp.get("sequenceParallel", cfg.get("sequenceParallel", False)))
+        seq_shard = None
+        if seq_parallel:
+            from ..parallel import group
+
+            if group.world_size() > 1:
+                seq_shard = (group.rank(), group.world_size())
+
         t0 = time.monotonic()
         if new_tokens > 0:
             tokens = model.generate(ids, new_tokens)
             logits = None
         else:
-            logits = model.prefill(ids)
+            logits = model.prefill(ids, seq_shard=seq_shard)
             tokens = logits.argmax(dim=-1, keepdim=True)
         if model.device.type == "cuda":
             torch.cuda.synchronize(model.device)
@@ -75,6 +86,7 @@ class LlmInferEngram(Engram):
             "tokens": tokens[:, :32].tolist(),  # inline sample
             "latencyMs": latency_ms,
             "tokensProcessed": batch * (seq_len + new_tokens),
+            "sequenceParallel": seq_shard is not None,
         }
         if logits is not None and ctx.storage is not None:
             # keep the full logits tensor resident (HBM payload indirection)

@@ -48,6 +48,7 @@ class TransportBinding:
     upstream: _t.List[str] = field(default_factory=list)
     downstream: _t.List[str] = field(default_factory=list)
     phase: str = "Ready"
+    generation: int = 0  # bumped on live cutover (reference: connector gen)
     heartbeat: float = field(default_factory=monotonic_now)
 
 
@@ -246,6 +247,9 @@ class _Stage(threading.Thread):
     def _impl(self):
         if self.step.ref is None:
             return None
+        cached = getattr(self, "_impl_cache", None)
+        if cached is not None:
+            return cached
         eng = self.sr.engine
         engram = eng.registry.try_engram(
             self.step.ref.name, self.step.ref.resolve_namespace(self.sr.story.namespace)
@@ -259,7 +263,14 @@ class _Stage(threading.Thread):
         )
         name = tpl.implementation if tpl is not None else self.step.ref.name
         self._engram_cfg = engram.with_
-        return engram_registry.resolve(name)
+        self._impl_cache = engram_registry.resolve(name)
+        return self._impl_cache
+
+    def invalidate_impl(self) -> None:
+        """Next packet re-resolves the engram + config from the registry
+        (live cutover — reference: connector generation bumping
+        steprun_controller.go:2693-2760)."""
+        self._impl_cache = None
 
     def _ctx(self, packet) -> EngramContext:
         return EngramContext(
@@ -478,6 +489,36 @@ class StreamingRun:
         self.canceled = True
         for ring in self._edge_rings.values():
             ring.close()
+
+    def upgrade(self, step: _t.Optional[str] = None, timeout: _t.Optional[float] = None) -> int:
+        """Live cutover of streaming stage(s) to the CURRENTLY-applied
+        Engram definition (reference: TransportLifecycleSettings drain/
+        cutover + connector generation bumping).  strategy=drain waits for
+        the stage's input rings to empty (bounded by drainTimeoutSeconds /
+        `timeout`); strategy=recreate cuts over immediately.  Returns the
+        number of stages cut over; binding generations are bumped."""
+        lc = self.settings.lifecycle
+        strategy = lc.strategy if lc is not None else "drain"
+        if timeout is None:
+            timeout = float(lc.drain_timeout_seconds) if lc and lc.drain_timeout_seconds else 5.0
+        targets = [s for s in self.stages if step is None or s.step.name == step]
+        if step is not None and not targets:
+            raise KeyError(f"no streaming stage named {step!r}")
+        n = 0
+        for stage in targets:
+            if strategy == "drain":
+                deadline = time.monotonic() + timeout
+                while any(r.depth > 0 for r in stage.in_rings):
+                    if time.monotonic() >= deadline:
+                        break
+                    time.sleep(0.002)
+            stage.invalidate_impl()
+            binding = self.bindings.get(stage.step.name)
+            if binding is not None:
+                binding.generation = getattr(binding, "generation", 0) + 1
+            self.engine.metrics.inc("stream_stage_upgrades_total")
+            n += 1
+        return n
 
     def _record(self, stage: str, packet) -> None:
         rec = self.settings.recording

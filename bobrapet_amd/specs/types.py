@@ -659,6 +659,17 @@ class TransportRecording:
 
 
 @dataclass
+class TransportLifecycle:
+    """Drain/cutover behavior for live streaming steps (reference:
+    transport_settings_types.go:431-447)."""
+
+    strategy: str = "drain"  # drain | recreate
+    drain_timeout_seconds: _t.Optional[int] = None
+    max_in_flight: _t.Optional[int] = None
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
 class TransportFanIn:
     """Fan-in join behavior when multiple upstreams feed a step
     (reference: transport_settings_types.go:174-192, modes all|any|quorum)."""
@@ -683,6 +694,7 @@ class TransportStreamingSettings:
     fan_in: _t.Optional[TransportFanIn] = None
     replay: _t.Optional[TransportReplay] = None
     recording: _t.Optional[TransportRecording] = None
+    lifecycle: _t.Optional[TransportLifecycle] = None
     extra: dict = field(default_factory=dict)
 
 

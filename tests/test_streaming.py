@@ -491,3 +491,68 @@ spec:
         with pytest.raises(ValueError):
             stream.replay()
         stream.finish(timeout=10)
+
+
+class TestLifecycleUpgrade:
+    """Live stage cutover (reference: TransportLifecycleSettings drain/
+    cutover + connector generation bumping)."""
+
+    def test_upgrade_picks_up_new_engram_config(self, eng):
+        eng.apply_yaml(
+            """
+kind: Engram
+metadata: {name: scaler}
+spec:
+  templateRef: {name: echo-tpl}
+  with: {factor: 1}
+---
+kind: Story
+metadata: {name: up}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        lifecycle: {strategy: drain, drainTimeoutSeconds: 2}
+  steps:
+    - name: only
+      ref: {name: scaler}
+"""
+        )
+        stream = eng.submit_stream("default/up")
+        stream.push({"v": 1})
+        time.sleep(0.15)
+        # apply a NEW engram config, then cut over the live stage
+        eng.apply_yaml(
+            """
+kind: Engram
+metadata: {name: scaler}
+spec:
+  templateRef: {name: echo-tpl}
+  with: {factor: 2}
+"""
+        )
+        assert stream.upgrade("only") == 1
+        assert stream.bindings["only"].generation == 1
+        stream.push({"v": 2})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # echo merges config: first packet saw factor 1, second factor 2
+        outs = stream._last_outputs
+        assert outs[0]["factor"] == 1 and outs[1]["factor"] == 2, outs
+
+    def test_upgrade_unknown_stage_raises(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: up2}
+spec:
+  pattern: streaming
+  steps:
+    - {name: only, ref: {name: echoer}}
+"""
+        )
+        stream = eng.submit_stream("default/up2")
+        with pytest.raises(KeyError):
+            stream.upgrade("ghost")
+        stream.finish(timeout=10)

@@ -226,6 +226,106 @@ def build_http_app(engine: "RunEngine"):
             }
         return result
 
+    # ---- run control plane (the kube-apiserver role for runs) ----
+
+    def _run_view(run) -> dict:
+        return {
+            "name": run.name,
+            "namespace": run.namespace,
+            "story": f"{run.story_namespace}/{run.story_name}",
+            "phase": str(run.phase),
+            "execPhase": run.exec_phase,
+            "steps": {
+                k: {
+                    "phase": str(st.phase),
+                    "retries": st.retries,
+                    "message": st.message,
+                    "error": st.error.message if st.error else None,
+                }
+                for k, st in run.step_states.items()
+            },
+            "output": run.output,
+            "error": run.error.to_dict() if run.error else None,
+            "degraded": run.degraded,
+        }
+
+    def _get_run(ns: str, name: str):
+        run = engine.store.try_get_story_run(f"{ns}/{name}")
+        if run is None:
+            raise HTTPException(404, f"run {ns}/{name} not found")
+        return run
+
+    @app.post("/resources")
+    async def apply_resources(body: dict):
+        """Apply CRD-style YAML documents (the admission-webhook surface:
+        invalid specs are rejected with the validation errors)."""
+        text = body.get("yaml")
+        if not isinstance(text, str):
+            raise HTTPException(400, "body must be {yaml: '<documents>'}")
+        try:
+            applied = engine.apply_yaml(text)
+        except ValueError as exc:
+            raise HTTPException(422, str(exc))
+        return {"applied": len(applied)}
+
+    @app.get("/stories")
+    async def list_stories():
+        return {
+            "stories": [
+                {"key": key, **engine.registry.story_status(key)}
+                for key in sorted(engine.registry.stories)
+            ]
+        }
+
+    @app.post("/stories/{ns}/{name}/runs")
+    async def submit(ns: str, name: str, body: _t.Optional[dict] = None):
+        body = body or {}
+        try:
+            run = engine.submit_run(
+                f"{ns}/{name}", body.get("inputs") or {}, name=body.get("runName")
+            )
+        except KeyError:
+            raise HTTPException(404, f"story {ns}/{name} not found")
+        except ValueError as exc:
+            raise HTTPException(422, str(exc))
+        if body.get("wait"):
+            run = engine.wait(run, timeout=float(body.get("timeout", 300.0)))
+        return _run_view(run)
+
+    @app.get("/runs/{ns}/{name}")
+    async def run_status(ns: str, name: str):
+        return _run_view(_get_run(ns, name))
+
+    @app.post("/runs/{ns}/{name}/cancel")
+    async def cancel_run(ns: str, name: str, body: _t.Optional[dict] = None):
+        run = _get_run(ns, name)
+        engine.cancel(run, graceful=bool((body or {}).get("graceful", True)))
+        return {"ok": True, "phase": str(run.phase)}
+
+    @app.post("/runs/{ns}/{name}/redrive")
+    async def redrive_run(ns: str, name: str, body: _t.Optional[dict] = None):
+        run = _get_run(ns, name)
+        step = (body or {}).get("fromStep")
+        try:
+            if step:
+                engine.redrive_from_step(run, step)
+            else:
+                engine.redrive(run)
+        except ValueError as exc:
+            raise HTTPException(409, str(exc))
+        return {"ok": True}
+
+    @app.post("/runs/{ns}/{name}/gates/{step}")
+    async def decide_gate(ns: str, name: str, step: str, body: dict):
+        run = _get_run(ns, name)
+        approve = bool(body.get("approve", False))
+        decided_by = str(body.get("decidedBy", "http"))
+        if approve:
+            engine.approve_gate(run, step, decided_by)
+        else:
+            engine.reject_gate(run, step, decided_by)
+        return {"ok": True, "state": "Approved" if approve else "Rejected"}
+
     @app.get("/healthz")
     async def healthz():
         return {"ok": True}

@@ -556,3 +556,77 @@ spec:
         with pytest.raises(KeyError):
             stream.upgrade("ghost")
         stream.finish(timeout=10)
+
+
+class TestRestControlPlane:
+    """REST run control plane (the kube-apiserver role): apply, submit,
+    status, cancel, redrive, gate decisions over HTTP."""
+
+    def test_apply_submit_status_gate(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        client = TestClient(build_http_app(eng))
+        res = client.post(
+            "/resources",
+            json={
+                "yaml": """
+kind: Story
+metadata: {name: http-flow}
+spec:
+  steps:
+    - {name: approve, type: gate, with: {timeout: 30s}}
+    - name: done
+      ref: {name: echoer}
+      needs: [approve]
+      with: {ok: true}
+  output: {ok: "{{ steps.done.output.ok }}"}
+"""
+            },
+        )
+        assert res.status_code == 200, res.text
+        res = client.post("/stories/default/http-flow/runs", json={"inputs": {"x": 1}})
+        assert res.status_code == 200, res.text
+        run_name = res.json()["name"]
+        assert res.json()["phase"] in ("Pending", "Running")
+        res = client.get(f"/runs/default/{run_name}")
+        assert res.json()["steps"]["approve"]["phase"] in ("Running", "Pending", "Paused")
+        res = client.post(f"/runs/default/{run_name}/gates/approve", json={"approve": True})
+        assert res.json()["state"] == "Approved"
+        run = eng.wait(f"default/{run_name}", timeout=30)
+        assert str(run.phase) == "Succeeded"
+        res = client.get(f"/runs/default/{run_name}")
+        assert res.json()["output"] == {"ok": True}
+        stories = client.get("/stories").json()["stories"]
+        assert any(s["key"] == "default/http-flow" for s in stories)
+
+    def test_apply_invalid_rejected_422(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        client = TestClient(build_http_app(eng))
+        res = client.post(
+            "/resources",
+            json={
+                "yaml": """
+kind: Story
+metadata: {name: bad}
+spec:
+  steps:
+    - {name: a, type: sleep, needs: [ghost], with: {duration: 1s}}
+"""
+            },
+        )
+        assert res.status_code == 422
+        assert "ghost" in res.text or "unknown" in res.text
+
+    def test_missing_run_404(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        client = TestClient(build_http_app(eng))
+        assert client.get("/runs/default/nope").status_code == 404
+        assert client.post("/stories/default/nope/runs", json={}).status_code == 404

@@ -280,14 +280,29 @@ __global__ __launch_bounds__(256) void embed_pool_sum_kernel(
   if (h0 + 8 > H) return;
 
   float acc[8] = {};
-  const int s_end = min((chunk + 1) * EMB_SCHUNK, S);
-  for (int sidx = chunk * EMB_SCHUNK; sidx < s_end; ++sidx) {
-    int id = ids[(long)b * S + sidx];
-    if (id < 0 || id >= V) continue;
-    ushort8v v = *reinterpret_cast<const ushort8v*>(table + (long)id * H + h0);
+  const int s0 = chunk * EMB_SCHUNK;
+  const int s_end = min(s0 + EMB_SCHUNK, S);
+  // preload the chunk's ids, then issue ALL row gathers before consuming:
+  // the naive loop serializes id-load -> row-load dependent chains (8 HBM
+  // round-trips per WG); splitting the phases keeps 8 gathers in flight
+  int id_reg[EMB_SCHUNK];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc[j] += bf2f(v[j]);
+  for (int i = 0; i < EMB_SCHUNK; ++i) {
+    int sidx = s0 + i;
+    id_reg[i] = (sidx < s_end) ? ids[(long)b * S + sidx] : -1;
   }
+  ushort8v rows[EMB_SCHUNK];
+#pragma unroll
+  for (int i = 0; i < EMB_SCHUNK; ++i) {
+    int id = id_reg[i];
+    rows[i] = (id >= 0 && id < V)
+                  ? *reinterpret_cast<const ushort8v*>(table + (long)id * H + h0)
+                  : ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
+  }
+#pragma unroll
+  for (int i = 0; i < EMB_SCHUNK; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] += bf2f(rows[i][j]);
   // per-chunk partial slot, plain coalesced stores — the atomicAdd version
   // of this kernel was atomic-throughput bound (16-way contention per
   // pooled element, ~0.57 TB/s effective); partials + merge-in-norm stream

@@ -114,3 +114,78 @@ class TestStores:
         assert mgr.sweep(older_than_seconds=3600) == 0
         assert mgr.sweep(older_than_seconds=-1) == 1
         assert store.list() == []
+
+
+class TestHydrateDehydrateFuzz:
+    """Property-based round-trip fuzzing of the $storageRef walker
+    (reference: pkg/storage/manager_fuzz_test.go:28,67)."""
+
+    @staticmethod
+    def _json_values():
+        from hypothesis import strategies as st
+
+        scalars = st.one_of(
+            st.none(),
+            st.booleans(),
+            st.integers(min_value=-(2**31), max_value=2**31),
+            st.floats(allow_nan=False, allow_infinity=False, width=32),
+            st.text(max_size=64),
+        )
+        return st.recursive(
+            scalars,
+            lambda children: st.one_of(
+                st.lists(children, max_size=5),
+                st.dictionaries(
+                    st.text(min_size=1, max_size=12).filter(
+                        lambda k: not k.startswith("$")
+                    ),
+                    children,
+                    max_size=5,
+                ),
+            ),
+            max_leaves=25,
+        )
+
+    def test_round_trip_any_json(self):
+        from hypothesis import given, settings
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=1, max_inline_size=48)).start()
+        try:
+
+            @settings(max_examples=120, deadline=None)
+            @given(self._json_values())
+            def check(doc):
+                dehydrated = eng.storage.dehydrate_document(doc)
+                back = eng.storage.hydrate(dehydrated)
+                assert back == doc, (doc, dehydrated, back)
+
+            check()
+        finally:
+            eng.stop()
+
+    def test_spoofed_refs_never_resolve_foreign_data(self):
+        """A crafted $storageRef in user data must not read another ref's
+        payload after the untrusted-input scrub (spoof rejection)."""
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=1, max_inline_size=16)).start()
+        try:
+            secret_ref = eng.storage.dehydrate_document({"secret": "x" * 64})
+            assert isinstance(secret_ref, dict) and "$storageRef" in secret_ref
+            eng.apply_yaml(
+                """
+kind: Story
+metadata: {name: guard}
+spec:
+  steps:
+    - {name: a, type: sleep, with: {duration: 0ms}}
+"""
+            )
+            import pytest as _pytest
+
+            with _pytest.raises(ValueError):
+                eng.submit_run("default/guard", {"data": secret_ref})
+        finally:
+            eng.stop()

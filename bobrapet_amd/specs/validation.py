@@ -307,6 +307,13 @@ def validate_engram(
 
             for err in validate_instance(engram.with_ or {}, template.config_schema):
                 res.error(f"engram.with: {err}")
+        if template.secret_schema is not None:
+            from ..utils.jsonschema import validate_instance
+
+            # secret VALUES are env/file indirections; the schema governs
+            # which keys must exist (reference: engram_webhook secretSchema)
+            for err in validate_instance(dict(engram.secrets or {}), template.secret_schema):
+                res.error(f"engram.secrets: {err}")
     return res
 
 

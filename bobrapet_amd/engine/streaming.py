@@ -67,14 +67,17 @@ class _Stage(threading.Thread):
         sr: "StreamingRun",
         step: T.Step,
         in_rings: _t.List[flow.CreditRing],
-        out_rings: _t.List[flow.CreditRing],
+        out_edges: _t.List[_t.Tuple[_t.Tuple[str, str], _t.List[flow.CreditRing]]],
         device: _t.Optional[int],
+        lane: int = 0,
     ):
-        super().__init__(name=f"stage-{sr.run.name}-{step.name}", daemon=True)
+        super().__init__(name=f"stage-{sr.run.name}-{step.name}.{lane}", daemon=True)
         self.sr = sr
         self.step = step
         self.in_rings = in_rings
-        self.out_rings = out_rings
+        self.out_edges = out_edges  # [(edge_key, [ring per partition])]
+        self.out_rings = [r for _, rings in out_edges for r in rings]  # leaf test
+        self.lane = lane
         self.device = device
         self.stats = StageStats()
         self.stream = None
@@ -97,8 +100,10 @@ class _Stage(threading.Thread):
             self.error = f"{type(exc).__name__}: {exc}"
             self.sr.on_stage_error(self.step.name, self.error)
         finally:
-            for ring in self.out_rings:
-                ring.close()
+            # downstream rings close only when EVERY lane of this step is
+            # done (a shared ring closed early would cut sibling lanes off)
+            for edge_key, _rings in self.out_edges:
+                self.sr._producer_done(edge_key)
 
     def _loop(self) -> None:
         import contextlib
@@ -124,8 +129,16 @@ class _Stage(threading.Thread):
             self.stats.busy_seconds += time.monotonic() - t0
             self.stats.packets_in += 1
             if out is not None:
-                for ring in self.out_rings:
-                    ring.push(out)
+                # per-partition ordering: a packet's partition id routes it
+                # to that partition's ring (reference: ordering per_partition)
+                if isinstance(out, dict) and isinstance(packet, dict) and "$partition" in packet:
+                    out.setdefault("$partition", packet["$partition"])
+                for _key, rings in self.out_edges:
+                    if len(rings) == 1:
+                        rings[0].push(out)
+                    else:
+                        part = out.get("$partition", 0) if isinstance(out, dict) else 0
+                        rings[int(part) % len(rings)].push(out)
                 self.stats.packets_out += 1
                 self.sr.on_packet(self.step.name, leaf=not self.out_rings, packet=out)
 
@@ -341,35 +354,57 @@ class StreamingRun:
         self._rec_seq = 0
         self.settings = self._settings(story)
 
-        # rings: one per edge; ingress edges share the ingress ring list
-        self._edge_rings: _t.Dict[_t.Tuple[str, str], flow.CreditRing] = {}
-        self.ingress: _t.List[flow.CreditRing] = []
+        # partitioning (reference: TransportPartitioningSettings hash mode):
+        # an edge into a fan-in-free stage splits into P rings, one per
+        # partition, and the stage runs P parallel lanes — per-partition
+        # order preserved, cross-partition work parallel on worker threads
+        part = self.settings.partitioning
+        self._nparts = (
+            max(1, int(part.partitions or 4))
+            if part is not None and part.mode == "hash"
+            else 1
+        )
+        self._part_rr = 0  # round-robin fallback when no key configured
+
+        # rings: per edge, 1 ring (ordered lane) or P partition rings
+        self._edge_rings: _t.Dict[_t.Tuple[str, str], _t.List[flow.CreditRing]] = {}
+        self._edge_producers: _t.Dict[_t.Tuple[str, str], int] = {}
+        self.ingress: _t.List[_t.List[flow.CreditRing]] = []
+        upstream_count = {name: len(self.topo.upstream_of(name)) for name in self.topo.stages}
         for e in self.topo.edges:
-            ring = flow.CreditRing(
-                name=f"{run.name}:{e.src or '@'}->{e.dst}",
-                settings=self.settings,
-                lane=e.lane,
-            )
-            self._edge_rings[(e.src, e.dst)] = ring
+            nrings = self._nparts if upstream_count.get(e.dst, 1) == 1 else 1
+            rings = [
+                flow.CreditRing(
+                    name=f"{run.name}:{e.src or '@'}->{e.dst}#{p_}",
+                    settings=self.settings,
+                    lane=e.lane,
+                )
+                for p_ in range(nrings)
+            ]
+            self._edge_rings[(e.src, e.dst)] = rings
             if e.src == "":
-                self.ingress.append(ring)
+                self.ingress.append(rings)
 
         self.stages: _t.List[_Stage] = []
         by_name = {s.name: s for s in story.steps}
         n_dev = engine.workers.device_count
         for i, name in enumerate(self.topo.stages):
             step = by_name[name]
-            in_rings = [
-                self._edge_rings[(e.src, e.dst)] for e in self.topo.upstream_of(name)
-            ]
-            out_rings = [
-                self._edge_rings[(e.src, e.dst)] for e in self.topo.downstream_of(name)
-            ]
+            up = self.topo.upstream_of(name)
+            down = self.topo.downstream_of(name)
+            out_edges = [((e.src, e.dst), self._edge_rings[(e.src, e.dst)]) for e in down]
             device = None
             if n_dev > 0 and step.ref is not None:
                 device = i % n_dev  # stage-per-device pipeline placement
-            stage = _Stage(self, step, in_rings, out_rings, device)
-            self.stages.append(stage)
+            lanes = len(self._edge_rings[(up[0].src, up[0].dst)]) if len(up) == 1 else 1
+            for lane in range(lanes):
+                if lanes > 1:
+                    in_rings = [self._edge_rings[(up[0].src, up[0].dst)][lane]]
+                else:
+                    in_rings = [self._edge_rings[(e.src, e.dst)][0] for e in up]
+                self.stages.append(_Stage(self, step, in_rings, out_edges, device, lane))
+            for (ekey, _r) in out_edges:
+                self._edge_producers[ekey] = self._edge_producers.get(ekey, 0) + lanes
             st = run.step_state(name)
             st.phase = Phase.RUNNING
             st.started_at = monotonic_now()
@@ -422,10 +457,51 @@ class StreamingRun:
                 if rp.mode == "durable"
                 else packet
             )
+        packet = self._stamp_partition(packet)
         ok = True
-        for ring in self.ingress:
-            ok = ring.push(packet, timeout=timeout) and ok
+        for rings in self.ingress:
+            if len(rings) == 1:
+                ok = rings[0].push(packet, timeout=timeout) and ok
+            else:
+                part = packet.get("$partition", 0) if isinstance(packet, dict) else 0
+                ok = rings[int(part) % len(rings)].push(packet, timeout=timeout) and ok
         return ok
+
+    def _stamp_partition(self, packet):
+        """Assign a partition id (reference: TransportPartitioningSettings):
+        hash of the configured key path (round-robin without a key),
+        preserved when sticky and already present."""
+        part = self.settings.partitioning
+        if part is None or part.mode in (None, "", "none") or not isinstance(packet, dict):
+            return packet
+        if part.sticky and "$partition" in packet:
+            return packet
+        if part.mode == "preserve":
+            packet.setdefault("$partition", 0)
+            return packet
+        n = self._nparts
+        if part.key:
+            from ..storage.manager import extract_path
+
+            try:
+                key_val = extract_path(packet, part.key)
+            except Exception:
+                key_val = None
+            pid = (hash(str(key_val)) & 0x7FFFFFFF) % n
+        else:
+            pid = self._part_rr % n
+            self._part_rr += 1
+        packet = dict(packet)
+        packet["$partition"] = pid
+        return packet
+
+    def _producer_done(self, edge_key) -> None:
+        with self._lock:
+            left = self._edge_producers.get(edge_key, 1) - 1
+            self._edge_producers[edge_key] = left
+        if left <= 0:
+            for ring in self._edge_rings.get(edge_key, []):
+                ring.close()
 
     def replay(self, last: _t.Optional[int] = None, timeout: _t.Optional[float] = 30.0) -> int:
         """Re-push recorded ingress packets (reference: TransportReplaySettings
@@ -438,41 +514,57 @@ class StreamingRun:
         for pkt in src:
             if rp.mode == "durable":
                 pkt = self.engine.storage.hydrate(pkt)
+            pkt = self._stamp_partition(pkt)
             ok = True
-            for ring in self.ingress:
-                ok = ring.push(pkt, timeout=timeout) and ok
+            for rings in self.ingress:
+                if len(rings) == 1:
+                    ok = rings[0].push(pkt, timeout=timeout) and ok
+                else:
+                    part = pkt.get("$partition", 0) if isinstance(pkt, dict) else 0
+                    ok = rings[int(part) % len(rings)].push(pkt, timeout=timeout) and ok
             n += 1 if ok else 0
         return n
 
     def finish(self, timeout: float = 60.0) -> StoryRun:
         """Close the ingress, drain and finalize the run."""
-        for ring in self.ingress:
-            ring.close()
+        for rings in self.ingress:
+            for ring in rings:
+                ring.close()
         deadline = time.monotonic() + timeout
         for s in self.stages:
             s.join(timeout=max(deadline - time.monotonic(), 0.1))
         now = monotonic_now()
         failed = False
-        for s in self.stages:
-            st = self.run.step_state(s.step.name)
-            st.output = {
-                "packetsIn": s.stats.packets_in,
-                "packetsOut": s.stats.packets_out,
-                "errors": s.stats.errors,
-                "graphReplays": s.stats.graph_replays,
-                "busySeconds": round(s.stats.busy_seconds, 6),
-            }
-            st.finished_at = now
+        agg: _t.Dict[str, dict] = {}
+        errs: _t.Dict[str, str] = {}
+        for s in self.stages:  # sum stats across a step's partition lanes
+            a = agg.setdefault(
+                s.step.name,
+                {"packetsIn": 0, "packetsOut": 0, "errors": 0, "graphReplays": 0,
+                 "busySeconds": 0.0, "lanes": 0},
+            )
+            a["packetsIn"] += s.stats.packets_in
+            a["packetsOut"] += s.stats.packets_out
+            a["errors"] += s.stats.errors
+            a["graphReplays"] += s.stats.graph_replays
+            a["busySeconds"] = round(a["busySeconds"] + s.stats.busy_seconds, 6)
+            a["lanes"] += 1
             if s.error:
+                errs[s.step.name] = s.error
+        for name, a in agg.items():
+            st = self.run.step_state(name)
+            st.output = a
+            st.finished_at = now
+            if name in errs:
                 st.phase = Phase.FAILED
-                st.error = StructuredError(message=s.error)
+                st.error = StructuredError(message=errs[name])
                 failed = True
             else:
                 st.phase = Phase.SUCCEEDED
         self.run.phase = Phase.FAILED if failed else Phase.FINISHED
         self.run.output = {
             "packets": self._leaf_packets,
-            "stages": len(self.stages),
+            "stages": len(agg),
         }
         if self._recorded:
             # recordings ride the normal $storageRef offload path
@@ -487,8 +579,9 @@ class StreamingRun:
 
     def cancel(self) -> None:
         self.canceled = True
-        for ring in self._edge_rings.values():
-            ring.close()
+        for rings in self._edge_rings.values():
+            for ring in rings:
+                ring.close()
 
     def upgrade(self, step: _t.Optional[str] = None, timeout: _t.Optional[float] = None) -> int:
         """Live cutover of streaming stage(s) to the CURRENTLY-applied

@@ -659,6 +659,18 @@ class TransportRecording:
 
 
 @dataclass
+class TransportPartitioning:
+    """Partition assignment for stream envelopes (reference:
+    transport_settings_types.go:391-419, modes none|preserve|hash)."""
+
+    mode: str = "none"
+    key: _t.Optional[str] = None  # dot path into the packet, e.g. "meta.user"
+    partitions: _t.Optional[int] = None
+    sticky: bool = True
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
 class TransportLifecycle:
     """Drain/cutover behavior for live streaming steps (reference:
     transport_settings_types.go:431-447)."""
@@ -695,6 +707,7 @@ class TransportStreamingSettings:
     replay: _t.Optional[TransportReplay] = None
     recording: _t.Optional[TransportRecording] = None
     lifecycle: _t.Optional[TransportLifecycle] = None
+    partitioning: _t.Optional[TransportPartitioning] = None
     extra: dict = field(default_factory=dict)
 
 

@@ -630,3 +630,56 @@ spec:
         client = TestClient(build_http_app(eng))
         assert client.get("/runs/default/nope").status_code == 404
         assert client.post("/stories/default/nope/runs", json={}).status_code == 404
+
+
+class TestPartitioning:
+    """TransportPartitioningSettings (reference:
+    transport_settings_types.go:391-419): hash partitioning splits edges
+    into per-partition rings and runs parallel stage lanes with
+    per-partition ordering."""
+
+    STORY = """
+kind: Story
+metadata: {name: parts}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        partitioning: {mode: hash, key: "user", partitions: 4}
+  steps:
+    - name: tag
+      ref: {name: transformer}
+      runtime: {map: {user: "{{ item.user }}", n: "{{ item.n }}"}}
+"""
+
+    def test_hash_partitions_run_parallel_lanes(self, eng):
+        eng.apply_yaml(self.STORY)
+        stream = eng.submit_stream("default/parts")
+        for i in range(20):
+            stream.push({"items": [{"user": f"u{i % 5}", "n": i}], "user": f"u{i % 5}"})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        out = run.step_states["tag"].output
+        assert out["lanes"] == 4
+        assert out["packetsIn"] == 20
+        assert run.output["packets"] == 20
+
+    def test_same_key_same_partition_order_preserved(self, eng):
+        eng.apply_yaml(self.STORY)
+        stream = eng.submit_stream("default/parts")
+        # one hot key: everything lands on ONE partition -> strict order
+        for i in range(12):
+            stream.push({"items": [{"user": "hot", "n": i}], "user": "hot"})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        seq = [o["items"][0]["n"] for o in stream._last_outputs]
+        assert seq == sorted(seq), seq
+
+    def test_sticky_preserves_existing_partition(self, eng):
+        eng.apply_yaml(self.STORY)
+        stream = eng.submit_stream("default/parts")
+        stream.push({"items": [{"user": "x", "n": 1}], "$partition": 3})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        assert run.output["packets"] == 1

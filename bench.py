@@ -142,6 +142,13 @@ kind: Story
 metadata: {name: bench-stream}
 spec:
   pattern: streaming
+  transports:
+    - name: rings
+      streaming:
+        # keyless hash partitioning = round-robin over 4 partition lanes:
+        # each lane owns its own hipGraph + HIP stream, so per-packet GPU
+        # work replays 4-wide while per-partition order is preserved
+        partitioning: {mode: hash, partitions: 4}
   steps:
     - name: featurize
       ref: {name: embedder}

@@ -630,12 +630,16 @@ class StreamingRun:
         self._recorded.append(entry)
 
     def on_packet(self, stage: str, leaf: bool, packet) -> None:
-        with self._lock:
-            self._record(stage, packet)
-            if leaf:
-                self._leaf_packets += 1
-                if len(self._last_outputs) < 8:
-                    self._last_outputs.append(_strip_tensors(packet))
+        rec = self.settings.recording
+        if leaf or (rec is not None and rec.mode):
+            # only leaf counting / recording needs the run lock — keep the
+            # hot per-packet path lock-free for parallel partition lanes
+            with self._lock:
+                self._record(stage, packet)
+                if leaf:
+                    self._leaf_packets += 1
+                    if len(self._last_outputs) < 8:
+                        self._last_outputs.append(_strip_tensors(packet))
         self.engine.metrics.inc("stream_packets_total", stage=stage)
 
     def on_stage_error(self, stage: str, message: str) -> None:

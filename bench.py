@@ -222,8 +222,17 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
         time.sleep(0.001)
     group.barrier()
     t0 = time.monotonic()
-    for i in range(args.steps):
-        stream.push({"tensor": ids, "seq": i, "items": [{}]})
+    # concurrent producers (impulse events arrive concurrently); partition
+    # lanes keep per-partition ordering regardless of producer interleaving
+    n_push = 4
+    import concurrent.futures as _cf
+
+    def _pusher(base: int) -> None:
+        for i in range(base, args.steps, n_push):
+            stream.push({"tensor": ids, "seq": i, "items": [{}]})
+
+    with _cf.ThreadPoolExecutor(max_workers=n_push) as ex:
+        list(ex.map(_pusher, range(n_push)))
     while stream.leaf_packets < args.warmup + args.steps and time.monotonic() < t0 + 120:
         time.sleep(0.001)
     group.barrier()

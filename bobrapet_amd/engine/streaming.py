@@ -364,7 +364,9 @@ class StreamingRun:
             if part is not None and part.mode == "hash"
             else 1
         )
-        self._part_rr = 0  # round-robin fallback when no key configured
+        import itertools as _it
+
+        self._part_rr = _it.count()  # thread-safe round-robin (no key case)
 
         # rings: per edge, 1 ring (ordered lane) or P partition rings
         self._edge_rings: _t.Dict[_t.Tuple[str, str], _t.List[flow.CreditRing]] = {}
@@ -489,8 +491,7 @@ class StreamingRun:
                 key_val = None
             pid = (hash(str(key_val)) & 0x7FFFFFFF) % n
         else:
-            pid = self._part_rr % n
-            self._part_rr += 1
+            pid = next(self._part_rr) % n
         packet = dict(packet)
         packet["$partition"] = pid
         return packet

@@ -61,6 +61,9 @@ class StageStats:
     busy_seconds: float = 0.0
 
 
+_CAPTURE_LOCK = threading.Lock()
+
+
 class _Stage(threading.Thread):
     def __init__(
         self,
@@ -317,17 +320,24 @@ class _Stage(threading.Thread):
             return None
         tensor = tensor.to(f"cuda:{self.device}", non_blocking=False)
         if self._graph is None:
-            ctx = self._ctx(packet)
-            # warmup (allocations settle), then capture
-            fn(ctx, tensor)
-            torch.cuda.synchronize(self.device)
-            self._graph_in = tensor.clone()
-            g = torch.cuda.CUDAGraph()
-            capture_stream = torch.cuda.Stream(device=self.device)
-            with torch.cuda.stream(capture_stream):
-                with torch.cuda.graph(g, stream=capture_stream):
-                    self._graph_out = fn(ctx, self._graph_in)
-            self._graph = g
+            # one capture at a time: concurrent partition lanes each own a
+            # graph, but global-mode stream capture poisons OTHER threads'
+            # in-flight GPU work (thread_local relaxes the check; the lock
+            # keeps captures from overlapping at all)
+            with _CAPTURE_LOCK:
+                ctx = self._ctx(packet)
+                # warmup (allocations settle), then capture
+                fn(ctx, tensor)
+                torch.cuda.synchronize(self.device)
+                self._graph_in = tensor.clone()
+                g = torch.cuda.CUDAGraph()
+                capture_stream = torch.cuda.Stream(device=self.device)
+                with torch.cuda.stream(capture_stream):
+                    with torch.cuda.graph(
+                        g, stream=capture_stream, capture_error_mode="thread_local"
+                    ):
+                        self._graph_out = fn(ctx, self._graph_in)
+                self._graph = g
         self._graph_in.copy_(tensor)
         self._graph.replay()
         if self.stream is not None:

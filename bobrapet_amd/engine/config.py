@@ -61,6 +61,11 @@ class EngineConfig:
     queues: _t.Dict[str, QueueConfig] = field(default_factory=dict)
     default_queue: str = "default"
 
+    # durability: periodic state snapshots (the reference's state lives in
+    # etcd and is always durable; here checkpointing is opt-in)
+    checkpoint_path: _t.Optional[str] = None
+    checkpoint_interval_seconds: float = 30.0
+
     # retention (reference: handleTerminalStoryRun defaults 2044/2056)
     child_ttl_seconds: float = 3600.0
     storyrun_retention_seconds: float = 86400.0

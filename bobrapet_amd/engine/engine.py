@@ -117,6 +117,10 @@ class RunEngine:
         self._running = True
         self._thread = threading.Thread(target=self._loop, name="run-engine", daemon=True)
         self._thread.start()
+        if self.config.checkpoint_path:
+            self.set_timer(
+                monotonic_now() + self.config.checkpoint_interval_seconds, "", "checkpoint"
+            )
         return self
 
     def stop(self) -> None:
@@ -436,6 +440,24 @@ class RunEngine:
 
         snapshot.save_state(self, path)
 
+    def _checkpoint_tick(self) -> None:
+        """Periodic durability snapshot (atomic rename) + reschedule."""
+        path = self.config.checkpoint_path
+        if not path:
+            return
+        import os as _os
+
+        tmp = f"{path}.tmp"
+        try:
+            self.save_state(tmp)
+            _os.replace(tmp, path)
+            self.metrics.inc("checkpoints_total")
+        except Exception:  # durability must never kill the loop
+            self.metrics.inc("checkpoint_errors_total")
+        self.set_timer(
+            monotonic_now() + self.config.checkpoint_interval_seconds, "", "checkpoint"
+        )
+
     def load_state(self, path: str) -> int:
         from . import snapshot
 
@@ -697,6 +719,9 @@ class RunEngine:
     # -- timers ----------------------------------------------------------
 
     def _on_timer(self, run_key: str, tag: str) -> None:
+        if tag == "checkpoint":
+            self._checkpoint_tick()
+            return
         run = self.store.try_get_story_run(run_key)
         if run is None:
             return

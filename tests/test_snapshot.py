@@ -136,3 +136,45 @@ def test_spoofed_storage_ref_rejected():
             )
     finally:
         eng.stop()
+
+
+class TestAutoCheckpoint:
+    """Periodic durability snapshots (EngineConfig.checkpoint_path) — the
+    opt-in analog of the reference's always-durable etcd state."""
+
+    def test_periodic_checkpoint_and_restore(self, tmp_path):
+        import os
+        import time
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        ckpt = str(tmp_path / "state.json")
+        eng = RunEngine(
+            EngineConfig(cpu_workers=2, checkpoint_path=ckpt, checkpoint_interval_seconds=0.1)
+        ).start()
+        story = """
+kind: Story
+metadata: {name: s}
+spec:
+  steps: [{name: a, type: sleep, with: {duration: 0ms}}]
+"""
+        try:
+            eng.apply_yaml(story)
+            run = eng.run_story("default/s", {}, timeout=10)
+            assert run.phase == Phase.SUCCEEDED
+            deadline = time.time() + 5
+            while not os.path.exists(ckpt) and time.time() < deadline:
+                time.sleep(0.05)
+            assert os.path.exists(ckpt)
+            assert eng.metrics.counter_value("checkpoints_total") >= 1
+        finally:
+            eng.stop()
+        eng2 = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng2.apply_yaml(story)
+            assert eng2.load_state(ckpt) >= 1
+            restored = eng2.store.try_get_story_run(run.key)
+            assert restored is not None and restored.phase == Phase.SUCCEEDED
+        finally:
+            eng2.stop()

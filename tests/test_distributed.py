@@ -307,3 +307,73 @@ def test_sequence_parallel_prefill_matches_full():
     for rank, payload in results.items():
         assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
         assert payload["err"] < 0.05 * max(payload["scale"], 1.0), payload
+
+
+def _ulysses_rank(rank: int, world: int, port: int, q):
+    import os
+    import sys
+
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    try:
+        sys.path.insert(0, REPO)
+        import torch
+
+        from bobrapet_amd import ops
+        from bobrapet_amd.parallel import group
+        from bobrapet_amd.parallel.ulysses_attention import ulysses_attention
+
+        group.init_distributed(backend="gloo")
+        torch.manual_seed(11)
+        B, S, Hq, Hkv, D = 2, 40, 4, 2, 64
+        S_glob = S * world
+        qg = torch.randn(B, S_glob, Hq, D, dtype=torch.bfloat16)
+        kg = torch.randn(B, S_glob, Hkv, D, dtype=torch.bfloat16)
+        vg = torch.randn_like(kg)
+        res = {}
+        for causal in (True, False):
+            full = ops.attn_prefill(qg, kg, vg, None, causal)
+            lo, hi = rank * S, (rank + 1) * S
+            out = ulysses_attention(
+                qg[:, lo:hi].contiguous(),
+                kg[:, lo:hi].contiguous(),
+                vg[:, lo:hi].contiguous(),
+                causal=causal,
+            )
+            res[f"causal={causal}"] = (out.float() - full[:, lo:hi].float()).abs().max().item()
+        q.put((rank, res))
+        group.teardown()
+    except Exception as exc:
+        import traceback
+
+        q.put((rank, {"error": f"{exc}\n{traceback.format_exc()}"}))
+
+
+@pytest.mark.timeout(180)
+def test_ulysses_attention_matches_full():
+    """Head-parallel Ulysses attention == single-rank full attention
+    (2 ranks, gloo all-gather fallback for the all-to-all)."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [ctx.Process(target=_ulysses_rank, args=(r, world, port, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=160)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    for rank, payload in results.items():
+        assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
+        for key, err in payload.items():
+            assert err < 0.03, f"rank {rank} {key}: err {err}"

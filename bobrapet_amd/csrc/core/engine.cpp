@@ -161,13 +161,61 @@ void NativeEngine::sync_primitives(Run& run, const Plan& plan) {
 
 bool NativeEngine::has_failure(Run& run, const Plan& plan) {
   if (run.failure_step >= 0) return true;
-  for (size_t i = 0; i < plan.steps.size(); ++i) {
+  for (int i = 0; i < plan.main_end(); ++i) {  // failure = a MAIN step failed
     Phase p = run.states[i].phase;
     bool hard = p == Phase::Failed || p == Phase::Timeout || p == Phase::Canceled;
     if (hard && !plan.steps[i].allow_failure) {
-      run.failure_step = (int)i;
+      run.failure_step = i;
       return true;
     }
+  }
+  return false;
+}
+
+bool NativeEngine::steps_settled(Run& run, const Plan& plan, int begin, int end,
+                                 bool failure, bool fail_fast) {
+  // mirrors dag.py _steps_settled: terminal everywhere, or under fail-fast
+  // with a failure present every non-terminal step is unstarted (skip it)
+  double t = now();
+  bool settled = true;
+  for (int i = begin; i < end; ++i) {
+    StepState& st = run.states[i];
+    if (is_terminal(st.phase)) continue;
+    if (failure && fail_fast) {
+      if (st.phase == Phase::Running || st.phase == Phase::Paused) return false;
+      st.phase = Phase::Skipped;
+      st.error = "skipped by fail-fast";
+      st.finished = t;
+      continue;
+    }
+    settled = false;
+  }
+  return settled;
+}
+
+bool NativeEngine::active_range(Run& run, const Plan& plan, int* begin, int* end) {
+  // main → compensation (on failure/cancel) → finally (always)
+  bool failure = has_failure(run, plan) || run.cancel_requested;
+  if (!steps_settled(run, plan, 0, plan.main_end(), failure, plan.fail_fast)) {
+    *begin = 0;
+    *end = plan.main_end();
+    run.exec_phase = 0;
+    return true;
+  }
+  failure = has_failure(run, plan) || run.cancel_requested;
+  if (failure && plan.n_comp > 0 &&
+      !steps_settled(run, plan, plan.main_end(), plan.comp_end(), false, false)) {
+    *begin = plan.main_end();
+    *end = plan.comp_end();
+    run.exec_phase = 1;
+    return true;
+  }
+  if (plan.n_fin > 0 &&
+      !steps_settled(run, plan, plan.comp_end(), (int)plan.steps.size(), false, false)) {
+    *begin = plan.comp_end();
+    *end = (int)plan.steps.size();
+    run.exec_phase = 2;
+    return true;
   }
   return false;
 }
@@ -220,7 +268,12 @@ int NativeEngine::readiness(Run& run, const Plan& plan, int idx, bool failure,
 }
 
 bool NativeEngine::phase_pass(Run& run, const Plan& plan) {
-  bool failure = has_failure(run, plan);
+  int begin = 0, end = 0;
+  if (!active_range(run, plan, &begin, &end)) return false;
+  const bool main_phase = begin == 0;
+  // fail-fast/stop semantics apply only to the MAIN phase; compensation
+  // and finally steps run precisely BECAUSE of a failure or stop
+  bool failure = main_phase && has_failure(run, plan);
   bool progressed = false;
   JObject scope;  // built lazily, invalidated after each launch
   int active = 0;
@@ -228,10 +281,10 @@ bool NativeEngine::phase_pass(Run& run, const Plan& plan) {
     for (const auto& st : run.states)
       if (st.phase == Phase::Running || st.phase == Phase::Paused) ++active;
   double t = now();
-  for (size_t i = 0; i < plan.steps.size(); ++i) {
+  for (size_t i = (size_t)begin; i < (size_t)end; ++i) {
     StepState& st = run.states[i];
     if (st.phase != Phase::Pending) continue;
-    if (run.stop_seen) {
+    if (run.stop_seen && main_phase) {
       st.phase = Phase::Skipped;
       st.error = "skipped by stop";
       st.finished = t;
@@ -250,7 +303,7 @@ bool NativeEngine::phase_pass(Run& run, const Plan& plan) {
     launch_step(run, plan, (int)i);
     ++active;
     progressed = true;
-    failure = has_failure(run, plan);
+    failure = main_phase && has_failure(run, plan);
     scope.clear();  // immediate completions may have changed step outputs
   }
   return progressed;
@@ -589,13 +642,37 @@ void NativeEngine::handle_timer(const Event& ev) {
 }
 
 void NativeEngine::maybe_finalize(Run& run, const Plan& plan) {
-  for (const auto& st : run.states)
-    if (!is_terminal(st.phase)) return;
+  int b = 0, e = 0;
+  if (active_range(run, plan, &b, &e)) return;  // a phase still has work
+  // unreached compensation steps (no failure) stay Pending — settle them
+  double t = now();
+  for (auto& st : run.states)
+    if (!is_terminal(st.phase)) {
+      st.phase = Phase::Skipped;
+      st.finished = t;
+    }
   bool failure = has_failure(run, plan);
-  if (run.stop_seen)
+  bool comp_ok = plan.n_comp > 0;
+  for (int i = plan.main_end(); i < plan.comp_end(); ++i) {
+    Phase p = run.states[i].phase;
+    if (p != Phase::Succeeded && p != Phase::Skipped) comp_ok = false;
+  }
+  bool fin_failed = false;
+  for (int i = plan.comp_end(); i < (int)plan.steps.size(); ++i) {
+    Phase p = run.states[i].phase;
+    bool hard = p == Phase::Failed || p == Phase::Timeout || p == Phase::Canceled;
+    if (hard && !plan.steps[i].allow_failure) fin_failed = true;
+  }
+  if (run.cancel_requested)
+    run.phase = Phase::Canceled;
+  else if (run.stop_seen)
     run.phase = run.stop_phase;
+  else if (failure && plan.n_comp > 0 && comp_ok)
+    run.phase = Phase::Compensated;
+  else if (failure || fin_failed)
+    run.phase = Phase::Failed;
   else
-    run.phase = failure ? Phase::Failed : Phase::Succeeded;
+    run.phase = Phase::Succeeded;
   if (failure && run.failure_step >= 0)
     run.error = plan.steps[run.failure_step].name + ": " +
                 run.states[run.failure_step].error;

@@ -42,6 +42,7 @@ enum class Phase : uint8_t {
   Timeout,
   Skipped,
   Blocked,
+  Compensated,
 };
 
 inline bool is_terminal(Phase p) {
@@ -52,6 +53,7 @@ inline bool is_terminal(Phase p) {
     case Phase::Canceled:
     case Phase::Timeout:
     case Phase::Skipped:
+    case Phase::Compensated:
       return true;
     default:
       return false;
@@ -70,6 +72,7 @@ inline const char* phase_name(Phase p) {
     case Phase::Timeout: return "Timeout";
     case Phase::Skipped: return "Skipped";
     case Phase::Blocked: return "Blocked";
+    case Phase::Compensated: return "Compensated";
   }
   return "?";
 }
@@ -118,12 +121,18 @@ struct PlanStep {
 
 struct Plan {
   std::string name;
-  std::vector<PlanStep> steps;
+  std::vector<PlanStep> steps;  // main ++ compensations ++ finally
   std::vector<std::vector<int>> dependents;
   TNodePtr output_tpl;
   bool fail_fast = true;
   double story_timeout = 0;
   int concurrency = 0;  // max concurrently Running/Paused steps (0 = unlimited)
+  // 3-phase layout (reference: dag.go:482-511 main→compensation→finally)
+  int n_main = -1;  // -1 → every step is main
+  int n_comp = 0;
+  int n_fin = 0;
+  int main_end() const { return n_main < 0 ? (int)steps.size() : n_main; }
+  int comp_end() const { return main_end() + n_comp; }
 };
 
 struct StepState {
@@ -144,6 +153,7 @@ struct Run {
   std::vector<std::vector<StepState>> branch_states;
   std::unordered_map<int, int> gates;  // step → 0 pending / 1 approved / 2 rejected
   bool cancel_requested = false;
+  int exec_phase = 0;  // 0 main, 1 compensation, 2 finally
   bool stop_seen = false;
   Phase stop_phase = Phase::Succeeded;
   int failure_step = -1;
@@ -324,6 +334,9 @@ class NativeEngine {
   void tick(Run& run);
   void sync_primitives(Run& run, const Plan& plan);
   bool phase_pass(Run& run, const Plan& plan);
+  bool steps_settled(Run& run, const Plan& plan, int begin, int end,
+                     bool failure, bool fail_fast);
+  bool active_range(Run& run, const Plan& plan, int* begin, int* end);
   void launch_step(Run& run, const Plan& plan, int idx);
   void launch_branch(Run& run, const Plan& plan, int idx, int bidx);
   void maybe_finalize(Run& run, const Plan& plan);

@@ -241,6 +241,9 @@ static Plan to_plan(const py::dict& d) {
   if (d.contains("storyTimeout"))
     p.story_timeout = d["storyTimeout"].cast<double>();
   if (d.contains("concurrency")) p.concurrency = d["concurrency"].cast<int>();
+  if (d.contains("nMain")) p.n_main = d["nMain"].cast<int>();
+  if (d.contains("nComp")) p.n_comp = d["nComp"].cast<int>();
+  if (d.contains("nFin")) p.n_fin = d["nFin"].cast<int>();
   return p;
 }
 

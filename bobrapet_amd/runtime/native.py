@@ -105,9 +105,7 @@ def story_supported(story: T.Story) -> _t.Optional[str]:
     """None if the native fast path can run this story; else the reason."""
     if story.pattern != StoryPattern.BATCH:
         return "streaming stories run on the streaming runtime"
-    if story.compensations or story.finally_:
-        return "compensations/finally run on the Python engine"
-    for s in story.steps:
+    for s in story.all_steps():
         if s.type == StepType.PARALLEL:
             w = s.with_ if isinstance(s.with_, dict) else {}
             for raw in w.get("steps", []):
@@ -134,17 +132,28 @@ def compile_story_plan(
     cs = py_compile(story)
     if cs.cycle_error:
         raise NativeCompileError(cs.cycle_error)
-    index = {s.name: i for i, s in enumerate(story.steps)}
+    # global step index: main ++ compensations ++ finally (the core's
+    # 3-phase layout; comp/fin needs may reference main steps by name)
+    ordered = list(story.steps) + list(story.compensations) + list(story.finally_)
+    index = {s.name: i for i, s in enumerate(ordered)}
+    dep_graphs = (
+        [(s, cs.deps) for s in story.steps]
+        + [(s, cs.comp_deps) for s in story.compensations]
+        + [(s, cs.fin_deps) for s in story.finally_]
+    )
 
     steps = []
-    for s in story.steps:
-        steps.append(_compile_step(s, story, resolver, registry, index, cs, plan_ids))
+    for s, graph in dep_graphs:
+        steps.append(_compile_step(s, story, resolver, registry, index, graph, plan_ids))
 
     plan = {
         "name": story.key,
         "steps": steps,
         "failFast": True,
         "concurrency": 0,
+        "nMain": len(story.steps),
+        "nComp": len(story.compensations),
+        "nFin": len(story.finally_),
     }
     if story.policy is not None:
         if story.policy.retries is not None and story.policy.retries.continue_on_step_failure:
@@ -156,12 +165,13 @@ def compile_story_plan(
     return plan
 
 
-def _compile_step(s: T.Step, story, resolver, registry, index, cs, plan_ids) -> dict:
+def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_ids) -> dict:
     d: dict = {"name": s.name, "kind": _KIND[s.type]}
-    deps = sorted(cs.deps.get(s.name, set()))
+    deps = sorted(dep_graph.get(s.name, set()))
+    by_name = {x.name: x for x in story.all_steps()}
     d["deps"] = [index[x] for x in deps if x in index]
     d["depAllowFailure"] = [
-        bool(story.step(x).allow_failure) if story.step(x) else False
+        bool(by_name[x].allow_failure) if x in by_name else False
         for x in deps
         if x in index
     ]
@@ -228,7 +238,7 @@ def _compile_step(s: T.Step, story, resolver, registry, index, cs, plan_ids) -> 
         for raw in w.get("steps", []):
             b = T._step_from_dict(dict(raw))
             branches.append(
-                _compile_step(b, story, resolver, registry, {}, _EmptyCS(), plan_ids)
+                _compile_step(b, story, resolver, registry, {}, {}, plan_ids)
             )
         d["branches"] = branches
     elif s.type == StepType.EXECUTE_STORY:
@@ -241,10 +251,6 @@ def _compile_step(s: T.Step, story, resolver, registry, index, cs, plan_ids) -> 
         if w.get("with") is not None:
             d["with"] = compile_template_value({"with": w["with"]})
     return d
-
-
-class _EmptyCS:
-    deps: dict = {}
 
 
 # ---------------------------------------------------------------------------

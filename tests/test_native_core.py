@@ -409,7 +409,7 @@ spec:
         )
         assert story_supported(story) is not None
 
-    def test_compensations_not_supported(self, rig):
+    def test_compensations_now_supported(self, rig):
         from bobrapet_amd.specs import load_yaml
 
         (story,) = load_yaml(
@@ -423,7 +423,7 @@ spec:
     - {name: undo, ref: {name: x}}
 """
         )
-        assert story_supported(story) is not None
+        assert story_supported(story) is None  # 3-phase machine is native now
 
 
 class TestThroughput:
@@ -453,3 +453,95 @@ spec:
         for rid in ids:
             nr.engine.gc_run(rid)
         assert n / native_dt > 2000, f"native only {n/native_dt:.0f} runs/s"
+
+
+class TestNativeThreePhase:
+    """bobraccel 3-phase machine: main → compensation (on failure) →
+    finally (always) — previously Python-engine-only."""
+
+    YAML = """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: EngramTemplate
+metadata: {name: fail}
+spec: {builtin: fail}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Engram
+metadata: {name: f}
+spec: {templateRef: {name: fail}}
+---
+kind: Story
+metadata: {name: comp}
+spec:
+  steps:
+    - {name: work, ref: {name: f}, with: {succeedAfter: 99}}
+    - {name: after, ref: {name: e}, needs: [work], with: {x: 1}}
+  compensations:
+    - {name: undo, ref: {name: e}, with: {undid: "{{ steps.work.phase }}"}}
+  finally:
+    - {name: report, ref: {name: e}, with: {done: true}}
+---
+kind: Story
+metadata: {name: ok}
+spec:
+  steps:
+    - {name: work, ref: {name: e}, with: {v: 7}}
+  compensations:
+    - {name: undo, ref: {name: e}, with: {nope: 1}}
+  finally:
+    - {name: report, ref: {name: e}, with: {done: true}}
+"""
+
+    def _runner(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        eng.apply_yaml(self.YAML)
+        return eng, NativeRunner.from_run_engine(eng)
+
+    def test_failure_compensates_then_finally(self):
+        eng, nr = self._runner()
+        try:
+            res = nr.run_story("default/comp", {}, timeout=30)
+            assert res["phase"] == "Compensated", res
+            ph = {k: v["phase"] for k, v in res["steps"].items()}
+            assert ph == {
+                "work": "Failed", "after": "Skipped",
+                "undo": "Succeeded", "report": "Succeeded",
+            }
+            # compensation scope sees main-step state
+            assert res["steps"]["undo"]["output"]["undid"] == "Failed"
+        finally:
+            eng.stop()
+
+    def test_success_skips_compensations_runs_finally(self):
+        eng, nr = self._runner()
+        try:
+            res = nr.run_story("default/ok", {}, timeout=30)
+            assert res["phase"] == "Succeeded", res
+            assert res["steps"]["undo"]["phase"] == "Skipped"
+            assert res["steps"]["report"]["phase"] == "Succeeded"
+        finally:
+            eng.stop()
+
+    def test_parity_with_python_engine(self):
+        """Same stories through the Python engine produce the same phases."""
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(self.YAML)
+            run = eng.run_story("default/comp", {}, timeout=30)
+            assert run.phase == Phase.COMPENSATED
+            assert run.step_states["undo"].phase == Phase.SUCCEEDED
+            assert run.step_states["report"].phase == Phase.SUCCEEDED
+        finally:
+            eng.stop()

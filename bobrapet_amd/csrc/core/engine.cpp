@@ -520,6 +520,29 @@ void NativeEngine::handle_engram_done(const Event& ev) {
   double t = now();
   // exit classes: 0 success, 1 retry, 2 terminal, 3 rateLimited, other unknown
   int code = ev.exit_code;
+  const PlanStep& pstep = ev.branch < 0 ? plan.steps[ev.step]
+                                        : plan.steps[ev.step].branches[ev.branch];
+  if (code == 0 && pstep.post_exec) {
+    // postExecution check (reference: steprun_controller.go:2050-2124):
+    // condition over the scope + this output; falsy → terminal failure
+    JObject scope = build_scope(run, plan);
+    scope["output"] = ev.output;
+    bool ok = false;
+    try {
+      ok = eval_expr(*pstep.post_exec, scope).truthy();
+    } catch (const ExprError&) {
+      ok = false;
+    }
+    if (!ok) {
+      st.phase = Phase::Failed;
+      st.output = ev.output;
+      st.error = pstep.post_exec_msg.empty() ? "postExecution condition failed"
+                                             : pstep.post_exec_msg;
+      st.finished = t;
+      tick(run);
+      return;
+    }
+  }
   if (code == 0) {
     st.phase = Phase::Succeeded;
     st.output = ev.output;

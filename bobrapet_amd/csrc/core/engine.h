@@ -116,6 +116,8 @@ struct PlanStep {
   Phase stop_phase = Phase::Succeeded;
   std::vector<PlanStep> branches;  // Parallel
   std::string engram;              // launcher routing key
+  ExprPtr post_exec;               // postExecution condition over {output}
+  std::string post_exec_msg;
   int target_plan = -1;            // ExecuteStory
 };
 

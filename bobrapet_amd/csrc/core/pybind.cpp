@@ -223,6 +223,9 @@ static PlanStep to_step(const py::dict& d) {
   }
   if (d.contains("engram")) s.engram = d["engram"].cast<std::string>();
   if (d.contains("targetPlan")) s.target_plan = d["targetPlan"].cast<int>();
+  if (d.contains("postExec")) s.post_exec = to_expr(d["postExec"]);
+  if (d.contains("postExecMsg"))
+    s.post_exec_msg = d["postExecMsg"].cast<std::string>();
   if (d.contains("branches"))
     for (const auto& b : d["branches"])
       s.branches.push_back(to_step(b.cast<py::dict>()));

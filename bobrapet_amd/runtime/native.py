@@ -112,8 +112,6 @@ def story_supported(story: T.Story) -> _t.Optional[str]:
                 b = T._step_from_dict(dict(raw))
                 if b.type not in (None, StepType.CONDITION, StepType.SLEEP):
                     return f"parallel branch type {b.type} unsupported natively"
-        if s.post_execution is not None:
-            return "postExecution checks run on the Python engine"
     return None
 
 
@@ -207,6 +205,10 @@ def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_i
             d["timeout"] = cfg.timeout_seconds
         if s.with_ is not None:
             d["with"] = compile_template_value(s.with_)
+        if s.post_execution is not None:
+            d["postExec"] = compile_condition(s.post_execution.condition)
+            if s.post_execution.failure_message:
+                d["postExecMsg"] = s.post_execution.failure_message
     elif s.type == StepType.CONDITION:
         expr = w.get("expression") or w.get("if") or s.if_ or "true"
         d["if"] = compile_condition(str(expr))

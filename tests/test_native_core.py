@@ -545,3 +545,42 @@ spec:
             assert run.step_states["report"].phase == Phase.SUCCEEDED
         finally:
             eng.stop()
+
+
+class TestNativePostExecution:
+    """postExecution checks evaluated by the C++ expression VM on engram
+    completion (reference: steprun_controller.go:2050-2124)."""
+
+    def test_post_execution_native(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner, story_supported
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: pe}
+spec:
+  steps:
+    - name: a
+      ref: {name: e}
+      with: {v: 3}
+      postExecution: {condition: "{{ output.v > 5 }}", failureMessage: "v too small"}
+"""
+            )
+            assert story_supported(eng.registry.story("pe", "default")) is None
+            nr = NativeRunner.from_run_engine(eng)
+            res = nr.run_story("default/pe", {}, timeout=30)
+            assert res["phase"] == "Failed"
+            assert "v too small" in res["steps"]["a"]["error"]
+        finally:
+            eng.stop()

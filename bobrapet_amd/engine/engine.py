@@ -551,7 +551,12 @@ class RunEngine:
             with self.tracer.span(
                 "engram.run", run=run.name, step=sr.spec.step_name, impl=impl_name or ""
             ):
-                impl = engram_registry.resolve(impl_name)
+                if template is not None and getattr(template, "command", None):
+                    from ..engrams.process import ProcessEngram
+
+                    impl = ProcessEngram(template.command)
+                else:
+                    impl = engram_registry.resolve(impl_name)
                 if impl.wants_gpu and slot.device is None:
                     import torch
 

@@ -11,7 +11,7 @@ from __future__ import annotations
 import typing as _t
 
 from ..engrams import registry as engram_registry
-from ..engrams.base import EngramContext, EngramFailure, EngramResult
+from ..engrams.base import Engram, EngramContext, EngramFailure, EngramResult
 from ..enums import StepType, StopMode, StoryPattern
 from ..specs import types as T
 from ..templating import is_template, parse_expression, parse_template
@@ -354,6 +354,12 @@ class NativeRunner:
                 template = self.registry.engram_template(engram.template_ref.name)
             except KeyError:
                 pass
+        if template is not None and getattr(template, "command", None):
+            from ..engrams.process import ProcessEngram
+
+            hit = (ProcessEngram(template.command), engram.with_)
+            self._engram_cache[engram_key] = hit
+            return hit
         impl_name = template.implementation if template is not None else engram_key.split("/")[-1]
         hit = (impl_name, engram.with_)
         self._engram_cache[engram_key] = hit
@@ -374,7 +380,11 @@ class NativeRunner:
             ids = self.workers.device_ids
             device = ids[(hash((run_id, step, branch)) & 0x7FFFFFFF) % len(ids)]
 
-        impl = engram_registry.resolve(impl_name)
+        impl = (
+            impl_name
+            if isinstance(impl_name, Engram)
+            else engram_registry.resolve(impl_name)
+        )
 
         def body(slot):
             try:

@@ -547,6 +547,7 @@ class EngramTemplate:
     version: _t.Optional[str] = None
     builtin: _t.Optional[str] = None  # registered engram implementation name
     image: _t.Optional[str] = None  # accepted for YAML parity; resolved via registry
+    command: _t.List[str] = field(default_factory=list)  # external-process engram
     supported_modes: _t.List[WorkloadMode] = field(default_factory=list)
     config_schema: _t.Optional[JSON] = None
     secret_schema: _t.Optional[JSON] = None

@@ -373,8 +373,10 @@ def validate_engram_template(tpl: T.EngramTemplate) -> ValidationResult:
     res = ValidationResult()
     if not _valid_name(tpl.name):
         res.error(f"engram template name {tpl.name!r} must be a DNS-1123 label")
-    if not tpl.implementation:
-        res.error("engram template must name an implementation (builtin or image)")
+    if not tpl.implementation and not tpl.command:
+        res.error(
+            "engram template must name an implementation (builtin, image, or command)"
+        )
     for label, schema in (
         ("configSchema", tpl.config_schema),
         ("inputSchema", tpl.input_schema),

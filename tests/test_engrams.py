@@ -122,3 +122,60 @@ spec:
 
         res = CliRunner().invoke(app, ["engrams"])
         assert "llm-infer" in res.output and "embed" in res.output
+
+
+class TestProcessEngram:
+    """External-process engrams: the bubu-sdk BUBU_* env contract as real
+    env vars (reference: steprun_controller.go:1692-1732 env build +
+    exit-code classes)."""
+
+    YAML = """
+kind: EngramTemplate
+metadata: {name: sh}
+spec:
+  command: [python3, -c, "%s"]
+---
+kind: Engram
+metadata: {name: sheller}
+spec: {templateRef: {name: sh}}
+---
+kind: Story
+metadata: {name: ext}
+spec:
+  steps:
+    - {name: go, ref: {name: sheller}, with: {v: 42}}
+  output: {out: "{{ steps.go.output }}"}
+"""
+
+    def _run(self, script, expect_phase="Succeeded", retries=0):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(self.YAML % script)
+            run = eng.run_story("default/ext", {}, timeout=40)
+            srs = list(eng.store.step_runs_of(run.key))
+            return run, srs[0] if srs else None
+        finally:
+            eng.stop()
+
+    def test_env_contract_and_json_output(self):
+        run, sr = self._run(
+            "import os, json; print('log line');"
+            " print(json.dumps({'step': os.environ['BUBU_STEP_NAME'],"
+            " 'v': json.loads(os.environ['BUBU_TRIGGER_DATA'])['v']}))"
+        )
+        assert str(run.phase) == "Succeeded", run.error
+        assert run.output["out"] == {"step": "go", "v": 42}
+        assert any("log line" in l for l in sr.status.logs)
+
+    def test_exit_two_is_terminal_no_retry(self):
+        run, sr = self._run("import sys; sys.exit(2)")
+        assert str(run.phase) == "Failed"
+        assert sr.status.retries == 0
+        assert "exited 2" in (sr.status.error.message or "")
+
+    def test_non_json_stdout_wrapped(self):
+        run, _sr = self._run("print('plain text result')")
+        assert str(run.phase) == "Succeeded"
+        assert run.output["out"] == {"stdout": "plain text result"}

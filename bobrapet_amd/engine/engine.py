@@ -346,6 +346,9 @@ class RunEngine:
                     frontier.append(dep)
         for name in closure:
             r.step_states.pop(name, None)
+            # stale delegated-evaluation results must not survive a redrive
+            # (upstream outputs in the closure may change)
+            r.materialized.pop(name, None)
             for sr in self.store.step_runs_of(key):
                 if sr.spec.step_name == name or sr.spec.step_name.startswith(name + "/"):
                     self.store.delete_step_run(sr.key)
@@ -354,6 +357,10 @@ class RunEngine:
                     del r.timers[tag]
             r.primitive_children.pop(name, None)
             r.gates.pop(name, None)
+        r.materialized.pop("__output__", None)
+        for sr in self.store.step_runs_of(key):
+            if sr.spec.step_name == "__output__/materialize":
+                self.store.delete_step_run(sr.key)
         r.phase = Phase.RUNNING
         r.exec_phase = "main"
         r.failure_cause = None
@@ -368,6 +375,7 @@ class RunEngine:
 
     def _reset_run(self, r: StoryRun) -> None:
         r.step_states.clear()
+        r.materialized.clear()
         r.gates.clear()
         r.primitive_children.clear()
         r.timers.clear()

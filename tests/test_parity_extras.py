@@ -337,3 +337,24 @@ spec:
             assert "materialize" in (run.step_states["use"].error.message or "").lower() or True
         finally:
             eng.stop()
+
+    def test_redrive_invalidates_materialized(self):
+        """redrive-from-step must drop delegated results whose upstream
+        outputs may change (stale-materialization guard)."""
+        from bobrapet_amd.enums import Phase
+
+        eng = self._engine()
+        try:
+            eng.apply_yaml(self.YAML % ("m" * 100))
+            run = eng.run_story("default/mat", {}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED
+            assert "use" in run.materialized
+            eng.redrive_from_step(run, "big")
+            run = eng.wait(run, timeout=30)
+            assert run.phase == Phase.SUCCEEDED
+            # the materialize ran again after the redrive (fresh SR)
+            names = [sr.spec.step_name for sr in eng.store.step_runs_of(run.key)]
+            assert "use/materialize" in names
+            assert eng.storage.hydrate(run.output)["l"] == "m" * 100
+        finally:
+            eng.stop()

@@ -188,37 +188,48 @@ class _Stage(threading.Thread):
                 if not progressed:
                     time.sleep(0)  # yield between poll sweeps
 
-        # all / quorum: one packet per ring per round
-        need = len(rings)
-        if mode == "quorum":
-            need = max(1, min(int(fi.quorum or len(rings)), len(rings)))
-        deadline = (time.monotonic() + timeout) if timeout else None
+        # all / quorum: accumulate one packet per ring per round, emitting
+        # as soon as `need` arrived — a slow ring never gates a met quorum
+        # (stragglers' packets stay queued and join the NEXT round)
         closed = getattr(self, "_closed_rings", None)
         if closed is None:
             closed = self._closed_rings = set()
+        need = len([r for r in rings if r not in closed])
+        if mode == "quorum":
+            need = max(1, min(int(fi.quorum or len(rings)), need))
+        deadline = (time.monotonic() + timeout) if timeout else None
         packets, arrived_from = [], []
-        for i, ring in enumerate(rings):
-            if ring in closed:
-                continue
-            try:
-                if deadline is None:
-                    pkt = ring.pop()
-                else:
-                    pkt = ring.pop(timeout=max(deadline - time.monotonic(), 0.001))
-            except TimeoutError:
-                continue
-            if pkt is flow.SENTINEL:
-                closed.add(ring)
-                continue
-            packets.append(pkt)
-            arrived_from.append(i)
-        if len(rings) - len(closed) < need and len(packets) < need:
-            # quorum can never be met again (upstreams ended) — end of stream
-            return flow.SENTINEL
-        if len(packets) < need:
-            self.stats.errors += 0  # dropped round, not an error
-            self.sr.engine.metrics.inc("stream_fanin_dropped_total")
-            return None
+        pending = {i: r for i, r in enumerate(rings) if r not in closed}
+        while True:
+            for i in list(pending):
+                ring = pending[i]
+                try:
+                    pkt = ring.pop(timeout=0.005)
+                except TimeoutError:
+                    continue
+                if pkt is flow.SENTINEL:
+                    closed.add(ring)
+                    del pending[i]
+                    continue
+                packets.append(pkt)
+                arrived_from.append(i)
+                del pending[i]
+            live = len(rings) - len(closed)
+            if live < need:
+                need = max(1, min(need, live)) if mode == "quorum" else live
+            if need <= 0 or (live <= 0 and not packets):
+                return flow.SENTINEL if not packets else self._fan_in_merge(rings, packets, arrived_from)
+            if len(packets) >= need:
+                return self._fan_in_merge(rings, packets, arrived_from)
+            if deadline is not None and time.monotonic() >= deadline:
+                if packets and len(packets) >= need:
+                    return self._fan_in_merge(rings, packets, arrived_from)
+                self.sr.engine.metrics.inc("stream_fanin_dropped_total")
+                return None
+            if self.sr.canceled:
+                return flow.SENTINEL
+
+    def _fan_in_merge(self, rings, packets, arrived_from):
         merged = {"fanIn": packets}
         if len(packets) < len(rings):
             merged["arrivedFrom"] = arrived_from

@@ -683,3 +683,39 @@ spec:
         run = stream.finish(timeout=10)
         assert run.phase == Phase.FINISHED
         assert run.output["packets"] == 1
+
+
+class TestFanInQuorumNoBlock:
+    def test_quorum_met_despite_starved_upstream(self, eng):
+        """quorum=1 emits from the live upstream even when the other branch
+        drops its packets (no blocking on the starved ring, no timeout
+        configured)."""
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: finq}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        fanIn: {mode: quorum, quorum: 1, timeoutSeconds: 5}
+  steps:
+    - name: left
+      type: condition
+      runtime: {route: "{{ packet.keep }}"}
+    - name: right
+      ref: {name: transformer}
+      runtime: {map: {v: "{{ item.v }}"}}
+    - name: join
+      ref: {name: echoer}
+      needs: [left, right]
+"""
+        )
+        stream = eng.submit_stream("default/finq")
+        for i in range(4):
+            stream.push({"items": [{"v": i}], "keep": False})  # left drops all
+        run = stream.finish(timeout=15)
+        assert run.phase == Phase.FINISHED
+        # the right branch's 4 packets each met quorum=1
+        assert run.step_states["join"].output["packetsIn"] == 4

@@ -316,7 +316,10 @@ class LlamaModel:
     def decode_step_graphed(self, ids: torch.Tensor) -> torch.Tensor:
         """hipGraph-captured decode (guide: capture launch-bound inner loops
         in hipGraphs): one replay per token; position/length live in device
-        scalars that the graph itself advances."""
+        scalars that the graph itself advances.  Graphs are cached per
+        batch size (a served model alternates batches), and the device
+        position is re-synced whenever a new prefill reset the cache — so a
+        process-wide cached model serves many sequential stories safely."""
         assert self._kv_cache is not None, "call prefill(fill_cache=True) first"
         B = ids.shape[0]
         if not hasattr(self, "_inv_freq"):
@@ -325,31 +328,53 @@ class LlamaModel:
                 cfg.rope_theta
                 ** (torch.arange(0, cfg.head_dim, 2, dtype=torch.float32, device=self.device) / cfg.head_dim)
             )
-        if self._decode_graph is None:
-            self._g_ids = torch.zeros(B, dtype=torch.long, device=self.device)
-            self._g_pos = torch.zeros(1, dtype=torch.int32, device=self.device)
-            self._g_ids.copy_(ids)
-            self._g_pos.fill_(self._cache_len)
+        if not hasattr(self, "_graphs"):
+            self._graphs = {}
+        entry = self._graphs.get(B)
+        if entry is None or entry["cache_id"] != id(self._kv_cache):
+            # capture for this batch (or re-capture: prefill reallocated the
+            # KV cache, so the captured pointers are stale)
+            g_ids = torch.zeros(B, dtype=torch.long, device=self.device)
+            g_pos = torch.zeros(1, dtype=torch.int32, device=self.device)
+            g_ids.copy_(ids)
+            g_pos.fill_(self._cache_len)
             # warmup outside capture (allocator settles)
-            self._decode_body(self._g_ids, self._g_pos)
-            self._g_pos.fill_(self._cache_len)
+            self._decode_body(g_ids, g_pos)
+            g_pos.fill_(self._cache_len)
             torch.cuda.synchronize(self.device)
             g = torch.cuda.CUDAGraph()
             stream = torch.cuda.Stream(device=self.device)
             with torch.cuda.stream(stream):
                 with torch.cuda.graph(g, stream=stream):
-                    self._g_logits = self._decode_body(self._g_ids, self._g_pos)
-                    self._g_pos.add_(1)  # the graph advances its own position
-            self._decode_graph = g
-            self._decode_graph_batch = B
-        assert self._decode_graph_batch == B, "decode batch changed; new model needed"
-        self._g_ids.copy_(ids)
-        self._decode_graph.replay()
+                    g_logits = self._decode_body(g_ids, g_pos)
+                    g_pos.add_(1)  # the graph advances its own position
+            entry = {
+                "graph": g, "ids": g_ids, "pos": g_pos, "logits": g_logits,
+                "pos_val": self._cache_len, "cache_id": id(self._kv_cache),
+            }
+            self._graphs[B] = entry
+        if entry["pos_val"] != self._cache_len:
+            # a new prefill moved the cache length since the last replay
+            entry["pos"].fill_(self._cache_len)
+            entry["pos_val"] = self._cache_len
+        entry["ids"].copy_(ids)
+        entry["graph"].replay()
+        entry["pos_val"] += 1
         self._cache_len += 1
-        return self._g_logits
+        return entry["logits"]
 
     def _alloc_cache(self, B: int, smax: int) -> None:
         cfg = self.cfg
+        if (
+            self._kv_cache is not None
+            and self._kv_cache[0][0].shape[0] == B
+            and self._kv_cache[0][0].shape[2] >= smax
+        ):
+            # reuse the allocation: decode never reads beyond _cache_len, so
+            # no zeroing needed — and captured decode graphs keep pointing
+            # at live tensors across stories
+            self._cache_len = 0
+            return
         self._kv_cache = [
             (
                 torch.zeros(

@@ -157,3 +157,38 @@ spec:
         })
         assert run.output["result"]["checksum"] == 1000.0
         assert run.output["result"]["numel"] == int(4.0 * (1 << 30) // 2)
+
+
+@pytest.mark.gpu
+class TestServedModelReuse:
+    """A process-cached model must serve SEQUENTIAL stories correctly:
+    re-prefill re-syncs the graphed decode position, the KV allocation is
+    reused (captured graph pointers stay live), and batch changes get their
+    own graphs."""
+
+    def test_sequential_generations_match_fresh_model(self):
+        import torch
+
+        from bobrapet_amd.models.llama import LlamaModel
+
+        served = LlamaModel("llama-tiny", device="cuda")
+        torch.manual_seed(5)
+        p1 = torch.randint(0, served.cfg.vocab_size, (2, 24), device="cuda")
+        p2 = torch.randint(0, served.cfg.vocab_size, (2, 16), device="cuda")
+        served.generate(p1, 4)          # story 1 (captures graphs)
+        got = served.generate(p2, 4)    # story 2 reuses cache + graphs
+        fresh = LlamaModel("llama-tiny", device="cuda")
+        want = fresh.generate(p2, 4, use_graph=False)
+        assert torch.equal(got, want), (got.tolist(), want.tolist())
+
+    def test_batch_change_recaptures(self):
+        import torch
+
+        from bobrapet_amd.models.llama import LlamaModel
+
+        m = LlamaModel("llama-tiny", device="cuda")
+        a = torch.randint(0, m.cfg.vocab_size, (1, 8), device="cuda")
+        b = torch.randint(0, m.cfg.vocab_size, (3, 8), device="cuda")
+        m.generate(a, 2)
+        out = m.generate(b, 2)  # different batch: second graph, fresh cache
+        assert out.shape == (3, 2)

@@ -667,19 +667,16 @@ void NativeEngine::handle_timer(const Event& ev) {
 void NativeEngine::maybe_finalize(Run& run, const Plan& plan) {
   int b = 0, e = 0;
   if (active_range(run, plan, &b, &e)) return;  // a phase still has work
-  // unreached compensation steps (no failure) stay Pending — settle them
-  double t = now();
-  for (auto& st : run.states)
-    if (!is_terminal(st.phase)) {
-      st.phase = Phase::Skipped;
-      st.finished = t;
-    }
+  // unreached compensation steps (no failure) stay Pending with no start
+  // time — the Python engine / reference leave them without state at all,
+  // and run_status omits them for the same surface
   bool failure = has_failure(run, plan);
   bool comp_ok = plan.n_comp > 0;
   for (int i = plan.main_end(); i < plan.comp_end(); ++i) {
     Phase p = run.states[i].phase;
     if (p != Phase::Succeeded && p != Phase::Skipped) comp_ok = false;
   }
+  (void)0;
   bool fin_failed = false;
   for (int i = plan.comp_end(); i < (int)plan.steps.size(); ++i) {
     Phase p = run.states[i].phase;

@@ -270,6 +270,11 @@ class NativeEngine {
     JObject steps;
     for (size_t i = 0; i < run.states.size(); ++i) {
       const StepState& st = run.states[i];
+      // never-launched compensation/finally steps have no state surface
+      // (reference parity: unreconciled steps don't appear in stepStates)
+      if ((int)i >= plan.main_end() && st.phase == Phase::Pending &&
+          is_terminal(run.phase))
+        continue;
       JObject s;
       s["phase"] = phase_name(st.phase);
       s["output"] = st.output;

@@ -41,11 +41,21 @@ def _random_story(rng: random.Random, idx: int) -> str:
         needs = f" needs: [{', '.join(deps)}]," if deps else ""
         kind = rng.random()
         extra = " allowFailure: true," if rng.random() < 0.25 else ""
-        if kind < 0.15:
+        if kind < 0.1:
             step = (
                 f"    - {{name: {name},{needs}{extra} ref: {{name: f}}, "
                 f"with: {{succeedAfter: 99}}}}"
             )
+        elif kind < 0.18:
+            # fails once then succeeds — exercises retry parity (the fail
+            # engram counts attempts per (storyRun, step))
+            step = (
+                f"    - {{name: {name},{needs}{extra} ref: {{name: f}}, "
+                f"retry: {{maxRetries: 2, delay: 1ms}}, "
+                f"with: {{succeedAfter: 1}}}}"
+            )
+        elif kind < 0.22 and i == n - 1:
+            step = f"    - {{name: {name},{needs} type: stop, with: {{phase: Succeeded}}}}"
         elif kind < 0.3 and deps:
             cond = f"steps.{deps[0]}.phase == 'Succeeded'"
             step = (
@@ -80,7 +90,7 @@ def test_python_and_native_engines_agree():
         eng.apply_yaml(RESOURCES)
         nr = NativeRunner.from_run_engine(eng)
         checked = 0
-        for idx in range(30):
+        for idx in range(60):
             yaml_text = _random_story(rng, idx)
             try:
                 eng.apply_yaml(yaml_text)
@@ -96,8 +106,19 @@ def test_python_and_native_engines_agree():
             assert str(run.phase) == res["phase"], (
                 idx, yaml_text, str(run.phase), res["phase"], py_phases, nat_phases,
             )
-            assert py_phases == nat_phases, (idx, yaml_text, py_phases, nat_phases)
+            run_failed = res["phase"] in ("Failed", "Compensated", "Canceled")
+            for k in set(py_phases) | set(nat_phases):
+                a, b = py_phases.get(k), nat_phases.get(k)
+                if a == b:
+                    continue
+                # fail-fast race window: a step whose deps were satisfied may
+                # launch (and succeed) before an unrelated failure lands, or
+                # be skipped after — both legal, in the reference too
+                # (findReadySteps skips PENDING steps only)
+                if run_failed and {a, b} <= {"Succeeded", "Skipped", None}:
+                    continue
+                raise AssertionError((idx, yaml_text, k, a, b, py_phases, nat_phases))
             checked += 1
-        assert checked >= 20, f"only {checked} stories compared"
+        assert checked >= 40, f"only {checked} stories compared"
     finally:
         eng.stop()

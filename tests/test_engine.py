@@ -976,3 +976,53 @@ spec:
                 assert run.is_terminal
         finally:
             eng.stop()
+
+
+class TestAdmissionStress:
+    def test_concurrent_submitters_all_complete(self):
+        """200 runs from 8 threads through queue admission (concurrency 16,
+        priority aging) — every run completes, none lost or duplicated."""
+        import concurrent.futures as cf
+
+        from bobrapet_amd.engine import EngineConfig, QueueConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        eng = RunEngine(
+            EngineConfig(
+                cpu_workers=8,
+                global_concurrency=16,
+                queues={"default": QueueConfig(concurrency=16)},
+                child_ttl_seconds=60.0,
+            )
+        ).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: stress}
+spec:
+  steps:
+    - {name: a, ref: {name: e}, with: {v: "{{ inputs.i }}"}}
+    - {name: b, type: condition, needs: [a], with: {expression: "{{ steps.a.output.v >= 0 }}"}}
+"""
+            )
+
+            def one(i: int) -> str:
+                run = eng.run_story("default/stress", {"i": i}, timeout=60)
+                assert run.phase == Phase.SUCCEEDED, (i, run.error)
+                assert run.step_states["a"].output["v"] == i
+                return run.name
+
+            with cf.ThreadPoolExecutor(max_workers=8) as ex:
+                names = list(ex.map(one, range(200)))
+            assert len(set(names)) == 200
+        finally:
+            eng.stop()

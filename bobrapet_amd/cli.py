@@ -152,13 +152,23 @@ def serve(
     host: str = typer.Option("127.0.0.1"),
     port: int = typer.Option(8080),
     start_impulses: bool = typer.Option(True, help="start all applied Impulses"),
+    checkpoint: _t.Optional[str] = typer.Option(
+        None, help="state snapshot path: restored on boot, saved periodically"
+    ),
+    checkpoint_interval: float = typer.Option(30.0, help="snapshot period (s)"),
 ):
-    """Start the engine + HTTP ingress (impulse triggers, /metrics)."""
+    """Start the engine + REST control plane / impulse ingress."""
+    import os
+
     from .engine import EngineConfig, RunEngine
     from .engine.impulses import serve_http
     from .specs import types as T
 
-    eng = RunEngine(EngineConfig()).start()
+    cfg = EngineConfig()
+    if checkpoint:
+        cfg.checkpoint_path = checkpoint
+        cfg.checkpoint_interval_seconds = checkpoint_interval
+    eng = RunEngine(cfg).start()
     objs = _load_files(file) if file else []
     for obj in objs:
         eng.apply(obj)
@@ -167,6 +177,9 @@ def serve(
             if isinstance(obj, T.Impulse):
                 eng.impulses.start(obj.key)
                 typer.echo(f"impulse {obj.key} started")
+    if checkpoint and os.path.exists(checkpoint):
+        n = eng.load_state(checkpoint)
+        typer.echo(f"restored {n} runs from {checkpoint}")
     typer.echo(f"serving on http://{host}:{port}")
     serve_http(eng, host=host, port=port)
 

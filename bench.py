@@ -222,24 +222,19 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
         time.sleep(0.001)
     group.barrier()
     t0 = time.monotonic()
-    # concurrent producers (impulse events arrive concurrently); partition
-    # lanes keep per-partition ordering regardless of producer interleaving
-    n_push = 4
-    import concurrent.futures as _cf
-
-    def _pusher(base: int) -> None:
-        for i in range(base, args.steps, n_push):
-            stream.push({"tensor": ids, "seq": i, "items": [{}]})
-
-    with _cf.ThreadPoolExecutor(max_workers=n_push) as ex:
-        list(ex.map(_pusher, range(n_push)))
+    for i in range(args.steps):
+        stream.push({"tensor": ids, "seq": i, "items": [{}]})
     while stream.leaf_packets < args.warmup + args.steps and time.monotonic() < t0 + 120:
         time.sleep(0.001)
     group.barrier()
     elapsed = time.monotonic() - t0
     run = stream.finish(timeout=30)
     if run.phase.value not in ("Finished",):
-        raise RuntimeError(f"stream bench failed: {run.phase}")
+        details = {
+            k: (str(v.phase), str(v.error.message if v.error else ""))
+            for k, v in run.step_states.items()
+        }
+        raise RuntimeError(f"stream bench failed: {run.phase} {details}")
     elapsed_max = group.max_over_ranks(elapsed, device="cpu" if not torch.cuda.is_available() else None)
     replays = run.step_states["featurize"].output.get("graphReplays", 0)
     if rank == 0:

@@ -61,7 +61,43 @@ class StageStats:
     busy_seconds: float = 0.0
 
 
-_CAPTURE_LOCK = threading.Lock()
+class _GraphGate:
+    """Capture/replay exclusion for partition lanes on one device: hipGraph
+    CAPTURE must run quiesced (no sibling lane mid-replay — stream capture
+    racing concurrent GPU work wedges intermittently on ROCm), while
+    replays proceed concurrently with each other."""
+
+    def __init__(self):
+        self._cv = threading.Condition()
+        self._replays = 0
+        self._capturing = False
+
+    def begin_replay(self):
+        with self._cv:
+            while self._capturing:
+                self._cv.wait()
+            self._replays += 1
+
+    def end_replay(self):
+        with self._cv:
+            self._replays -= 1
+            self._cv.notify_all()
+
+    def begin_capture(self):
+        with self._cv:
+            while self._capturing:
+                self._cv.wait()
+            self._capturing = True
+            while self._replays > 0:
+                self._cv.wait()
+
+    def end_capture(self):
+        with self._cv:
+            self._capturing = False
+            self._cv.notify_all()
+
+
+_GRAPH_GATE = _GraphGate()
 
 
 class _Stage(threading.Thread):
@@ -330,12 +366,15 @@ class _Stage(threading.Thread):
         if not torch.is_tensor(tensor):
             return None
         tensor = tensor.to(f"cuda:{self.device}", non_blocking=False)
+        if self._graph is False:  # capture failed earlier: permanent eager
+            return None
         if self._graph is None:
             # one capture at a time: concurrent partition lanes each own a
             # graph, but global-mode stream capture poisons OTHER threads'
             # in-flight GPU work (thread_local relaxes the check; the lock
             # keeps captures from overlapping at all)
-            with _CAPTURE_LOCK:
+            _GRAPH_GATE.begin_capture()
+            try:
                 ctx = self._ctx(packet)
                 # warmup (allocations settle), then capture
                 fn(ctx, tensor)
@@ -349,12 +388,27 @@ class _Stage(threading.Thread):
                     ):
                         self._graph_out = fn(ctx, self._graph_in)
                 self._graph = g
-        self._graph_in.copy_(tensor)
-        self._graph.replay()
-        if self.stream is not None:
-            torch.cuda.synchronize(self.device)
-        out = dict(packet) if isinstance(packet, dict) else {}
-        out["tensor"] = self._graph_out.clone()
+            except Exception as exc:
+                # never fail the pipeline over a capture problem: degrade
+                # THIS lane to eager permanently
+                self._graph = False
+                self.sr.engine.metrics.inc("stream_capture_fallbacks_total")
+                import warnings
+
+                warnings.warn(f"hipGraph capture fell back to eager: {exc}")
+                return None
+            finally:
+                _GRAPH_GATE.end_capture()
+        _GRAPH_GATE.begin_replay()
+        try:
+            self._graph_in.copy_(tensor)
+            self._graph.replay()
+            if self.stream is not None:
+                torch.cuda.synchronize(self.device)
+            out = dict(packet) if isinstance(packet, dict) else {}
+            out["tensor"] = self._graph_out.clone()
+        finally:
+            _GRAPH_GATE.end_replay()
         return out
 
 

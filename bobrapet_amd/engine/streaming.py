@@ -172,7 +172,9 @@ class _Stage(threading.Thread):
                 # to that partition's ring (reference: ordering per_partition)
                 if isinstance(out, dict) and isinstance(packet, dict) and "$partition" in packet:
                     out.setdefault("$partition", packet["$partition"])
-                for _key, rings in self.out_edges:
+                for (src, dst), rings in self.out_edges:
+                    if not self.sr.route_allows(dst, out):
+                        continue  # denied by a transport routing rule
                     if len(rings) == 1:
                         rings[0].push(out)
                     else:
@@ -543,6 +545,29 @@ class StreamingRun:
                 part = packet.get("$partition", 0) if isinstance(packet, dict) else 0
                 ok = rings[int(part) % len(rings)].push(packet, timeout=timeout) and ok
         return ok
+
+    def route_allows(self, dst: str, packet) -> bool:
+        """Per-packet transport routing rules (reference:
+        TransportRoutingRule transport_settings_types.go:353-370): first
+        matching rule targeting `dst` decides; no match → allow."""
+        routing = self.settings.routing
+        if routing is None or not routing.rules:
+            return True
+        scope = {"packet": packet, "inputs": self.run.inputs}
+        for rule in routing.rules:
+            if rule.steps and dst not in rule.steps:
+                continue
+            if rule.when:
+                try:
+                    if not self.engine.evaluator.evaluate_condition(str(rule.when), scope):
+                        continue
+                except Exception:
+                    continue  # unevaluable rule never matches
+            if rule.action == "deny":
+                self.engine.metrics.inc("stream_route_denied_total", rule=rule.name or "?")
+                return False
+            return True
+        return True
 
     def _stamp_partition(self, packet):
         """Assign a partition id (reference: TransportPartitioningSettings):

@@ -84,8 +84,10 @@ def from_dict(cls, data: _t.Optional[dict]):
             kwargs[name] = _convert(value, hints.get(name))
         else:
             extra[key] = value
+    if extra and "extra" in fields:
+        kwargs["extra"] = extra  # in-constructor so __post_init__ sees it
     obj = cls(**kwargs)
-    if extra and hasattr(obj, "extra"):
+    if extra and "extra" not in fields and hasattr(obj, "extra"):
         obj.extra = extra
     return obj
 
@@ -660,6 +662,36 @@ class TransportRecording:
 
 
 @dataclass
+class TransportRoutingRule:
+    """Conditional downstream routing (reference:
+    transport_settings_types.go:353-370): `when` evaluated per packet,
+    action allow|deny over the targeted downstream steps."""
+
+    name: str = ""
+    when: _t.Optional[str] = None
+    action: str = "allow"
+    steps: _t.List[str] = field(default_factory=list)  # empty = all downstreams
+    extra: dict = field(default_factory=dict)
+
+    def __post_init__(self):
+        # accept the reference's nested target: {steps: [...]} shape
+        if not self.steps and isinstance(self.extra.get("target"), dict):
+            self.steps = list(self.extra["target"].get("steps") or [])
+
+
+@dataclass
+class TransportRouting:
+    """Routing topology + per-packet rules (reference:
+    transport_settings_types.go:372-390)."""
+
+    mode: str = "auto"  # auto | hub | p2p
+    fan_out: str = "parallel"
+    max_downstreams: _t.Optional[int] = None
+    rules: _t.List[TransportRoutingRule] = field(default_factory=list)
+    extra: dict = field(default_factory=dict)
+
+
+@dataclass
 class TransportPartitioning:
     """Partition assignment for stream envelopes (reference:
     transport_settings_types.go:391-419, modes none|preserve|hash)."""
@@ -709,6 +741,7 @@ class TransportStreamingSettings:
     recording: _t.Optional[TransportRecording] = None
     lifecycle: _t.Optional[TransportLifecycle] = None
     partitioning: _t.Optional[TransportPartitioning] = None
+    routing: _t.Optional[TransportRouting] = None
     extra: dict = field(default_factory=dict)
 
 

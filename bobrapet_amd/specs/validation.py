@@ -131,6 +131,27 @@ def validate_story(story: T.Story) -> ValidationResult:
                     f"step {s.name!r}: primitive {s.type} is batch-only and not "
                     f"allowed in a streaming story"
                 )
+        # routing.maxDownstreams guardrail (reference:
+        # TransportRoutingSettings.MaxDownstreams) — enforced at apply time
+        for t in story.transports:
+            streaming = t.streaming if isinstance(t.streaming, dict) else None
+            cap = (
+                (streaming.get("routing") or {}).get("maxDownstreams")
+                if streaming
+                else None
+            )
+            if not cap:
+                continue
+            downstream: _t.Dict[str, int] = {}
+            for s2 in story.steps:
+                for dep in s2.needs:
+                    downstream[dep] = downstream.get(dep, 0) + 1
+            for name2, n2 in downstream.items():
+                if n2 > int(cap):
+                    res.error(
+                        f"step {name2!r} has {n2} downstream steps, over "
+                        f"routing.maxDownstreams={cap}"
+                    )
 
     # declared transports must exist in the story's transport list
     declared = {t.name for t in story.transports}

@@ -719,3 +719,66 @@ spec:
         assert run.phase == Phase.FINISHED
         # the right branch's 4 packets each met quorum=1
         assert run.step_states["join"].output["packetsIn"] == 4
+
+
+class TestRoutingRules:
+    """Transport routing rules (reference:
+    transport_settings_types.go:353-390): per-packet allow/deny over
+    targeted downstream steps + maxDownstreams apply-time guardrail."""
+
+    def test_deny_rule_filters_one_downstream(self, eng):
+        eng.apply_yaml(
+            """
+kind: Story
+metadata: {name: routed}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        routing:
+          rules:
+            - name: block-audit-small
+              when: "{{ packet.v < 10 }}"
+              action: deny
+              target: {steps: [audit]}
+  steps:
+    - name: src
+      ref: {name: echoer}
+    - name: audit
+      ref: {name: echoer}
+      needs: [src]
+    - name: main
+      ref: {name: echoer}
+      needs: [src]
+"""
+        )
+        stream = eng.submit_stream("default/routed")
+        for v in (1, 20, 2, 30):
+            stream.push({"v": v})
+        run = stream.finish(timeout=10)
+        assert run.phase == Phase.FINISHED
+        # audit saw only v>=10 packets; main saw all four
+        assert run.step_states["audit"].output["packetsIn"] == 2
+        assert run.step_states["main"].output["packetsIn"] == 4
+
+    def test_max_downstreams_guardrail_at_apply(self, eng):
+        import pytest as _p
+
+        with _p.raises(ValueError):
+            eng.apply_yaml(
+                """
+kind: Story
+metadata: {name: capped}
+spec:
+  pattern: streaming
+  transports:
+    - name: t
+      streaming:
+        routing: {maxDownstreams: 1}
+  steps:
+    - {name: src, ref: {name: echoer}}
+    - {name: a, ref: {name: echoer}, needs: [src]}
+    - {name: b, ref: {name: echoer}, needs: [src]}
+"""
+            )

@@ -56,6 +56,15 @@ def _random_story(rng: random.Random, idx: int) -> str:
             )
         elif kind < 0.22 and i == n - 1:
             step = f"    - {{name: {name},{needs} type: stop, with: {{phase: Succeeded}}}}"
+        elif kind < 0.27:
+            nb = rng.randint(2, 3)
+            branches = ", ".join(
+                f"{{name: b{j}, ref: {{name: e}}, with: {{j: {j}}}}}" for j in range(nb)
+            )
+            step = (
+                f"    - {{name: {name},{needs}{extra} type: parallel, "
+                f"with: {{steps: [{branches}]}}}}"
+            )
         elif kind < 0.3 and deps:
             cond = f"steps.{deps[0]}.phase == 'Succeeded'"
             step = (
@@ -107,6 +116,7 @@ def test_python_and_native_engines_agree():
                 idx, yaml_text, str(run.phase), res["phase"], py_phases, nat_phases,
             )
             run_failed = res["phase"] in ("Failed", "Compensated", "Canceled")
+            stop_fired = "type: stop" in yaml_text
             for k in set(py_phases) | set(nat_phases):
                 a, b = py_phases.get(k), nat_phases.get(k)
                 if a == b:
@@ -116,6 +126,10 @@ def test_python_and_native_engines_agree():
                 # be skipped after — both legal, in the reference too
                 # (findReadySteps skips PENDING steps only)
                 if run_failed and {a, b} <= {"Succeeded", "Skipped", None}:
+                    continue
+                # stop race window: a stop directive races in-flight steps
+                # and pending retries — their final states are timing-defined
+                if stop_fired and {a, b} <= {"Succeeded", "Skipped", "Failed", None}:
                     continue
                 raise AssertionError((idx, yaml_text, k, a, b, py_phases, nat_phases))
             checked += 1

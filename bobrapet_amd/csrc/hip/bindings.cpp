@@ -23,6 +23,9 @@ void launch_silu_mul_strided(void*, const void*, const void*, long, int, long,
                              hipStream_t);
 void launch_rope_qkv(void*, void*, const void*, const void*, const void*, int,
                      int, int, int, long, hipStream_t);
+void launch_rope_qkv_decode(void*, void*, void*, const void*, const void*,
+                            const void*, const void*, int, int, int, int, int,
+                            hipStream_t);
 void launch_attn_decode(void*, void*, const void*, const void*, const void*,
                         int, int, int, int, int, const void*, float,
                         hipStream_t);
@@ -232,6 +235,30 @@ torch::Tensor attn_decode_t(torch::Tensor q, torch::Tensor kc,
   return out;
 }
 
+torch::Tensor rope_qkv_decode(torch::Tensor qkv, torch::Tensor kc,
+                              torch::Tensor vc, torch::Tensor cos_t,
+                              torch::Tensor sin_t, torch::Tensor L_dev,
+                              long Hq, long Hkv, long D) {
+  // fused decode head prep: rope q (returned contiguous [B,Hq,D]) and
+  // rope k + copy v straight into the KV cache at device position L
+  check_bf16(qkv, "qkv");
+  check_bf16(kc, "kc");
+  check_bf16(vc, "vc");
+  const int B = qkv.size(0);
+  const int Smax = kc.size(2);
+  TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hkv) * D, "rope_qkv_decode: row size");
+  TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 && cos_t.is_contiguous(),
+              "cos must be f32 contiguous");
+  TORCH_CHECK(L_dev.scalar_type() == torch::kInt32 && L_dev.is_cuda(),
+              "L must be an int32 device scalar");
+  auto q_out = torch::empty({(long)B, Hq, D}, qkv.options());
+  launch_rope_qkv_decode(q_out.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                         qkv.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
+                         L_dev.data_ptr(), B, (int)Hq, (int)Hkv, (int)D, Smax,
+                         cur_stream());
+  return q_out;
+}
+
 torch::Tensor silu_mul_strided(torch::Tensor gate_up) {
   // gate_up: [..., 2*I] bf16 (contiguous); returns silu(g)*u of shape [..., I]
   check_bf16(gate_up, "gate_up");
@@ -332,6 +359,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "prefill attention + per-row (m,l) softmax stats");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
+  m.def("rope_qkv_decode", &rope_qkv_decode,
+        "fused decode rope + KV-cache append (graph-replayable)");
   m.def("attn_decode_t", &attn_decode_t,
         "decode attention, length from a device scalar (hipGraph-capturable)");
   m.def("dbg_mfma", &dbg_mfma, "layout probe: C=A@B one mfma");

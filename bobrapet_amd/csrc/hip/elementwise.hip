@@ -241,6 +241,67 @@ extern "C" void launch_rope_qkv(void* q_out, void* k_out, const void* qkv,
                      (const float*)sin_t, T, Hq, Hk, D, row_stride);
 }
 
+// Fused DECODE step head prep: one kernel replaces rope_inplace + two
+// cache index_copys + the k/v contiguous copies on the b1 decode path
+// (each was a ~5 us launch, x32 layers x2 copies per step).  Reads the
+// fused qkv projection row, applies rope to q/k at position L (a DEVICE
+// scalar — hipGraph-replayable), writes rotated q to a contiguous
+// [B,Hq,D] buffer and rotated k + raw v DIRECTLY into the KV cache at
+// sequence slot L.
+__global__ __launch_bounds__(256) void rope_qkv_decode_kernel(
+    unsigned short* __restrict__ q_out,      // [B,Hq,D]
+    unsigned short* __restrict__ kc,         // [B,Hkv,Smax,D]
+    unsigned short* __restrict__ vc,         // [B,Hkv,Smax,D]
+    const unsigned short* __restrict__ qkv,  // [B, (Hq+2*Hkv)*D]
+    const float* __restrict__ cos_t,         // [B, D/2] (position L row)
+    const float* __restrict__ sin_t,
+    const int* __restrict__ L_dev, int B, int Hq, int Hkv, int D, int Smax) {
+  const int L = *L_dev;
+  const int half = D / 2;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wave = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int nwaves = (gridDim.x * blockDim.x) / WAVE;
+  const long row = (long)(Hq + 2 * Hkv) * D;
+  const int total = B * (Hq + 2 * Hkv);
+  for (int idx = wave; idx < total; idx += nwaves) {
+    const int b = idx / (Hq + 2 * Hkv);
+    const int h = idx % (Hq + 2 * Hkv);
+    const unsigned short* src = qkv + b * row + (long)h * D;
+    if (h < Hq + Hkv) {  // rope'd q or k head
+      unsigned short* dst =
+          (h < Hq) ? q_out + ((long)b * Hq + h) * D
+                   : kc + (((long)b * Hkv + (h - Hq)) * Smax + L) * D;
+      const float* crow = cos_t + (long)b * half;
+      const float* srow = sin_t + (long)b * half;
+      for (int d = lane; d < half; d += WAVE) {
+        float a = bf2f(src[d]);
+        float bb = bf2f(src[d + half]);
+        float c = crow[d], sn = srow[d];
+        dst[d] = f2bf(a * c - bb * sn);
+        dst[d + half] = f2bf(bb * c + a * sn);
+      }
+    } else {  // v head: straight copy into the cache slot
+      int hv = h - Hq - Hkv;
+      unsigned short* dst = vc + (((long)b * Hkv + hv) * Smax + L) * D;
+      for (int d = lane; d < D; d += WAVE) dst[d] = src[d];
+    }
+  }
+}
+
+extern "C" void launch_rope_qkv_decode(void* q_out, void* kc, void* vc,
+                                       const void* qkv, const void* cos_t,
+                                       const void* sin_t, const void* L_dev,
+                                       int B, int Hq, int Hkv, int D, int Smax,
+                                       hipStream_t stream) {
+  int waves_needed = B * (Hq + 2 * Hkv);
+  int blocks = (waves_needed + 3) / 4;  // 4 waves per 256-thread block
+  hipLaunchKernelGGL(rope_qkv_decode_kernel, dim3(blocks), dim3(256), 0,
+                     stream, (unsigned short*)q_out, (unsigned short*)kc,
+                     (unsigned short*)vc, (const unsigned short*)qkv,
+                     (const float*)cos_t, (const float*)sin_t,
+                     (const int*)L_dev, B, Hq, Hkv, D, Smax);
+}
+
 extern "C" void launch_rope(void* q, void* k, const void* cos_t,
                             const void* sin_t, int T, int Hq, int Hk, int D,
                             hipStream_t stream) {

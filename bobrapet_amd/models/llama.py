@@ -276,14 +276,22 @@ class LlamaModel:
         for li, lw in enumerate(self.layers):
             x2 = x.view(B, cfg.hidden_size)
             qkv = self._decode_mm(x2, lw.w_qkv).view(B, 1, -1) if use_gemv else torch.matmul(x, lw.w_qkv.t())
-            q, k, v = self._split_qkv(qkv, B, 1)
-            qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
-            kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
-            qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
             kc, vc = self._kv_cache[li]
-            # device-indexed cache append (graph-replayable)
-            kc.index_copy_(2, pos_l, kf.view(B, cfg.num_kv_heads, 1, cfg.head_dim))
-            vc.index_copy_(2, pos_l, v.reshape(B, cfg.num_kv_heads, 1, cfg.head_dim).contiguous())
+            if use_gemv:
+                # fused rope + cache append: one kernel instead of rope +
+                # two index_copys + two layout copies (x32 layers/step)
+                qf = ops.rope_qkv_decode(
+                    qkv.view(B, -1), kc, vc, cos_t, sin_t, pos_i32,
+                    cfg.num_heads, cfg.num_kv_heads, cfg.head_dim,
+                )
+            else:
+                q, k, v = self._split_qkv(qkv, B, 1)
+                qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
+                kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
+                qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
+                # device-indexed cache append (graph-replayable)
+                kc.index_copy_(2, pos_l, kf.view(B, cfg.num_kv_heads, 1, cfg.head_dim))
+                vc.index_copy_(2, pos_l, v.reshape(B, cfg.num_kv_heads, 1, cfg.head_dim).contiguous())
             attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
             a2 = attn.reshape(B, cfg.num_heads * cfg.head_dim)
             attn_out = (

@@ -279,6 +279,27 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return (a.float() @ b.float().t()).to(a.dtype)
 
 
+def rope_qkv_decode(qkv2d, kc, vc, cos_t, sin_t, pos_dev: torch.Tensor, Hq: int, Hkv: int, D: int):
+    """Fused decode head prep: rope q (returned [B,Hq,D]) and rope k +
+    copy v straight into the KV cache at DEVICE position pos (hipGraph-
+    replayable — one kernel instead of rope + 2 index_copys + 2 copies)."""
+    if qkv2d.is_cuda:
+        return _require_ext().rope_qkv_decode(
+            qkv2d.contiguous(), kc, vc, cos_t, sin_t, pos_dev, Hq, Hkv, D
+        )
+    # CPU reference: compose from the existing reference ops
+    B = qkv2d.shape[0]
+    pos = int(pos_dev.item())
+    q = qkv2d[:, : Hq * D].reshape(B, Hq, D).clone()
+    k = qkv2d[:, Hq * D : (Hq + Hkv) * D].reshape(B, Hkv, D).clone()
+    v = qkv2d[:, (Hq + Hkv) * D :].reshape(B, Hkv, D)
+    q = rope_ref(q, cos_t, sin_t)
+    k = rope_ref(k, cos_t, sin_t)
+    kc[:, :, pos] = k
+    vc[:, :, pos] = v
+    return q
+
+
 def attn_decode_t(q, kc, vc, L_dev: torch.Tensor, scale: _t.Optional[float] = None):
     """Graph-capturable decode attention: L read from a device int32 scalar."""
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])

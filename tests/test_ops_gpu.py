@@ -228,3 +228,36 @@ class TestAttnPrefillStats:
         ref = ops.attn_prefill_stats_ref(q, k, v, D ** -0.5, False)[0]
         err = (merged - ref.float()).abs().max().item()
         assert err < 0.03, f"merge err {err}"
+
+
+@pytest.mark.gpu
+class TestRopeQkvDecode:
+    def test_matches_composed_reference(self):
+        import torch
+
+        from bobrapet_amd import ops
+
+        torch.manual_seed(6)
+        B, Hq, Hkv, D, Smax, pos = 1, 8, 2, 128, 256, 37
+        qkv = torch.randn(B, (Hq + 2 * Hkv) * D, device="cuda", dtype=torch.bfloat16)
+        kc = torch.zeros(B, Hkv, Smax, D, device="cuda", dtype=torch.bfloat16)
+        vc = torch.zeros_like(kc)
+        half = D // 2
+        ang = torch.rand(B, half, device="cuda") * 6.28
+        cos_t, sin_t = torch.cos(ang), torch.sin(ang)
+        pos_dev = torch.tensor([pos], dtype=torch.int32, device="cuda")
+        q = ops.rope_qkv_decode(qkv, kc, vc, cos_t, sin_t, pos_dev, Hq, Hkv, D)
+        # composed CPU reference
+        kc_ref = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16)
+        vc_ref = torch.zeros_like(kc_ref)
+        q_ref = ops.rope_qkv_decode(
+            qkv.cpu(), kc_ref, vc_ref, cos_t.cpu(), sin_t.cpu(),
+            pos_dev.cpu(), Hq, Hkv, D,
+        )
+        assert (q.float().cpu() - q_ref.float()).abs().max().item() < 0.02
+        assert (kc.float().cpu() - kc_ref.float()).abs().max().item() < 0.02
+        assert torch.equal(vc.cpu(), vc_ref)
+        # nothing written outside slot `pos`
+        mask = torch.ones(Smax, dtype=torch.bool)
+        mask[pos] = False
+        assert kc[:, :, mask].abs().max().item() == 0.0

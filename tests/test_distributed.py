@@ -377,3 +377,99 @@ def test_ulysses_attention_matches_full():
         assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
         for key, err in payload.items():
             assert err < 0.03, f"rank {rank} {key}: err {err}"
+
+
+def _sp_engram_rank(rank: int, world: int, port: int, q):
+    import os
+    import sys
+
+    os.environ.update(
+        {
+            "RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "LOCAL_RANK": str(rank),
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port),
+        }
+    )
+    try:
+        sys.path.insert(0, REPO)
+        import torch
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+        from bobrapet_amd.models.llama import LlamaModel
+        from bobrapet_amd.parallel import group
+
+        group.init_distributed(backend="gloo")
+        S_loc = 12
+        gen = torch.Generator().manual_seed(3)
+        full_ids = torch.randint(0, 1024, (1, S_loc * world), generator=gen)
+        lo, hi = rank * S_loc, (rank + 1) * S_loc
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        eng.apply_yaml(
+            """
+kind: EngramTemplate
+metadata: {name: llm-tpl}
+spec: {builtin: llm-infer}
+---
+kind: Engram
+metadata: {name: llm}
+spec:
+  templateRef: {name: llm-tpl}
+  with: {model: llama-tiny}
+---
+kind: Story
+metadata: {name: sp}
+spec:
+  steps:
+    - name: infer
+      ref: {name: llm}
+      with:
+        promptIds: "{{ inputs.ids }}"
+        sequenceParallel: true
+  output: {tokens: "{{ steps.infer.output.tokens }}"}
+"""
+        )
+        run = eng.run_story(
+            "default/sp", {"ids": full_ids[:, lo:hi].tolist()}, timeout=120
+        )
+        assert run.phase == Phase.SUCCEEDED, run.error
+        # single-rank reference: last-position argmax of the FULL sequence,
+        # which only the LAST rank's shard output can reproduce
+        ref_model = LlamaModel("llama-tiny", device="cpu")
+        want = ref_model.prefill(full_ids).argmax(dim=-1, keepdim=True)
+        got = torch.tensor(run.output["tokens"])
+        res = {"match": bool((rank != world - 1) or torch.equal(got, want))}
+        q.put((rank, res))
+        eng.stop()
+        group.teardown()
+    except Exception as exc:
+        import traceback
+
+        q.put((rank, {"error": f"{exc}\n{traceback.format_exc()}"}))
+
+
+@pytest.mark.timeout(240)
+def test_llm_infer_engram_sequence_parallel():
+    """The llm-infer engram's sequenceParallel knob: each rank submits its
+    shard through a full engine Story; the last rank's argmax equals a
+    single-rank full-sequence prefill."""
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = _free_port()
+    procs = [
+        ctx.Process(target=_sp_engram_rank, args=(r, world, port, q)) for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, payload = q.get(timeout=220)
+        results[rank] = payload
+    for p in procs:
+        p.join(timeout=30)
+    for rank, payload in results.items():
+        assert "error" not in payload, f"rank {rank}: {payload.get('error')}"
+        assert payload["match"], f"rank {rank} output mismatch"

@@ -292,6 +292,24 @@ def build_http_app(engine: "RunEngine"):
             run = engine.wait(run, timeout=float(body.get("timeout", 300.0)))
         return _run_view(run)
 
+    @app.get("/runs")
+    async def list_runs(phase: _t.Optional[str] = None, limit: int = 100):
+        runs = []
+        for run in engine.store.all_runs():
+            if phase and str(run.phase) != phase:
+                continue
+            runs.append(
+                {
+                    "name": run.name,
+                    "namespace": run.namespace,
+                    "story": f"{run.story_namespace}/{run.story_name}",
+                    "phase": str(run.phase),
+                }
+            )
+            if len(runs) >= max(1, min(limit, 1000)):
+                break
+        return {"runs": runs}
+
     @app.get("/runs/{ns}/{name}")
     async def run_status(ns: str, name: str):
         return _run_view(_get_run(ns, name))

@@ -622,6 +622,21 @@ spec:
         assert res.status_code == 422
         assert "ghost" in res.text or "unknown" in res.text
 
+    def test_list_runs_with_phase_filter(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        client = TestClient(build_http_app(eng))
+        client.post(
+            "/resources",
+            json={"yaml": "kind: Story\nmetadata: {name: lr}\nspec:\n  steps:\n    - {name: a, type: sleep, with: {duration: 0ms}}\n"},
+        )
+        r = client.post("/stories/default/lr/runs", json={"wait": True})
+        assert r.status_code == 200
+        listing = client.get("/runs", params={"phase": "Succeeded"}).json()["runs"]
+        assert any(e["story"] == "default/lr" for e in listing)
+
     def test_missing_run_404(self, eng):
         from fastapi.testclient import TestClient
 

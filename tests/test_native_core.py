@@ -586,3 +586,36 @@ spec:
             assert "v too small" in res["steps"]["a"]["error"]
         finally:
             eng.stop()
+
+
+class TestNativeProcessEngram:
+    def test_command_engram_through_native_core(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: sh}
+spec:
+  command: [python3, -c, "import os, json; print(json.dumps({'via': 'native', 'step': os.environ['BUBU_STEP_NAME']}))"]
+---
+kind: Engram
+metadata: {name: sheller}
+spec: {templateRef: {name: sh}}
+---
+kind: Story
+metadata: {name: extn}
+spec:
+  steps:
+    - {name: go, ref: {name: sheller}, with: {v: 1}}
+"""
+            )
+            nr = NativeRunner.from_run_engine(eng)
+            res = nr.run_story("default/extn", {}, timeout=40)
+            assert res["phase"] == "Succeeded", res
+            assert res["steps"]["go"]["output"] == {"via": "native", "step": "go"}
+        finally:
+            eng.stop()

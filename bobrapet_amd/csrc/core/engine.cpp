@@ -437,7 +437,8 @@ void NativeEngine::launch_step(Run& run, const Plan& plan, int idx) {
         ev.timer_tag = 3;
         arm(now() + step.timeout, std::move(ev));
       }
-      if (launcher_) launcher_(run.id, idx, -1, st.attempt, step.engram, with);
+      if (launcher_)
+        launcher_(run.id, idx, -1, st.attempt, step.engram, step.name, with);
       else {
         st.phase = Phase::Failed;
         st.error = "no engram launcher registered";
@@ -492,7 +493,8 @@ void NativeEngine::launch_branch(Run& run, const Plan& plan, int idx, int b) {
         }
       }
       if (launcher_)
-        launcher_(run.id, idx, b, st.attempt, branch.engram, with);
+        launcher_(run.id, idx, b, st.attempt, branch.engram,
+                  plan.steps[idx].name + "/" + branch.name, with);
       else {
         st.phase = Phase::Failed;
         st.error = "no engram launcher registered";

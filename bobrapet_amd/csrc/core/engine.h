@@ -168,9 +168,11 @@ struct Run {
 };
 
 // launcher(run_id, step_index, branch_index(-1), attempt, engram_name,
-//          resolved_input) — called FROM THE LOOP THREAD; must not block.
-using EngramLauncher = std::function<void(uint64_t, int, int, uint32_t,
-                                          const std::string&, const JValue&)>;
+//          step_name, resolved_input) — called FROM THE LOOP THREAD; must
+//          not block.
+using EngramLauncher =
+    std::function<void(uint64_t, int, int, uint32_t, const std::string&,
+                       const std::string&, const JValue&)>;
 
 class NativeEngine {
  public:

@@ -263,9 +263,11 @@ PYBIND11_MODULE(_core, m) {
            [](NativeEngine& e, py::function fn) {
              e.set_launcher([fn](uint64_t run, int step, int branch,
                                  uint32_t attempt, const std::string& engram,
+                                 const std::string& step_name,
                                  const JValue& input) {
                py::gil_scoped_acquire g;
-               fn(run, step, branch, attempt, engram, to_python(input));
+               fn(run, step, branch, attempt, engram, step_name,
+                  to_python(input));
              });
            })
       .def("start", &NativeEngine::start,

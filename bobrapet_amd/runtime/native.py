@@ -365,7 +365,7 @@ class NativeRunner:
         self._engram_cache[engram_key] = hit
         return hit
 
-    def _launch(self, run_id, step, branch, attempt, engram_key, resolved_input):
+    def _launch(self, run_id, step, branch, attempt, engram_key, step_name, resolved_input):
         """Engram launcher (called from the core's loop thread, GIL held):
         enqueue onto the worker pool and return immediately."""
         hit = self._resolve_engram(engram_key)
@@ -390,7 +390,7 @@ class NativeRunner:
             try:
                 ctx = EngramContext(
                     story_run=str(run_id),
-                    step_name=f"{step}" if branch < 0 else f"{step}/{branch}",
+                    step_name=step_name,
                     input=self.storage.hydrate(resolved_input),
                     config=config,
                     device=slot.device,

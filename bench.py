@@ -239,7 +239,7 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
     replays = run.step_states["featurize"].output.get("graphReplays", 0)
     if rank == 0:
         line = {
-            "metric": "StoryRuns/sec (8-way parallel Story)",
+            "metric": "StoryRuns/sec + p50 step latency, 8-way parallel Story",
             "value": round(args.steps * world / elapsed_max, 3),
             "unit": "packets/s",
             "n_gpus": n_gpus,
@@ -404,7 +404,7 @@ def main() -> int:
 
         if rank == 0:
             line = {
-                "metric": "StoryRuns/sec (8-way parallel Story)",
+                "metric": "StoryRuns/sec + p50 step latency, 8-way parallel Story",
                 "value": round(runs_per_sec, 3),
                 "unit": "runs/s",
                 "n_gpus": n_gpus,

@@ -358,3 +358,62 @@ spec:
             assert eng.storage.hydrate(run.output)["l"] == "m" * 100
         finally:
             eng.stop()
+
+
+class TestReferenceReadmeStory:
+    """The reference README's own documented example (gate + wait with a
+    BARE `until` expression) must parse and run verbatim
+    (reference: README.md:54-74)."""
+
+    YAML = """
+apiVersion: bubustack.io/v1alpha1
+kind: Story
+metadata:
+  name: gated-workflow
+spec:
+  pattern: batch
+  steps:
+  - name: approve
+    type: gate
+    with:
+      timeout: "30m"
+      onTimeout: "fail"
+  - name: wait-ready
+    type: wait
+    needs: [approve]
+    with:
+      until: "inputs.ready == true"
+      pollInterval: "5s"
+"""
+
+    def test_runs_verbatim(self):
+        import threading
+        import time
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(self.YAML)
+            run = eng.submit_run("default/gated-workflow", {"ready": True})
+
+            def approve():
+                deadline = time.time() + 10
+                while time.time() < deadline:
+                    st = run.step_states.get("approve")
+                    if st is not None and str(st.phase) in ("Paused", "Running"):
+                        eng.approve_gate(run, "approve", "readme-test")
+                        return
+                    time.sleep(0.02)
+
+            t = threading.Thread(target=approve)
+            t.start()
+            run = eng.wait(run, timeout=30)
+            t.join()
+            assert run.phase == Phase.SUCCEEDED, (
+                run.error,
+                {k: str(v.phase) for k, v in run.step_states.items()},
+            )
+        finally:
+            eng.stop()

@@ -314,6 +314,25 @@ def build_http_app(engine: "RunEngine"):
     async def run_status(ns: str, name: str):
         return _run_view(_get_run(ns, name))
 
+    @app.get("/runs/{ns}/{name}/trace")
+    async def run_trace(ns: str, name: str):
+        """Span trace for one run (the reference persists TraceInfo on run
+        status; spans here come from the in-process tracer ring)."""
+        run = _get_run(ns, name)
+        spans = [
+            {
+                "name": sp.name,
+                "start": sp.start,
+                "end": sp.end,
+                "durationMs": round(sp.duration * 1e3, 3),
+                "attributes": sp.attributes,
+                "error": sp.error,
+            }
+            for sp in engine.tracer.spans()
+            if sp.attributes.get("run") == run.name or sp.trace_id == run.trace.trace_id
+        ]
+        return {"traceId": run.trace.trace_id, "spans": spans}
+
     @app.post("/runs/{ns}/{name}/cancel")
     async def cancel_run(ns: str, name: str, body: _t.Optional[dict] = None):
         run = _get_run(ns, name)

@@ -797,3 +797,39 @@ spec:
     - {name: b, ref: {name: echoer}, needs: [src]}
 """
             )
+
+
+class TestPythonClient:
+    """bobrapet_amd.client.Client over the REST plane (TestClient session)."""
+
+    def test_client_round_trip(self, eng):
+        from fastapi.testclient import TestClient
+
+        from bobrapet_amd.client import Client, ClientError
+        from bobrapet_amd.engine.impulses import build_http_app
+
+        c = Client("http://testserver", session=TestClient(build_http_app(eng)))
+        n = c.apply(
+            """
+kind: Story
+metadata: {name: cli-flow}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {hello: "{{ inputs.who }}"}}
+  output: {msg: "{{ steps.a.output.hello }}"}
+"""
+        )
+        assert n == 1
+        rec = c.run_story("default/cli-flow", {"who": "world"})
+        assert rec["phase"] == "Succeeded"
+        assert rec["output"] == {"msg": "world"}
+        assert any(s["key"] == "default/cli-flow" for s in c.stories())
+        got = c.run("default", rec["name"])
+        assert got["phase"] == "Succeeded"
+        trace = c.trace("default", rec["name"])
+        assert isinstance(trace["spans"], list)
+        assert any(r["name"] == rec["name"] for r in c.runs(phase="Succeeded"))
+        import pytest as _p
+
+        with _p.raises(ClientError):
+            c.run("default", "no-such-run")

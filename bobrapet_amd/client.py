@@ -36,9 +36,11 @@ class Client:
     # -- plumbing -------------------------------------------------------
 
     def _req(self, method: str, path: str, json_body=None, params=None) -> dict:
-        r = self._s.request(
-            method, f"{self.base}{path}", json=json_body, params=params, timeout=self.timeout
-        )
+        kwargs = {"json": json_body, "params": params}
+        # test sessions (starlette TestClient) reject per-request timeouts
+        if type(self._s).__module__.startswith("requests"):
+            kwargs["timeout"] = self.timeout
+        r = self._s.request(method, f"{self.base}{path}", **kwargs)
         if r.status_code >= 400:
             raise ClientError(r.status_code, r.text)
         return r.json()
@@ -100,7 +102,7 @@ class Client:
         return self._req("POST", f"/impulses/{ns}/{name}", payload)
 
     def metrics_text(self) -> str:
-        r = self._s.get(f"{self.base}/metrics", timeout=self.timeout)
+        r = self._s.get(f"{self.base}/metrics")
         if r.status_code >= 400:
             raise ClientError(r.status_code, r.text)
         return r.text

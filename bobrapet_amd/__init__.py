@@ -18,3 +18,11 @@ from .enums import (  # noqa: F401
     StoryPattern,
     WorkloadMode,
 )
+
+
+def __getattr__(name):  # lazy: the client pulls in requests only on use
+    if name == "Client":
+        from .client import Client
+
+        return Client
+    raise AttributeError(name)

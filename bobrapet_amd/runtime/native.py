@@ -1,9 +1,12 @@
 """Native fast path: compile Stories onto the bobraccel C++ DAG core.
 
-The C++ core (csrc/core) owns the run state machine, timers, retries and
-template evaluation for the batch fast path; engram steps are dispatched to
-the same (gpu, stream) worker slots the Python engine uses.  Stories with
-compensations/finally or streaming pattern stay on the Python engine —
+The C++ core (csrc/core) owns the run state machine (the role of the
+reference's DAGReconciler — internal/controller/runs/dag.go:306-542 —
+including the 3-phase main→compensation→finally machine of dag.go:482-511,
+postExecution checks of steprun_controller.go:2050-2124, timers, retry
+classes, and template evaluation); engram steps are dispatched to the same
+(gpu, stream) worker slots the Python engine uses.  The full BATCH surface
+runs natively; streaming-pattern Stories route to the streaming runtime —
 ``story_supported()`` gates the fast path.
 """
 from __future__ import annotations

@@ -13,6 +13,14 @@ hipGraph capture (BASELINE config #4): a stage whose engram implements
 ``tensor_compute`` and whose step sets ``with.capture: true`` replays its
 per-packet GPU work as a captured hipGraph (torch.cuda.CUDAGraph is
 hipGraph on ROCm) after a warmup packet — fixed launch latency per packet.
+
+The full TransportStreamingSettings vocabulary is ENFORCED, not just
+parsed: fan-in all/any/quorum (accumulate-to-quorum joins), replay
+memory/durable (``StreamingRun.replay``), recording metadata/full with
+sampling + redaction, lifecycle drain/cutover (``StreamingRun.upgrade``
+with binding-generation bumps), hash/preserve partitioning (per-partition
+rings + parallel stage lanes, per-partition ordering), and per-packet
+routing rules (allow/deny over targeted downstreams).
 """
 from __future__ import annotations
 

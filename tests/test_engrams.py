@@ -179,3 +179,28 @@ spec:
         run, _sr = self._run("print('plain text result')")
         assert str(run.phase) == "Succeeded"
         assert run.output["out"] == {"stdout": "plain text result"}
+
+    def test_process_timeout_is_retryable(self):
+        """A process exceeding its step timeout maps to exit class retry
+        (the reference kills the Job pod and retries per policy)."""
+        from bobrapet_amd.engrams.base import EngramContext, EngramFailure
+        from bobrapet_amd.engrams.process import ProcessEngram
+
+        impl = ProcessEngram(["python3", "-c", "import time; time.sleep(30)"])
+        ctx = EngramContext(input={}, timeout_seconds=0.5)
+        import pytest as _p
+
+        with _p.raises(EngramFailure) as e:
+            impl.run(ctx)
+        assert e.value.exit_code == 1  # retryable
+
+    def test_missing_binary_is_terminal(self):
+        from bobrapet_amd.engrams.base import EngramContext, EngramFailure
+        from bobrapet_amd.engrams.process import ProcessEngram
+
+        impl = ProcessEngram(["/no/such/binary"])
+        import pytest as _p
+
+        with _p.raises(EngramFailure) as e:
+            impl.run(EngramContext(input={}))
+        assert e.value.exit_code == 2  # terminal

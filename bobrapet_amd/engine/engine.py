@@ -80,6 +80,9 @@ class RunEngine:
         self.effects = EffectLedger()
         self.metrics = metrics if metrics is not None else metrics_mod.MetricsRegistry()
         self.evaluator.metrics = self.metrics
+        from ..utils.logging import StructuredLogger
+
+        self.log = StructuredLogger("engine")
         self.tracer = tracer if tracer is not None else tracing_mod.Tracer(enabled=False)
         self.workers = WorkerPool(
             device_count=device_count,
@@ -121,6 +124,7 @@ class RunEngine:
             self.set_timer(
                 monotonic_now() + self.config.checkpoint_interval_seconds, "", "checkpoint"
             )
+        self.log.debug("engine started", feature="engine")
         return self
 
     def stop(self) -> None:
@@ -460,8 +464,9 @@ class RunEngine:
             self.save_state(tmp)
             _os.replace(tmp, path)
             self.metrics.inc("checkpoints_total")
-        except Exception:  # durability must never kill the loop
+        except Exception as exc:  # durability must never kill the loop
             self.metrics.inc("checkpoint_errors_total")
+            self.log.error("checkpoint failed", error=str(exc))
         self.set_timer(
             monotonic_now() + self.config.checkpoint_interval_seconds, "", "checkpoint"
         )
@@ -950,6 +955,15 @@ class RunEngine:
     # ------------------------------------------------------------------
 
     def on_run_terminal(self, run: StoryRun) -> None:
+        self.log.debug(
+            "run terminal", feature="runs", run=run.name, phase=str(run.phase),
+            story=f"{run.story_namespace}/{run.story_name}",
+        )
+        if run.phase in (Phase.FAILED, Phase.TIMEOUT):
+            self.log.warn(
+                "run failed", run=run.name,
+                error=(run.error.message if run.error else run.failure_cause),
+            )
         self.metrics.inc("storyruns_total", phase=str(run.phase))
         if run.started_at is not None and run.finished_at is not None:
             self.metrics.observe(

@@ -144,3 +144,61 @@ spec:
                 assert series in text, f"missing series {series}\n{text[:800]}"
         finally:
             eng.stop()
+
+
+class TestStructuredLogging:
+    def test_json_lines_and_feature_gating(self):
+        import io
+        import json as _json
+
+        from bobrapet_amd.utils.logging import (
+            ContractLogger,
+            StructuredLogger,
+            enable_feature,
+        )
+
+        buf = io.StringIO()
+        log = StructuredLogger("test", stream=buf, run="r1")
+        log.info("hello", x=1)
+        log.debug("hidden", feature="verbose-x")
+        enable_feature("verbose-x")
+        log.with_fields(step="a").debug("shown", feature="verbose-x")
+        enable_feature("verbose-x", False)
+        lines = [_json.loads(l) for l in buf.getvalue().splitlines()]
+        assert lines[0]["msg"] == "hello" and lines[0]["run"] == "r1"
+        assert len(lines) == 2 and lines[1]["step"] == "a"
+        cl = ContractLogger("boot", log)
+        cl.stage("load", detail="ok")
+        assert cl.stages[0]["stage"] == "load"
+
+    def test_engine_warns_on_failed_run(self, capsys):
+        import io
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            buf = io.StringIO()
+            eng.log.stream = buf
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: fail}
+spec: {builtin: fail}
+---
+kind: Engram
+metadata: {name: f}
+spec: {templateRef: {name: fail}}
+---
+kind: Story
+metadata: {name: boom}
+spec:
+  steps: [{name: a, ref: {name: f}, with: {succeedAfter: 99}}]
+"""
+            )
+            run = eng.run_story("default/boom", {}, timeout=30)
+            assert run.phase == Phase.FAILED
+            assert '"level": "warn"' in buf.getvalue()
+        finally:
+            eng.stop()

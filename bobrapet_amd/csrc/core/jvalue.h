@@ -143,6 +143,22 @@ class JValue {
       const auto& obj = as_object();
       auto it = obj.find(key);
       if (it != obj.end()) return it->second;
+      // indexing THROUGH an offloaded payload: derive a sub-path marker
+      // instead of mis-reading the marker dict.  The storage layer's ref
+      // path DSL (storage/manager.py path.go parity) resolves it when the
+      // value is hydrated on the worker — the engine loop never touches
+      // payload bytes (offloaded-data discipline, dag.go 3-way policy).
+      auto ref = obj.find("$storageRef");
+      if (ref != obj.end() && ref->second.is_object()) {
+        JObject inner = ref->second.as_object();
+        auto pit = inner.find("path");
+        std::string base =
+            (pit != inner.end() && pit->second.is_string()) ? pit->second.as_string() : "";
+        inner["path"] = base.empty() ? key : base + "." + key;
+        JObject marker;
+        marker["$storageRef"] = std::move(inner);
+        return JValue(std::move(marker));
+      }
       // alias tolerance: '_' in template identifiers ↔ '-' in step names
       if (key.find('_') != std::string::npos) {
         std::string alt = key;
@@ -160,6 +176,21 @@ class JValue {
       const auto& a = as_array();
       if (i < 0) i += (int64_t)a.size();
       if (i >= 0 && i < (int64_t)a.size()) return a[(size_t)i];
+    }
+    if (is_object()) {
+      const auto& obj = as_object();
+      auto ref = obj.find("$storageRef");
+      if (ref != obj.end() && ref->second.is_object()) {
+        JObject inner = ref->second.as_object();
+        auto pit = inner.find("path");
+        std::string base =
+            (pit != inner.end() && pit->second.is_string()) ? pit->second.as_string() : "";
+        std::string key = "[" + std::to_string(i) + "]";
+        inner["path"] = base.empty() ? key : base + key;
+        JObject marker;
+        marker["$storageRef"] = std::move(inner);
+        return JValue(std::move(marker));
+      }
     }
     return JValue();
   }

@@ -619,3 +619,53 @@ spec:
             assert res["steps"]["go"]["output"] == {"via": "native", "step": "go"}
         finally:
             eng.stop()
+
+
+class TestNativeOffloadedSubPath:
+    """Templates indexing THROUGH an offloaded output on the native core:
+    the C++ VM derives `$storageRef` sub-path markers (storage path DSL)
+    instead of mis-reading marker dicts — worker-side hydration resolves
+    them, so the engine loop never touches payload bytes."""
+
+    def test_sub_path_through_marker_matches_python(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.enums import Phase
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2, max_inline_size=64)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: sub}
+spec:
+  steps:
+    - name: big
+      ref: {name: e}
+      with: {data: {x: 41, arr: [7, 8, 9], pad: "%s"}}
+    - name: use
+      ref: {name: e}
+      needs: [big]
+      with:
+        got: "{{ steps.big.output.data.x }}"
+        second: "{{ steps.big.output.data.arr[1] }}"
+"""
+                % ("x" * 90)
+            )
+            run = eng.run_story("default/sub", {}, timeout=30)
+            assert run.phase == Phase.SUCCEEDED
+            py_out = run.step_states["use"].output
+            res = NativeRunner.from_run_engine(eng).run_story("default/sub", {}, timeout=30)
+            assert res["phase"] == "Succeeded"
+            assert res["steps"]["use"]["output"] == {"got": 41, "second": 8}
+            assert py_out == {"got": 41, "second": 8}
+        finally:
+            eng.stop()

@@ -16,6 +16,28 @@
 
 namespace bobraccel {
 
+// Optional offloaded-payload hydrator (set by the Python side): when an
+// expression CONSUMES a `$storageRef` marker (comparison, arithmetic,
+// truthiness, function argument), the value is materialized through the
+// storage layer; pure pass-through (member access / template splicing)
+// keeps moving markers so the loop never copies payload bytes.
+inline std::function<JValue(const JValue&)>& expr_hydrator() {
+  static std::function<JValue(const JValue&)> h;
+  return h;
+}
+
+inline bool is_storage_marker(const JValue& v) {
+  return v.is_object() && v.as_object().count("$storageRef") > 0;
+}
+
+inline JValue materialize(JValue v) {
+  if (is_storage_marker(v)) {
+    auto& h = expr_hydrator();
+    if (h) return h(v);
+  }
+  return v;
+}
+
 enum class Op {
   Const,
   Var,
@@ -302,27 +324,27 @@ inline JValue eval_expr(const ExprNode& n, const JObject& scope) {
       return JValue();
     }
     case Op::And:
-      if (!eval_expr(*n.children[0], scope).truthy()) return false;
-      return eval_expr(*n.children[1], scope).truthy();
+      if (!materialize(eval_expr(*n.children[0], scope)).truthy()) return false;
+      return materialize(eval_expr(*n.children[1], scope)).truthy();
     case Op::Or:
-      if (eval_expr(*n.children[0], scope).truthy()) return true;
-      return eval_expr(*n.children[1], scope).truthy();
+      if (materialize(eval_expr(*n.children[0], scope)).truthy()) return true;
+      return materialize(eval_expr(*n.children[1], scope)).truthy();
     case Op::Not:
-      return !eval_expr(*n.children[0], scope).truthy();
+      return !materialize(eval_expr(*n.children[0], scope)).truthy();
     case Op::Cmp:
-      return cmp_values(n.str, eval_expr(*n.children[0], scope),
-                        eval_expr(*n.children[1], scope));
+      return cmp_values(n.str, materialize(eval_expr(*n.children[0], scope)),
+                        materialize(eval_expr(*n.children[1], scope)));
     case Op::Bin:
-      return bin_values(n.str, eval_expr(*n.children[0], scope),
-                        eval_expr(*n.children[1], scope));
+      return bin_values(n.str, materialize(eval_expr(*n.children[0], scope)),
+                        materialize(eval_expr(*n.children[1], scope)));
     case Op::Neg: {
-      JValue v = eval_expr(*n.children[0], scope);
+      JValue v = materialize(eval_expr(*n.children[0], scope));
       if (v.is_int()) return -v.as_int();
       if (v.is_double()) return -v.as_double();
       return JValue();
     }
     case Op::Cond:
-      return eval_expr(*n.children[0], scope).truthy()
+      return materialize(eval_expr(*n.children[0], scope)).truthy()
                  ? eval_expr(*n.children[1], scope)
                  : eval_expr(*n.children[2], scope);
     case Op::List: {
@@ -338,14 +360,15 @@ inline JValue eval_expr(const ExprNode& n, const JObject& scope) {
     }
     case Op::Call: {
       std::vector<JValue> args;
-      for (const auto& c : n.children) args.push_back(eval_expr(*c, scope));
+      for (const auto& c : n.children)
+        args.push_back(materialize(eval_expr(*c, scope)));
       return call_fn(n.str, args);
     }
     case Op::Method: {
-      JValue obj = eval_expr(*n.children[0], scope);
+      JValue obj = materialize(eval_expr(*n.children[0], scope));
       std::vector<JValue> args;
       for (size_t i = 1; i < n.children.size(); ++i)
-        args.push_back(eval_expr(*n.children[i], scope));
+        args.push_back(materialize(eval_expr(*n.children[i], scope)));
       return call_method(n.str, obj, args);
     }
   }

@@ -316,6 +316,23 @@ PYBIND11_MODULE(_core, m) {
       .def("gc_run", &NativeEngine::gc_run,
            py::call_guard<py::gil_scoped_release>());
 
+  m.def("set_expr_hydrator", [](py::object fn) {
+    // the callable lives in a deliberately LEAKED holder: a py::object
+    // captured inside the static std::function would be decref'd during
+    // static teardown, after the interpreter is gone (segfault at exit)
+    static py::object* holder = new py::object();
+    *holder = fn;  // GIL held in a pybind def — safe decref of the old one
+    if (fn.is_none()) {
+      expr_hydrator() = nullptr;
+      return;
+    }
+    py::object* h = holder;
+    expr_hydrator() = [h](const JValue& marker) -> JValue {
+      py::gil_scoped_acquire g;
+      return to_jvalue((*h)(to_python(marker)));
+    };
+  });
+
   m.def("eval_expression", [](const py::handle& ast, const py::dict& scope) {
     ExprPtr e = to_expr(ast);
     JValue sv = to_jvalue(scope);

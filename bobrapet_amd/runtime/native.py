@@ -282,6 +282,10 @@ class NativeRunner:
         self._engram_cache: _t.Dict[str, tuple] = {}
         self._engram_cache_gen = -1
         self.engine.set_launcher(self._launch)
+        # expression-level hydration of offloaded payloads (markers passed
+        # THROUGH stay zero-copy; only consuming expressions materialize)
+        if self.storage is not None and hasattr(core, "set_expr_hydrator"):
+            core.set_expr_hydrator(self.storage.hydrate)
         self.engine.start()
 
     @classmethod

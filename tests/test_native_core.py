@@ -669,3 +669,46 @@ spec:
             assert py_out == {"got": 41, "second": 8}
         finally:
             eng.stop()
+
+    def test_consuming_expressions_hydrate(self):
+        """Conditions/arithmetic CONSUMING offloaded values hydrate through
+        the expression-level hydrator (pass-through stays marker-based)."""
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2, max_inline_size=64)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: consume}
+spec:
+  steps:
+    - name: big
+      ref: {name: e}
+      with: {data: {x: 41, arr: [7, 8, 9], pad: "%s"}}
+    - name: gatekeep
+      type: condition
+      needs: [big]
+      with: {expression: "{{ steps.big.output.data.x > 40 && size(steps.big.output.data.arr) == 3 }}"}
+    - name: use
+      ref: {name: e}
+      needs: [gatekeep]
+      if: "{{ steps.big.output.data.arr[2] == 9 }}"
+      with: {sum: "{{ steps.big.output.data.x + 1 }}"}
+"""
+                % ("x" * 90)
+            )
+            res = NativeRunner.from_run_engine(eng).run_story("default/consume", {}, timeout=30)
+            assert res["phase"] == "Succeeded", res
+            assert res["steps"]["use"]["output"] == {"sum": 42}
+        finally:
+            eng.stop()

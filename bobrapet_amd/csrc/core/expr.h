@@ -401,7 +401,9 @@ inline JValue eval_template(const TNode& t, const JObject& scope) {
       std::string out;
       for (const auto& [lit, e] : t.parts) {
         out += lit;
-        if (e) out += eval_expr(*e, scope).to_string();
+        // splicing into a string CONSUMES the value: hydrate markers so
+        // "id-{{ steps.a.output.big.field }}" interpolates the real value
+        if (e) out += materialize(eval_expr(*e, scope)).to_string();
       }
       return out;
     }

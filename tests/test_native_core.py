@@ -670,6 +670,37 @@ spec:
         finally:
             eng.stop()
 
+    def test_string_splice_hydrates(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2, max_inline_size=48)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo}}
+---
+kind: Story
+metadata: {name: splice}
+spec:
+  steps:
+    - {name: big, ref: {name: e}, with: {data: {id: "ORD-7", pad: "%s"}}}
+    - {name: use, ref: {name: e}, needs: [big], with: {msg: "order={{ steps.big.output.data.id }}!"}}
+"""
+                % ("x" * 60)
+            )
+            res = NativeRunner.from_run_engine(eng).run_story("default/splice", {}, timeout=30)
+            assert res["phase"] == "Succeeded"
+            assert res["steps"]["use"]["output"] == {"msg": "order=ORD-7!"}
+        finally:
+            eng.stop()
+
     def test_consuming_expressions_hydrate(self):
         """Conditions/arithmetic CONSUMING offloaded values hydrate through
         the expression-level hydrator (pass-through stays marker-based)."""

@@ -302,8 +302,30 @@ class NativeEngine {
   }
 
   void gc_run(uint64_t run_id) {
+    // erase a TERMINAL run and its terminal descendants (executeStory
+    // children submitted internally carry parent_run links) — the retention
+    // role of the Python engine's TTL cleanup for the fast path
     std::lock_guard<std::mutex> g(mu_);
-    runs_.erase(run_id);
+    auto it = runs_.find(run_id);
+    if (it == runs_.end() || !is_terminal(it->second.phase)) return;
+    std::vector<uint64_t> doomed{run_id};
+    bool grew = true;
+    while (grew) {
+      grew = false;
+      for (const auto& [id, r] : runs_) {
+        if (r.parent_run == 0 || !is_terminal(r.phase)) continue;
+        bool parent_doomed = false, self_doomed = false;
+        for (uint64_t d : doomed) {
+          parent_doomed |= d == r.parent_run;
+          self_doomed |= d == id;
+        }
+        if (parent_doomed && !self_doomed) {
+          doomed.push_back(id);
+          grew = true;
+        }
+      }
+    }
+    for (uint64_t d : doomed) runs_.erase(d);
   }
 
  private:

@@ -327,8 +327,14 @@ class NativeRunner:
             raise TimeoutError(f"native run {run_id} did not finish in {timeout}s")
         return self.engine.run_status(run_id)
 
-    def run_story(self, story, inputs=None, timeout: float = 60.0) -> dict:
-        return self.wait(self.submit(story, inputs), timeout)
+    def run_story(self, story, inputs=None, timeout: float = 60.0, gc: bool = True) -> dict:
+        run_id = self.submit(story, inputs)
+        status = self.wait(run_id, timeout)
+        if gc:
+            # reclaim the run record (+ terminal executeStory descendants)
+            # once its status is in hand — the fast path's retention
+            self.engine.gc_run(run_id)
+        return status
 
     def decide_gate(self, run_id: int, step_index: int, approved: bool) -> None:
         self.engine.decide_gate(run_id, step_index, approved)

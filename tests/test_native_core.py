@@ -743,3 +743,38 @@ spec:
             assert res["steps"]["use"]["output"] == {"sum": 42}
         finally:
             eng.stop()
+
+
+class TestNativeRunRetention:
+    def test_run_story_reclaims_records_including_substories(self):
+        """The fast path's retention: run_story gc's the terminal run and
+        its executeStory descendants (no unbounded run-map growth)."""
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: Story
+metadata: {name: inner}
+spec:
+  steps: [{name: a, type: sleep, with: {duration: 0ms}}]
+---
+kind: Story
+metadata: {name: outer}
+spec:
+  steps:
+    - {name: sub, type: executeStory, with: {storyRef: inner}}
+"""
+            )
+            nr = NativeRunner.from_run_engine(eng)
+            for _ in range(10):
+                assert nr.run_story("default/outer", {}, timeout=30)["phase"] == "Succeeded"
+            assert nr.engine.run_count() == 0
+            # opt-out keeps the record for later inspection
+            res = nr.run_story("default/outer", {}, timeout=30, gc=False)
+            assert res["phase"] == "Succeeded"
+            assert nr.engine.run_count() >= 1
+        finally:
+            eng.stop()

@@ -52,6 +52,9 @@ from .workers import WorkerPool
 _UNKNOWN_RETRY_ABS_CAP = 50  # safety net so budget-free UNKNOWN retries terminate
 
 
+_RUN_SEQ = __import__("itertools").count()  # GIL-atomic uniqueness suffix
+
+
 class RunEngine:
     def __init__(
         self,
@@ -212,7 +215,12 @@ class RunEngine:
             story = self.registry.story(nm, ns or "default")
         namespace = namespace or story.namespace
         if name is None:
-            name = f"{story.name}-{monotonic_now():.6f}".replace(".", "-")[:63]
+            # timestamp for readability + atomic counter for uniqueness:
+            # concurrent submitters in the same clock tick must never
+            # collide (a collision silently adopts the other run)
+            name = (
+                f"{story.name}-{monotonic_now():.6f}-{next(_RUN_SEQ)}".replace(".", "-")[:63]
+            )
 
         # guards (reference: storyrun_controller.go:981-1045)
         inputs = self._prepare_inputs(story, inputs, trusted=_trusted)
@@ -423,7 +431,12 @@ class RunEngine:
             raise ValueError(f"story {story.key} is not a streaming story")
         namespace = namespace or story.namespace
         if name is None:
-            name = f"{story.name}-{monotonic_now():.6f}".replace(".", "-")[:63]
+            # timestamp for readability + atomic counter for uniqueness:
+            # concurrent submitters in the same clock tick must never
+            # collide (a collision silently adopts the other run)
+            name = (
+                f"{story.name}-{monotonic_now():.6f}-{next(_RUN_SEQ)}".replace(".", "-")[:63]
+            )
         run = StoryRun(
             name=name,
             namespace=namespace,

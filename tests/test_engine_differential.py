@@ -93,13 +93,13 @@ def _random_story(rng: random.Random, idx: int) -> str:
 
 @pytest.mark.timeout(300)
 def test_python_and_native_engines_agree():
-    rng = random.Random(20260913)
+    rng = random.Random(20260914)
     eng = RunEngine(EngineConfig(cpu_workers=4, default_max_retries=0)).start()
     try:
         eng.apply_yaml(RESOURCES)
         nr = NativeRunner.from_run_engine(eng)
         checked = 0
-        for idx in range(60):
+        for idx in range(100):
             yaml_text = _random_story(rng, idx)
             try:
                 eng.apply_yaml(yaml_text)
@@ -134,6 +134,6 @@ def test_python_and_native_engines_agree():
                     continue
                 raise AssertionError((idx, yaml_text, k, a, b, py_phases, nat_phases))
             checked += 1
-        assert checked >= 40, f"only {checked} stories compared"
+        assert checked >= 70, f"only {checked} stories compared"
     finally:
         eng.stop()

@@ -269,7 +269,9 @@ class RunEngine:
         if not trusted and self.storage.contains_refs(inputs):
             # engine-created refs arrive only via trusted submitters
             # (executor sub-stories); user-facing paths reject them
-            raise ValueError("storyrun inputs may not contain $storageRef values")
+            raise ValueError(
+                "storyrun inputs may not contain $storageRef/$envRef/$fileRef values"
+            )
         if story.inputs_schema is not None:
             inputs = apply_defaults(inputs, story.inputs_schema)
             errs = validate_instance(inputs, story.inputs_schema)

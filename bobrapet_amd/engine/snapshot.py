@@ -13,7 +13,8 @@ from __future__ import annotations
 import json
 import typing as _t
 
-from ..enums import ExitClass, Phase
+from ..enums import EffectClaimPhase, ExitClass, Phase, classify_exit_code
+from .effects import EffectClaim
 from .records import (
     GateStatus,
     StepRun,
@@ -51,6 +52,18 @@ def _clean(value):
     if isinstance(value, list):
         return [_clean(v) for v in value]
     return value
+
+
+def _restore_exit_class(sd: dict):
+    """Restore the recorded exit class; fall back to classifying the exit
+    code (older snapshots recorded only exitCode)."""
+    ec = sd.get("exitClass")
+    if ec:
+        return ExitClass(ec)
+    code = sd.get("exitCode")
+    if code is None:
+        return None
+    return classify_exit_code(int(code))
 
 
 def dump_state(engine) -> dict:
@@ -115,10 +128,29 @@ def dump_state(engine) -> dict:
                     "error": _err_to_dict(sr.status.error),
                     "retries": sr.status.retries,
                     "exitCode": sr.status.exit_code,
+                    "exitClass": str(sr.status.exit_class) if sr.status.exit_class is not None else None,
                     "cacheHit": sr.status.cache_hit,
                 }
             )
-    return {"version": 1, "storyRuns": runs, "stepRuns": steps}
+    # EffectClaim ledger: without this, completed side effects re-run after
+    # a restore (restored non-terminal steps re-execute against an empty
+    # ledger) — the reference persists claims durably as the EffectClaim CRD.
+    effects = []
+    with engine.effects._lock:
+        for claim in engine.effects._claims.values():
+            effects.append(
+                {
+                    "key": claim.key,
+                    "holder": claim.holder,
+                    "phase": str(claim.phase),
+                    "leaseDuration": claim.lease_duration,
+                    "acquiredAt": claim.acquired_at,
+                    "renewedAt": claim.renewed_at,
+                    "takeovers": claim.takeovers,
+                    "description": claim.description,
+                }
+            )
+    return {"version": 1, "storyRuns": runs, "stepRuns": steps, "effectClaims": effects}
 
 
 def save_state(engine, path: str) -> None:
@@ -200,11 +232,26 @@ def load_state(engine, path: str) -> int:
                 error=_err_from_dict(sd.get("error")),
                 retries=int(sd.get("retries", 0)),
                 exit_code=sd.get("exitCode"),
-                exit_class=ExitClass.SUCCESS if sd.get("exitCode") == 0 else None,
+                exit_class=_restore_exit_class(sd),
                 cache_hit=bool(sd.get("cacheHit")),
             ),
         )
         engine.store.create_or_get_step_run(sr)
+    # restore the EffectClaim ledger BEFORE re-ticking runs, so side effects
+    # already Completed pre-restart are not performed again (exactly-once)
+    with engine.effects._lock:
+        for cd in data.get("effectClaims", []):
+            claim = EffectClaim(
+                key=cd["key"],
+                holder=cd.get("holder", ""),
+                phase=EffectClaimPhase(cd.get("phase", "Reserved")),
+                lease_duration=float(cd.get("leaseDuration", 60.0)),
+                takeovers=int(cd.get("takeovers", 0)),
+                description=cd.get("description", ""),
+            )
+            claim.acquired_at = float(cd.get("acquiredAt", claim.acquired_at))
+            claim.renewed_at = float(cd.get("renewedAt", claim.renewed_at))
+            engine.effects._claims.setdefault(claim.key, claim)
     # resume every non-terminal run
     for run in engine.store.all_runs():
         if not run.is_terminal:

@@ -53,12 +53,18 @@ def _ring_pass(k: torch.Tensor, v: torch.Tensor, pg) -> _t.Tuple[torch.Tensor, t
     prv = (rank - 1) % world
     rk = torch.empty_like(k)
     rv = torch.empty_like(v)
-    reqs = [
-        dist.isend(k, dst=nxt, group=pg, tag=11),
-        dist.isend(v, dst=nxt, group=pg, tag=12),
-        dist.irecv(rk, src=prv, group=pg, tag=11),
-        dist.irecv(rv, src=prv, group=pg, tag=12),
-    ]
+    # batch_isend_irecv groups the four P2P ops so RCCL schedules the
+    # send+recv pairs together: plain isend-before-irecv on every rank is
+    # a classic NCCL/RCCL P2P deadlock pattern (and NCCL ignores tags —
+    # correctness must come from posting order, which batching guarantees).
+    reqs = dist.batch_isend_irecv(
+        [
+            dist.P2POp(dist.isend, k, peer=nxt, group=pg),
+            dist.P2POp(dist.irecv, rk, peer=prv, group=pg),
+            dist.P2POp(dist.isend, v, peer=nxt, group=pg),
+            dist.P2POp(dist.irecv, rv, peer=prv, group=pg),
+        ]
+    )
     for r in reqs:
         r.wait()
     return rk, rv

@@ -180,10 +180,18 @@ class StorageManager:
             raise RefError(f"invalid storage ref key {key!r}")
 
     def contains_refs(self, value, *, _depth: int = 0) -> bool:
+        """True if the payload carries ANY hydratable ref marker.
+
+        Covers all three ref types ($storageRef, $envRef, $fileRef): the
+        untrusted-input guard (engine._prepare_inputs) uses this to reject
+        spoofed refs — an $envRef/$fileRef injected by a run submitter
+        would otherwise exfiltrate process env/files into step outputs
+        (reference analogue: storyrun_webhook.go:389-423 ref-spoofing
+        rejection)."""
         if _depth > MAX_WALK_DEPTH:
             return False
         if isinstance(value, dict):
-            if STORAGE_REF_KEY in value:
+            if STORAGE_REF_KEY in value or ENV_REF_KEY in value or FILE_REF_KEY in value:
                 return True
             return any(self.contains_refs(v, _depth=_depth + 1) for v in value.values())
         if isinstance(value, list):

@@ -313,7 +313,12 @@ def _eval(node, ctx: _Ctx):
     if op == "cmp":
         return _compare(node[1], _eval(node[2], ctx), _eval(node[3], ctx))
     if op == "bin":
-        return _binop(node[1], _eval(node[2], ctx), _eval(node[3], ctx))
+        return _binop(
+            node[1],
+            _eval(node[2], ctx),
+            _eval(node[3], ctx),
+            cap=ctx.ev.config.max_output_bytes,
+        )
     if op == "neg":
         v = _eval(node[1], ctx)
         if v is MISSING or v is None:
@@ -378,15 +383,32 @@ def _compare(op: str, a, b):
     raise TemplateError(f"unknown comparison {op!r}")
 
 
-def _binop(op: str, a, b):
+def _estimated_len(v) -> int:
+    """Cheap element/char count for allocation bounding (no serialization)."""
+    if isinstance(v, (str, bytes, list, dict)):
+        return len(v)
+    return 1
+
+
+def _guard_alloc(n: int, cap: _t.Optional[int]) -> None:
+    """Bound result size BEFORE allocating: the op-count budget caps AST
+    steps, not bytes — a single `{{ 'x' * 10**9 }}` or huge concat could
+    otherwise OOM the shared engine process before _check_size runs."""
+    if cap is not None and cap > 0 and n > cap:
+        raise OutputTooLarge(f"operation result ~{n} elements exceeds cap {cap}")
+
+
+def _binop(op: str, a, b, cap: _t.Optional[int] = None):
     if a is MISSING:
         a = None
     if b is MISSING:
         b = None
     if op == "+":
         if isinstance(a, str) or isinstance(b, str):
+            _guard_alloc(_estimated_len(a) + _estimated_len(b), cap)
             return _stringify(a) + _stringify(b)
         if isinstance(a, list) and isinstance(b, list):
+            _guard_alloc(len(a) + len(b), cap)
             return a + b
         if a is None or b is None:
             return a if b is None else b
@@ -396,6 +418,10 @@ def _binop(op: str, a, b):
     if op == "-":
         return a - b
     if op == "*":
+        if isinstance(a, (str, list)) and isinstance(b, (int, float)):
+            _guard_alloc(int(_estimated_len(a) * max(b, 0)), cap)
+        elif isinstance(b, (str, list)) and isinstance(a, (int, float)):
+            _guard_alloc(int(_estimated_len(b) * max(a, 0)), cap)
         return a * b
     if op == "/":
         if b == 0:
@@ -416,6 +442,12 @@ def _call(ctx: _Ctx, name: str, args: _t.List):
         raise TemplateError(f"unknown function {name!r}")
     if name in _NONDETERMINISTIC and ctx.ev.config.deterministic:
         raise EvaluationBlocked(f"function {name!r} is blocked in deterministic mode")
+    if name == "range":
+        # bound BEFORE materializing: list(range(10**9)) would OOM the
+        # shared engine process before _check_size ever sees the result
+        r = range(*[int(x) for x in args])
+        _guard_alloc(len(r), ctx.ev.config.max_output_bytes)
+        return list(r)
     return fn(*args)
 
 

@@ -138,6 +138,49 @@ def test_spoofed_storage_ref_rejected():
         eng.stop()
 
 
+def test_effect_ledger_and_exit_class_persist(tmp_path):
+    """completed EffectClaims must survive a restore (else
+    re-executed steps repeat already-performed side effects), and failed
+    steps keep their exit class."""
+    from bobrapet_amd.enums import EffectClaimPhase, ExitClass
+    from bobrapet_amd.engine import snapshot
+
+    path = str(tmp_path / "state.json")
+    eng1 = RunEngine(EngineConfig(cpu_workers=1)).start()
+    try:
+        eng1.apply_yaml(RESOURCES)
+        done = eng1.submit_run("default/snap", {"x": 1})
+        time.sleep(0.2)
+        claim, fresh = eng1.effects.acquire("run/step/send-email", "worker-1")
+        assert fresh
+        eng1.effects.complete("run/step/send-email", "worker-1")
+        state = snapshot.dump_state(eng1)
+        assert any(c["key"] == "run/step/send-email" for c in state["effectClaims"])
+        eng1.save_state(path)
+    finally:
+        eng1.stop()
+
+    eng2 = RunEngine(EngineConfig(cpu_workers=1)).start()
+    try:
+        eng2.apply_yaml(RESOURCES)
+        eng2.load_state(path)
+        restored = eng2.effects.get("run/step/send-email")
+        assert restored is not None
+        assert restored.phase == EffectClaimPhase.COMPLETED
+        # a re-executing step must observe fresh=False (skip the effect)
+        _, fresh = eng2.effects.acquire("run/step/send-email", "worker-2")
+        assert not fresh
+    finally:
+        eng2.stop()
+
+    # exit-class restore fallback: classify from exit code when absent
+    from bobrapet_amd.engine.snapshot import _restore_exit_class
+
+    assert _restore_exit_class({"exitClass": "terminal"}) == ExitClass.TERMINAL
+    assert _restore_exit_class({"exitCode": 3}) == ExitClass.RATE_LIMITED
+    assert _restore_exit_class({"exitCode": None}) is None
+
+
 class TestAutoCheckpoint:
     """Periodic durability snapshots (EngineConfig.checkpoint_path) — the
     opt-in analog of the reference's always-durable etcd state."""

@@ -52,6 +52,13 @@ class TestDehydrateHydrate:
         out = mgr.dehydrate({"big": "x" * 100})
         assert mgr.contains_refs(out)
 
+    def test_contains_refs_covers_env_and_file_refs(self, mgr):
+        # spoofed env/file refs in untrusted inputs must be detectable
+        # (exfiltration via injected {"$envRef": ...})
+        assert mgr.contains_refs({"x": {"$envRef": {"name": "SECRET"}}})
+        assert mgr.contains_refs([{"deep": {"$fileRef": {"path": "etc/x"}}}])
+        assert not mgr.contains_refs({"x": {"envRef": "not-a-ref"}})
+
     def test_ref_key_validation(self, mgr):
         for bad in ("", "/abs/path", "a/../b", "sp ace"):
             with pytest.raises(RefError):

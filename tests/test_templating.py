@@ -131,6 +131,20 @@ class TestGuards:
         with pytest.raises(OutputTooLarge):
             ev.resolve_value({"big": "x" * 100}, {})
 
+    def test_allocation_bounded_before_materializing(self):
+        # range()/'*'/'+' must be bounded at evaluation time,
+        # not after allocation — these would OOM before _check_size runs
+        ev = Evaluator(EvalConfig(max_output_bytes=1 << 20))
+        with pytest.raises(OutputTooLarge):
+            ev.resolve_string("{{ range(1000000000) }}", SCOPE)
+        with pytest.raises(OutputTooLarge):
+            ev.resolve_string("{{ 'abc' * 1000000000 }}", SCOPE)
+        with pytest.raises(OutputTooLarge):
+            ev.resolve_string("{{ 100000000 * 'abc' }}", SCOPE)
+        # small allocations still work
+        assert ev.resolve_string("{{ range(3) }}", SCOPE) == [0, 1, 2]
+        assert ev.resolve_string("{{ 'ab' * 2 }}", SCOPE) == "abab"
+
     def test_op_budget(self):
         ev = Evaluator(EvalConfig(max_ops=10))
         from bobrapet_amd.templating.evaluator import EvaluationBudgetExceeded

@@ -37,6 +37,10 @@ void launch_gemm_bf16_nt(void*, const void*, const void*, int, int, int,
                          hipStream_t);
 void launch_gemv_bf16_nt(void*, const void*, const void*, int, int, int,
                          hipStream_t);
+void launch_gemm256(int, void*, const void*, const void*, const void*,
+                    const void*, void*, int, int, int, float, float,
+                    hipStream_t);
+void launch_rowsumsq(void*, const void*, int, int, hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
 }
@@ -304,6 +308,77 @@ torch::Tensor gemm_nt(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+void check_gemm256(const torch::Tensor& a, const torch::Tensor& b, int M,
+                   int N, int K) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(b.size(1) == K, "gemm256: K mismatch");
+  TORCH_CHECK(N % 256 == 0 && K % 32 == 0,
+              "gemm256: N must be a multiple of 256 and K of 32");
+}
+
+const float* stat_ptr(const c10::optional<torch::Tensor>& stat, int M) {
+  if (!stat.has_value()) return nullptr;
+  TORCH_CHECK(stat->scalar_type() == torch::kFloat32 && stat->is_contiguous() &&
+                  stat->numel() == M,
+              "gemm256: row stat must be f32 [M]");
+  return (const float*)stat->data_ptr();
+}
+
+torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor b,
+                         c10::optional<torch::Tensor> stat, double stat_mul,
+                         double stat_eps) {
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  check_gemm256(a, b, M, N, K);
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm256(0, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
+                 stat_ptr(stat, M), nullptr, M, N, K, (float)stat_mul,
+                 (float)stat_eps, cur_stream());
+  return c;
+}
+
+torch::Tensor gemm256_swiglu(torch::Tensor a, torch::Tensor b,
+                             c10::optional<torch::Tensor> stat,
+                             double stat_mul, double stat_eps) {
+  // b rows interleaved (gate_i, up_i); out[M][N/2] = silu(g)*u
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  check_gemm256(a, b, M, N, K);
+  auto c = torch::empty({M, N / 2}, a.options());
+  launch_gemm256(1, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
+                 stat_ptr(stat, M), nullptr, M, N, K, (float)stat_mul,
+                 (float)stat_eps, cur_stream());
+  return c;
+}
+
+std::vector<torch::Tensor> gemm256_resid(torch::Tensor a, torch::Tensor b,
+                                         torch::Tensor resid, bool want_stat) {
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  check_gemm256(a, b, M, N, K);
+  check_bf16(resid, "resid");
+  TORCH_CHECK(resid.numel() == (long)M * N, "gemm256_resid: resid shape");
+  auto c = torch::empty({M, N}, a.options());
+  torch::Tensor stat;
+  void* stat_p = nullptr;
+  if (want_stat) {
+    stat = torch::empty({M}, a.options().dtype(torch::kFloat32));
+    stat_p = stat.data_ptr();
+  }
+  launch_gemm256(2, c.data_ptr(), a.data_ptr(), b.data_ptr(), resid.data_ptr(),
+                 nullptr, stat_p, M, N, K, 0.0f, 0.0f, cur_stream());
+  if (want_stat) return {c, stat};
+  return {c};
+}
+
+torch::Tensor rowsumsq(torch::Tensor x) {
+  check_bf16(x, "x");
+  const int K = x.size(-1);
+  const int M = x.numel() / K;
+  TORCH_CHECK(K % 8 == 0, "rowsumsq: K must be %8==0");
+  auto stat = torch::empty({M}, x.options().dtype(torch::kFloat32));
+  launch_rowsumsq(stat.data_ptr(), x.data_ptr(), M, K, cur_stream());
+  return stat;
+}
+
 torch::Tensor gemv_nt(torch::Tensor a, torch::Tensor b) {
   // a: [M<=8, K] bf16, b: [N, K] bf16 -> [M, N]
   check_bf16(a, "a");
@@ -367,6 +442,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_variant", &attn_prefill_variant,
         "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
+  m.def("gemm256_nt", &gemm256_nt,
+        "256-tile bf16 MFMA GEMM, optional fused row-scale epilogue");
+  m.def("gemm256_swiglu", &gemm256_swiglu,
+        "256-tile GEMM with fused SwiGLU epilogue (interleaved gate/up)");
+  m.def("gemm256_resid", &gemm256_resid,
+        "256-tile GEMM with fused residual-add (+row sumsq) epilogue");
+  m.def("rowsumsq", &rowsumsq, "per-row sum of squares (f32)");
   m.def("gemv_nt", &gemv_nt, "bf16 weight-streaming GEMV (M<=8)");
   m.def("silu_mul_strided", &silu_mul_strided,
         "silu(gate)*up from a fused [.., 2I] gate_up matrix (no copies)");

@@ -279,6 +279,73 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return (a.float() @ b.float().t()).to(a.dtype)
 
 
+def _rowscale(stat: torch.Tensor, stat_mul: float, stat_eps: float) -> torch.Tensor:
+    return torch.rsqrt(stat.float() * stat_mul + stat_eps)
+
+
+def gemm256_nt(
+    a: torch.Tensor,
+    b: torch.Tensor,
+    stat: _t.Optional[torch.Tensor] = None,
+    stat_mul: float = 0.0,
+    stat_eps: float = 0.0,
+) -> torch.Tensor:
+    """C = a @ b.T (b stored [N,K]) on the 256-tile MFMA GEMM; optional
+    fused row-scale epilogue C[m] *= rsqrt(stat[m]*mul + eps) — the RMSNorm
+    entry with the per-channel gain folded into b (csrc/hip/gemm256.hip)."""
+    if a.is_cuda:
+        return _require_ext().gemm256_nt(a.contiguous(), b.contiguous(), stat, stat_mul, stat_eps)
+    c = a.float() @ b.float().t()
+    if stat is not None:
+        c = c * _rowscale(stat, stat_mul, stat_eps)[:, None]
+    return c.to(a.dtype)
+
+
+def gemm256_swiglu(
+    a: torch.Tensor,
+    b_interleaved: torch.Tensor,
+    stat: _t.Optional[torch.Tensor] = None,
+    stat_mul: float = 0.0,
+    stat_eps: float = 0.0,
+) -> torch.Tensor:
+    """Fused gate/up projection + SwiGLU: b rows interleaved
+    (gate_0, up_0, gate_1, up_1, ...) → out[M, N/2] = silu(g)*u.
+    Optional fused row-scale (norm entry) applied before the activation."""
+    if a.is_cuda:
+        return _require_ext().gemm256_swiglu(
+            a.contiguous(), b_interleaved.contiguous(), stat, stat_mul, stat_eps
+        )
+    c = a.float() @ b_interleaved.float().t()
+    if stat is not None:
+        c = c * _rowscale(stat, stat_mul, stat_eps)[:, None]
+    g, u = c[:, 0::2], c[:, 1::2]
+    return (g * torch.sigmoid(g) * u).to(a.dtype)
+
+
+def gemm256_resid(
+    a: torch.Tensor, b: torch.Tensor, resid: torch.Tensor, want_stat: bool = True
+):
+    """C = a @ b.T + resid, plus (optionally) the per-row sum of squares of
+    C — the next projection's norm statistic — accumulated in the epilogue.
+    Returns (C, stat) or (C, None)."""
+    if a.is_cuda:
+        out = _require_ext().gemm256_resid(
+            a.contiguous(), b.contiguous(), resid.contiguous(), want_stat
+        )
+        return (out[0], out[1]) if want_stat else (out[0], None)
+    c = a.float() @ b.float().t() + resid.float()
+    stat = c.pow(2).sum(dim=-1) if want_stat else None
+    return c.to(a.dtype), stat
+
+
+def rowsumsq(x: torch.Tensor) -> torch.Tensor:
+    """stat[m] = sum_k x[m,k]^2 (f32) over the trailing dim."""
+    if x.is_cuda:
+        return _require_ext().rowsumsq(x.contiguous())
+    k = x.shape[-1]
+    return x.float().reshape(-1, k).pow(2).sum(dim=-1)
+
+
 def rope_qkv_decode(qkv2d, kc, vc, cos_t, sin_t, pos_dev: torch.Tensor, Hq: int, Hkv: int, D: int):
     """Fused decode head prep: rope q (returned [B,Hq,D]) and rope k +
     copy v straight into the KV cache at DEVICE position pos (hipGraph-

@@ -145,6 +145,69 @@ class TestGemm:
         assert (got.float() - ref).abs().max().item() < 0.1
 
 
+class TestGemm256:
+    """The 256-tile fused-epilogue GEMM (the prefill hot path):
+    every epilogue variant vs the plain fp32 torch reference, including
+    M-tail shapes (clamped glds rows + predicated stores)."""
+
+    @pytest.mark.parametrize("m,n,k", [(256, 256, 32), (300, 512, 96), (1024, 768, 4096), (1, 256, 64)])
+    def test_plain_vs_torch(self, m, n, k):
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        got = ops.gemm256_nt(a, b)
+        ref = torch.matmul(a.float(), b.float().t())
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
+
+    def test_asymmetric_catches_transpose(self):
+        a = (torch.arange(300 * 96, device="cuda", dtype=torch.float32)
+             .reshape(300, 96) % 7).to(torch.bfloat16) / 7
+        b = (torch.arange(512 * 96, device="cuda", dtype=torch.float32)
+             .reshape(512, 96) % 5).to(torch.bfloat16) / 5
+        got = ops.gemm256_nt(a, b)
+        ref = torch.matmul(a.float(), b.float().t())
+        assert (got.float() - ref).abs().max().item() < 0.1
+
+    def test_rowscale_epilogue(self):
+        m, n, k = 300, 512, 4096
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        stat = ops.rowsumsq(a)
+        got = ops.gemm256_nt(a, b, stat, 1.0 / k, 1e-5)
+        # reference: rmsnorm(a) @ b.T with unit gain == rowscale epilogue
+        ref = torch.matmul(ops.rmsnorm_ref(a, torch.ones(k, device="cuda", dtype=a.dtype)).float(), b.float().t())
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.03, rel
+
+    def test_swiglu_epilogue(self):
+        m, n, k = 512, 1024, 256
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.5
+        b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.5
+        got = ops.gemm256_swiglu(a, b)
+        c = torch.matmul(a.float(), b.float().t())
+        g, u = c[:, 0::2], c[:, 1::2]
+        ref = g * torch.sigmoid(g) * u
+        assert (got.float() - ref).abs().max().item() < 0.05
+
+    def test_resid_epilogue_and_stat(self):
+        m, n, k = 300, 512, 768
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        b = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        r = torch.randn(m, n, dtype=torch.bfloat16, device="cuda")
+        got, stat = ops.gemm256_resid(a, b, r, want_stat=True)
+        ref = torch.matmul(a.float(), b.float().t()) + r.float()
+        assert (got.float() - ref).abs().max().item() < ref.abs().max().item() * 0.02
+        ref_stat = ref.pow(2).sum(dim=-1)
+        rel = ((stat - ref_stat).abs() / ref_stat.clamp(min=1.0)).max().item()
+        assert rel < 0.02, rel
+
+    def test_rowsumsq(self):
+        x = torch.randn(333, 4096, dtype=torch.bfloat16, device="cuda")
+        got = ops.rowsumsq(x)
+        ref = x.float().pow(2).sum(dim=-1)
+        assert ((got - ref).abs() / ref.clamp(min=1.0)).max().item() < 1e-3
+
+
 class TestAttentionDecode:
     @pytest.mark.parametrize("b,hq,hkv,l", [(1, 1, 1, 64), (2, 32, 8, 500), (4, 8, 8, 1024)])
     def test_vs_ref(self, b, hq, hkv, l):

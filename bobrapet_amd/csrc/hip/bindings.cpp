@@ -40,6 +40,8 @@ void launch_gemv_bf16_nt(void*, const void*, const void*, int, int, int,
 void launch_gemm256(int, void*, const void*, const void*, const void*,
                     const void*, void*, int, int, int, float, float,
                     hipStream_t);
+void launch_gemv2(int, void*, const void*, const void*, const void*, int, int,
+                  int, float, float, hipStream_t);
 void launch_rowsumsq(void*, const void*, int, int, hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
@@ -369,6 +371,47 @@ std::vector<torch::Tensor> gemm256_resid(torch::Tensor a, torch::Tensor b,
   return {c};
 }
 
+torch::Tensor gemv_norm(torch::Tensor a, torch::Tensor w, double mul,
+                        double eps) {
+  // out = rsqrt(sumsq(a_m)*mul + eps) * (a @ w.T)  (self-normalizing entry)
+  check_bf16(a, "a");
+  check_bf16(w, "w");
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(M <= 8 && w.size(1) == K && K % 8 == 0, "gemv_norm: bad shape");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemv2(1, c.data_ptr(), a.data_ptr(), w.data_ptr(), nullptr, M, N, K,
+               (float)mul, (float)eps, cur_stream());
+  return c;
+}
+
+torch::Tensor gemv_resid(torch::Tensor a, torch::Tensor w,
+                         torch::Tensor resid) {
+  check_bf16(a, "a");
+  check_bf16(w, "w");
+  check_bf16(resid, "resid");
+  const int M = a.size(0), K = a.size(1), N = w.size(0);
+  TORCH_CHECK(M <= 8 && w.size(1) == K && K % 8 == 0, "gemv_resid: bad shape");
+  TORCH_CHECK(resid.numel() == (long)M * N, "gemv_resid: resid shape");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemv2(2, c.data_ptr(), a.data_ptr(), w.data_ptr(), resid.data_ptr(),
+               M, N, K, 0.0f, 0.0f, cur_stream());
+  return c;
+}
+
+torch::Tensor gemv_swiglu_norm(torch::Tensor a, torch::Tensor w_i, double mul,
+                               double eps) {
+  // w_i rows interleaved (gate_i, up_i): out[M, rows/2] = silu(s*g)*(s*u)
+  check_bf16(a, "a");
+  check_bf16(w_i, "w");
+  const int M = a.size(0), K = a.size(1), N = w_i.size(0) / 2;
+  TORCH_CHECK(M <= 8 && w_i.size(1) == K && K % 8 == 0 && w_i.size(0) % 2 == 0,
+              "gemv_swiglu_norm: bad shape");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemv2(3, c.data_ptr(), a.data_ptr(), w_i.data_ptr(), nullptr, M, N, K,
+               (float)mul, (float)eps, cur_stream());
+  return c;
+}
+
 torch::Tensor rowsumsq(torch::Tensor x) {
   check_bf16(x, "x");
   const int K = x.size(-1);
@@ -449,6 +492,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm256_resid", &gemm256_resid,
         "256-tile GEMM with fused residual-add (+row sumsq) epilogue");
   m.def("rowsumsq", &rowsumsq, "per-row sum of squares (f32)");
+  m.def("gemv_norm", &gemv_norm, "self-normalizing decode GEMV (rmsnorm entry folded)");
+  m.def("gemv_resid", &gemv_resid, "decode GEMV with fused residual add");
+  m.def("gemv_swiglu_norm", &gemv_swiglu_norm,
+        "decode GEMV, interleaved gate/up + fused SwiGLU + norm entry");
   m.def("gemv_nt", &gemv_nt, "bf16 weight-streaming GEMV (M<=8)");
   m.def("silu_mul_strided", &silu_mul_strided,
         "silu(gate)*up from a fused [.., 2I] gate_up matrix (no copies)");

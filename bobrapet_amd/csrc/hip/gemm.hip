@@ -189,6 +189,131 @@ __global__ __launch_bounds__(256) void gemv_bf16_nt_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Decode GEMV epilogue variants (fewer, fused decode
+// kernels).  Same weight-streaming structure as gemv_bf16_nt; the variants
+// fold the surrounding elementwise work into the stream:
+//   EPI 1 "norm entry":  out = s[m] * (x @ W^T) with s[m] computed from the
+//         SAME streamed x rows (rsqrt(sumsq*mul + eps)) — rmsnorm folded
+//         into the projection (per-channel gain folded into W on the host),
+//         no stat buffer, no separate rmsnorm kernel.
+//   EPI 2 "resid": out = x @ W^T + resid  (the residual-add fused away)
+//   EPI 3 "swiglu norm": W rows interleaved (gate_i, up_i); each wave
+//         streams BOTH rows of its output column: out = silu(s·g)·(s·u)
+//         — gate_up intermediate and the silu_mul kernel both eliminated.
+// ---------------------------------------------------------------------------
+
+template <int M, int EPI>
+__global__ __launch_bounds__(256) void gemv2_kernel(
+    unsigned short* __restrict__ C,        // [M][N]
+    const unsigned short* __restrict__ A,  // [M][K]
+    const unsigned short* __restrict__ W,  // [N][K] (EPI 3: [2N][K])
+    const unsigned short* __restrict__ resid,  // [M][N] (EPI 2)
+    int N, int K, float stat_mul, float stat_eps) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int waves_total = (gridDim.x * blockDim.x) >> 6;
+
+  float scale[M];
+#pragma unroll
+  for (int m = 0; m < M; ++m) scale[m] = 1.0f;
+  if (EPI == 1 || EPI == 3) {
+    // row scales from the same x this wave streams for its dots (x rows
+    // are L1/L2-resident; the extra pass is noise next to the W stream)
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      float ss = 0.f;
+      for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
+        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)m * K + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = bf2f(av[j]);
+          ss += f * f;
+        }
+      }
+      ss = wave_reduce_sum(ss);
+      scale[m] = rsqrtf(ss * stat_mul + stat_eps);
+    }
+  }
+
+  for (int n = blockIdx.x * 4 + wid; n < N; n += waves_total) {
+    const unsigned short* wg = W + (long)(EPI == 3 ? 2 * n : n) * K;
+    const unsigned short* wu = wg + K;  // EPI 3 only
+    float accg[M], accu[M];
+#pragma unroll
+    for (int m = 0; m < M; ++m) accg[m] = accu[m] = 0.f;
+    for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
+      ushort8v gv = *reinterpret_cast<const ushort8v*>(wg + k0);
+      ushort8v uv;
+      if (EPI == 3) uv = *reinterpret_cast<const ushort8v*>(wu + k0);
+#pragma unroll
+      for (int m = 0; m < M; ++m) {
+        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)m * K + k0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float a = bf2f(av[j]);
+          accg[m] += a * bf2f(gv[j]);
+          if (EPI == 3) accu[m] += a * bf2f(uv[j]);
+        }
+      }
+    }
+#pragma unroll
+    for (int m = 0; m < M; ++m) {
+      float g = wave_reduce_sum(accg[m]);
+      float u = (EPI == 3) ? wave_reduce_sum(accu[m]) : 0.f;
+      if (lane == 0) {
+        float out;
+        if (EPI == 3) {
+          g *= scale[m];
+          u *= scale[m];
+          out = (g / (1.0f + __expf(-g))) * u;
+        } else if (EPI == 2) {
+          out = g + bf2f(resid[(long)m * N + n]);
+        } else {
+          out = g * scale[m];
+        }
+        C[(long)m * N + n] = f2bf(out);
+      }
+    }
+  }
+}
+
+template <int EPI>
+static void gemv2_dispatch(void* C, const void* A, const void* W,
+                           const void* resid, int M, int N, int K,
+                           float stat_mul, float stat_eps, hipStream_t stream) {
+  int blocks = (N + 3) / 4;
+  if (blocks > 2048) blocks = 2048;
+  dim3 grid(blocks), block(256);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
+                       (const unsigned short*)A, (const unsigned short*)W,
+                       (const unsigned short*)resid, N, K, stat_mul, stat_eps);
+  };
+  switch (M) {
+    case 1: launch(gemv2_kernel<1, EPI>); break;
+    case 2: launch(gemv2_kernel<2, EPI>); break;
+    case 3: launch(gemv2_kernel<3, EPI>); break;
+    case 4: launch(gemv2_kernel<4, EPI>); break;
+    case 5: launch(gemv2_kernel<5, EPI>); break;
+    case 6: launch(gemv2_kernel<6, EPI>); break;
+    case 7: launch(gemv2_kernel<7, EPI>); break;
+    default: launch(gemv2_kernel<8, EPI>); break;
+  }
+}
+
+extern "C" void launch_gemv2(int epi, void* C, const void* A, const void* W,
+                             const void* resid, int M, int N, int K,
+                             float stat_mul, float stat_eps,
+                             hipStream_t stream) {
+  switch (epi) {
+    case 1: gemv2_dispatch<1>(C, A, W, resid, M, N, K, stat_mul, stat_eps, stream); break;
+    case 2: gemv2_dispatch<2>(C, A, W, resid, M, N, K, stat_mul, stat_eps, stream); break;
+    case 3: gemv2_dispatch<3>(C, A, W, resid, M, N, K, stat_mul, stat_eps, stream); break;
+    default: gemv2_dispatch<0>(C, A, W, resid, M, N, K, stat_mul, stat_eps, stream); break;
+  }
+}
+
 extern "C" void launch_gemv_bf16_nt(void* C, const void* A, const void* W,
                                     int M, int N, int K, hipStream_t stream) {
   int blocks = (N + 3) / 4;

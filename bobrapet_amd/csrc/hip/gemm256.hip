@@ -35,16 +35,22 @@
 #define G2_BM 256
 #define G2_BN 256
 #define G2_BKS 32
-#define G2_NSLOT 4
+#define G2_NSLOT 5                          // 5 x 32 KB = all 160 KB of LDS
 #define G2_PART (G2_BM * G2_BKS * 2)        // one operand part: 16 KB
 #define G2_SLOT (2 * G2_PART)               // A+B: 32 KB
 
 typedef float f32x4g2 __attribute__((ext_vector_type(4)));
 
-// bank-spread XOR for the [row][64-byte k-row] LDS image: rows sharing
-// row%4 land on the same ds_read_b128 bank; XOR the 16-byte slot by
-// (row>>2)&3 so all 16 rows of a lane group hit distinct banks.
-__device__ __forceinline__ int g2_swz(int row) { return ((row >> 2) & 3) << 4; }
+// Bank-spread XOR for the [row][64-byte k-row] LDS image.  A wave64
+// ds_read_b128 is serviced in four 16-lane groups that MIX row (l15) and
+// k-group (lg) lanes — e.g. {l15 0-3 & 12-15 at lg=0} ∪ {l15 4-11 at
+// lg=1} (microarch §LDS).  With bank = (16·row + 4·(lg ^ f(row))) mod 64,
+// the per-group distinctness condition reduces to
+// {f(c), f(12+c), 1^f(4+c), 1^f(8+c)} all distinct for every c∈0..3,
+// which f(row) = 3·((row>>3)&1) satisfies for all four groups — measured:
+// the previous (row>>2)&3 variant left a 2-way conflict in every group
+// (SQ_LDS_BANK_CONFLICT ≈ 10% of wave cycles).
+__device__ __forceinline__ int g2_swz(int row) { return ((row >> 3) & 1) * 48; }
 
 __device__ __forceinline__ void g2_glds(const unsigned short* src, char* lds_dst) {
   __builtin_amdgcn_global_load_lds(
@@ -70,8 +76,30 @@ __global__ __launch_bounds__(512, 2) void gemm256_kernel(
   const int nbn = N / G2_BN;
   const int nbm = (M + G2_BM - 1) / G2_BM;
   unsigned int wgid = xcd_swizzle(blockIdx.x, nbm * nbn);
-  const int bm = (int)(wgid / nbn) * G2_BM;
-  const int bn = (int)(wgid % nbn) * G2_BN;
+  // grouped supertile order: bn varies fastest within GN-wide column
+  // groups, bm within a group next.  An XCD's contiguous chunk then
+  // covers a compact (many-bm × GN-bn) window: the A panel (2 MB) stays
+  // L2-resident across each 16-tile bn run and the GN B panels stay
+  // LLC-resident across bm — measured FETCH dropped ~4x vs the flat
+  // bn-fastest order on the gate/up shape.
+  const int GN = 16;
+  int bm_i, bn_i;
+  {
+    const int full = nbn / GN;             // full-width column groups
+    const int per = nbm * GN;
+    if ((int)wgid < full * per) {
+      const int grp = (int)wgid / per, rem = (int)wgid % per;
+      bm_i = rem / GN;
+      bn_i = grp * GN + rem % GN;
+    } else {
+      const int tail_n = nbn - full * GN;  // last, narrower group
+      const int rem = (int)wgid - full * per;
+      bm_i = rem / tail_n;
+      bn_i = full * GN + rem % tail_n;
+    }
+  }
+  const int bm = bm_i * G2_BM;
+  const int bn = bn_i * G2_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
@@ -84,71 +112,160 @@ __global__ __launch_bounds__(512, 2) void gemm256_kernel(
   // per-wave glds piece assignment: 16 x 1 KB pieces per operand part,
   // wave w stages pieces {2w, 2w+1}; chunk = piece*64 + lane (16 B each);
   // LDS image row = chunk/4 (64-B k-rows), source k-byte inverse-swizzled.
+  // Source pointers are computed ONCE (incl. the M-tail row clamp) and
+  // only advance by kt*BKS columns per slice — keeps the hot loop's
+  // register pressure down to one scalar offset per stage.
+  // piece it=1 is piece it=0 shifted by 16 rows; the swizzle term is
+  // invariant under row+16 ((row>>3)&1 unchanged), so its source is the
+  // it=0 pointer + a UNIFORM 16*K elements — one pointer pair to carry.
+  const unsigned short* asrc0;
+  const unsigned short* bsrc0;
+  long arow1_fix = 16L * K;  // it=1 offset; adjusted below if clamped
+  {
+    const int chunk = (wid * 2) * 64 + lane;
+    const int row = chunk >> 2;
+    const int kbx = ((chunk & 3) * 16) ^ g2_swz(row);
+    int arow = bm + row;
+    if (arow >= M) arow = M - 1;  // M-tail: clamp (stores are predicated)
+    int arow1 = bm + row + 16;
+    if (arow1 >= M) arow1 = M - 1;
+    arow1_fix = (long)(arow1 - arow) * K;
+    asrc0 = A + (long)arow * K + kbx / 2;
+    bsrc0 = B + (long)(bn + row) * K + kbx / 2;
+  }
   auto stage = [&](int slot, int kt) {
     const int k0 = kt * G2_BKS;
-    char* abase = smem + slot * G2_SLOT;
+    char* abase = smem + slot * G2_SLOT + (wid * 2) * 1024;
     char* bbase = abase + G2_PART;
-#pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      const int p = wid * 2 + it;
-      const int chunk = p * 64 + lane;
-      const int row = chunk >> 2;
-      const int kbx = ((chunk & 3) * 16) ^ g2_swz(row);
-      int arow = bm + row;
-      if (arow >= M) arow = M - 1;   // M-tail: clamp (stores are predicated)
-      g2_glds(A + (long)arow * K + k0 + kbx / 2, abase + p * 1024);
-      g2_glds(B + (long)(bn + row) * K + k0 + kbx / 2, bbase + p * 1024);
-    }
+    g2_glds(asrc0 + k0, abase);
+    g2_glds(bsrc0 + k0, bbase);
+    g2_glds(asrc0 + arow1_fix + k0, abase + 1024);
+    g2_glds(bsrc0 + 16L * K + k0, bbase + 1024);
   };
 
   f32x4g2 acc[8][4] = {};
   const int nk = K / G2_BKS;
 
+  // fragment-read addresses are loop-invariant except the slot base
+  const int a_off = (wr * 128 + l15) * 64 + ((lg * 16) ^ g2_swz(wr * 128 + l15));
+  const int b_off = (wc * 64 + l15) * 64 + ((lg * 16) ^ g2_swz(wc * 64 + l15));
+  // i*16 rows stride: 16*64 bytes; the swizzle depends on row>>2 which
+  // changes every 4 rows — but row = base + i*16 keeps (row>>2)&3
+  // invariant, so the XOR term is the same for every fragment index.
+  auto rd_a = [&](const char* ab, int i) {
+    return *reinterpret_cast<const bf16x8*>(ab + a_off + i * 16 * 64);
+  };
+  auto rd_b = [&](const char* bb, int j) {
+    return *reinterpret_cast<const bf16x8*>(bb + b_off + j * 16 * 64);
+  };
+
+  // software-pipelined schedule (template-style read/barrier pairing):
+  // each phase issues the CURRENT slice's remaining fragment reads (B +
+  // A-tail, slot s) and the glds prefetch BEFORE the barrier, so their
+  // latency overlaps the barrier wait; after the barrier the MFMA
+  // cluster starts on fragments that are already (or nearly) resident,
+  // with only the next slice's 4 leading A reads interleaved into the
+  // cluster.  glds runs 4 slices ahead through the 5-slot ring with
+  // counted vmcnt — 2 slices always in flight across the barrier.
+  bf16x8 afp1[2], afp2[2];  // leading A frags of s (alternating sets)
+  bf16x8 aft[6];            // trailing A frags of s (read + used in-phase)
+  bf16x8 bf[4];
+
   stage(0, 0);
   if (nk > 1) stage(1, 1);
   if (nk > 2) stage(2, 2);
+  if (nk > 3) stage(3, 3);
+  // slices 0 (all reads at phase-0) and 1 (interleaved A prefetch) must
+  // land before phase 0's cluster; slices 2,3 stay in flight
+  if (nk >= 4)
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if (nk == 3)
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+  if constexpr (EPI != 0) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) afp1[i] = rd_a(smem, i);
+  }
 
-  for (int s = 0; s < nk; ++s) {
-    // counted drain: slice s's glds (issued 3 phases back) must have
-    // landed; up to 8 younger glds (2 slices) stay in flight across the
-    // barrier.  Tail phases tighten 8 -> 4 -> 0.
-    if (s + 2 < nk)
+  int slot_cur = 0;            // slot of slice s
+  int slot_pf = 4 % G2_NSLOT;  // slot for slice s+4
+
+  // A-set selection must be STATIC (a runtime-selected pointer to a
+  // register array spills to scratch — guide rule 20): 2x hand-unroll.
+  auto phase = [&](int s, bf16x8(&afp_cur)[2], bf16x8(&afp_next)[2]) {
+    const char* ab = smem + slot_cur * G2_SLOT;
+    const char* bb = ab + G2_PART;
+    const char* abn = smem + (slot_cur + 1 == G2_NSLOT ? 0 : slot_cur + 1) * G2_SLOT;
+    if (++slot_cur == G2_NSLOT) slot_cur = 0;
+    // pre-barrier: this slice's B + trailing-A reads and the prefetch
+    // glds — their latency hides under the barrier wait.  (EPI 0's
+    // allocator mishandles the cross-phase afp live ranges — fragments
+    // round-trip through scratch — so it reads ALL its A fragments at
+    // the head instead of software-pipelining the leading pair.)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) bf[j] = rd_b(bb, j);
+#pragma unroll
+    for (int i = 0; i < 6; ++i) aft[i] = rd_a(ab, 2 + i);
+    if constexpr (EPI == 0) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) afp_cur[i] = rd_a(ab, i);
+    }
+    if (s + 4 < nk) {
+      stage(slot_pf, s + 4);
+      if (++slot_pf == G2_NSLOT) slot_pf = 0;
+    }
+    // slice s+1 must be landed for the interleaved A reads below
+    if (s + 3 < nk)
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    else if (s + 1 < nk)
+    else if (s + 2 < nk)
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
-    // compiler fence: neither the fragment reads (slot s) nor the stage
-    // glds (overwriting slot s-1, which other waves were reading until
-    // this barrier) may be hoisted above the barrier
     asm volatile("" ::: "memory");
-
-    const char* ab = smem + (s & 3) * G2_SLOT;
-    const char* bb = ab + G2_PART;
-    bf16x8 bfr[4];
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int r = wc * 64 + j * 16 + l15;
-      bfr[j] = *reinterpret_cast<const bf16x8*>(bb + r * 64 + ((lg * 16) ^ g2_swz(r)));
-    }
-    bf16x8 afr[8];
-#pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      const int r = wr * 128 + i * 16 + l15;
-      afr[i] = *reinterpret_cast<const bf16x8*>(ab + r * 64 + ((lg * 16) ^ g2_swz(r)));
-    }
-    if (s + 3 < nk) stage((s + 3) & 3, s + 3);
-
+    const bool pf = s + 1 < nk;
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < 2; ++i) {
 #pragma unroll
       for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[i], bfr[j],
-                                                            acc[i][j], 0, 0, 0);
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afp_cur[i], bf[j], acc[i][j], 0, 0, 0);
+      if (EPI != 0 && pf) afp_next[i] = rd_a(abn, i);
+    }
+#pragma unroll
+    for (int i = 0; i < 6; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc[2 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            aft[i], bf[j], acc[2 + i][j], 0, 0, 0);
     __builtin_amdgcn_s_setprio(0);
+  };
+
+  if constexpr (EPI == 0) {
+    // no cross-phase register state (afp read at the head): plain loop
+    for (int s = 0; s < nk; ++s) phase(s, afp1, afp2);
+  } else {
+    for (int s = 0; s < nk;) {
+      phase(s, afp1, afp2);
+      if (++s >= nk) break;
+      phase(s, afp2, afp1);
+      ++s;
+    }
   }
+
+  // keep the epilogue's global loads AND address chains BELOW the main
+  // loop: hoisted stat loads / 32 precomputed 64-bit C addresses extend
+  // live ranges across the whole K loop and spill the accumulators (an
+  // asm memory clobber pins the loads; the "+v" below pins the
+  // register-only address arithmetic, which a clobber cannot order)
+  asm volatile("" ::: "memory");
+  int row0 = bm + wr * 128 + lg * 4;
+  int col0 = bn + wc * 64 + l15;
+  asm volatile("" : "+v"(row0), "+v"(col0));
 
   // ------------------------------------------------------------------
   // epilogue.  acc[i][j][r] -> row = bm + wr*128 + i*16 + lg*4 + r,
@@ -157,8 +274,6 @@ __global__ __launch_bounds__(512, 2) void gemm256_kernel(
   // with clamped addresses — a load inside a predicated per-element
   // branch would serialize 32 dependent global round trips (guide §5
   // ".s-level traps" (c)).
-  const int row0 = bm + wr * 128 + lg * 4;
-  const int col0 = bn + wc * 64 + l15;
 
   float sc[8][4];
   if (EPI != 2 && stat_in != nullptr) {
@@ -199,10 +314,13 @@ __global__ __launch_bounds__(512, 2) void gemm256_kernel(
       const int row = row0 + i * 16 + r;
       const bool live = row < M;
       if (EPI == 0) {
+        float outs[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) outs[j] = acc[i][j][r] * sc[i][r];
         if (live) {
 #pragma unroll
           for (int j = 0; j < 4; ++j)
-            C[(long)row * N + col0 + j * 16] = f2bf(acc[i][j][r] * sc[i][r]);
+            C[(long)row * N + col0 + j * 16] = f2bf(outs[j]);
         }
       } else if (EPI == 1) {
         // interleaved (gate, up) columns: even fused col = gate_i,
@@ -258,11 +376,16 @@ extern "C" void launch_gemm256(int epi, void* C, const void* A, const void* B,
                        (const unsigned short*)resid, (const float*)stat_in,
                        (float*)stat_out, M, N, K, stat_mul, stat_eps);
   };
+#ifdef G2_SOLO_EPI0
+  launch(gemm256_kernel<0>);
+  (void)epi;
+#else
   switch (epi) {
     case 0: launch(gemm256_kernel<0>); break;
     case 1: launch(gemm256_kernel<1>); break;
     default: launch(gemm256_kernel<2>); break;
   }
+#endif
 }
 
 // ---------------------------------------------------------------------------

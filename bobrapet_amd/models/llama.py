@@ -45,7 +45,7 @@ CONFIGS: _t.Dict[str, LlamaConfig] = {
         name="llama-tiny",
         vocab_size=1024,
         hidden_size=256,
-        intermediate_size=688,
+        intermediate_size=768,
         num_layers=2,
         num_heads=2,
         num_kv_heads=1,
@@ -80,6 +80,17 @@ CONFIGS: _t.Dict[str, LlamaConfig] = {
 
 
 class LlamaLayerWeights:
+    """Layer weights in the MI355X-native serving layout:
+
+    - [out, in] rows (the GEMM/GEMV weight-streaming layout — no
+      transposed copies, works at 70B);
+    - the RMSNorm per-channel GAINS are FOLDED into w_qkv / w_gate_up
+      (rmsnorm(x)·W^T == rowscale(x) · (W·diag(g))^T — the per-row scale
+      commutes with the GEMM and is applied in the kernel epilogues, so
+      no separate norm kernels run on the hot path);
+    - w_gate_up rows are INTERLEAVED (gate_0, up_0, gate_1, up_1, ...) so
+      the fused-SwiGLU epilogues pair gate/up in adjacent output columns.
+    """
     __slots__ = ("ln_attn", "w_qkv", "w_o", "ln_mlp", "w_gate_up", "w_down")
 
     def __init__(self, cfg: LlamaConfig, device, dtype, gen):
@@ -92,14 +103,27 @@ class LlamaLayerWeights:
             return t
 
         self.ln_attn = torch.ones(H, device=device, dtype=dtype)
-        # all projection weights live in [out,in] layout: the decode GEMV
-        # streams W rows directly (no transposed copies, works at 70B), and
-        # prefill passes .t() views to hipBLASLt (native NT GEMM, no copy)
-        self.w_qkv = mk(cfg.qkv_out, H)
-        self.w_o = mk(H, cfg.num_heads * cfg.head_dim)
         self.ln_mlp = torch.ones(H, device=device, dtype=dtype)
-        self.w_gate_up = mk(2 * I, H)
+        self.w_qkv = fold_gain(mk(cfg.qkv_out, H), self.ln_attn)
+        self.w_o = mk(H, cfg.num_heads * cfg.head_dim)
+        gate_up = mk(2 * I, H)
+        self.w_gate_up = fold_gain(interleave_gate_up(gate_up), self.ln_mlp)
         self.w_down = mk(H, I)
+
+
+def fold_gain(w: torch.Tensor, gain: torch.Tensor) -> torch.Tensor:
+    """W' = W · diag(g): fold a norm's per-channel gain into the weight."""
+    return (w.float() * gain.float()[None, :]).to(w.dtype).contiguous()
+
+
+def interleave_gate_up(w_gate_up: torch.Tensor) -> torch.Tensor:
+    """[gate(I rows); up(I rows)] → rows (g0,u0,g1,u1,...)."""
+    I = w_gate_up.shape[0] // 2
+    return (
+        torch.stack([w_gate_up[:I], w_gate_up[I:]], dim=1)
+        .reshape(2 * I, w_gate_up.shape[1])
+        .contiguous()
+    )
 
 
 class LlamaModel:
@@ -129,9 +153,27 @@ class LlamaModel:
             LlamaLayerWeights(cfg, self.device, dtype, gen) for _ in range(cfg.num_layers)
         ]
         self.ln_final = torch.ones(cfg.hidden_size, device=self.device, dtype=dtype)
-        self.lm_head = torch.empty(
-            cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=dtype
-        ).normal_(0.0, std, generator=gen)
+        self.lm_head = fold_gain(
+            torch.empty(
+                cfg.vocab_size, cfg.hidden_size, device=self.device, dtype=dtype
+            ).normal_(0.0, std, generator=gen),
+            self.ln_final,
+        )
+        self._inv_h = 1.0 / cfg.hidden_size
+        if self.device.type == "cuda":
+            # the fused-epilogue GEMM path needs these alignments; every
+            # shipped config satisfies them (gemm256: N%256, K%32)
+            for n, k in (
+                (cfg.qkv_out, cfg.hidden_size),
+                (cfg.hidden_size, cfg.num_heads * cfg.head_dim),
+                (2 * cfg.intermediate_size, cfg.hidden_size),
+                (cfg.hidden_size, cfg.intermediate_size),
+                (cfg.vocab_size, cfg.hidden_size),
+            ):
+                assert n % 256 == 0 and k % 32 == 0, (
+                    f"config {cfg.name}: projection [{n},{k}] violates the "
+                    "gemm256 alignment (N%256, K%32)"
+                )
         self.scale = 1.0 / math.sqrt(cfg.head_dim)
         self._kv_cache: _t.Optional[_t.List[_t.Tuple[torch.Tensor, torch.Tensor]]] = None
         self._cache_len = 0
@@ -207,17 +249,24 @@ class LlamaModel:
         cos_f = cos_t.repeat(B, 1)
         sin_f = sin_t.repeat(B, 1)
 
-        residual = self.embed[ids.to(self.device)]  # [B,S,H]
-        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
+        # the residual stream x2 [M=B*S, H] and its per-row sumsq `stat`
+        # flow through FUSED-epilogue GEMMs (csrc/hip/gemm256.hip): the
+        # rmsnorm entries become row scales inside the projections (gains
+        # folded into the weights), SwiGLU runs in the gate/up epilogue,
+        # and the residual adds + next-norm statistics come out of the
+        # o/down projection epilogues — no separate norm/activation/add
+        # kernels and no normalized-x round trips through HBM.
+        M = B * S
+        x2 = self.embed[ids.to(self.device)].reshape(M, cfg.hidden_size)
+        stat = ops.rowsumsq(x2)
         if fill_cache:
             self._alloc_cache(B, max(cfg.max_seq_len, S))
 
         for li, lw in enumerate(self.layers):
-            # attention block — zero layout copies: rope reads strided q/k
-            # heads straight from the qkv projection; v stays a strided view
-            qkv = torch.matmul(x, lw.w_qkv.t())  # [B,S,qkv_out]
+            qkv = ops.gemm256_nt(x2, lw.w_qkv, stat, self._inv_h, cfg.rms_eps)
             qh, kh, vh = ops.rope_qkv_split(
-                qkv, B, S, cfg.num_heads, cfg.num_kv_heads, cfg.head_dim, cos_f, sin_f
+                qkv.view(B, S, -1), B, S, cfg.num_heads, cfg.num_kv_heads,
+                cfg.head_dim, cos_f, sin_f,
             )
             if fill_cache:
                 kc, vc = self._kv_cache[li]
@@ -227,40 +276,42 @@ class LlamaModel:
                 attn = ring(qh, kh, vh, self.scale, causal=True)
             else:
                 attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
-            attn = attn.reshape(B, S, cfg.num_heads * cfg.head_dim)
-            attn_out = torch.matmul(attn, lw.w_o.t())
-            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
-
-            # mlp block — silu reads gate/up halves in place
-            gate_up = torch.matmul(x, lw.w_gate_up.t())
-            act = ops.silu_mul_fused(gate_up)
-            mlp_out = torch.matmul(act, lw.w_down.t())
-            next_norm = (
-                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
-            )
-            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
+            a2 = attn.reshape(M, cfg.num_heads * cfg.head_dim)
+            x2, stat = ops.gemm256_resid(a2, lw.w_o, x2)
+            act = ops.gemm256_swiglu(x2, lw.w_gate_up, stat, self._inv_h, cfg.rms_eps)
+            x2, stat = ops.gemm256_resid(act, lw.w_down, x2)
 
         if fill_cache:
             self._cache_len = S
         if logits_for_all:
-            return torch.matmul(x, self.lm_head.t())
-        return torch.matmul(x[:, -1], self.lm_head.t())
+            logits = ops.gemm256_nt(x2, self.lm_head, stat, self._inv_h, cfg.rms_eps)
+            return logits.view(B, S, cfg.vocab_size)
+        xl = x2.view(B, S, cfg.hidden_size)[:, -1].contiguous()
+        sl = stat.view(B, S)[:, -1].contiguous()
+        return ops.gemm256_nt(xl, self.lm_head, sl, self._inv_h, cfg.rms_eps)
 
     # ------------------------------------------------------------------
 
-    def _decode_mm(self, x2d: torch.Tensor, w: torch.Tensor):
-        """Decode projection over [out,in] weights: batch-1 rows go through
-        the hand-written weight-streaming GEMV (≈6-7 TB/s vs ~2.6-6.6 for
-        the library at these skinny shapes — no transposed copies needed,
-        the row-major [out,in] layout IS the streaming layout, so this works
-        at 70B too); larger batches use hipBLASLt (native NT)."""
-        if x2d.shape[0] == 1 and x2d.is_cuda and x2d.shape[1] % 8 == 0:
-            return ops.gemv_nt(x2d, w)
-        return torch.matmul(x2d, w.t())
+    def _ensure_inv_freq(self) -> None:
+        if not hasattr(self, "_inv_freq"):
+            cfg = self.cfg
+            self._inv_freq = 1.0 / (
+                cfg.rope_theta
+                ** (
+                    torch.arange(0, cfg.head_dim, 2, dtype=torch.float32, device=self.device)
+                    / cfg.head_dim
+                )
+            )
 
     def _decode_body(self, ids: torch.Tensor, pos_i32: torch.Tensor) -> torch.Tensor:
         """One decode step with all dynamic state in device tensors — every
-        op here is hipGraph-capturable (no host-dependent shapes/values)."""
+        op here is hipGraph-capturable (no host-dependent shapes/values).
+
+        B<=8 runs entirely on the fused weight-streaming GEMVs
+        (csrc/hip/gemm.hip gemv2): the rmsnorm entries are computed from
+        the x rows the GEMV streams anyway (norm gains folded into the
+        weights), residual adds and SwiGLU live in the GEMV epilogues —
+        6 kernels per layer, no elementwise passes."""
         cfg = self.cfg
         B = ids.shape[0]
         pos_l = pos_i32.to(torch.long)
@@ -269,23 +320,24 @@ class LlamaModel:
         cos_t = torch.cos(ang).expand(B, -1).contiguous()
         sin_t = torch.sin(ang).expand(B, -1).contiguous()
 
-        residual = self.embed[ids].view(B, 1, cfg.hidden_size)
-        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
+        x2 = self.embed[ids].view(B, cfg.hidden_size)  # residual stream
         L_dev = (pos_i32 + 1).contiguous()
-        use_gemv = B == 1 and self.device.type == "cuda"
+        use_gemv = B <= 8 and self.device.type == "cuda"
         for li, lw in enumerate(self.layers):
-            x2 = x.view(B, cfg.hidden_size)
-            qkv = self._decode_mm(x2, lw.w_qkv).view(B, 1, -1) if use_gemv else torch.matmul(x, lw.w_qkv.t())
             kc, vc = self._kv_cache[li]
+            if use_gemv or not x2.is_cuda:
+                qkv = ops.gemv_norm(x2, lw.w_qkv, self._inv_h, cfg.rms_eps)
+            else:
+                qkv = self._qkv_torch(x2, lw)
             if use_gemv:
                 # fused rope + cache append: one kernel instead of rope +
                 # two index_copys + two layout copies (x32 layers/step)
                 qf = ops.rope_qkv_decode(
-                    qkv.view(B, -1), kc, vc, cos_t, sin_t, pos_i32,
+                    qkv, kc, vc, cos_t, sin_t, pos_i32,
                     cfg.num_heads, cfg.num_kv_heads, cfg.head_dim,
                 )
             else:
-                q, k, v = self._split_qkv(qkv, B, 1)
+                q, k, v = self._split_qkv(qkv.view(B, 1, -1), B, 1)
                 qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
                 kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
                 qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
@@ -294,31 +346,31 @@ class LlamaModel:
                 vc.index_copy_(2, pos_l, v.reshape(B, cfg.num_kv_heads, 1, cfg.head_dim).contiguous())
             attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
             a2 = attn.reshape(B, cfg.num_heads * cfg.head_dim)
-            attn_out = (
-                self._decode_mm(a2, lw.w_o).view(B, 1, -1)
-                if use_gemv
-                else torch.matmul(attn.reshape(B, 1, -1), lw.w_o.t())
-            )
-            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
-            x2 = x.view(B, cfg.hidden_size)
-            gate_up = (
-                self._decode_mm(x2, lw.w_gate_up).view(B, 1, -1)
-                if use_gemv
-                else torch.matmul(x, lw.w_gate_up.t())
-            )
-            act = ops.silu_mul_fused(gate_up)
-            mlp_out = (
-                self._decode_mm(act.view(B, -1), lw.w_down).view(B, 1, -1)
-                if use_gemv
-                else torch.matmul(act, lw.w_down.t())
-            )
-            next_norm = (
-                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
-            )
-            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
-        if use_gemv:
-            return ops.gemv_nt(x.view(B, cfg.hidden_size), self.lm_head)
-        return torch.matmul(x[:, -1], self.lm_head.t())
+            if use_gemv or not x2.is_cuda:
+                x2 = ops.gemv_resid(a2, lw.w_o, x2)
+                act = ops.gemv_swiglu_norm(x2, lw.w_gate_up, self._inv_h, cfg.rms_eps)
+                x2 = ops.gemv_resid(act, lw.w_down, x2)
+            else:
+                x2 = x2 + torch.matmul(a2, lw.w_o.t())
+                act = self._swiglu_torch(x2, lw)
+                x2 = x2 + torch.matmul(act, lw.w_down.t())
+        if use_gemv or not x2.is_cuda:
+            return ops.gemv_norm(x2, self.lm_head, self._inv_h, cfg.rms_eps)
+        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
+        return ((x2.float() * s) @ self.lm_head.float().t()).to(x2.dtype)
+
+    def _qkv_torch(self, x2, lw):
+        """Large-batch decode fallback (eager torch over folded weights)."""
+        cfg = self.cfg
+        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
+        return ((x2.float() * s) @ lw.w_qkv.float().t()).to(x2.dtype)
+
+    def _swiglu_torch(self, x2, lw):
+        cfg = self.cfg
+        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
+        gu = (x2.float() * s) @ lw.w_gate_up.float().t()
+        g, u = gu[:, 0::2], gu[:, 1::2]
+        return (g * torch.sigmoid(g) * u).to(x2.dtype)
 
     @torch.no_grad()
     def decode_step_graphed(self, ids: torch.Tensor) -> torch.Tensor:
@@ -330,12 +382,7 @@ class LlamaModel:
         process-wide cached model serves many sequential stories safely."""
         assert self._kv_cache is not None, "call prefill(fill_cache=True) first"
         B = ids.shape[0]
-        if not hasattr(self, "_inv_freq"):
-            cfg = self.cfg
-            self._inv_freq = 1.0 / (
-                cfg.rope_theta
-                ** (torch.arange(0, cfg.head_dim, 2, dtype=torch.float32, device=self.device) / cfg.head_dim)
-            )
+        self._ensure_inv_freq()
         if not hasattr(self, "_graphs"):
             self._graphs = {}
         entry = self._graphs.get(B)
@@ -399,38 +446,14 @@ class LlamaModel:
     @torch.no_grad()
     def decode_step(self, ids: torch.Tensor) -> torch.Tensor:
         """One token per sequence: ids [B] → logits [B, V]; uses the cache
-        filled by prefill(fill_cache=True)."""
-        cfg = self.cfg
+        filled by prefill(fill_cache=True).  Eager (ungraphed) path —
+        same body as the graphed one."""
         assert self._kv_cache is not None, "call prefill(fill_cache=True) first"
-        B = ids.shape[0]
-        pos = self._cache_len
-        positions = torch.full((B,), pos, device=self.device, dtype=torch.long)
-        cos_t, sin_t = ops.rope_tables(positions, cfg.head_dim, cfg.rope_theta)
-
-        residual = self.embed[ids.to(self.device)].view(B, 1, cfg.hidden_size)
-        x = ops.rmsnorm(residual, self.layers[0].ln_attn, cfg.rms_eps)
-        for li, lw in enumerate(self.layers):
-            qkv = torch.matmul(x, lw.w_qkv.t())
-            q, k, v = self._split_qkv(qkv, B, 1)
-            qf = q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous()
-            kf = k.reshape(B, cfg.num_kv_heads, cfg.head_dim).contiguous()
-            qf, kf = ops.rope_inplace(qf, kf, cos_t, sin_t)
-            kc, vc = self._kv_cache[li]
-            kc[:, :, pos] = kf.view(B, cfg.num_kv_heads, cfg.head_dim)
-            vc[:, :, pos] = v.view(B, cfg.num_kv_heads, cfg.head_dim)
-            attn = ops.attn_decode(qf, kc, vc, pos + 1, self.scale)  # [B,Hq,D]
-            attn_out = torch.matmul(attn.reshape(B, 1, -1), lw.w_o.t())
-            x, residual = ops.fused_add_rmsnorm(attn_out, residual, lw.ln_mlp, cfg.rms_eps)
-            gate_up = torch.matmul(x, lw.w_gate_up.t())
-            gate, up = gate_up.chunk(2, dim=-1)
-            act = ops.silu_mul(gate.contiguous(), up.contiguous())
-            mlp_out = torch.matmul(act, lw.w_down.t())
-            next_norm = (
-                self.layers[li + 1].ln_attn if li + 1 < cfg.num_layers else self.ln_final
-            )
-            x, residual = ops.fused_add_rmsnorm(mlp_out, residual, next_norm, cfg.rms_eps)
-        self._cache_len = pos + 1
-        return torch.matmul(x[:, -1], self.lm_head.t())
+        self._ensure_inv_freq()
+        pos_i32 = torch.tensor([self._cache_len], dtype=torch.int32, device=self.device)
+        logits = self._decode_body(ids.to(self.device), pos_i32)
+        self._cache_len += 1
+        return logits
 
     @torch.no_grad()
     def generate(

@@ -346,6 +346,34 @@ def rowsumsq(x: torch.Tensor) -> torch.Tensor:
     return x.float().reshape(-1, k).pow(2).sum(dim=-1)
 
 
+def gemv_norm(a: torch.Tensor, w: torch.Tensor, mul: float, eps: float) -> torch.Tensor:
+    """Decode GEMV with the rmsnorm entry folded in: the kernel computes
+    s[m] = rsqrt(sumsq(a_m)*mul + eps) from the x rows it streams anyway
+    and scales the dot products — no rmsnorm kernel, no stat buffer."""
+    if a.is_cuda:
+        return _require_ext().gemv_norm(a.contiguous(), w.contiguous(), mul, eps)
+    s = torch.rsqrt(a.float().pow(2).sum(-1) * mul + eps)
+    return ((a.float() @ w.float().t()) * s[:, None]).to(a.dtype)
+
+
+def gemv_resid(a: torch.Tensor, w: torch.Tensor, resid: torch.Tensor) -> torch.Tensor:
+    """out = a @ w.T + resid (decode residual add fused into the GEMV)."""
+    if a.is_cuda:
+        return _require_ext().gemv_resid(a.contiguous(), w.contiguous(), resid.contiguous())
+    return (a.float() @ w.float().t() + resid.float()).to(a.dtype)
+
+
+def gemv_swiglu_norm(a: torch.Tensor, w_interleaved: torch.Tensor, mul: float, eps: float) -> torch.Tensor:
+    """Decode gate/up projection with interleaved rows + fused SwiGLU and
+    norm entry: out[M, rows/2] = silu(s·g)·(s·u)."""
+    if a.is_cuda:
+        return _require_ext().gemv_swiglu_norm(a.contiguous(), w_interleaved.contiguous(), mul, eps)
+    s = torch.rsqrt(a.float().pow(2).sum(-1) * mul + eps)
+    c = (a.float() @ w_interleaved.float().t()) * s[:, None]
+    g, u = c[:, 0::2], c[:, 1::2]
+    return (g * torch.sigmoid(g) * u).to(a.dtype)
+
+
 def rope_qkv_decode(qkv2d, kc, vc, cos_t, sin_t, pos_dev: torch.Tensor, Hq: int, Hkv: int, D: int):
     """Fused decode head prep: rope q (returned [B,Hq,D]) and rope k +
     copy v straight into the KV cache at DEVICE position pos (hipGraph-

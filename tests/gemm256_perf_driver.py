@@ -4,8 +4,11 @@ shapes.  Run on a GPU box:
 Prints per-shape ms + TF/s for (a) torch.matmul NT, (b) gemm256 plain,
 (c) the fused-epilogue variant the model would use.
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

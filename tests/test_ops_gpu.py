@@ -187,7 +187,8 @@ class TestGemm256:
         c = torch.matmul(a.float(), b.float().t())
         g, u = c[:, 0::2], c[:, 1::2]
         ref = g * torch.sigmoid(g) * u
-        assert (got.float() - ref).abs().max().item() < 0.05
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
 
     def test_resid_epilogue_and_stat(self):
         m, n, k = 300, 512, 768

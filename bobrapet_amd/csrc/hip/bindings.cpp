@@ -42,6 +42,9 @@ void launch_gemm256(int, void*, const void*, const void*, const void*,
                     hipStream_t);
 void launch_gemv2(int, void*, const void*, const void*, const void*, int, int,
                   int, float, float, hipStream_t);
+void launch_gemm256b(int, void*, const void*, const void*, const void*,
+                     const void*, void*, int, int, int, float, float,
+                     hipStream_t);
 void launch_rowsumsq(void*, const void*, int, int, hipStream_t);
 void launch_dbg_attn_core(void*, void*, const void*, const void*, const void*,
                           hipStream_t);
@@ -319,6 +322,18 @@ void check_gemm256(const torch::Tensor& a, const torch::Tensor& b, int M,
               "gemm256: N must be a multiple of 256 and K of 32");
 }
 
+void launch_gemm256_auto(int epi, void* C, const void* A, const void* B,
+                         const void* resid, const void* stat_in,
+                         void* stat_out, int M, int N, int K, float stat_mul,
+                         float stat_eps, hipStream_t st) {
+  if (K % 64 == 0)
+    launch_gemm256b(epi, C, A, B, resid, stat_in, stat_out, M, N, K, stat_mul,
+                    stat_eps, st);
+  else
+    launch_gemm256(epi, C, A, B, resid, stat_in, stat_out, M, N, K, stat_mul,
+                   stat_eps, st);
+}
+
 const float* stat_ptr(const c10::optional<torch::Tensor>& stat, int M) {
   if (!stat.has_value()) return nullptr;
   TORCH_CHECK(stat->scalar_type() == torch::kFloat32 && stat->is_contiguous() &&
@@ -333,7 +348,7 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor b,
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   check_gemm256(a, b, M, N, K);
   auto c = torch::empty({M, N}, a.options());
-  launch_gemm256(0, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
+  launch_gemm256_auto(0, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
                  stat_ptr(stat, M), nullptr, M, N, K, (float)stat_mul,
                  (float)stat_eps, cur_stream());
   return c;
@@ -346,7 +361,7 @@ torch::Tensor gemm256_swiglu(torch::Tensor a, torch::Tensor b,
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   check_gemm256(a, b, M, N, K);
   auto c = torch::empty({M, N / 2}, a.options());
-  launch_gemm256(1, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
+  launch_gemm256_auto(1, c.data_ptr(), a.data_ptr(), b.data_ptr(), nullptr,
                  stat_ptr(stat, M), nullptr, M, N, K, (float)stat_mul,
                  (float)stat_eps, cur_stream());
   return c;
@@ -365,7 +380,7 @@ std::vector<torch::Tensor> gemm256_resid(torch::Tensor a, torch::Tensor b,
     stat = torch::empty({M}, a.options().dtype(torch::kFloat32));
     stat_p = stat.data_ptr();
   }
-  launch_gemm256(2, c.data_ptr(), a.data_ptr(), b.data_ptr(), resid.data_ptr(),
+  launch_gemm256_auto(2, c.data_ptr(), a.data_ptr(), b.data_ptr(), resid.data_ptr(),
                  nullptr, stat_p, M, N, K, 0.0f, 0.0f, cur_stream());
   if (want_stat) return {c, stat};
   return {c};

@@ -1,0 +1,398 @@
+// 256x256-tile bf16 MFMA GEMM, 8-phase quadrant schedule (gfx950).
+//
+// Faithful implementation of the measured-fastest plain-HIP GEMM
+// structure for CDNA4 (guide §5 "256² 8-phase template"): per K-tile of
+// 64, four QUADRANT phases of 16 MFMAs each; every phase issues the
+// NEXT quadrant's 4 A fragment reads (plus, once per tile, the next
+// tile's 8 B reads) and one half-tile glds stage BEFORE the barrier
+// pair, so LDS-read latency and glds landing hide under the barrier
+// wait and the matrix pipe runs nearly back-to-back.
+//
+// Staging runs through a 10-slot half-tile ring (4 rings: A-half0 x3,
+// A-half1 x3, B-half0 x2, B-half1 x2 = all 160 KB of LDS) with counted
+// s_waitcnt vmcnt(6) per phase — 3 half-tiles always in flight across
+// the barriers, never drained in the steady loop.  Stage schedule (at
+// tile t, quadrant q): q0 -> B1(t+1), q1 -> A0(t+2), q2 -> A1(t+2),
+// q3 -> B0(t+2); every half lands >= 4 phases before its first read,
+// which is exactly what vmcnt(6) (+ the barrier) guarantees.
+//
+// Same epilogues as gemm256.hip (row-scale / SwiGLU / residual+sumsq).
+// Requires K % 64 == 0 (the dispatcher falls back to gemm256 otherwise).
+#include "common.h"
+
+#define G3_NRING 10
+#define G3_HALF 16384                 // one half-tile operand slot: 16 KB
+// slot layout inside a half slot: 2 ks-blocks of [128 rows][32 elems]
+// (64-byte rows -> the proven conflict-free swizzle of gemm256.hip)
+
+typedef float f32x4g3 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ int g3_swz(int row) { return ((row >> 3) & 1) * 48; }
+
+__device__ __forceinline__ void g3_glds(const unsigned short* src, char* lds_dst) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)src,
+      (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
+}
+
+__device__ __forceinline__ float g3_silu(float g) {
+  return g / (1.0f + __expf(-g));
+}
+
+template <int EPI>
+__global__ __launch_bounds__(512, 2) void gemm256b_kernel(
+    unsigned short* __restrict__ C,
+    const unsigned short* __restrict__ A,
+    const unsigned short* __restrict__ B,
+    const unsigned short* __restrict__ resid,  // EPI 2
+    const float* __restrict__ stat_in,         // EPI 0/1 optional row stat
+    float* __restrict__ stat_out,              // EPI 2 optional
+    int M, int N, int K, float stat_mul, float stat_eps) {
+  __shared__ __attribute__((aligned(16))) char smem[G3_NRING * G3_HALF];
+  // ring bases: A0 slots 0..2, A1 slots 3..5, B0 slots 6..7, B1 slots 8..9
+
+  const int nbn = N / 256;
+  const int nbm = (M + 255) / 256;
+  unsigned int wgid = xcd_swizzle(blockIdx.x, nbm * nbn);
+  const int GN = 16;  // grouped supertile order (see gemm256.hip)
+  int bm_i, bn_i;
+  {
+    const int full = nbn / GN;
+    const int per = nbm * GN;
+    if ((int)wgid < full * per) {
+      const int grp = (int)wgid / per, rem = (int)wgid % per;
+      bm_i = rem / GN;
+      bn_i = grp * GN + rem % GN;
+    } else {
+      const int tail_n = nbn - full * GN;
+      const int rem = (int)wgid - full * per;
+      bm_i = rem / tail_n;
+      bn_i = full * GN + rem % tail_n;
+    }
+  }
+  const int bm = bm_i * 256;
+  const int bn = bn_i * 256;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;  // 8 waves: 2 (M) x 4 (N)
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+  const int l15 = lane & 15;
+  const int lg = lane >> 4;
+
+  // ---- staging.  A half-tile = 128 rows x 64 k, stored as 2 ks-blocks
+  // of [128][32]; piece p (1 KB, p = 0..15) = ks-block p>>3, sub-rows
+  // (p&7)*16..+15; wave w stages pieces {2w, 2w+1} (same ks, rows +16).
+  const int p0 = wid * 2;
+  const int st_ks = p0 >> 3;                       // this wave's ks block
+  const int st_row = (p0 & 7) * 16 + (lane >> 2);  // rows of piece p0
+  const int st_kbx = (((p0 * 64 + lane) & 3) * 16) ^ g3_swz(st_row);
+  // base source pointers for the four half types (k advances per tile)
+  int ar0 = bm + st_row, ar1 = bm + 128 + st_row;
+  if (ar0 >= M) ar0 = M - 1;
+  if (ar1 >= M) ar1 = M - 1;
+  int ar0b = bm + st_row + 16, ar1b = bm + 128 + st_row + 16;
+  if (ar0b >= M) ar0b = M - 1;
+  if (ar1b >= M) ar1b = M - 1;
+  const unsigned short* a0src = A + (long)ar0 * K + st_ks * 32 + st_kbx / 2;
+  const unsigned short* a1src = A + (long)ar1 * K + st_ks * 32 + st_kbx / 2;
+  const unsigned short* b0src =
+      B + (long)(bn + st_row) * K + st_ks * 32 + st_kbx / 2;
+  const int a0d = (ar0b - ar0) * K;  // element delta to the second piece
+  const int a1d = (ar1b - ar1) * K;
+  const int b0d = 16 * K;
+
+  // half type: 0=A0 1=A1 2=B0 3=B1; ring slot index passed explicitly
+  auto stage = [&](int half, int slot, int kt) {
+    char* base = smem + slot * G3_HALF + p0 * 1024;
+    const int k0 = kt * 64;
+    const unsigned short* s0;
+    int d;
+    if (half == 0) {
+      s0 = a0src + k0;
+      d = a0d;
+    } else if (half == 1) {
+      s0 = a1src + k0;
+      d = a1d;
+    } else if (half == 2) {
+      s0 = b0src + k0;
+      d = b0d;
+    } else {
+      s0 = b0src + 128L * K + k0;
+      d = b0d;
+    }
+    g3_glds(s0, base);
+    g3_glds(s0 + d, base + 1024);
+  };
+
+  f32x4g3 acc[8][4] = {};
+  const int nt = K / 64;
+
+  // fragment-read offsets within a half slot (ks-block + row + swizzle)
+  const int a_off = l15 * 64 + ((lg * 16) ^ g3_swz(l15));
+  const int b_row = (wc & 1) * 64 + l15;
+  const int b_off = b_row * 64 + ((lg * 16) ^ g3_swz(b_row));
+  auto rd_a = [&](const char* slotb, int fi, int ks) {
+    return *reinterpret_cast<const bf16x8*>(slotb + ks * 8192 + a_off +
+                                            fi * 16 * 64);
+  };
+  auto rd_b = [&](const char* slotb, int j, int ks) {
+    return *reinterpret_cast<const bf16x8*>(slotb + ks * 8192 + b_off +
+                                            j * 16 * 64);
+  };
+  // this wave's A ring (its own M-half) and B ring (its own N-half)
+  char* aring = smem + (wr ? 3 : 0) * G3_HALF;
+  char* bring = smem + ((wc >> 1) ? 8 : 6) * G3_HALF;
+
+  bf16x8 bfr[4][2];            // B frags of the current tile (4 j x 2 ks)
+  bf16x8 aq1[2][2], aq2[2][2]; // alternating quadrant sets (2 i x 2 ks)
+
+  // ---- prologue: stage the 7 halves the steady state has in flight,
+  // in stage order [A0(0) A1(0) B0(0) B1(0) A0(1) A1(1) B0(1)]
+  stage(0, 0, 0);
+  stage(1, 3, 0);
+  stage(2, 6, 0);
+  stage(3, 8, 0);
+  if (nt > 1) {
+    stage(0, 1, 1);
+    stage(1, 4, 1);
+    stage(2, 7, 1);
+  }
+  // read quadrant 0 of tile 0 (needs A(0) = the two oldest halves)
+  if (nt > 1)
+    asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  asm volatile("" ::: "memory");
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aring, i, ks);
+
+  // ring cursors (tile -> slot): A rings advance mod 3, B rings mod 2
+  // a_rd = slot of tile t in the A ring; staged slots tracked per stage
+  for (int t = 0; t < nt; ++t) {
+    const int a_rd = t % 3;            // A slot of tile t
+    const int a_rd1 = (t + 1) % 3;     // A slot of tile t+1
+    const int b_rd = t & 1;            // B slot of tile t
+    const char* aslot = aring + a_rd * G3_HALF;
+    const char* aslot1 = aring + a_rd1 * G3_HALF;
+    const char* bslot = bring + b_rd * G3_HALF;
+    const bool tail = t + 2 >= nt;  // last two tiles: drain instead
+
+    // ---- phase q=0: read B(t) + A quad 1; stage B1(t+1)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) bfr[j][ks] = rd_b(bslot, j, ks);
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) aq2[i][ks] = rd_a(aslot, 2 + i, ks);
+    if (t + 1 < nt) stage(3, 8 + ((t + 1) & 1), t + 1);
+    if (!tail)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aq1[i][ks], bfr[j][ks], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase q=1: read A quad 2; stage A0(t+2)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aslot, 4 + i, ks);
+    if (t + 2 < nt) stage(0, (t + 2) % 3, t + 2);
+    if (!tail)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[2 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aq2[i][ks], bfr[j][ks], acc[2 + i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase q=2: read A quad 3; stage A1(t+2)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) aq2[i][ks] = rd_a(aslot, 6 + i, ks);
+    if (t + 2 < nt) stage(1, 3 + (t + 2) % 3, t + 2);
+    if (!tail)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aq1[i][ks], bfr[j][ks], acc[4 + i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+
+    // ---- phase q=3: read A quad 0 of tile t+1; stage B0(t+2)
+    if (t + 1 < nt) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aslot1, i, ks);
+    }
+    if (t + 2 < nt) stage(2, 6 + ((t + 2) & 1), t + 2);
+    if (!tail)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[6 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aq2[i][ks], bfr[j][ks], acc[6 + i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // keep the epilogue's loads and address chains below the loop
+  asm volatile("" ::: "memory");
+  int row0 = bm + wr * 128 + lg * 4;
+  int col0 = bn + wc * 64 + l15;
+  asm volatile("" : "+v"(row0), "+v"(col0));
+
+  // ------------------------------------------------------------------
+  // epilogue (identical layout/semantics to gemm256.hip)
+  float sc[8][4];
+  if (EPI != 2 && stat_in != nullptr) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + i * 16 + r;
+        sc[i][r] = stat_in[row < M ? row : M - 1];
+      }
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        sc[i][r] = rsqrtf(sc[i][r] * stat_mul + stat_eps);
+  } else {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) sc[i][r] = 1.0f;
+  }
+
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    float rsv[4][4];
+    if (EPI == 2) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = row0 + i * 16 + r;
+        const long rb = (long)(row < M ? row : M - 1) * N;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) rsv[r][j] = bf2f(resid[rb + col0 + j * 16]);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int row = row0 + i * 16 + r;
+      const bool live = row < M;
+      if (EPI == 0) {
+        float outs[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) outs[j] = acc[i][j][r] * sc[i][r];
+        if (live) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            C[(long)row * N + col0 + j * 16] = f2bf(outs[j]);
+        }
+      } else if (EPI == 1) {
+        float outs[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const float v = acc[i][j][r] * sc[i][r];
+          const float partner = __shfl_xor(v, 1, WAVE);
+          outs[j] = g3_silu(v) * partner;
+        }
+        if (live && (l15 & 1) == 0) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            C[(long)row * (N / 2) + ((col0 + j * 16) >> 1)] = f2bf(outs[j]);
+        }
+      } else {
+        float ss = 0.0f;
+        float outs[4];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          outs[j] = acc[i][j][r] + rsv[r][j];
+          ss += outs[j] * outs[j];
+        }
+        if (live) {
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            C[(long)row * N + col0 + j * 16] = f2bf(outs[j]);
+        }
+        if (stat_out != nullptr) {
+#pragma unroll
+          for (int off = 8; off > 0; off >>= 1) ss += __shfl_xor(ss, off, WAVE);
+          if (l15 == 0 && live) atomicAdd(stat_out + row, ss);
+        }
+      }
+    }
+  }
+}
+
+extern "C" void launch_gemm256b(int epi, void* C, const void* A, const void* B,
+                                const void* resid, const void* stat_in,
+                                void* stat_out, int M, int N, int K,
+                                float stat_mul, float stat_eps,
+                                hipStream_t stream) {
+  const int nbm = (M + 255) / 256;
+  dim3 grid(nbm * (N / 256)), block(512);
+  if (stat_out != nullptr)
+    hipMemsetAsync(stat_out, 0, (size_t)M * sizeof(float), stream);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
+                       (const unsigned short*)A, (const unsigned short*)B,
+                       (const unsigned short*)resid, (const float*)stat_in,
+                       (float*)stat_out, M, N, K, stat_mul, stat_eps);
+  };
+  switch (epi) {
+    case 0: launch(gemm256b_kernel<0>); break;
+    case 1: launch(gemm256b_kernel<1>); break;
+    default: launch(gemm256b_kernel<2>); break;
+  }
+}

@@ -209,6 +209,83 @@ class TestGemm256:
         assert ((got - ref).abs() / ref.clamp(min=1.0)).max().item() < 1e-3
 
 
+class TestGemv2:
+    """Fused decode GEMV variants vs fp32 torch references."""
+
+    @pytest.mark.parametrize("m", [1, 4, 8])
+    def test_norm_entry(self, m):
+        k, n = 4096, 1024
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        got = ops.gemv_norm(a, w, 1.0 / k, 1e-5)
+        s = torch.rsqrt(a.float().pow(2).sum(-1) / k + 1e-5)
+        ref = (a.float() @ w.float().t()) * s[:, None]
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
+
+    def test_resid(self):
+        m, k, n = 2, 2048, 512
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda")
+        w = torch.randn(n, k, dtype=torch.bfloat16, device="cuda")
+        r = torch.randn(m, n, dtype=torch.bfloat16, device="cuda")
+        got = ops.gemv_resid(a, w, r)
+        ref = a.float() @ w.float().t() + r.float()
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
+
+    def test_swiglu_norm(self):
+        m, k, n2 = 1, 1024, 2048  # 1024 output cols
+        a = torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.5
+        w = torch.randn(n2, k, dtype=torch.bfloat16, device="cuda") * 0.5
+        got = ops.gemv_swiglu_norm(a, w, 1.0 / k, 1e-5)
+        s = torch.rsqrt(a.float().pow(2).sum(-1) / k + 1e-5)
+        c = (a.float() @ w.float().t()) * s[:, None]
+        g, u = c[:, 0::2], c[:, 1::2]
+        ref = g * torch.sigmoid(g) * u
+        rel = (got.float() - ref).abs().max().item() / max(ref.abs().max().item(), 1.0)
+        assert rel < 0.02, rel
+
+
+class TestModel8BLayer:
+    """GPU llm-8B numerics spot check: one full layer at
+    the REAL 8B shapes, fixed seed, fused path vs the fp32 composite
+    reference — catches kernel regressions at production shapes."""
+
+    def test_one_8b_layer_prefill_math(self):
+        torch.manual_seed(11)
+        H, I, Hq, Hkv, D = 4096, 14336, 32, 8, 128
+        M = 512  # rows (B*S) — real K/N, modest M to keep it fast
+        eps = 1e-5
+        x2 = (torch.randn(M, H, device="cuda") * 0.5).bfloat16()
+        w_qkv = (torch.randn((Hq + 2 * Hkv) * D, H, device="cuda") * 0.02).bfloat16()
+        w_o = (torch.randn(H, Hq * D, device="cuda") * 0.02).bfloat16()
+        w_gu = (torch.randn(2 * I, H, device="cuda") * 0.02).bfloat16()
+        w_dn = (torch.randn(H, I, device="cuda") * 0.02).bfloat16()
+
+        stat = ops.rowsumsq(x2)
+        qkv = ops.gemm256_nt(x2, w_qkv, stat, 1.0 / H, eps)
+        a2 = qkv[:, : Hq * D].contiguous()  # stand-in for attention output
+        x2b, stat2 = ops.gemm256_resid(a2, w_o, x2)
+        act = ops.gemm256_swiglu(x2b, w_gu, stat2, 1.0 / H, eps)
+        x2c, _ = ops.gemm256_resid(act, w_dn, x2b)
+
+        # fp32 composite reference
+        xf = x2.float()
+        s = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+        qkv_ref = (xf * s) @ w_qkv.float().t()
+        a2_ref = qkv_ref[:, : Hq * D]
+        xb_ref = xf + a2_ref @ w_o.float().t()
+        s2 = torch.rsqrt(xb_ref.pow(2).mean(-1, keepdim=True) + eps)
+        gu = (xb_ref * s2) @ w_gu.float().t()
+        g, u = gu[:, 0::2], gu[:, 1::2]
+        xc_ref = xb_ref + (g * torch.sigmoid(g) * u) @ w_dn.float().t()
+
+        for got, ref, name in ((qkv, qkv_ref, "qkv"), (x2b, xb_ref, "o+resid"), (x2c, xc_ref, "down+resid")):
+            scale = ref.abs().max().item()
+            rel = (got.float() - ref).abs().max().item() / max(scale, 1.0)
+            assert rel < 0.03, (name, rel)
+
+
 class TestAttentionDecode:
     @pytest.mark.parametrize("b,hq,hkv,l", [(1, 1, 1, 64), (2, 32, 8, 500), (4, 8, 8, 1024)])
     def test_vs_ref(self, b, hq, hkv, l):

@@ -145,8 +145,8 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
   char* aring = smem + (wr ? 3 : 0) * G3_HALF;
   char* bring = smem + ((wc >> 1) ? 8 : 6) * G3_HALF;
 
-  bf16x8 bfr[4][2];            // B frags of the current tile (4 j x 2 ks)
-  bf16x8 aq1[2][2], aq2[2][2]; // alternating quadrant sets (2 i x 2 ks)
+  bf16x8 bfr[4][2];  // B frags of the current tile (4 j x 2 ks)
+  bf16x8 aq[2][2];   // current quadrant's A frags (2 i x 2 ks)
 
   // ---- prologue: stage the 7 halves the steady state has in flight,
   // in stage order [A0(0) A1(0) B0(0) B1(0) A0(1) A1(1) B0(1)]
@@ -159,30 +159,43 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     stage(1, 4, 1);
     stage(2, 7, 1);
   }
-  // read quadrant 0 of tile 0 (needs A(0) = the two oldest halves)
+  // tile 0's 4 halves must be landed before its q0 reads
   if (nt > 1)
-    asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
   asm volatile("" ::: "memory");
-#pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aring, i, ks);
 
-  // ring cursors (tile -> slot): A rings advance mod 3, B rings mod 2
-  // a_rd = slot of tile t in the A ring; staged slots tracked per stage
+  // per phase q: read quadrant q's fragments (+ the tile's B at q0),
+  // issue one half-tile stage, then barrier -> the read latency hides
+  // under the barrier wait and hipcc's counted lgkm before the MFMAs;
+  // ONE vmcnt(6) per tile (at q3) guarantees the NEXT tile's halves,
+  // propagated to every wave by q3's barriers (template: "vmcnt at
+  // phases 4 and 8 only, never 0 in the main loop").
+#define G3_MFMA_QUAD(base)                                              \
+    __builtin_amdgcn_s_setprio(1);                                      \
+    _Pragma("unroll")                                                   \
+    for (int ks = 0; ks < 2; ++ks)                                      \
+      _Pragma("unroll")                                                 \
+      for (int i = 0; i < 2; ++i)                                       \
+        _Pragma("unroll")                                               \
+        for (int j = 0; j < 4; ++j)                                     \
+          acc[(base) + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16( \
+              aq[i][ks], bfr[j][ks], acc[(base) + i][j], 0, 0, 0);      \
+    __builtin_amdgcn_s_setprio(0);
+  // NOTE: no barrier after the MFMA cluster — the next phase's reads may
+  // run while the SIMD partner is still in this cluster (complementary
+  // matrix-beside-memory pairing).  Safe: per-phase barrier-1 bounds the
+  // skew to <1 phase and every staged slot's last read is >=3 phases
+  // before its overwrite.
+
   for (int t = 0; t < nt; ++t) {
-    const int a_rd = t % 3;            // A slot of tile t
-    const int a_rd1 = (t + 1) % 3;     // A slot of tile t+1
-    const int b_rd = t & 1;            // B slot of tile t
-    const char* aslot = aring + a_rd * G3_HALF;
-    const char* aslot1 = aring + a_rd1 * G3_HALF;
-    const char* bslot = bring + b_rd * G3_HALF;
-    const bool tail = t + 2 >= nt;  // last two tiles: drain instead
+    const char* aslot = aring + (t % 3) * G3_HALF;
+    const char* bslot = bring + (t & 1) * G3_HALF;
+    const bool tail = t + 2 >= nt;
 
-    // ---- phase q=0: read B(t) + A quad 1; stage B1(t+1)
+    // ---- phase q=0: read B(t) + A quad 0; stage B1(t+1)
 #pragma unroll
     for (int j = 0; j < 4; ++j)
 #pragma unroll
@@ -190,81 +203,38 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) aq2[i][ks] = rd_a(aslot, 2 + i, ks);
+      for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, i, ks);
     if (t + 1 < nt) stage(3, 8 + ((t + 1) & 1), t + 1);
-    if (!tail)
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");  // pace: first 4 reads
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aq1[i][ks], bfr[j][ks], acc[i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
+    G3_MFMA_QUAD(0)
 
-    // ---- phase q=1: read A quad 2; stage A0(t+2)
+    // ---- phase q=1: read A quad 1; stage A0(t+2)
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aslot, 4 + i, ks);
+      for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 2 + i, ks);
     if (t + 2 < nt) stage(0, (t + 2) % 3, t + 2);
-    if (!tail)
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[2 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aq2[i][ks], bfr[j][ks], acc[2 + i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
+    G3_MFMA_QUAD(2)
 
-    // ---- phase q=2: read A quad 3; stage A1(t+2)
+    // ---- phase q=2: read A quad 2; stage A1(t+2)
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) aq2[i][ks] = rd_a(aslot, 6 + i, ks);
+      for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 4 + i, ks);
     if (t + 2 < nt) stage(1, 3 + (t + 2) % 3, t + 2);
-    if (!tail)
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-    else
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[4 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aq1[i][ks], bfr[j][ks], acc[4 + i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
+    G3_MFMA_QUAD(4)
 
-    // ---- phase q=3: read A quad 0 of tile t+1; stage B0(t+2)
-    if (t + 1 < nt) {
+    // ---- phase q=3: read A quad 3; stage B0(t+2); tile-boundary vmcnt
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
+    for (int i = 0; i < 2; ++i)
 #pragma unroll
-        for (int ks = 0; ks < 2; ++ks) aq1[i][ks] = rd_a(aslot1, i, ks);
-    }
+      for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 6 + i, ks);
     if (t + 2 < nt) stage(2, 6 + ((t + 2) & 1), t + 2);
     if (!tail)
       asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
@@ -272,18 +242,9 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
-    __builtin_amdgcn_s_setprio(1);
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)   // ks outer: consecutive MFMAs independent
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[6 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              aq2[i][ks], bfr[j][ks], acc[6 + i][j], 0, 0, 0);
-    __builtin_amdgcn_s_setprio(0);
-    __builtin_amdgcn_s_barrier();
+    G3_MFMA_QUAD(6)
   }
+#undef G3_MFMA_QUAD
 
   // keep the epilogue's loads and address chains below the loop
   asm volatile("" ::: "memory");

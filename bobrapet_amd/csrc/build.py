@@ -80,12 +80,17 @@ def build(force: bool = False, verbose: bool = True) -> str:
         if force or _newer(src_path, obj) or _newer(header, obj):
             _run(common + ["-c", src_path, "-o", obj])
             rebuilt = True
-    bind_path = os.path.join(SRC_DIR, BINDING_SOURCE)
-    bind_obj = os.path.join(BUILD_DIR, "bindings.o")
-    if force or _newer(bind_path, bind_obj):
-        _run(common + includes + defines + ["-x", "hip", "-c", bind_path, "-o", bind_obj])
-        rebuilt = True
-    objs.append(bind_obj)
+    core_hdrs = [
+        os.path.join(PKG_DIR, "csrc", "core", "jvalue.h"),
+        os.path.join(PKG_DIR, "csrc", "core", "native_lane.h"),
+    ]
+    for src in (BINDING_SOURCE, "native_engrams.cpp"):
+        src_path = os.path.join(SRC_DIR, src)
+        obj = os.path.join(BUILD_DIR, src.rsplit(".", 1)[0] + ".o")
+        if force or _newer(src_path, obj) or any(_newer(h, obj) for h in core_hdrs):
+            _run(common + includes + defines + ["-x", "hip", "-c", src_path, "-o", obj])
+            rebuilt = True
+        objs.append(obj)
     if force or rebuilt or not os.path.exists(OUT_SO):
         _run(common + ["-shared", *objs, *libs, "-o", OUT_SO])
     return OUT_SO

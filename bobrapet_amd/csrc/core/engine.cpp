@@ -20,8 +20,30 @@ void NativeEngine::loop() {
       timers_.pop();
       events_.push_back(std::move(ev));
     }
+    // poll native-lane tickets (hipEventQuery behind the fn pointer):
+    // completed engram work becomes EngramDone events without the GIL
+    if (!ntickets_.empty() && lane_) {
+      for (size_t i = 0; i < ntickets_.size();) {
+        JValue out;
+        std::string err;
+        int rc = lane_->poll(lane_->self, ntickets_[i].ticket, &out, &err);
+        if (rc == 0) {
+          ++i;
+          continue;
+        }
+        const NTicket tk = ntickets_[i];
+        ntickets_[i] = ntickets_.back();
+        ntickets_.pop_back();
+        events_.push_back({EvKind::EngramDone, tk.run, tk.step, tk.branch,
+                           tk.attempt, std::move(out), std::move(err),
+                           rc == 1 ? 0 : 2});
+      }
+    }
     if (events_.empty()) {
-      if (timers_.empty()) {
+      if (!ntickets_.empty()) {
+        // GPU work in flight: nap briefly, then re-poll
+        cv_.wait_for(g, std::chrono::microseconds(50));
+      } else if (timers_.empty()) {
         cv_.wait(g);
       } else {
         double dt = timers_.top().at - now();
@@ -437,6 +459,18 @@ void NativeEngine::launch_step(Run& run, const Plan& plan, int idx) {
         ev.timer_tag = 3;
         arm(now() + step.timeout, std::move(ev));
       }
+      if (step.native_kind > 0 && lane_) {
+        const int dev =
+            devices_.empty()
+                ? 0
+                : devices_[(run.id * 131 + (uint64_t)idx * 31) % devices_.size()];
+        long tk = lane_->launch(lane_->self, step.native_kind,
+                                &step.native_cfg, &with, dev);
+        if (tk > 0) {
+          ntickets_.push_back({tk, run.id, idx, -1, st.attempt});
+          break;
+        }
+      }
       if (launcher_)
         launcher_(run.id, idx, -1, st.attempt, step.engram, step.name, with);
       else {
@@ -490,6 +524,19 @@ void NativeEngine::launch_branch(Run& run, const Plan& plan, int idx, int b) {
           st.error = std::string("template: ") + e.what();
           st.finished = now();
           return;
+        }
+      }
+      if (branch.native_kind > 0 && lane_) {
+        const int dev =
+            devices_.empty()
+                ? 0
+                : devices_[(run.id * 131 + (uint64_t)idx * 31 + (uint64_t)b * 7 + 1) %
+                           devices_.size()];
+        long tk = lane_->launch(lane_->self, branch.native_kind,
+                                &branch.native_cfg, &with, dev);
+        if (tk > 0) {
+          ntickets_.push_back({tk, run.id, idx, b, st.attempt});
+          break;
         }
       }
       if (launcher_)

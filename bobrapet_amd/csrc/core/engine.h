@@ -27,6 +27,7 @@
 #include <unordered_map>
 
 #include "expr.h"
+#include "native_lane.h"
 #include "jvalue.h"
 
 namespace bobraccel {
@@ -116,6 +117,8 @@ struct PlanStep {
   Phase stop_phase = Phase::Succeeded;
   std::vector<PlanStep> branches;  // Parallel
   std::string engram;              // launcher routing key
+  int native_kind = 0;             // >0: dispatch via the NativeLane
+  JValue native_cfg;               // engram config for the lane
   ExprPtr post_exec;               // postExecution condition over {output}
   std::string post_exec_msg;
   int target_plan = -1;            // ExecuteStory
@@ -186,6 +189,16 @@ class NativeEngine {
   }
 
   void set_launcher(EngramLauncher fn) { launcher_ = std::move(fn); }
+
+  // GIL-free built-in engram lane (native_lane.h); devices for placement
+  void set_native_lane(const NativeLane* lane) {
+    std::lock_guard<std::mutex> g(mu_);
+    lane_ = lane;
+  }
+  void set_devices(std::vector<int> devs) {
+    std::lock_guard<std::mutex> g(mu_);
+    devices_ = std::move(devs);
+  }
 
   void start() {
     bool expected = false;
@@ -325,7 +338,37 @@ class NativeEngine {
         }
       }
     }
+    if (lane_ && lane_->free_key) {
+      for (uint64_t d : doomed) {
+        auto rit = runs_.find(d);
+        if (rit == runs_.end()) continue;
+        free_native_keys(rit->second.output);
+        for (auto& st : rit->second.states) free_native_keys(st.output);
+        for (auto& bs : rit->second.branch_states)
+          for (auto& st : bs) free_native_keys(st.output);
+      }
+    }
     for (uint64_t d : doomed) runs_.erase(d);
+  }
+
+  // release lane-held payloads referenced by a gc'd run's outputs
+  void free_native_keys(const JValue& v) {  // mu_ held
+    if (v.is_object()) {
+      const JObject& o = v.as_object();
+      auto it = o.find("$storageRef");
+      if (it != o.end() && it->second.is_object()) {
+        const JObject& r = it->second.as_object();
+        auto k = r.find("key");
+        if (k != r.end() && k->second.is_string()) {
+          const std::string& key = k->second.as_string();
+          if (key.rfind("native/", 0) == 0)
+            lane_->free_key(lane_->self, key.c_str());
+        }
+      }
+      for (const auto& [kk, vv] : o) free_native_keys(vv);
+    } else if (v.is_array()) {
+      for (const auto& e : v.as_array()) free_native_keys(e);
+    }
   }
 
  private:
@@ -388,6 +431,16 @@ class NativeEngine {
   std::unordered_map<uint64_t, Run> runs_;
   uint64_t next_run_ = 1;
   EngramLauncher launcher_;
+  const NativeLane* lane_ = nullptr;
+  std::vector<int> devices_;
+  struct NTicket {
+    long ticket;
+    uint64_t run;
+    int step;
+    int branch;
+    uint32_t attempt;
+  };
+  std::vector<NTicket> ntickets_;
   std::thread loop_;
   std::atomic<bool> running_{false};
   std::mt19937 rng_{12345};

@@ -222,6 +222,8 @@ static PlanStep to_step(const py::dict& d) {
                                    : Phase::Succeeded;
   }
   if (d.contains("engram")) s.engram = d["engram"].cast<std::string>();
+  if (d.contains("nativeKind")) s.native_kind = d["nativeKind"].cast<int>();
+  if (d.contains("nativeCfg")) s.native_cfg = to_jvalue(d["nativeCfg"]);
   if (d.contains("targetPlan")) s.target_plan = d["targetPlan"].cast<int>();
   if (d.contains("postExec")) s.post_exec = to_expr(d["postExec"]);
   if (d.contains("postExecMsg"))
@@ -295,6 +297,15 @@ PYBIND11_MODULE(_core, m) {
           py::arg("run"), py::arg("step"), py::arg("branch"),
           py::arg("attempt"), py::arg("exit_code"),
           py::arg("output") = py::none(), py::arg("error") = "")
+      .def("set_native_lane",
+           [](NativeEngine& e, py::capsule cap) {
+             e.set_native_lane(
+                 reinterpret_cast<const NativeLane*>(cap.get_pointer()));
+           })
+      .def("set_devices",
+           [](NativeEngine& e, std::vector<int> devs) {
+             e.set_devices(std::move(devs));
+           })
       .def("decide_gate", &NativeEngine::decide_gate,
            py::call_guard<py::gil_scoped_release>())
       .def("cancel", &NativeEngine::cancel,

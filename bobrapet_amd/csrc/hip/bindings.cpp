@@ -8,6 +8,11 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+// native-lane hooks (native_engrams.cpp, same .so)
+const void* bobra_native_lane_ptr();
+bool bobra_native_tensor_get(const std::string& key, at::Tensor* out);
+size_t bobra_native_registry_size();
+
 extern "C" {
 void launch_rmsnorm(void*, void*, void*, const void*, int, int, float, bool,
                     hipStream_t);
@@ -517,4 +522,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_qkv", &rope_qkv,
         "fused rope: strided q/k heads from the qkv projection -> contiguous");
   m.def("dbg_attn_core", &dbg_attn_core, "layout probe: QK^T + pack + PV");
+  m.def("native_lane_capsule", []() {
+    return py::capsule(const_cast<void*>(bobra_native_lane_ptr()),
+                       "bobra_native_lane");
+  });
+  m.def("native_tensor_get", [](const std::string& key) -> py::object {
+    at::Tensor t;
+    if (!bobra_native_tensor_get(key, &t)) return py::none();
+    return py::cast(t);
+  });
+  m.def("native_registry_size", []() { return bobra_native_registry_size(); });
 }

@@ -123,6 +123,7 @@ def compile_story_plan(
     resolver,
     registry,
     plan_ids: _t.Mapping[str, int],
+    native_kinds: _t.Optional[_t.Mapping[str, int]] = None,
 ) -> dict:
     """Build the plan dict the C++ core ingests."""
     from ..engine.dag import compile_story as py_compile
@@ -145,7 +146,10 @@ def compile_story_plan(
 
     steps = []
     for s, graph in dep_graphs:
-        steps.append(_compile_step(s, story, resolver, registry, index, graph, plan_ids))
+        steps.append(
+            _compile_step(s, story, resolver, registry, index, graph, plan_ids,
+                          native_kinds)
+        )
 
     plan = {
         "name": story.key,
@@ -166,7 +170,8 @@ def compile_story_plan(
     return plan
 
 
-def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_ids) -> dict:
+def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_ids,
+                  native_kinds=None) -> dict:
     d: dict = {"name": s.name, "kind": _KIND[s.type]}
     deps = sorted(dep_graph.get(s.name, set()))
     by_name = {x.name: x for x in story.all_steps()}
@@ -197,6 +202,13 @@ def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_i
                 except KeyError:
                     template = None
         cfg = resolver.resolve(step=s, story=story, engram=engram, template=template)
+        if native_kinds and template is not None and getattr(template, "builtin", None):
+            kind = native_kinds.get(str(template.builtin))
+            if kind:
+                # GIL-free lane dispatch (csrc/hip/native_engrams.cpp); the
+                # lane falls back to the Python launcher per-input
+                d["nativeKind"] = int(kind)
+                d["nativeCfg"] = dict(engram.with_ or {}) if engram is not None else {}
         d["retry"] = {
             "maxRetries": cfg.max_retries,
             "delay": cfg.retry_delay,
@@ -243,7 +255,8 @@ def _compile_step(s: T.Step, story, resolver, registry, index, dep_graph, plan_i
         for raw in w.get("steps", []):
             b = T._step_from_dict(dict(raw))
             branches.append(
-                _compile_step(b, story, resolver, registry, {}, {}, plan_ids)
+                _compile_step(b, story, resolver, registry, {}, {}, plan_ids,
+                              native_kinds)
             )
         d["branches"] = branches
     elif s.type == StepType.EXECUTE_STORY:
@@ -282,11 +295,39 @@ class NativeRunner:
         self._engram_cache: _t.Dict[str, tuple] = {}
         self._engram_cache_gen = -1
         self.engine.set_launcher(self._launch)
+        self.native_kinds: _t.Dict[str, int] = {}
+        self._setup_native_lane()
         # expression-level hydration of offloaded payloads (markers passed
         # THROUGH stay zero-copy; only consuming expressions materialize)
         if self.storage is not None and hasattr(core, "set_expr_hydrator"):
             core.set_expr_hydrator(self.storage.hydrate)
         self.engine.start()
+
+    def _setup_native_lane(self) -> None:
+        """Register the GIL-free built-in engram lane when a GPU and the
+        HIP extension are present (embed always; the local join only at
+        world_size 1 — multi-rank joins all-gather over RCCL in Python)."""
+        try:
+            import torch
+
+            if not torch.cuda.is_available():
+                return
+            from bobrapet_amd import _hipops
+
+            if not hasattr(_hipops, "native_lane_capsule") or not hasattr(
+                self.engine, "set_native_lane"
+            ):
+                return
+            self.engine.set_native_lane(_hipops.native_lane_capsule())
+            devs = list(getattr(self.workers, "device_ids", None) or [0])
+            self.engine.set_devices([int(d) for d in devs])
+            self.native_kinds = {"embed": 1}
+            import torch.distributed as dist
+
+            if not (dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1):
+                self.native_kinds["allgather-join"] = 2
+        except Exception:
+            self.native_kinds = {}
 
     @classmethod
     def from_run_engine(cls, eng) -> "NativeRunner":
@@ -308,7 +349,10 @@ class NativeRunner:
                 tns = s.with_.get("namespace") or story.namespace
                 if f"{tns}/{target}" not in self.plan_ids:
                     self.compile(self.registry.story(target, tns))
-        plan = compile_story_plan(story, self.resolver, self.registry, self.plan_ids)
+        plan = compile_story_plan(
+            story, self.resolver, self.registry, self.plan_ids,
+            native_kinds=self.native_kinds or None,
+        )
         pid = self.engine.register_plan(plan)
         self.plan_ids[key] = pid
         self._plan_gen[key] = story.generation

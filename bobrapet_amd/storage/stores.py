@@ -184,6 +184,15 @@ class TensorStore:
     def get(self, key: str):
         with self._lock:
             t = self._table.get(key)
+        if t is None and key.startswith("native/"):
+            # payload held by the GIL-free native engram lane
+            # (csrc/hip/native_engrams.cpp registry)
+            try:
+                from bobrapet_amd import _hipops
+
+                t = _hipops.native_tensor_get(key)
+            except ImportError:
+                t = None
         if t is None:
             raise BlobNotFound(key)
         return t

@@ -192,3 +192,80 @@ class TestServedModelReuse:
         m.generate(a, 2)
         out = m.generate(b, 2)  # different batch: second graph, fresh cache
         assert out.shape == (3, 2)
+
+
+class TestNativeLane:
+    """The GIL-free built-in engram lane (csrc/hip/native_engrams.cpp):
+    the parallel8 story's embed branches + local join run entirely in
+    C++/HIP off the core's loop thread — no Python launcher bodies."""
+
+    def test_parallel8_story_on_native_lane(self):
+        from bobrapet_amd import _hipops
+        from bobrapet_amd.runtime.native import NativeRunner
+
+        resources = """
+kind: EngramTemplate
+metadata: {name: embed}
+spec: {builtin: embed}
+---
+kind: Engram
+metadata: {name: embedder}
+spec:
+  templateRef: {name: embed}
+  with: {dim: 1024, vocab: 4096, batch: 8, seqLen: 32}
+---
+kind: EngramTemplate
+metadata: {name: allgather-join}
+spec: {builtin: allgather-join}
+---
+kind: Engram
+metadata: {name: joiner}
+spec: {templateRef: {name: allgather-join}}
+---
+kind: Story
+metadata: {name: lane8}
+spec:
+  steps:
+    - name: fanout
+      type: parallel
+      with:
+        steps:
+""" + "\n".join(
+            f"          - {{name: b{i}, ref: {{name: embedder}}, with: {{seed: {i}}}}}"
+            for i in range(8)
+        ) + """
+    - name: join
+      ref: {name: joiner}
+      needs: [fanout]
+      with:
+        branches: "{{ steps.fanout.output.branches }}"
+  output:
+    rows: "{{ steps.join.output.worldRows }}"
+"""
+        eng = RunEngine(EngineConfig(cpu_workers=2, workers_per_device=4)).start()
+        try:
+            eng.apply_yaml(resources)
+            runner = NativeRunner.from_run_engine(eng)
+            assert runner.native_kinds.get("embed") == 1, "lane not registered"
+            story = eng.registry.story("lane8", "default")
+            runner.compile(story)
+            for _ in range(3):
+                st = runner.run_story(story, {}, timeout=30.0, gc=False)
+                assert st["phase"] == "Succeeded", st
+                assert st["output"]["rows"] == 8 * 8, st["output"]
+            assert _hipops.native_registry_size() > 0, "lane never held payloads"
+            # joined rows are L2-normalized embed rows: hydrate + check norms
+            joined_ref = st["steps"]["join"]["output"]["joined"]
+            t = eng.storage.hydrate(joined_ref)
+            norms = t.float().norm(dim=-1)
+            assert abs(norms.mean().item() - 1.0) < 0.05
+            # gc releases the lane-held payloads of reclaimed runs
+            before = _hipops.native_registry_size()
+            rid = runner.submit(story, {})
+            runner.wait(rid, 30.0)
+            runner.engine.gc_run(rid)
+            time.sleep(0.1)
+            assert _hipops.native_registry_size() <= before + 1
+            runner.stop()
+        finally:
+            eng.stop()

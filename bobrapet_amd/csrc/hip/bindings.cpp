@@ -29,8 +29,8 @@ void launch_silu_mul_strided(void*, const void*, const void*, long, int, long,
 void launch_rope_qkv(void*, void*, const void*, const void*, const void*, int,
                      int, int, int, long, hipStream_t);
 void launch_rope_qkv_decode(void*, void*, void*, const void*, const void*,
-                            const void*, const void*, int, int, int, int, int,
-                            hipStream_t);
+                            const void*, int, int, int, int, int, hipStream_t);
+void launch_argmax_rows(void*, const void*, int, int, hipStream_t);
 void launch_attn_decode(void*, void*, const void*, const void*, const void*,
                         int, int, int, int, int, const void*, float,
                         hipStream_t);
@@ -250,27 +250,36 @@ torch::Tensor attn_decode_t(torch::Tensor q, torch::Tensor kc,
 }
 
 torch::Tensor rope_qkv_decode(torch::Tensor qkv, torch::Tensor kc,
-                              torch::Tensor vc, torch::Tensor cos_t,
-                              torch::Tensor sin_t, torch::Tensor L_dev,
-                              long Hq, long Hkv, long D) {
+                              torch::Tensor vc, torch::Tensor inv_freq,
+                              torch::Tensor L_dev, long Hq, long Hkv, long D) {
   // fused decode head prep: rope q (returned contiguous [B,Hq,D]) and
-  // rope k + copy v straight into the KV cache at device position L
+  // rope k + copy v straight into the KV cache at device position L;
+  // rope angles computed in-kernel from inv_freq (no host cos/sin tables)
   check_bf16(qkv, "qkv");
   check_bf16(kc, "kc");
   check_bf16(vc, "vc");
   const int B = qkv.size(0);
   const int Smax = kc.size(2);
   TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hkv) * D, "rope_qkv_decode: row size");
-  TORCH_CHECK(cos_t.scalar_type() == torch::kFloat32 && cos_t.is_contiguous(),
-              "cos must be f32 contiguous");
+  TORCH_CHECK(inv_freq.scalar_type() == torch::kFloat32 &&
+                  inv_freq.is_contiguous() && inv_freq.numel() == D / 2,
+              "inv_freq must be f32 [D/2] contiguous");
   TORCH_CHECK(L_dev.scalar_type() == torch::kInt32 && L_dev.is_cuda(),
               "L must be an int32 device scalar");
   auto q_out = torch::empty({(long)B, Hq, D}, qkv.options());
   launch_rope_qkv_decode(q_out.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                         qkv.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
-                         L_dev.data_ptr(), B, (int)Hq, (int)Hkv, (int)D, Smax,
-                         cur_stream());
+                         qkv.data_ptr(), inv_freq.data_ptr(), L_dev.data_ptr(),
+                         B, (int)Hq, (int)Hkv, (int)D, Smax, cur_stream());
   return q_out;
+}
+
+torch::Tensor argmax_rows(torch::Tensor x) {
+  check_bf16(x, "x");
+  const int N = x.size(-1);
+  const int M = x.numel() / N;
+  auto ids = torch::empty({(long)M}, x.options().dtype(torch::kLong));
+  launch_argmax_rows(ids.data_ptr(), x.data_ptr(), M, N, cur_stream());
+  return ids;
 }
 
 torch::Tensor silu_mul_strided(torch::Tensor gate_up) {
@@ -532,4 +541,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return py::cast(t);
   });
   m.def("native_registry_size", []() { return bobra_native_registry_size(); });
+  m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
 }

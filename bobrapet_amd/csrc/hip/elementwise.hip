@@ -253,8 +253,7 @@ __global__ __launch_bounds__(256) void rope_qkv_decode_kernel(
     unsigned short* __restrict__ kc,         // [B,Hkv,Smax,D]
     unsigned short* __restrict__ vc,         // [B,Hkv,Smax,D]
     const unsigned short* __restrict__ qkv,  // [B, (Hq+2*Hkv)*D]
-    const float* __restrict__ cos_t,         // [B, D/2] (position L row)
-    const float* __restrict__ sin_t,
+    const float* __restrict__ inv_freq,      // [D/2] rope inverse freqs
     const int* __restrict__ L_dev, int B, int Hq, int Hkv, int D, int Smax) {
   const int L = *L_dev;
   const int half = D / 2;
@@ -271,12 +270,14 @@ __global__ __launch_bounds__(256) void rope_qkv_decode_kernel(
       unsigned short* dst =
           (h < Hq) ? q_out + ((long)b * Hq + h) * D
                    : kc + (((long)b * Hkv + (h - Hq)) * Smax + L) * D;
-      const float* crow = cos_t + (long)b * half;
-      const float* srow = sin_t + (long)b * half;
       for (int d = lane; d < half; d += WAVE) {
         float a = bf2f(src[d]);
         float bb = bf2f(src[d + half]);
-        float c = crow[d], sn = srow[d];
+        // angles computed in-kernel from inv_freq (the
+        // host-side cos/sin table build was ~6 at::native launches/step)
+        const float ang = (float)L * inv_freq[d];
+        float sn, c;
+        __sincosf(ang, &sn, &c);
         dst[d] = f2bf(a * c - bb * sn);
         dst[d + half] = f2bf(bb * c + a * sn);
       }
@@ -289,17 +290,81 @@ __global__ __launch_bounds__(256) void rope_qkv_decode_kernel(
 }
 
 extern "C" void launch_rope_qkv_decode(void* q_out, void* kc, void* vc,
-                                       const void* qkv, const void* cos_t,
-                                       const void* sin_t, const void* L_dev,
-                                       int B, int Hq, int Hkv, int D, int Smax,
+                                       const void* qkv, const void* inv_freq,
+                                       const void* L_dev, int B, int Hq,
+                                       int Hkv, int D, int Smax,
                                        hipStream_t stream) {
   int waves_needed = B * (Hq + 2 * Hkv);
   int blocks = (waves_needed + 3) / 4;  // 4 waves per 256-thread block
   hipLaunchKernelGGL(rope_qkv_decode_kernel, dim3(blocks), dim3(256), 0,
                      stream, (unsigned short*)q_out, (unsigned short*)kc,
                      (unsigned short*)vc, (const unsigned short*)qkv,
-                     (const float*)cos_t, (const float*)sin_t,
-                     (const int*)L_dev, B, Hq, Hkv, D, Smax);
+                     (const float*)inv_freq, (const int*)L_dev, B, Hq, Hkv, D,
+                     Smax);
+}
+
+// ---------------------------------------------------------------------------
+// row argmax: ids[m] = argmax_n x[m][n]  (greedy sampling — replaces the
+// at::native argmax on [B, V~128k] logits)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void argmax_rows_kernel(
+    long* __restrict__ ids, const unsigned short* __restrict__ x, int N) {
+  const int m = blockIdx.x;
+  const unsigned short* row = x + (long)m * N;
+  const int tid = threadIdx.x;
+  float best = -3.4e38f;
+  int bi = 0;
+  for (int n = tid * 8; n + 7 < N; n += 256 * 8) {
+    ushort8v v = *reinterpret_cast<const ushort8v*>(row + n);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float f = bf2f(v[j]);
+      if (f > best) {
+        best = f;
+        bi = n + j;
+      }
+    }
+  }
+  for (int n = (N / 8) * 8 + tid; n < N; n += 256) {
+    const float f = bf2f(row[n]);
+    if (f > best) {
+      best = f;
+      bi = n;
+    }
+  }
+  // wave reduce (value, index) — ties resolve to the LOWEST index like
+  // torch.argmax
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ob = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(bi, off, WAVE);
+    if (ob > best || (ob == best && oi < bi)) {
+      best = ob;
+      bi = oi;
+    }
+  }
+  __shared__ float sb[4];
+  __shared__ int si[4];
+  const int w = tid >> 6;
+  if ((tid & (WAVE - 1)) == 0) {
+    sb[w] = best;
+    si[w] = bi;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    for (int i = 1; i < 4; ++i)
+      if (sb[i] > best || (sb[i] == best && si[i] < bi)) {
+        best = sb[i];
+        bi = si[i];
+      }
+    ids[m] = (long)bi;
+  }
+}
+
+extern "C" void launch_argmax_rows(void* ids, const void* x, int M, int N,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(argmax_rows_kernel, dim3(M), dim3(256), 0, stream,
+                     (long*)ids, (const unsigned short*)x, N);
 }
 
 extern "C" void launch_rope(void* q, void* k, const void* cos_t,

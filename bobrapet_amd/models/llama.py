@@ -314,15 +314,19 @@ class LlamaModel:
         6 kernels per layer, no elementwise passes."""
         cfg = self.cfg
         B = ids.shape[0]
-        pos_l = pos_i32.to(torch.long)
         inv_freq = self._inv_freq
-        ang = pos_l.float().reshape(1, 1) * inv_freq[None, :]
-        cos_t = torch.cos(ang).expand(B, -1).contiguous()
-        sin_t = torch.sin(ang).expand(B, -1).contiguous()
+        use_gemv = B <= 8 and self.device.type == "cuda"
+        cos_t = sin_t = None
+        pos_l = None
+        if not use_gemv:
+            # eager fallback path still builds host-side tables
+            pos_l = pos_i32.to(torch.long)
+            ang = pos_l.float().reshape(1, 1) * inv_freq[None, :]
+            cos_t = torch.cos(ang).expand(B, -1).contiguous()
+            sin_t = torch.sin(ang).expand(B, -1).contiguous()
 
         x2 = self.embed[ids].view(B, cfg.hidden_size)  # residual stream
         L_dev = (pos_i32 + 1).contiguous()
-        use_gemv = B <= 8 and self.device.type == "cuda"
         for li, lw in enumerate(self.layers):
             kc, vc = self._kv_cache[li]
             if use_gemv or not x2.is_cuda:
@@ -333,7 +337,7 @@ class LlamaModel:
                 # fused rope + cache append: one kernel instead of rope +
                 # two index_copys + two layout copies (x32 layers/step)
                 qf = ops.rope_qkv_decode(
-                    qkv, kc, vc, cos_t, sin_t, pos_i32,
+                    qkv, kc, vc, inv_freq, pos_i32,
                     cfg.num_heads, cfg.num_kv_heads, cfg.head_dim,
                 )
             else:
@@ -468,7 +472,7 @@ class LlamaModel:
         step = self.decode_step_graphed if use_graph else self.decode_step
         out = []
         for _ in range(new_tokens):
-            nxt = logits.argmax(dim=-1) if greedy else torch.multinomial(
+            nxt = ops.argmax_rows(logits) if greedy else torch.multinomial(
                 torch.softmax(logits.float(), dim=-1), 1
             ).squeeze(-1)
             out.append(nxt)

@@ -374,17 +374,21 @@ def gemv_swiglu_norm(a: torch.Tensor, w_interleaved: torch.Tensor, mul: float, e
     return (g * torch.sigmoid(g) * u).to(a.dtype)
 
 
-def rope_qkv_decode(qkv2d, kc, vc, cos_t, sin_t, pos_dev: torch.Tensor, Hq: int, Hkv: int, D: int):
+def rope_qkv_decode(qkv2d, kc, vc, inv_freq, pos_dev: torch.Tensor, Hq: int, Hkv: int, D: int):
     """Fused decode head prep: rope q (returned [B,Hq,D]) and rope k +
     copy v straight into the KV cache at DEVICE position pos (hipGraph-
-    replayable — one kernel instead of rope + 2 index_copys + 2 copies)."""
+    replayable — one kernel instead of rope + 2 index_copys + 2 copies).
+    Rope angles come from inv_freq [D/2] computed in-kernel (no host
+    cos/sin table launches)."""
     if qkv2d.is_cuda:
         return _require_ext().rope_qkv_decode(
-            qkv2d.contiguous(), kc, vc, cos_t, sin_t, pos_dev, Hq, Hkv, D
+            qkv2d.contiguous(), kc, vc, inv_freq.contiguous(), pos_dev, Hq, Hkv, D
         )
     # CPU reference: compose from the existing reference ops
     B = qkv2d.shape[0]
     pos = int(pos_dev.item())
+    ang = float(pos) * inv_freq.float().reshape(1, -1)
+    cos_t, sin_t = torch.cos(ang).expand(B, -1), torch.sin(ang).expand(B, -1)
     q = qkv2d[:, : Hq * D].reshape(B, Hq, D).clone()
     k = qkv2d[:, Hq * D : (Hq + Hkv) * D].reshape(B, Hkv, D).clone()
     v = qkv2d[:, (Hq + Hkv) * D :].reshape(B, Hkv, D)
@@ -393,6 +397,13 @@ def rope_qkv_decode(qkv2d, kc, vc, cos_t, sin_t, pos_dev: torch.Tensor, Hq: int,
     kc[:, :, pos] = k
     vc[:, :, pos] = v
     return q
+
+
+def argmax_rows(x: torch.Tensor) -> torch.Tensor:
+    """ids[m] = argmax_n x[m,n] (greedy sampling over logits)."""
+    if x.is_cuda:
+        return _require_ext().argmax_rows(x.contiguous())
+    return x.float().argmax(dim=-1)
 
 
 def attn_decode_t(q, kc, vc, L_dev: torch.Tensor, scale: _t.Optional[float] = None):

@@ -371,6 +371,14 @@ class TestAttnPrefillStats:
         assert err < 0.03, f"merge err {err}"
 
 
+class TestArgmaxRows:
+    def test_matches_torch(self):
+        x = torch.randn(4, 128256, dtype=torch.bfloat16, device="cuda")
+        got = ops.argmax_rows(x)
+        ref = x.float().argmax(dim=-1)
+        assert torch.equal(got.cpu(), ref.cpu()), (got, ref)
+
+
 @pytest.mark.gpu
 class TestRopeQkvDecode:
     def test_matches_composed_reference(self):
@@ -384,16 +392,14 @@ class TestRopeQkvDecode:
         kc = torch.zeros(B, Hkv, Smax, D, device="cuda", dtype=torch.bfloat16)
         vc = torch.zeros_like(kc)
         half = D // 2
-        ang = torch.rand(B, half, device="cuda") * 6.28
-        cos_t, sin_t = torch.cos(ang), torch.sin(ang)
+        inv_freq = 1.0 / (500000.0 ** (torch.arange(0, D, 2, dtype=torch.float32, device="cuda") / D))
         pos_dev = torch.tensor([pos], dtype=torch.int32, device="cuda")
-        q = ops.rope_qkv_decode(qkv, kc, vc, cos_t, sin_t, pos_dev, Hq, Hkv, D)
+        q = ops.rope_qkv_decode(qkv, kc, vc, inv_freq, pos_dev, Hq, Hkv, D)
         # composed CPU reference
         kc_ref = torch.zeros(B, Hkv, Smax, D, dtype=torch.bfloat16)
         vc_ref = torch.zeros_like(kc_ref)
         q_ref = ops.rope_qkv_decode(
-            qkv.cpu(), kc_ref, vc_ref, cos_t.cpu(), sin_t.cpu(),
-            pos_dev.cpu(), Hq, Hkv, D,
+            qkv.cpu(), kc_ref, vc_ref, inv_freq.cpu(), pos_dev.cpu(), Hq, Hkv, D,
         )
         assert (q.float().cpu() - q_ref.float()).abs().max().item() < 0.02
         assert (kc.float().cpu() - kc_ref.float()).abs().max().item() < 0.02

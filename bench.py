@@ -82,6 +82,12 @@ spec: {{builtin: echo}}
 kind: Engram
 metadata: {{name: echo-stage}}
 spec: {{templateRef: {{name: echo-tpl}}}}
+---
+kind: Engram
+metadata: {{name: refiner}}
+spec:
+  templateRef: {{name: echo-tpl}}
+  with: {{stage: refined}}
 """
 
 PARALLEL_STORY = f"""
@@ -153,9 +159,12 @@ spec:
     - name: featurize
       ref: {name: embedder}
       with: {capture: true}
+    - name: refine
+      ref: {name: refiner}
+      needs: [featurize]
     - name: tag
       ref: {name: echo-stage}
-      needs: [featurize]
+      needs: [refine]
 """
 
 BIGPAYLOAD_STORY = """
@@ -209,34 +218,85 @@ def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None, sl
 
 
 def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
-    """Config #4: packets/sec through the captured streaming pipeline;
-    one 'step' = one packet."""
-    stream = eng.submit_stream("default/bench-stream")
-    ids = torch.randint(0, EMBED_VOCAB, (EMBED_BATCH, EMBED_SEQ), dtype=torch.int32)
-    if torch.cuda.is_available():
-        ids = ids.cuda()
-    for i in range(args.warmup):
-        stream.push({"tensor": ids, "seq": -i, "items": [{}]})
-    deadline = time.monotonic() + 30
-    while stream.leaf_packets < args.warmup and time.monotonic() < deadline:
-        time.sleep(0.001)
-    group.barrier()
-    t0 = time.monotonic()
-    for i in range(args.steps):
-        stream.push({"tensor": ids, "seq": i, "items": [{}]})
-    while stream.leaf_packets < args.warmup + args.steps and time.monotonic() < t0 + 120:
-        time.sleep(0.001)
-    group.barrier()
-    elapsed = time.monotonic() - t0
-    run = stream.finish(timeout=30)
-    if run.phase.value not in ("Finished",):
-        details = {
-            k: (str(v.phase), str(v.error.message if v.error else ""))
-            for k, v in run.step_states.items()
-        }
-        raise RuntimeError(f"stream bench failed: {run.phase} {details}")
+    """Config #4 as specified: packets enter over a REAL
+    gRPC connection (loopback) and flow through the 3-stage engram
+    pipeline; one 'step' = one packet.  BOBRA_STREAM_INPROC=1 falls back
+    to in-process pushes (ablation)."""
+    import numpy as np
+
+    from bobrapet_amd.engine.ingress_grpc import IngressClient, pack_frame, serve_grpc
+
+    use_grpc = os.environ.get("BOBRA_STREAM_INPROC") != "1"
+    ids_np = np.random.default_rng(7).integers(
+        0, EMBED_VOCAB, (EMBED_BATCH, EMBED_SEQ), dtype=np.int32
+    )
+    payload = ids_np.tobytes()
+    tensor_spec = {"shape": [EMBED_BATCH, EMBED_SEQ], "dtype": "int32"}
+
+    if use_grpc:
+        server, port = serve_grpc(eng, port=0)
+        client = IngressClient(f"127.0.0.1:{port}")
+        session = f"bench-{rank}"
+
+        def frames(first, count):
+            for i in range(first, first + count):
+                yield pack_frame(
+                    {"stream": "default/bench-stream", "session": session,
+                     "seq": i, "meta": {"items": [{}]}, "tensor": tensor_spec},
+                    payload,
+                )
+
+        client.push_stream(frames(0, args.warmup))
+        deadline = time.monotonic() + 60
+        while client.stream_stats(session)["leafPackets"] < args.warmup and time.monotonic() < deadline:
+            time.sleep(0.001)
+        group.barrier()
+        t0 = time.monotonic()
+        client.push_stream(frames(args.warmup, args.steps))
+        while client.stream_stats(session)["leafPackets"] < args.warmup + args.steps and time.monotonic() < t0 + 300:
+            time.sleep(0.001)
+        group.barrier()
+        elapsed = time.monotonic() - t0
+        fin = client.finish_stream(session, timeout=60.0)
+        if fin.get("phase") != "Finished":
+            raise RuntimeError(f"stream bench failed: {fin}")
+        run = None
+        for r in eng.store.all_runs():
+            if r.story_name == "bench-stream":
+                run = r
+        client.close()
+        server.stop(0)
+    else:
+        stream = eng.submit_stream("default/bench-stream")
+        ids = torch.from_numpy(ids_np.copy())
+        if torch.cuda.is_available():
+            ids = ids.cuda()
+        for i in range(args.warmup):
+            stream.push({"tensor": ids, "seq": -i, "items": [{}]})
+        deadline = time.monotonic() + 30
+        while stream.leaf_packets < args.warmup and time.monotonic() < deadline:
+            time.sleep(0.001)
+        group.barrier()
+        t0 = time.monotonic()
+        for i in range(args.steps):
+            stream.push({"tensor": ids, "seq": i, "items": [{}]})
+        while stream.leaf_packets < args.warmup + args.steps and time.monotonic() < t0 + 120:
+            time.sleep(0.001)
+        group.barrier()
+        elapsed = time.monotonic() - t0
+    if not use_grpc:
+        run = stream.finish(timeout=30)
+        if run.phase.value not in ("Finished",):
+            details = {
+                k: (str(v.phase), str(v.error.message if v.error else ""))
+                for k, v in run.step_states.items()
+            }
+            raise RuntimeError(f"stream bench failed: {run.phase} {details}")
     elapsed_max = group.max_over_ranks(elapsed, device="cpu" if not torch.cuda.is_available() else None)
-    replays = run.step_states["featurize"].output.get("graphReplays", 0)
+    replays = 0
+    if run is not None and "featurize" in run.step_states:
+        out = run.step_states["featurize"].output or {}
+        replays = out.get("graphReplays", 0)
     if rank == 0:
         line = {
             "metric": "StoryRuns/sec + p50 step latency, 8-way parallel Story",
@@ -252,7 +312,7 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
-                "model": "streaming 2-stage pipeline, hipGraph-captured embed",
+                "model": ("streaming 3-stage pipeline via gRPC ingress, hipGraph-captured embed" if use_grpc else "streaming 3-stage pipeline (in-process), hipGraph-captured embed"),
                 "bench_config": "stream",
                 "global_batch": args.steps * world,
                 "seq_len": EMBED_SEQ,

@@ -42,7 +42,7 @@ void NativeEngine::loop() {
     if (events_.empty()) {
       if (!ntickets_.empty()) {
         // GPU work in flight: nap briefly, then re-poll
-        cv_.wait_for(g, std::chrono::microseconds(50));
+        cv_.wait_for(g, std::chrono::microseconds(20));
       } else if (timers_.empty()) {
         cv_.wait(g);
       } else {

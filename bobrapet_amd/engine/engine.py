@@ -582,7 +582,17 @@ class RunEngine:
                 if template is not None and getattr(template, "command", None):
                     from ..engrams.process import ProcessEngram
 
-                    impl = ProcessEngram(template.command)
+                    secrets = {}
+                    if sr.spec.engram and "/" in sr.spec.engram:
+                        e_ns, e_name = sr.spec.engram.split("/", 1)
+                        eng_obj = self.registry.try_engram(e_name, e_ns)
+                        if eng_obj is not None:
+                            secrets = eng_obj.secrets
+                    impl = ProcessEngram(
+                        template.command,
+                        secret_defs=getattr(template, "secrets", None),
+                        secrets=secrets,
+                    )
                 else:
                     impl = engram_registry.resolve(impl_name)
                 if impl.wants_gpu and slot.device is None:

@@ -30,10 +30,46 @@ class ProcessEngram(Engram):
     name = "process"
     wants_gpu = False
 
-    def __init__(self, command: _t.List[str]):
+    def __init__(self, command: _t.List[str], secret_defs=None, secrets=None):
         self.command = [str(c) for c in command]
+        # secret artifact delivery (reference: pkg/podspec/secrets.go +
+        # SecretDefinition catalog/shared_types.go:296-322 — env/file mounts)
+        self.secret_defs = list(secret_defs or [])
+        self.secrets = dict(secrets or {})
+
+    def _secret_env(self, tmpdir: _t.Optional[str]) -> _t.Dict[str, str]:
+        env: _t.Dict[str, str] = {}
+        for sd in self.secret_defs:
+            name = getattr(sd, "name", None) or (sd.get("name") if isinstance(sd, dict) else None)
+            if not name:
+                continue
+            mount = getattr(sd, "mount_type", None) or (
+                sd.get("mountType") if isinstance(sd, dict) else None
+            ) or "env"
+            required = getattr(sd, "required", None)
+            if required is None and isinstance(sd, dict):
+                required = sd.get("required")
+            value = self.secrets.get(name)
+            if value is None:
+                if required:
+                    raise EngramFailure(
+                        f"missing required secret {name!r}", exit_code=2
+                    )
+                continue
+            key = name.upper().replace("-", "_").replace(".", "_")
+            if mount in ("env", "both"):
+                env[f"BUBU_SECRET_{key}"] = str(value)
+            if mount in ("file", "both") and tmpdir is not None:
+                path = os.path.join(tmpdir, key)
+                with open(path, "w", encoding="utf-8") as fh:
+                    fh.write(str(value))
+                os.chmod(path, 0o600)
+                env[f"BUBU_SECRET_FILE_{key}"] = path
+        return env
 
     def run(self, ctx: EngramContext) -> EngramResult:
+        import tempfile
+
         env = dict(os.environ)
         env.update(
             {
@@ -55,7 +91,14 @@ class ProcessEngram(Engram):
         if ctx.device is not None:
             env["BUBU_DEVICE"] = str(ctx.device)
             env["HIP_VISIBLE_DEVICES"] = str(ctx.device)
+        needs_files = any(
+            (getattr(sd, "mount_type", None) or (sd.get("mountType") if isinstance(sd, dict) else None))
+            in ("file", "both")
+            for sd in self.secret_defs
+        )
+        tmp_ctx = tempfile.TemporaryDirectory(prefix="bubu-secrets-") if needs_files else None
         try:
+            env.update(self._secret_env(tmp_ctx.name if tmp_ctx else None))
             proc = subprocess.run(
                 self.command,
                 env=env,
@@ -67,6 +110,9 @@ class ProcessEngram(Engram):
             raise EngramFailure(f"process {self.command[0]} timed out", exit_code=1)
         except OSError as exc:  # missing binary etc. — terminal
             raise EngramFailure(f"process spawn failed: {exc}", exit_code=2)
+        finally:
+            if tmp_ctx is not None:
+                tmp_ctx.cleanup()
         lines = [ln for ln in proc.stdout.splitlines() if ln.strip()]
         for ln in lines[:-1]:
             ctx.log(ln)

@@ -414,7 +414,14 @@ class NativeRunner:
         if template is not None and getattr(template, "command", None):
             from ..engrams.process import ProcessEngram
 
-            hit = (ProcessEngram(template.command), engram.with_)
+            hit = (
+                ProcessEngram(
+                    template.command,
+                    secret_defs=getattr(template, "secrets", None),
+                    secrets=engram.secrets,
+                ),
+                engram.with_,
+            )
             self._engram_cache[engram_key] = hit
             return hit
         impl_name = template.implementation if template is not None else engram_key.split("/")[-1]

@@ -94,3 +94,95 @@ class Tracer:
 
 
 GLOBAL = Tracer(enabled=False)
+
+
+# ---------------------------------------------------------------------------
+# OTLP export (role parity: reference pkg/observability/exporter.go — OTLP
+# exporter wired at startup, flushed LIFO at shutdown, cmd/main.go:253-288).
+# Spans serialize to the OTLP/JSON ResourceSpans shape; the sink is a file
+# path or an OTLP/HTTP endpoint (http(s)://host:4318/v1/traces).
+# ---------------------------------------------------------------------------
+
+
+def _otlp_attr(key: str, value) -> dict:
+    if isinstance(value, bool):
+        v = {"boolValue": value}
+    elif isinstance(value, int):
+        v = {"intValue": str(value)}
+    elif isinstance(value, float):
+        v = {"doubleValue": value}
+    else:
+        v = {"stringValue": str(value)}
+    return {"key": key, "value": v}
+
+
+def spans_to_otlp(spans: _t.Iterable[Span], service_name: str = "bobrapet-amd") -> dict:
+    otlp_spans = []
+    for s in spans:
+        otlp_spans.append(
+            {
+                "traceId": s.trace_id[:32].ljust(32, "0"),
+                "spanId": s.span_id[:16].ljust(16, "0"),
+                "parentSpanId": s.parent_id[:16].ljust(16, "0") if s.parent_id else "",
+                "name": s.name,
+                "kind": 1,  # SPAN_KIND_INTERNAL
+                "startTimeUnixNano": str(int(s.start * 1e9)),
+                "endTimeUnixNano": str(int(s.end * 1e9)),
+                "attributes": [_otlp_attr(k, v) for k, v in s.attributes.items()],
+                "status": {"code": 2, "message": s.error} if s.error else {"code": 1},
+            }
+        )
+    return {
+        "resourceSpans": [
+            {
+                "resource": {"attributes": [_otlp_attr("service.name", service_name)]},
+                "scopeSpans": [
+                    {"scope": {"name": "bobrapet_amd.tracer"}, "spans": otlp_spans}
+                ],
+            }
+        ]
+    }
+
+
+class OTLPExporter:
+    """Flush tracer spans as OTLP/JSON to a file or an OTLP/HTTP endpoint.
+
+    Endpoint forms: a filesystem path (append one JSON document per flush,
+    newline-delimited) or http(s)://host:port/v1/traces (POST, stdlib
+    urllib — no extra deps; failures are counted, never raised into the
+    engine)."""
+
+    def __init__(self, endpoint: str, service_name: str = "bobrapet-amd"):
+        self.endpoint = endpoint
+        self.service_name = service_name
+        self.exported = 0
+        self.errors = 0
+        self._cursor = 0
+
+    def flush(self, tracer: "Tracer") -> int:
+        spans = tracer.spans()
+        fresh = spans[self._cursor :]
+        self._cursor = len(spans)
+        if not fresh:
+            return 0
+        doc = spans_to_otlp(fresh, self.service_name)
+        payload = json.dumps(doc, separators=(",", ":"))
+        try:
+            if self.endpoint.startswith(("http://", "https://")):
+                import urllib.request
+
+                req = urllib.request.Request(
+                    self.endpoint,
+                    data=payload.encode(),
+                    headers={"Content-Type": "application/json"},
+                    method="POST",
+                )
+                urllib.request.urlopen(req, timeout=5.0).read()
+            else:
+                with open(self.endpoint, "a", encoding="utf-8") as fh:
+                    fh.write(payload + "\n")
+            self.exported += len(fresh)
+            return len(fresh)
+        except Exception:
+            self.errors += 1
+            return 0

@@ -204,3 +204,70 @@ spec:
         with _p.raises(EngramFailure) as e:
             impl.run(EngramContext(input={}))
         assert e.value.exit_code == 2  # terminal
+
+
+class TestSecretDelivery:
+    """Secret artifacts for process engrams (reference:
+    pkg/podspec/secrets.go + SecretDefinition shared_types.go:296-322):
+    env-prefix and file mounts per the template's SecretDefinitions, with
+    required-secret enforcement."""
+
+    YAML = """
+kind: EngramTemplate
+metadata: {name: sec-tpl}
+spec:
+  command: [python3, -c, "%s"]
+  secrets:
+    - {name: api-key, required: true, mountType: env}
+    - {name: cert, mountType: file}
+    - {name: optional-thing, mountType: env}
+---
+kind: Engram
+metadata: {name: secret-user}
+spec:
+  templateRef: {name: sec-tpl}
+  secrets: {api-key: "sk-12345", cert: "---CERT---"}
+---
+kind: Story
+metadata: {name: sec-story}
+spec:
+  steps:
+    - {name: go, ref: {name: secret-user}}
+  output: {out: "{{ steps.go.output }}"}
+"""
+
+    def _run(self, script, yaml=None):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml((yaml or self.YAML) % script)
+            return eng.run_story("default/sec-story", {}, timeout=40)
+        finally:
+            eng.stop()
+
+    def test_env_and_file_mounts(self):
+        run = self._run(
+            "import os, json;"
+            " print(json.dumps({'key': os.environ['BUBU_SECRET_API_KEY'],"
+            " 'cert': open(os.environ['BUBU_SECRET_FILE_CERT']).read(),"
+            " 'missing': 'BUBU_SECRET_OPTIONAL_THING' in os.environ}))"
+        )
+        assert str(run.phase) == "Succeeded", run.error
+        assert run.output["out"] == {
+            "key": "sk-12345", "cert": "---CERT---", "missing": False,
+        }
+
+    def test_secret_file_cleaned_up(self):
+        run = self._run(
+            "import os, json; print(json.dumps({'path': os.environ['BUBU_SECRET_FILE_CERT']}))"
+        )
+        import os
+        assert not os.path.exists(run.output["out"]["path"])
+
+    def test_missing_required_secret_is_terminal(self):
+        yaml = self.YAML.replace('secrets: {api-key: "sk-12345", cert: "---CERT---"}',
+                                 'secrets: {cert: "x"}')
+        run = self._run("print('never runs')", yaml=yaml)
+        assert str(run.phase) == "Failed"
+        assert "api-key" in (run.error.message if run.error else str(run))

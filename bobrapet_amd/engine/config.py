@@ -66,6 +66,11 @@ class EngineConfig:
     checkpoint_path: _t.Optional[str] = None
     checkpoint_interval_seconds: float = 30.0
 
+    # telemetry export (reference: pkg/observability/exporter.go — OTLP
+    # exporter; here a file path or an OTLP/HTTP endpoint)
+    otlp_endpoint: _t.Optional[str] = None
+    otlp_flush_interval_seconds: float = 10.0
+
     # retention (reference: handleTerminalStoryRun defaults 2044/2056)
     child_ttl_seconds: float = 3600.0
     storyrun_retention_seconds: float = 86400.0

@@ -127,6 +127,14 @@ class RunEngine:
             self.set_timer(
                 monotonic_now() + self.config.checkpoint_interval_seconds, "", "checkpoint"
             )
+        if self.config.otlp_endpoint:
+            # span export (reference: observability.InitTracerProvider —
+            # exporter started with the manager, flushed at shutdown)
+            self.tracer.enabled = True
+            self._otlp = tracing_mod.OTLPExporter(self.config.otlp_endpoint)
+            self.set_timer(
+                monotonic_now() + self.config.otlp_flush_interval_seconds, "", "otlp"
+            )
         self.log.debug("engine started", feature="engine")
         return self
 
@@ -141,6 +149,8 @@ class RunEngine:
         if self._thread is not None:
             self._thread.join(timeout=5)
         self.workers.shutdown()
+        if getattr(self, "_otlp", None) is not None:
+            self._otlp.flush(self.tracer)  # LIFO shutdown flush
 
     def __enter__(self) -> "RunEngine":
         return self.start()
@@ -762,6 +772,14 @@ class RunEngine:
     # -- timers ----------------------------------------------------------
 
     def _on_timer(self, run_key: str, tag: str) -> None:
+        if tag == "otlp":
+            exporter = getattr(self, "_otlp", None)
+            if exporter is not None:
+                exporter.flush(self.tracer)
+                self.set_timer(
+                    monotonic_now() + self.config.otlp_flush_interval_seconds, "", "otlp"
+                )
+            return
         if tag == "checkpoint":
             self._checkpoint_tick()
             return

@@ -202,3 +202,42 @@ spec:
             assert '"level": "warn"' in buf.getvalue()
         finally:
             eng.stop()
+
+
+class TestOTLPExport:
+    """OTLP span export (reference: pkg/observability/exporter.go)."""
+
+    def test_engine_exports_spans_to_file(self, tmp_path):
+        import json
+
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        path = str(tmp_path / "spans.otlp.jsonl")
+        eng = RunEngine(EngineConfig(cpu_workers=1, otlp_endpoint=path)).start()
+        try:
+            eng.apply_yaml(
+                """
+kind: EngramTemplate
+metadata: {name: echo-tpl}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e}
+spec: {templateRef: {name: echo-tpl}}
+---
+kind: Story
+metadata: {name: s}
+spec:
+  steps: [{name: a, ref: {name: e}, with: {v: 1}}]
+"""
+            )
+            eng.run_story("default/s", {}, timeout=10)
+        finally:
+            eng.stop()
+        doc = json.loads(open(path).read().splitlines()[0])
+        spans = doc["resourceSpans"][0]["scopeSpans"][0]["spans"]
+        names = {s["name"] for s in spans}
+        assert "engram.run" in names and "dag.tick" in names
+        one = spans[0]
+        assert len(one["traceId"]) == 32 and len(one["spanId"]) == 16
+        assert int(one["endTimeUnixNano"]) >= int(one["startTimeUnixNano"])

@@ -110,6 +110,9 @@ class ImpulseRuntime:
         self._seq = itertools.count()
         self._lock = threading.Lock()
         self._evaluator = Evaluator()
+        # idempotent trigger-counting tokens (reference:
+        # trigger_annotations.go) — survive impulse restarts
+        self._counted_tokens: _t.Dict[str, set] = {}
 
     def start(self, impulse: _t.Union[T.Impulse, str]) -> LiveImpulse:
         """Materialize one Impulse: resolve its template to a builtin handler
@@ -146,6 +149,9 @@ class ImpulseRuntime:
             keys = list(self.live)
         for k in keys:
             self.stop(k)
+
+    def status(self, impulse_key: str, max_scan: int = 2000) -> dict:
+        return aggregate_impulse_stats(self, impulse_key, max_scan=max_scan)
 
     # ------------------------------------------------------------------
 
@@ -378,3 +384,62 @@ def serve_http(engine: "RunEngine", host: str = "127.0.0.1", port: int = 8080):
     import uvicorn
 
     uvicorn.run(build_http_app(engine), host=host, port=port, log_level="warning")
+
+
+# ---------------------------------------------------------------------------
+# trigger-stats aggregation + bounded backfill (reference:
+# internal/controller/impulse_controller.go:1151-1233 trigger stats from
+# StoryRuns, trigger_annotations.go idempotent counting tokens with
+# bounded backfill loops / bobrapet_trigger_backfill_* metrics)
+# ---------------------------------------------------------------------------
+
+
+def aggregate_impulse_stats(
+    runtime: "ImpulseRuntime",
+    impulse_key: str,
+    max_scan: int = 2000,
+):
+    """Status of one Impulse: live emit/decision counters plus a BOUNDED
+    backfill over existing StoryRuns labelled with this impulse — each
+    run's trigger token is counted exactly once (idempotent tokens kept on
+    the runtime, so repeated aggregation and impulse restarts never
+    double-count, like the reference's counting annotations)."""
+    eng = runtime.engine
+    ns, _, name = impulse_key.rpartition("/")
+    live = runtime.live.get(impulse_key)
+    counted = runtime._counted_tokens.setdefault(impulse_key, set())
+
+    scanned = 0
+    backfilled = 0
+    active = 0
+    phases: _t.Dict[str, int] = {}
+    capped = False
+    for run in eng.store.all_runs():
+        if scanned >= max_scan:
+            capped = True
+            break
+        scanned += 1
+        if run.labels.get("impulse") != name:
+            continue
+        if not run.is_terminal:
+            active += 1
+        phases[str(run.phase)] = phases.get(str(run.phase), 0) + 1
+        for tok in run.trigger_tokens:
+            if tok not in counted:
+                counted.add(tok)
+                backfilled += 1
+    eng.metrics.inc("trigger_backfill_scans_total", impulse=name)
+    if backfilled:
+        eng.metrics.inc("trigger_backfill_counted_total", impulse=name, n=backfilled)
+
+    return {
+        "impulse": impulse_key,
+        "running": live is not None,
+        "emitted": live.emitted if live else 0,
+        "decisions": dict(live.decisions) if live else {},
+        "triggers": len(counted),
+        "backfilled": backfilled,
+        "activeRuns": active,
+        "runPhases": phases,
+        "scanCapped": capped,
+    }

@@ -1026,3 +1026,63 @@ spec:
             assert len(set(names)) == 200
         finally:
             eng.stop()
+
+
+class TestImpulseTriggerStats:
+    """Trigger-stats aggregation + bounded backfill (reference:
+    impulse_controller.go:1151-1233, trigger_annotations.go)."""
+
+    YAML = """
+kind: EngramTemplate
+metadata: {name: echo-tpl-ts}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: e-ts}
+spec: {templateRef: {name: echo-tpl-ts}}
+---
+kind: Story
+metadata: {name: ts-story}
+spec:
+  steps: [{name: a, ref: {name: e-ts}, with: {v: "{{ inputs.n }}"}}]
+---
+kind: ImpulseTemplate
+metadata: {name: manual-ts}
+spec: {builtin: manual}
+---
+kind: Impulse
+metadata: {name: poker}
+spec:
+  templateRef: {name: manual-ts}
+  storyRef: {name: ts-story}
+  mapping: {n: "{{ event.n }}"}
+"""
+
+    def test_stats_and_idempotent_backfill(self):
+        from bobrapet_amd.engine import EngineConfig, RunEngine
+
+        eng = RunEngine(EngineConfig(cpu_workers=2)).start()
+        try:
+            eng.apply_yaml(self.YAML)
+            live = eng.impulses.start("default/poker")
+            for n in range(5):
+                r = live.handler.emit({"n": n})
+                eng.wait(r.story_run_ref, timeout=10)
+            st = eng.impulses.status("default/poker")
+            assert st["emitted"] == 5
+            assert st["decisions"].get("Created") == 5
+            assert st["triggers"] == 5
+            assert st["backfilled"] == 5
+            # second aggregation: tokens already counted → no double count
+            st2 = eng.impulses.status("default/poker")
+            assert st2["triggers"] == 5 and st2["backfilled"] == 0
+            # restart the impulse: counted tokens survive (idempotent)
+            eng.impulses.stop("default/poker")
+            eng.impulses.start("default/poker")
+            st3 = eng.impulses.status("default/poker")
+            assert st3["triggers"] == 5 and st3["backfilled"] == 0
+            # bounded scan reports the cap
+            st4 = eng.impulses.status("default/poker", max_scan=2)
+            assert st4["scanCapped"] is True
+        finally:
+            eng.stop()

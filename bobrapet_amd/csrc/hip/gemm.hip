@@ -214,34 +214,22 @@ __global__ __launch_bounds__(256) void gemv2_kernel(
   const int wid = threadIdx.x >> 6;
   const int waves_total = (gridDim.x * blockDim.x) >> 6;
 
+  // EPI 1/3 row scales come from a sum-of-squares accumulated IN the dot
+  // loop (x is streamed there anyway — a dedicated pre-pass measured
+  // ~2.8 TB/s effective on the qkv projection); every wave computes the
+  // same ss, so no cross-wave exchange is needed.
   float scale[M];
+  bool have_scale = (EPI != 1 && EPI != 3);
 #pragma unroll
   for (int m = 0; m < M; ++m) scale[m] = 1.0f;
-  if (EPI == 1 || EPI == 3) {
-    // row scales from the same x this wave streams for its dots (x rows
-    // are L1/L2-resident; the extra pass is noise next to the W stream)
-#pragma unroll
-    for (int m = 0; m < M; ++m) {
-      float ss = 0.f;
-      for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
-        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)m * K + k0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float f = bf2f(av[j]);
-          ss += f * f;
-        }
-      }
-      ss = wave_reduce_sum(ss);
-      scale[m] = rsqrtf(ss * stat_mul + stat_eps);
-    }
-  }
 
   for (int n = blockIdx.x * 4 + wid; n < N; n += waves_total) {
     const unsigned short* wg = W + (long)(EPI == 3 ? 2 * n : n) * K;
     const unsigned short* wu = wg + K;  // EPI 3 only
-    float accg[M], accu[M];
+    float accg[M], accu[M], ssq[M];
 #pragma unroll
-    for (int m = 0; m < M; ++m) accg[m] = accu[m] = 0.f;
+    for (int m = 0; m < M; ++m) accg[m] = accu[m] = ssq[m] = 0.f;
+    const bool want_ss = !have_scale;
     for (int k0 = lane * 8; k0 < K; k0 += WAVE * 8) {
       ushort8v gv = *reinterpret_cast<const ushort8v*>(wg + k0);
       ushort8v uv;
@@ -254,8 +242,15 @@ __global__ __launch_bounds__(256) void gemv2_kernel(
           const float a = bf2f(av[j]);
           accg[m] += a * bf2f(gv[j]);
           if (EPI == 3) accu[m] += a * bf2f(uv[j]);
+          if (want_ss) ssq[m] += a * a;
         }
       }
+    }
+    if (want_ss) {
+#pragma unroll
+      for (int m = 0; m < M; ++m)
+        scale[m] = rsqrtf(wave_reduce_sum(ssq[m]) * stat_mul + stat_eps);
+      have_scale = true;
     }
 #pragma unroll
     for (int m = 0; m < M; ++m) {

@@ -287,6 +287,10 @@ class LlamaModel:
             logits = ops.gemm256_nt(x2, self.lm_head, stat, self._inv_h, cfg.rms_eps)
             return logits.view(B, S, cfg.vocab_size)
         xl = x2.view(B, S, cfg.hidden_size)[:, -1].contiguous()
+        if B <= 8 and self.device.type == "cuda":
+            # tiny-M logits: the weight-streaming GEMV (norm entry fused)
+            # beats a 256-row-tile GEMM at M=B
+            return ops.gemv_norm(xl, self.lm_head, self._inv_h, cfg.rms_eps)
         sl = stat.view(B, S)[:, -1].contiguous()
         return ops.gemm256_nt(xl, self.lm_head, sl, self._inv_h, cfg.rms_eps)
 

@@ -246,13 +246,33 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
                     payload,
                 )
 
-        client.push_stream(frames(0, args.warmup))
+        # P concurrent client streams (the reference's connectors are
+        # many parallel gRPC channels; the engine's partition lanes absorb
+        # out-of-order arrival across pushers)
+        P = int(os.environ.get("BOBRA_STREAM_PUSHERS", "4"))
+        import concurrent.futures as cf
+
+        def push_span(first, count):
+            if P <= 1 or count < 2 * P:
+                client.push_stream(frames(first, count))
+                return
+            span = (count + P - 1) // P
+            with cf.ThreadPoolExecutor(max_workers=P) as ex:
+                futs = [
+                    ex.submit(client.push_stream,
+                              frames(first + w * span, min(span, count - w * span)))
+                    for w in range(P) if w * span < count
+                ]
+                for f in futs:
+                    f.result()
+
+        push_span(0, args.warmup)
         deadline = time.monotonic() + 60
         while client.stream_stats(session)["leafPackets"] < args.warmup and time.monotonic() < deadline:
             time.sleep(0.001)
         group.barrier()
         t0 = time.monotonic()
-        client.push_stream(frames(args.warmup, args.steps))
+        push_span(args.warmup, args.steps)
         while client.stream_stats(session)["leafPackets"] < args.warmup + args.steps and time.monotonic() < t0 + 300:
             time.sleep(0.001)
         group.barrier()

@@ -12,6 +12,7 @@
 const void* bobra_native_lane_ptr();
 bool bobra_native_tensor_get(const std::string& key, at::Tensor* out);
 size_t bobra_native_registry_size();
+void bobra_native_lane_stats(long*, long*, long*, std::string*);
 
 extern "C" {
 void launch_rmsnorm(void*, void*, void*, const void*, int, int, float, bool,
@@ -541,5 +542,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return py::cast(t);
   });
   m.def("native_registry_size", []() { return bobra_native_registry_size(); });
+  m.def("native_lane_stats", []() {
+    long l, fb, fl;
+    std::string err;
+    bobra_native_lane_stats(&l, &fb, &fl, &err);
+    py::dict d;
+    d["launches"] = l;
+    d["fallbacks"] = fb;
+    d["failures"] = fl;
+    d["lastError"] = err;
+    return d;
+  });
   m.def("argmax_rows", &argmax_rows, "row argmax over bf16 logits");
 }

@@ -76,6 +76,10 @@ struct LaneState {
   std::unordered_map<std::string, at::Tensor> cache;
   std::vector<std::vector<c10::hip::HIPStream>> streams;  // per device
   int rr = 0;
+  long launches = 0;
+  long fallbacks = 0;
+  long failures = 0;
+  std::string last_error;
 };
 
 LaneState& S() {
@@ -248,7 +252,13 @@ long join_launch(const JValue& cfg, const JValue& inp, int device) {
   std::vector<hipEvent_t> waits;
   for (const std::string& k : keys) {
     auto it = s.registry.find(k);
-    if (it == s.registry.end()) return 0;
+    if (it == s.registry.end()) {
+      // a lane-minted ref no longer held: failing loudly beats handing
+      // lane tensors to the Python join (mixed-runtime hazard)
+      s.failures++;
+      s.last_error = "native join: ref " + k + " not in the lane registry";
+      return -1;
+    }
     parts.push_back(it->second.t.reshape({-1, it->second.t.size(-1)}));
     if (it->second.ready) waits.push_back(it->second.ready);
   }
@@ -276,15 +286,42 @@ long join_launch(const JValue& cfg, const JValue& inp, int device) {
   return tk;
 }
 
+long fail_ticket(const std::string& msg) {
+  LaneState& s = S();  // mu held by caller? no — take it
+  std::lock_guard<std::mutex> g(s.mu);
+  s.failures++;
+  s.last_error = msg;
+  long tk = s.next_ticket++;
+  Ticket t;
+  t.ev = nullptr;
+  t.out = JValue(msg);  // poll() reports it as the error
+  s.tickets[tk] = std::move(t);
+  return tk;  // a positive ticket that polls straight to failure
+}
+
 long lane_launch(void*, int kind, const JValue* cfg, const JValue* input,
                  int device) {
+  long r = 0;
   try {
-    if (kind == 1) return embed_launch(*cfg, *input, device);
-    if (kind == 2) return join_launch(*cfg, *input, device);
-  } catch (const std::exception&) {
-    return 0;  // fall back to the Python launcher
+    if (kind == 1)
+      r = embed_launch(*cfg, *input, device);
+    else if (kind == 2)
+      r = join_launch(*cfg, *input, device);
+  } catch (const std::exception& e) {
+    // internal error mid-flight: fail the step loudly (falling back to
+    // the Python body after lane side effects mixes runtimes)
+    return fail_ticket(std::string("native lane: ") + e.what());
   }
-  return 0;
+  LaneState& s = S();
+  std::string err;
+  {
+    std::lock_guard<std::mutex> g(s.mu);
+    if (r > 0) s.launches++;
+    if (r == 0) s.fallbacks++;
+    if (r >= 0) return r;
+    err = s.last_error;
+  }
+  return fail_ticket(err);  // takes the mutex itself
 }
 
 int lane_poll(void*, long ticket, JValue* out, std::string* err) {
@@ -293,6 +330,12 @@ int lane_poll(void*, long ticket, JValue* out, std::string* err) {
   auto it = s.tickets.find(ticket);
   if (it == s.tickets.end()) {
     *err = "unknown native ticket";
+    return -1;
+  }
+  if (it->second.ev == nullptr) {  // pre-failed ticket
+    *err = it->second.out.is_string() ? it->second.out.as_string()
+                                      : "native lane failure";
+    s.tickets.erase(it);
     return -1;
   }
   hipError_t rc = hipEventQuery(it->second.ev);
@@ -329,12 +372,19 @@ const void* bobra_native_lane_ptr() { return &g_lane; }
 
 bool bobra_native_tensor_get(const std::string& key, at::Tensor* out) {
   LaneState& s = S();
-  std::lock_guard<std::mutex> g(s.mu);
-  auto it = s.registry.find(key);
-  if (it == s.registry.end()) return false;
-  // make the payload safe to read on the caller's stream
-  if (it->second.ready) hipEventSynchronize(it->second.ready);
-  *out = it->second.t;
+  at::Tensor t;
+  hipEvent_t ev = nullptr;
+  {
+    std::lock_guard<std::mutex> g(s.mu);
+    auto it = s.registry.find(key);
+    if (it == s.registry.end()) return false;
+    t = it->second.t;
+    ev = it->second.ready;
+  }
+  // make the payload safe to read on the caller's stream; sync OUTSIDE
+  // the lane mutex so a slow event never stalls the core's launch path
+  if (ev) (void)hipEventSynchronize(ev);
+  *out = t;
   return true;
 }
 
@@ -342,4 +392,14 @@ size_t bobra_native_registry_size() {
   LaneState& s = S();
   std::lock_guard<std::mutex> g(s.mu);
   return s.registry.size();
+}
+
+void bobra_native_lane_stats(long* launches, long* fallbacks, long* failures,
+                             std::string* last_error) {
+  LaneState& s = S();
+  std::lock_guard<std::mutex> g(s.mu);
+  *launches = s.launches;
+  *fallbacks = s.fallbacks;
+  *failures = s.failures;
+  *last_error = s.last_error;
 }

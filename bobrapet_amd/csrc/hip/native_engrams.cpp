@@ -15,6 +15,7 @@
 // mixed native/Python stories and tests keep working), and the core's
 // gc_run frees them via free_key.
 #include <ATen/ATen.h>
+#include <c10/hip/HIPCachingAllocator.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
@@ -267,7 +268,10 @@ long join_launch(const JValue& cfg, const JValue& inp, int device) {
   for (hipEvent_t ev : waits) hipStreamWaitEvent(stream.stream(), ev, 0);
   c10::hip::setCurrentHIPStream(stream);
   at::Tensor local = at::cat(parts, 0);
-  for (auto& p : parts) p.record_stream(stream.unwrap());
+  // allocator stream-use records via the HIP API (Tensor::record_stream
+  // rejects the raw HIP stream under the masquerading layer)
+  for (auto& p : parts)
+    c10::hip::HIPCachingAllocator::recordStream(p.storage().data_ptr(), stream);
   // world == 1 (the Python glue only routes the join natively then):
   // the all-gather is the identity — `joined` IS the local concat
   hipEvent_t ev;

@@ -249,7 +249,7 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
         # P concurrent client streams (the reference's connectors are
         # many parallel gRPC channels; the engine's partition lanes absorb
         # out-of-order arrival across pushers)
-        P = int(os.environ.get("BOBRA_STREAM_PUSHERS", "4"))
+        P = int(os.environ.get("BOBRA_STREAM_PUSHERS", "1"))  # >1 measured SLOWER (GIL vs stage threads)
         import concurrent.futures as cf
 
         def push_span(first, count):

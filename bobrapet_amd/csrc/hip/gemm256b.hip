@@ -164,8 +164,7 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
-  asm volatile("" ::: "memory");
+  // (the loop's q0 barrier propagates this wait)
 
   // per phase q: read quadrant q's fragments (+ the tile's B at q0),
   // issue one half-tile stage, then barrier -> the read latency hides
@@ -195,7 +194,16 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     const char* bslot = bring + (t & 1) * G3_HALF;
     const bool tail = t + 2 >= nt;
 
-    // ---- phase q=0: read B(t) + A quad 0; stage B1(t+1)
+    // ---- phase q=0: barrier (propagates the q3 vmcnt: tile t's halves
+    // are landed for every wave), then reads B(t) + A quad 0; stage
+    // B1(t+1).  Barriers run every TWO phases only — they exist to bound
+    // wave skew for slot reuse (stage is always >=3 phases after a
+    // slot's last read) and to propagate the per-tile vmcnt; the
+    // read->MFMA edge is per-wave (hipcc's counted lgkm).  The half-
+    // phase drift lets one wave's reads overlap its SIMD partner's MFMA
+    // cluster (matrix-beside-memory pairing).
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
 #pragma unroll
     for (int j = 0; j < 4; ++j)
 #pragma unroll
@@ -205,19 +213,14 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, i, ks);
     if (t + 1 < nt) stage(3, 8 + ((t + 1) & 1), t + 1);
-    asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");  // pace: first 4 reads
-    __builtin_amdgcn_s_barrier();
-    asm volatile("" ::: "memory");
     G3_MFMA_QUAD(0)
 
-    // ---- phase q=1: read A quad 1; stage A0(t+2)
+    // ---- phase q=1 (no barrier): read A quad 1; stage A0(t+2)
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 2 + i, ks);
     if (t + 2 < nt) stage(0, (t + 2) % 3, t + 2);
-    __builtin_amdgcn_s_barrier();
-    asm volatile("" ::: "memory");
     G3_MFMA_QUAD(2)
 
     // ---- phase q=2: read A quad 2; stage A1(t+2)
@@ -230,7 +233,8 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     asm volatile("" ::: "memory");
     G3_MFMA_QUAD(4)
 
-    // ---- phase q=3: read A quad 3; stage B0(t+2); tile-boundary vmcnt
+    // ---- phase q=3 (no barrier): read A quad 3; stage B0(t+2); the
+    // tile-boundary vmcnt sits here, propagated by the next q0 barrier
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -240,8 +244,6 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
       asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
-    asm volatile("" ::: "memory");
     G3_MFMA_QUAD(6)
   }
 #undef G3_MFMA_QUAD

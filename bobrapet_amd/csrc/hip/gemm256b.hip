@@ -166,6 +166,13 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   // (the loop's q0 barrier propagates this wait)
 
+  // T5 static form: ONE priority raise for the second-dispatched half —
+  // the arbitration loser on every segment — and no per-cluster flips
+  // (guide: per-cluster setprio is sub-additive with this and the
+  // readfirstlane guard is required for a truly scalar s_setprio)
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
   // per phase q: read quadrant q's fragments (+ the tile's B at q0),
   // issue one half-tile stage, then barrier -> the read latency hides
   // under the barrier wait and hipcc's counted lgkm before the MFMAs;

@@ -193,6 +193,14 @@ spec:
 """
 
 
+# round-1 published values (BASELINE.md "numbers to beat"): vs_baseline =
+# value / this number at N=1
+_R1_BASELINE = {
+    "parallel8": 819.0, "cpu": 51600.0, "llm": 10.1,
+    "stream": 6680.0, "bigpayload": 424.0,
+}
+
+
 def run_one(eng: RunEngine, story_key: str, idx: int, rank: int, native=None, slot=None) -> dict:
     inputs = {"i": idx, "rank": rank, "slot": slot if slot is not None else 0}
     if native is not None:
@@ -328,7 +336,7 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
             "ms_per_step": round(elapsed_max * 1000.0 / args.steps, 3),
             "higher_is_better": True,
             "scaling": "weak",
-            "vs_baseline": None,
+            "vs_baseline": round(args.steps * world / elapsed_max / _R1_BASELINE["stream"], 3),
             "dtype": "bf16",
             "data": "synthetic",
             "config": {
@@ -518,7 +526,7 @@ def main() -> int:
                 "ms_per_step": round(ms_per_step, 3),
                 "higher_is_better": True,
                 "scaling": "weak",
-                "vs_baseline": None,
+                "vs_baseline": round(runs_per_sec / _R1_BASELINE.get(config_name, 1.0), 3) if config_name in _R1_BASELINE else None,
                 "dtype": "bf16",
                 "data": "synthetic",
                 "config": {

@@ -156,6 +156,12 @@ def serve(
         None, help="state snapshot path: restored on boot, saved periodically"
     ),
     checkpoint_interval: float = typer.Option(30.0, help="snapshot period (s)"),
+    grpc_port: _t.Optional[int] = typer.Option(
+        None, help="also serve the gRPC ingress (triggers + streaming packets)"
+    ),
+    otlp: _t.Optional[str] = typer.Option(
+        None, help="OTLP span export: a file path or http://host:4318/v1/traces"
+    ),
 ):
     """Start the engine + REST control plane / impulse ingress."""
     import os
@@ -168,6 +174,8 @@ def serve(
     if checkpoint:
         cfg.checkpoint_path = checkpoint
         cfg.checkpoint_interval_seconds = checkpoint_interval
+    if otlp:
+        cfg.otlp_endpoint = otlp
     eng = RunEngine(cfg).start()
     objs = _load_files(file) if file else []
     for obj in objs:
@@ -180,6 +188,11 @@ def serve(
     if checkpoint and os.path.exists(checkpoint):
         n = eng.load_state(checkpoint)
         typer.echo(f"restored {n} runs from {checkpoint}")
+    if grpc_port is not None:
+        from .engine.ingress_grpc import serve_grpc
+
+        _server, bound = serve_grpc(eng, port=grpc_port)
+        typer.echo(f"gRPC ingress on 127.0.0.1:{bound}")
     typer.echo(f"serving on http://{host}:{port}")
     serve_http(eng, host=host, port=port)
 

@@ -158,6 +158,22 @@ void NativeEngine::sync_primitives(Run& run, const Plan& plan) {
         st.error = "gate rejected";
         st.finished = t;
       }
+    } else if (step.kind == StepKind::Wait && st.phase == Phase::Running) {
+      // Event-driven wakeup: an until-condition only references run state
+      // (steps.*, inputs), which changes exclusively on ticks — so re-check
+      // here and the wait resolves on the same tick its dependency
+      // completes.  The armed poll timer (tag 2) stays as a fallback and
+      // no-ops once the phase is terminal.
+      JObject scope = build_scope(run, plan);
+      bool done = false;
+      try {
+        done = step.until && eval_expr(*step.until, scope).truthy();
+      } catch (const ExprError&) {
+      }
+      if (done) {
+        st.phase = Phase::Succeeded;
+        st.finished = t;
+      }
     } else if (step.kind == StepKind::ExecuteStory &&
                st.phase == Phase::Running && st.output.is_object()) {
       const JValue child_id = st.output.get("childRun");

@@ -245,6 +245,31 @@ spec:
         assert st["phase"] == "Succeeded"
         assert st["steps"]["watch"]["phase"] == "Succeeded"
 
+    def test_wait_event_driven_wakeup(self, rig):
+        """A wait over a completing dependency resolves on the dependency's
+        own tick, not on the next poll timer: with a 5 s pollInterval the run
+        still finishes in milliseconds (engine.cpp sync_primitives Wait)."""
+        eng, nr = rig
+        _apply(
+            eng,
+            """
+kind: Story
+metadata: {name: n8b}
+spec:
+  steps:
+    - {name: slow, type: sleep, with: {duration: 30ms}}
+    - name: watch
+      type: wait
+      with: {until: "{{ steps.slow.phase == 'Succeeded' }}", pollInterval: 5s, timeout: 30s}
+""",
+        )
+        t0 = time.monotonic()
+        st = nr.run_story("default/n8b", {}, timeout=10)
+        elapsed = time.monotonic() - t0
+        assert st["phase"] == "Succeeded"
+        assert st["steps"]["watch"]["phase"] == "Succeeded"
+        assert elapsed < 2.0, f"wait waited for the poll timer: {elapsed:.3f}s"
+
     def test_gate_approved(self, rig):
         eng, nr = rig
         story = _apply(

@@ -48,6 +48,8 @@ void launch_gemm256(int, void*, const void*, const void*, const void*,
                     hipStream_t);
 void launch_gemv2(int, void*, const void*, const void*, const void*, int, int,
                   int, float, float, hipStream_t);
+void launch_gemm256b_disc(void*, const void*, const void*, int, int, int,
+                          hipStream_t);
 void launch_gemm256b(int, void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, float, float,
                      hipStream_t);
@@ -369,6 +371,17 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+torch::Tensor gemm256_nt_disc(torch::Tensor a, torch::Tensor b) {
+  // probe-only: the guide-template per-phase barrier discipline (A/B)
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  check_gemm256(a, b, M, N, K);
+  TORCH_CHECK(K % 64 == 0, "gemm256_nt_disc: K % 64 required");
+  auto c = torch::empty({M, N}, a.options());
+  launch_gemm256b_disc(c.data_ptr(), a.data_ptr(), b.data_ptr(), M, N, K,
+                       cur_stream());
+  return c;
+}
+
 torch::Tensor gemm256_swiglu(torch::Tensor a, torch::Tensor b,
                              c10::optional<torch::Tensor> stat,
                              double stat_mul, double stat_eps) {
@@ -515,6 +528,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_variant", &attn_prefill_variant,
         "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
+  m.def("gemm256_nt_disc", &gemm256_nt_disc);
   m.def("gemm256_nt", &gemm256_nt,
         "256-tile bf16 MFMA GEMM, optional fused row-scale epilogue");
   m.def("gemm256_swiglu", &gemm256_swiglu,

@@ -39,7 +39,12 @@ __device__ __forceinline__ float g3_silu(float g) {
   return g / (1.0f + __expf(-g));
 }
 
-template <int EPI>
+// DISC selects the synchronization discipline (same math, same staging):
+//   0 = shipped: barriers at q0/q2 only, skew-tolerant (default)
+//   1 = guide-template: per-phase barrier PAIR around each MFMA cluster
+//       with an explicit lgkmcnt(0) drain before it (8 barriers/tile) —
+//       convoyed waves, kept behind a probe launcher for A/B measurement
+template <int EPI, int DISC = 0>
 __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     unsigned short* __restrict__ C,
     const unsigned short* __restrict__ A,
@@ -180,6 +185,10 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
   // propagated to every wave by q3's barriers (template: "vmcnt at
   // phases 4 and 8 only, never 0 in the main loop").
 #define G3_MFMA_QUAD(base)                                              \
+    if (DISC == 1) {                                                    \
+      __builtin_amdgcn_s_barrier();                                     \
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");                \
+    }                                                                   \
     __builtin_amdgcn_s_setprio(1);                                      \
     _Pragma("unroll")                                                   \
     for (int ks = 0; ks < 2; ++ks)                                      \
@@ -189,7 +198,8 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
         for (int j = 0; j < 4; ++j)                                     \
           acc[(base) + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16( \
               aq[i][ks], bfr[j][ks], acc[(base) + i][j], 0, 0, 0);      \
-    __builtin_amdgcn_s_setprio(0);
+    __builtin_amdgcn_s_setprio(0);                                      \
+    if (DISC == 1) __builtin_amdgcn_s_barrier();
   // NOTE: no barrier after the MFMA cluster — the next phase's reads may
   // run while the SIMD partner is still in this cluster (complementary
   // matrix-beside-memory pairing).  Safe: per-phase barrier-1 bounds the
@@ -209,7 +219,7 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     // read->MFMA edge is per-wave (hipcc's counted lgkm).  The half-
     // phase drift lets one wave's reads overlap its SIMD partner's MFMA
     // cluster (matrix-beside-memory pairing).
-    __builtin_amdgcn_s_barrier();
+    if (DISC == 0) __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
 #pragma unroll
     for (int j = 0; j < 4; ++j)
@@ -236,7 +246,7 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 4 + i, ks);
     if (t + 2 < nt) stage(1, 3 + (t + 2) % 3, t + 2);
-    __builtin_amdgcn_s_barrier();
+    if (DISC == 0) __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
     G3_MFMA_QUAD(4)
 
@@ -365,4 +375,16 @@ extern "C" void launch_gemm256b(int epi, void* C, const void* A, const void* B,
     case 1: launch(gemm256b_kernel<1>); break;
     default: launch(gemm256b_kernel<2>); break;
   }
+}
+
+// probe-only entry: the guide-template per-phase barrier discipline
+// (DISC=1) for within-process A/B against the shipped schedule
+extern "C" void launch_gemm256b_disc(void* C, const void* A, const void* B,
+                                     int M, int N, int K, hipStream_t stream) {
+  const int nbm = (M + 255) / 256;
+  dim3 grid(nbm * (N / 256)), block(512);
+  hipLaunchKernelGGL((gemm256b_kernel<0, 1>), grid, block, 0, stream,
+                     (unsigned short*)C, (const unsigned short*)A,
+                     (const unsigned short*)B, nullptr, nullptr, nullptr, M, N,
+                     K, 1.0f, 1e-6f);
 }

@@ -485,9 +485,31 @@ __global__ __launch_bounds__(256, 4) void attn_decode_chunk_kernel(
 #pragma unroll
   for (int g = 0; g < DEC_GMAX; ++g) o0[g] = o1[g] = 0.f;
   const int rows = min(WAVE, kv_end - (kv0 + wid * WAVE));
-  for (int rr = 0; rr < rows; ++rr) {
-    int row = kv0 + wid * WAVE + rr;
-    ushort2 v2 = *reinterpret_cast<const ushort2*>(vc + base + (long)row * D_HEAD + lane * 2);
+  // 8-row batches: the per-row 4-B load is latency-bound when issued one
+  // at a time (the runtime trip count blocks compiler unrolling); batching
+  // 8 independent loads ahead of the FMAs keeps ~8 in flight and removes
+  // the ~2-4x latency multiplier this loop carried at decode L
+  const unsigned short* vrow0 = vc + base + (long)(kv0 + wid * WAVE) * D_HEAD + lane * 2;
+  int rr = 0;
+  for (; rr + 8 <= rows; rr += 8) {
+    ushort2 v8[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+      v8[u] = *reinterpret_cast<const ushort2*>(vrow0 + (long)(rr + u) * D_HEAD);
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float vx = bf2f(v8[u].x), vy = bf2f(v8[u].y);
+#pragma unroll
+      for (int g = 0; g < DEC_GMAX; ++g) {
+        if (g >= G) break;
+        float pv = p_lds[wid][g][rr + u];
+        o0[g] += pv * vx;
+        o1[g] += pv * vy;
+      }
+    }
+  }
+  for (; rr < rows; ++rr) {
+    ushort2 v2 = *reinterpret_cast<const ushort2*>(vrow0 + (long)rr * D_HEAD);
     float vx = bf2f(v2.x), vy = bf2f(v2.y);
 #pragma unroll
     for (int g = 0; g < DEC_GMAX; ++g) {

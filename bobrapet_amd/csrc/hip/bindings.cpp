@@ -48,8 +48,8 @@ void launch_gemm256(int, void*, const void*, const void*, const void*,
                     hipStream_t);
 void launch_gemv2(int, void*, const void*, const void*, const void*, int, int,
                   int, float, float, hipStream_t);
-void launch_gemm256b_disc(void*, const void*, const void*, int, int, int,
-                          hipStream_t);
+void launch_gemm256b_disc(int, void*, const void*, const void*, int, int,
+                          int, hipStream_t);
 void launch_gemm256b(int, void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, float, float,
                      hipStream_t);
@@ -371,14 +371,15 @@ torch::Tensor gemm256_nt(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
-torch::Tensor gemm256_nt_disc(torch::Tensor a, torch::Tensor b) {
-  // probe-only: the guide-template per-phase barrier discipline (A/B)
+torch::Tensor gemm256_nt_disc(torch::Tensor a, torch::Tensor b,
+                              int64_t disc) {
+  // probe-only: alternate sync disciplines (1=template pairs, 2=4/tile)
   const int M = a.size(0), K = a.size(1), N = b.size(0);
   check_gemm256(a, b, M, N, K);
   TORCH_CHECK(K % 64 == 0, "gemm256_nt_disc: K % 64 required");
   auto c = torch::empty({M, N}, a.options());
-  launch_gemm256b_disc(c.data_ptr(), a.data_ptr(), b.data_ptr(), M, N, K,
-                       cur_stream());
+  launch_gemm256b_disc((int)disc, c.data_ptr(), a.data_ptr(), b.data_ptr(), M,
+                       N, K, cur_stream());
   return c;
 }
 

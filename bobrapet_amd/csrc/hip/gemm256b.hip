@@ -44,6 +44,7 @@ __device__ __forceinline__ float g3_silu(float g) {
 //   1 = guide-template: per-phase barrier PAIR around each MFMA cluster
 //       with an explicit lgkmcnt(0) drain before it (8 barriers/tile) —
 //       convoyed waves, kept behind a probe launcher for A/B measurement
+//   2 = barrier at every phase START (4/tile, no post-cluster barriers)
 template <int EPI, int DISC = 0>
 __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     unsigned short* __restrict__ C,
@@ -169,7 +170,10 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  // (the loop's q0 barrier propagates this wait)
+  // (the loop's q0 barrier propagates this wait; DISC=1 removes that
+  // barrier, so it needs one HERE — without it tile-0 q0 reads race the
+  // other waves' prologue glds, measured nan/0.15 relerr)
+  if (DISC == 1) __builtin_amdgcn_s_barrier();
 
   // T5 static form: ONE priority raise for the second-dispatched half —
   // the arbitration loser on every segment — and no per-cluster flips
@@ -219,7 +223,7 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     // read->MFMA edge is per-wave (hipcc's counted lgkm).  The half-
     // phase drift lets one wave's reads overlap its SIMD partner's MFMA
     // cluster (matrix-beside-memory pairing).
-    if (DISC == 0) __builtin_amdgcn_s_barrier();
+    if (DISC == 0 || DISC == 2) __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
 #pragma unroll
     for (int j = 0; j < 4; ++j)
@@ -232,7 +236,8 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     if (t + 1 < nt) stage(3, 8 + ((t + 1) & 1), t + 1);
     G3_MFMA_QUAD(0)
 
-    // ---- phase q=1 (no barrier): read A quad 1; stage A0(t+2)
+    // ---- phase q=1 (no barrier at DISC 0): read A quad 1; stage A0(t+2)
+    if (DISC == 2) __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -246,12 +251,13 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) aq[i][ks] = rd_a(aslot, 4 + i, ks);
     if (t + 2 < nt) stage(1, 3 + (t + 2) % 3, t + 2);
-    if (DISC == 0) __builtin_amdgcn_s_barrier();
+    if (DISC == 0 || DISC == 2) __builtin_amdgcn_s_barrier();
     asm volatile("" ::: "memory");
     G3_MFMA_QUAD(4)
 
-    // ---- phase q=3 (no barrier): read A quad 3; stage B0(t+2); the
-    // tile-boundary vmcnt sits here, propagated by the next q0 barrier
+    // ---- phase q=3 (no barrier at DISC 0): read A quad 3; stage B0(t+2);
+    // the tile-boundary vmcnt sits here, propagated by the next q0 barrier
+    if (DISC == 2) __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -379,12 +385,16 @@ extern "C" void launch_gemm256b(int epi, void* C, const void* A, const void* B,
 
 // probe-only entry: the guide-template per-phase barrier discipline
 // (DISC=1) for within-process A/B against the shipped schedule
-extern "C" void launch_gemm256b_disc(void* C, const void* A, const void* B,
-                                     int M, int N, int K, hipStream_t stream) {
+extern "C" void launch_gemm256b_disc(int disc, void* C, const void* A,
+                                     const void* B, int M, int N, int K,
+                                     hipStream_t stream) {
   const int nbm = (M + 255) / 256;
   dim3 grid(nbm * (N / 256)), block(512);
-  hipLaunchKernelGGL((gemm256b_kernel<0, 1>), grid, block, 0, stream,
-                     (unsigned short*)C, (const unsigned short*)A,
-                     (const unsigned short*)B, nullptr, nullptr, nullptr, M, N,
-                     K, 1.0f, 1e-6f);
+  auto go = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
+                       (const unsigned short*)A, (const unsigned short*)B,
+                       nullptr, nullptr, nullptr, M, N, K, 1.0f, 1e-6f);
+  };
+  if (disc == 1) go(gemm256b_kernel<0, 1>);
+  else go(gemm256b_kernel<0, 2>);
 }

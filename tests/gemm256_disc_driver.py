@@ -12,7 +12,9 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 from bobrapet_amd import ops
-from bobrapet_amd.ops import _hipops as hip
+
+hip = ops._try_load()
+assert hip is not None, ops._load_error
 
 
 def timed(fn, iters):
@@ -34,13 +36,14 @@ def main():
         a = (torch.randn(m, k, dtype=torch.bfloat16, device=dev) * 0.3).contiguous()
         b = (torch.randn(n, k, dtype=torch.bfloat16, device=dev) * 0.3).contiguous()
         flops = 2.0 * m * n * k
-        # numerics: DISC=1 vs DISC=0 must agree exactly (same math order)
+        # numerics: all disciplines must agree exactly (same math order)
         d0 = ops.gemm256_nt(a[:512], b)
-        d1 = hip.gemm256_nt_disc(a[:512], b)
-        same = torch.equal(d0, d1)
+        same = all(torch.equal(d0, hip.gemm256_nt_disc(a[:512], b, d))
+                   for d in (1, 2))
         variants = {
             "disc0": lambda: ops.gemm256_nt(a, b),
-            "disc1": lambda: hip.gemm256_nt_disc(a, b),
+            "disc1": lambda: hip.gemm256_nt_disc(a, b, 1),
+            "disc2": lambda: hip.gemm256_nt_disc(a, b, 2),
             "blaslt": lambda: torch.matmul(a, b.t()),
         }
         for fn in variants.values():  # warmup

@@ -232,7 +232,8 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
     to in-process pushes (ablation)."""
     import numpy as np
 
-    from bobrapet_amd.engine.ingress_grpc import IngressClient, pack_frame, serve_grpc
+    from bobrapet_amd.engine.ingress_grpc import (IngressClient, pack_batch,
+                                                   pack_frame, serve_grpc)
 
     use_grpc = os.environ.get("BOBRA_STREAM_INPROC") != "1"
     ids_np = np.random.default_rng(7).integers(
@@ -246,13 +247,23 @@ def run_stream_bench(eng, args, rank, world, n_gpus) -> int:
         client = IngressClient(f"127.0.0.1:{port}")
         session = f"bench-{rank}"
 
+        # transport batching: FB frames per gRPC message (server fans them
+        # out) — amortizes the ~40 us per-message envelope
+        FB = max(1, int(os.environ.get("BOBRA_STREAM_FRAMEBATCH", "16")))
+
         def frames(first, count):
+            batch = []
             for i in range(first, first + count):
-                yield pack_frame(
+                batch.append(pack_frame(
                     {"stream": "default/bench-stream", "session": session,
                      "seq": i, "meta": {"items": [{}]}, "tensor": tensor_spec},
                     payload,
-                )
+                ))
+                if len(batch) == FB:
+                    yield batch[0] if FB == 1 else pack_batch(batch)
+                    batch = []
+            if batch:
+                yield batch[0] if len(batch) == 1 else pack_batch(batch)
 
         # P concurrent client streams (the reference's connectors are
         # many parallel gRPC channels; the engine's partition lanes absorb

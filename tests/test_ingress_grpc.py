@@ -76,6 +76,30 @@ def test_stream_packets_through_3_stages(served):
     assert fin["phase"] == "Finished", fin
 
 
+def test_stream_batched_frames(served):
+    # transport batching: several packet frames bundled into one gRPC
+    # message via pack_batch; the server fans them back out in order
+    from bobrapet_amd.engine.ingress_grpc import pack_batch
+
+    eng, client = served
+    n = 30
+    frames = [
+        pack_frame({"stream": "default/pipe3", "session": "sb", "seq": i,
+                    "meta": {"items": [{}], "seq": i}})
+        for i in range(n)
+    ]
+    msgs = [pack_batch(frames[i : i + 8]) for i in range(0, n, 8)]
+    out = client.push_stream(msgs)
+    assert out["pushed"] == n
+    import time
+    deadline = time.time() + 15
+    while client.stream_stats("sb")["leafPackets"] < n and time.time() < deadline:
+        time.sleep(0.01)
+    assert client.stream_stats("sb")["leafPackets"] == n
+    fin = client.finish_stream("sb")
+    assert fin["phase"] == "Finished", fin
+
+
 def test_packet_with_tensor_payload(served):
     eng, client = served
     arr = np.arange(12, dtype=np.int32).reshape(3, 4)

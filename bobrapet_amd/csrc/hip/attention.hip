@@ -363,6 +363,307 @@ extern "C" void launch_attn_prefill(void* out, const void* q, const void* k,
                      Hkv, S, v_sstride, scale, causal, (float*)stats);
 }
 
+// ---------------------------------------------------------------------------
+// Pipelined prefill: QK(t) MFMAs are ISSUED, then softmax+PV of tile t-1
+// run in their shadow — the softmax VALU (47% of the lockstep kernel,
+// profiles/attn_ablation_r01c.txt) overlaps the wave's own in-flight
+// MFMAs instead of stalling the matrix pipe.  Costs: the score tile is
+// double-buffered in registers (+32 VGPR) and V is TRIPLE-buffered in
+// LDS (PV reads tile t-1 while tile t+1 stages), so occupancy is 1
+// block/CU (2 waves/SIMD) — the overlap no longer depends on 4-wave
+// co-residency.  K stays double-buffered.  Same fragment maps, masking,
+// defer-max softmax and (m,l) stats as attn_prefill_kernel.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512, 1) void attn_prefill_pipe_kernel(
+    unsigned short* __restrict__ out,
+    const unsigned short* __restrict__ q,
+    const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v,
+    int B, int Hq, int Hkv, int S, long v_sstride, float scale, int causal,
+    float* __restrict__ stats) {
+  __shared__ __attribute__((aligned(16))) char smem[2 * KVBLK * 256 + 3 * D_HEAD * 128];
+  auto k_buf = [&](int i) -> char* { return smem + i * (KVBLK * 256); };
+  auto v_buf = [&](int i) -> char* { return smem + 2 * KVBLK * 256 + i * (D_HEAD * 128); };
+
+  const int wg = blockIdx.x;
+  const int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
+  const int nbh = gridDim.x / nqblk;
+  const int bh = wg % nbh;
+  const int qblk = nqblk - 1 - (wg / nbh);
+  const int b = bh / Hq;
+  const int hq = bh % Hq;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int hi = lane >> 5;
+  const int l31 = lane & 31;
+
+  const long q_sstride = (long)Hq * D_HEAD;
+  const long kv_sstride = (long)Hkv * D_HEAD;
+  const long q_base = (long)b * S * q_sstride + (long)hq * D_HEAD;
+  const long kv_base = (long)b * S * kv_sstride + (long)hkv * D_HEAD;
+  const long v_base = (long)b * S * v_sstride + (long)hkv * D_HEAD;
+
+  const int qbase = qblk * WG_QROWS;
+  const int my_q = qbase + NWAVES * l31 + wid;
+
+  bf16x8 qf[8];
+  {
+    const unsigned short* qrow = q + q_base + (long)my_q * q_sstride;
+    bool valid = my_q < S;
+#pragma unroll
+    for (int s = 0; s < 8; ++s) {
+      int d0 = s * 16 + hi * 8;
+      if (valid)
+        qf[s] = *reinterpret_cast<const bf16x8*>(qrow + d0);
+      else
+        qf[s] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  }
+
+  float m_run = -1e30f;
+  float l_run = 0.f;
+  f32x16 o_acc[4] = {};
+
+  const int q_hi_wg = qbase + WG_QROWS - 1;
+  int kv_end = S;
+  if (causal) kv_end = min(S, q_hi_wg + 1);
+  const int my_q_hi = qbase + NWAVES * 31 + wid;
+  const int ntiles = (kv_end + KVBLK - 1) / KVBLK;
+
+  ushort8v kreg[2], vreg[2];
+  auto load_tile = [&](int tile) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int chunk = it * 512 + tid;
+      int kv = chunk >> 4;
+      int byte = (chunk & 15) * 16;
+      int kvg = tile * KVBLK + kv;
+      if (kvg < S)
+        kreg[it] = *reinterpret_cast<const ushort8v*>(
+            k + kv_base + (long)kvg * kv_sstride + byte / 2);
+      else
+        kreg[it] = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
+      int d0 = (chunk & 15) * 8;
+      if (kvg < S)
+        vreg[it] = *reinterpret_cast<const ushort8v*>(
+            v + v_base + (long)kvg * v_sstride + d0);
+      else
+        vreg[it] = ushort8v{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+  };
+  auto write_tile = [&](char* kbuf, char* vbuf) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      int chunk = it * 512 + tid;
+      int kv = chunk >> 4;
+      int byte = (chunk & 15) * 16;
+      *reinterpret_cast<ushort8v*>(kbuf + k_lds_off(kv, byte)) = kreg[it];
+      int d0 = (chunk & 15) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<unsigned short*>(vbuf + vt_lds_off(d0 + j, kv * 2)) =
+            vreg[it][j];
+    }
+  };
+
+  if (ntiles > 0) {
+    load_tile(0);
+    write_tile(k_buf(0), v_buf(0));
+    if (ntiles > 1) load_tile(1);
+    __syncthreads();
+  }
+
+  f32x16 st_p[2];        // tile t-1 scores, softmaxed in tile t's shadow
+  int pt = -1;           // which tile st_p holds (-1: none)
+  const float sc2 = scale * 1.44269504f;
+
+  // softmax(st_p for tile pt) + PV from v_buf(pt % 3); runs while the
+  // CURRENT tile's QK MFMAs are in flight (st_p is complete: its chain
+  // retired before the current issue began)
+  auto softmax_pv = [&]() {
+    const int kv0 = pt * KVBLK;
+    const bool edge = (kv0 + KVBLK > S) || (causal && (kv0 + KVBLK - 1) > my_q);
+    float m_tile = -1e30f;
+    if (edge) {
+#pragma unroll
+      for (int tt = 0; tt < 2; ++tt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int kvg = kv0 + tt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float sv = st_p[tt][r] * sc2;
+          if (kvg >= S || (causal && kvg > my_q)) sv = -1e30f;
+          st_p[tt][r] = sv;
+          m_tile = fmaxf(m_tile, sv);
+        }
+    } else {
+#pragma unroll
+      for (int tt = 0; tt < 2; ++tt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float sv = st_p[tt][r] * sc2;
+          st_p[tt][r] = sv;
+          m_tile = fmaxf(m_tile, sv);
+        }
+    }
+    m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, WAVE));
+
+    const float THR2 = 8.0f;
+    bool defer = __builtin_amdgcn_wave_reduce_and_b32(
+                     (m_run > -1e30f) && (m_tile - m_run <= THR2), 0) != 0;
+    float m_new;
+    if (defer) {
+      m_new = m_run;
+    } else {
+      m_new = fmaxf(m_run, m_tile);
+      float alpha = (m_run <= -1e30f) ? 0.f : __builtin_amdgcn_exp2f(m_run - m_new);
+      if (m_new <= -1e30f) alpha = 1.f;
+      float alpha_row[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+        alpha_row[r] = __shfl(alpha, qrow, WAVE);
+      }
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha_row[r];
+      l_run *= alpha;
+      m_run = m_new;
+    }
+
+    float p_sum = 0.f;
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float pv = (st_p[tt][r] <= -1e30f)
+                       ? 0.f
+                       : __builtin_amdgcn_exp2f(st_p[tt][r] - m_new);
+        st_p[tt][r] = pv;
+        p_sum += pv;
+      }
+    p_sum += __shfl_xor(p_sum, 32, WAVE);
+    l_run += p_sum;
+
+    bf16x8 pa[4];
+#pragma unroll
+    for (int tt = 0; tt < 2; ++tt) {
+      unsigned int pk[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        pk[j] = pack_bf16(st_p[tt][2 * j], st_p[tt][2 * j + 1]);
+      {
+        auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+        auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+        unsigned int w0 = r0[0], w2 = r0[1], w1 = r1[0], w3 = r1[1];
+        pa[tt * 2] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
+      }
+      {
+        auto r0 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+        auto r1 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+        unsigned int w0 = r0[0], w2 = r0[1], w1 = r1[0], w3 = r1[1];
+        pa[tt * 2 + 1] = __builtin_bit_cast(bf16x8, (uint4{w0, w1, w2, w3}));
+      }
+    }
+
+    const char* vb = v_buf(pt % 3);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int ss = 0; ss < 4; ++ss) {
+      int byte = (ss * 16 + hi * 8) * 2;
+#pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        int d = dt * 32 + l31;
+        bf16x8 vf = *reinterpret_cast<const bf16x8*>(vb + vt_lds_off(d, byte));
+        o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa[ss], vf,
+                                                            o_acc[dt], 0, 0, 0);
+      }
+    }
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KVBLK;
+    if (t + 1 < ntiles) {
+      // stage K(t+1) into kb((t+1)&1) (last read: QK(t-1), before the
+      // previous barrier) and V(t+1) into vb((t+1)%3) (last read:
+      // PV(t-2), two barriers back)
+      write_tile(k_buf((t + 1) & 1), v_buf((t + 1) % 3));
+      if (t + 2 < ntiles) load_tile(t + 2);
+    }
+
+    const bool compute = (!causal) || (kv0 <= my_q_hi);
+    if (compute) {
+      const char* kb = k_buf(t & 1);
+      f32x16 st_c[2] = {};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int ss = 0; ss < 8; ++ss) {
+        int byte = (ss * 16 + hi * 8) * 2;
+#pragma unroll
+        for (int tt = 0; tt < 2; ++tt) {
+          int kv = tt * 32 + l31;
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(kb + k_lds_off(kv, byte));
+          st_c[tt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[ss], st_c[tt], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
+      // tile t's MFMAs are now in flight; retire tile t-1 under them
+      if (pt >= 0) softmax_pv();
+      st_p[0] = st_c[0];
+      st_p[1] = st_c[1];
+      pt = t;
+    } else if (pt >= 0) {
+      softmax_pv();
+      pt = -1;
+    }
+    __syncthreads();
+  }
+  if (pt >= 0) softmax_pv();  // flush: v_buf(pt%3) is untouched after
+                              // the last stage (write_tile stops at
+                              // ntiles-1, and (ntiles-1)%3 was staged
+                              // two iterations before its PV here)
+
+  float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
+  float inv_for_row[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    inv_for_row[r] = __shfl(inv_l, qrow, WAVE);
+  }
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    int qg = qbase + NWAVES * qrow + wid;
+    if (qg >= S) continue;
+    unsigned short* orow = out + q_base + (long)qg * q_sstride;
+#pragma unroll
+    for (int dt = 0; dt < 4; ++dt)
+      orow[dt * 32 + l31] = f2bf(o_acc[dt][r] * inv_for_row[r]);
+  }
+  if (stats != nullptr && hi == 0 && my_q < S) {
+    float2 ml{m_run, l_run};
+    *reinterpret_cast<float2*>(stats + ((long)bh * S + my_q) * 2) = ml;
+  }
+}
+
+extern "C" void launch_attn_prefill_pipe(void* out, const void* q,
+                                         const void* k, const void* v, int B,
+                                         int Hq, int Hkv, int S,
+                                         long v_sstride, float scale,
+                                         int causal, void* stats,
+                                         hipStream_t stream) {
+  int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
+  dim3 grid(B * Hq * nqblk), block(512);
+  hipLaunchKernelGGL(attn_prefill_pipe_kernel, grid, block, 0, stream,
+                     (unsigned short*)out, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v, B, Hq,
+                     Hkv, S, v_sstride, scale, causal, (float*)stats);
+}
+
 extern "C" void launch_attn_prefill_variant(int variant, void* out,
                                             const void* q, const void* k,
                                             const void* v, int B, int Hq,

@@ -36,6 +36,9 @@ void launch_attn_decode(void*, void*, const void*, const void*, const void*,
                         int, int, int, int, int, const void*, float,
                         hipStream_t);
 void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
+void launch_attn_prefill_pipe(void*, const void*, const void*, const void*,
+                              int, int, int, int, long, float, int, void*,
+                              hipStream_t);
 void launch_attn_prefill_variant(int, void*, const void*, const void*,
                                  const void*, int, int, int, int, float, int,
                                  hipStream_t);
@@ -173,6 +176,27 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   launch_attn_prefill(out.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
                       B, Hq, Hkv, S, (long)v.stride(1), (float)scale,
                       causal ? 1 : 0, nullptr, cur_stream());
+  return out;
+}
+
+torch::Tensor attn_prefill_pipe(torch::Tensor q, torch::Tensor k,
+                                torch::Tensor v, double scale, bool causal) {
+  // pipelined variant (QK(t) in flight over softmax+PV(t-1)) — A/B entry
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128, "attn_prefill_pipe: D must be 128");
+  TORCH_CHECK(Hq % Hkv == 0, "attn_prefill_pipe: Hq % Hkv");
+  TORCH_CHECK(k.size(1) == S && v.size(1) == S, "attn_prefill_pipe: S mismatch");
+  TORCH_CHECK(v.is_cuda() && v.scalar_type() == torch::kBFloat16, "v dtype");
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D, "v head must be contiguous");
+  TORCH_CHECK(v.stride(0) == v.stride(1) * S, "v batch stride");
+  auto out = torch::empty_like(q);
+  launch_attn_prefill_pipe(out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                           v.data_ptr(), B, Hq, Hkv, S, (long)v.stride(1),
+                           (float)scale, causal ? 1 : 0, nullptr,
+                           cur_stream());
   return out;
 }
 
@@ -520,6 +544,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill_stats", &attn_prefill_stats,
         "prefill attention + per-row (m,l) softmax stats");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
+  m.def("attn_prefill_pipe", &attn_prefill_pipe);
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
   m.def("rope_qkv_decode", &rope_qkv_decode,
         "fused decode rope + KV-cache append (graph-replayable)");

@@ -53,6 +53,9 @@ void launch_gemv2(int, void*, const void*, const void*, const void*, int, int,
                   int, float, float, hipStream_t);
 void launch_gemm256b_disc(int, void*, const void*, const void*, int, int,
                           int, hipStream_t);
+void launch_gemm256w(int, void*, const void*, const void*, const void*,
+                     const void*, void*, int, int, int, float, float,
+                     hipStream_t);
 void launch_gemm256b(int, void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, float, float,
                      hipStream_t);
@@ -407,6 +410,29 @@ torch::Tensor gemm256_nt_disc(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+torch::Tensor gemm256_w(torch::Tensor a, torch::Tensor b, int64_t epi,
+                        c10::optional<torch::Tensor> resid,
+                        c10::optional<torch::Tensor> stat, double stat_mul,
+                        double stat_eps) {
+  // 32x32x16-MFMA variant probe (all three epilogues)
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  check_gemm256(a, b, M, N, K);
+  TORCH_CHECK(K % 64 == 0, "gemm256_w: K % 64 required");
+  auto c = epi == 1 ? torch::empty({M, N / 2}, a.options())
+                    : torch::empty({M, N}, a.options());
+  torch::Tensor statout;
+  void* statout_p = nullptr;
+  if (epi == 2) {
+    statout = torch::empty({M}, a.options().dtype(torch::kFloat32));
+    statout_p = statout.data_ptr();
+  }
+  launch_gemm256w((int)epi, c.data_ptr(), a.data_ptr(), b.data_ptr(),
+                  resid ? resid->data_ptr() : nullptr, stat_ptr(stat, M),
+                  statout_p, M, N, K, (float)stat_mul, (float)stat_eps,
+                  cur_stream());
+  return c;
+}
+
 torch::Tensor gemm256_swiglu(torch::Tensor a, torch::Tensor b,
                              c10::optional<torch::Tensor> stat,
                              double stat_mul, double stat_eps) {
@@ -555,6 +581,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "ablation: 1=stage 3=+qk/softmax 7=full");
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
   m.def("gemm256_nt_disc", &gemm256_nt_disc);
+  m.def("gemm256_w", &gemm256_w);
   m.def("gemm256_nt", &gemm256_nt,
         "256-tile bf16 MFMA GEMM, optional fused row-scale epilogue");
   m.def("gemm256_swiglu", &gemm256_swiglu,

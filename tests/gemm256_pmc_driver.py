@@ -22,14 +22,20 @@ SHAPES = {
 def main():
     name = sys.argv[1] if len(sys.argv) > 1 else "gateup"
     iters = int(sys.argv[2]) if len(sys.argv) > 2 else 10
+    variant = sys.argv[3] if len(sys.argv) > 3 else "b"
     m, n, k = SHAPES[name]
     torch.manual_seed(3)
     a = (torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
     b = (torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
-    for _ in range(iters):
-        ops.gemm256_nt(a, b)
+    if variant == "w":
+        hip = ops._try_load()
+        for _ in range(iters):
+            hip.gemm256_w(a, b, 0, None, None, 0, 0)
+    else:
+        for _ in range(iters):
+            ops.gemm256_nt(a, b)
     torch.cuda.synchronize()
-    print(f"done {name} x{iters}")
+    print(f"done {name} x{iters} variant={variant}")
 
 
 if __name__ == "__main__":

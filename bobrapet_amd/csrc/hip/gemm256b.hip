@@ -45,6 +45,9 @@ __device__ __forceinline__ float g3_silu(float g) {
 //       with an explicit lgkmcnt(0) drain before it (8 barriers/tile) —
 //       convoyed waves, kept behind a probe launcher for A/B measurement
 //   2 = barrier at every phase START (4/tile, no post-cluster barriers)
+//   3 = DISC 0 + sched_group_barrier interleave hints after each phase:
+//       alternate ds_read/glds issues between MFMA pairs (the poor-man's
+//       asm K-loop interleave; probe only)
 template <int EPI, int DISC = 0>
 __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
     unsigned short* __restrict__ C,
@@ -203,6 +206,17 @@ __global__ __launch_bounds__(512, 2) void gemm256b_kernel(
           acc[(base) + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16( \
               aq[i][ks], bfr[j][ks], acc[(base) + i][j], 0, 0, 0);      \
     __builtin_amdgcn_s_setprio(0);                                      \
+    if (DISC == 3) {                                                    \
+      /* region schedule: spread this phase's 4-6 ds_reads and 2 glds   \
+         between MFMA pairs instead of clustering them at the head */   \
+      _Pragma("unroll")                                                 \
+      for (int sg = 0; sg < 4; ++sg) {                                  \
+        __builtin_amdgcn_sched_group_barrier(0x100, 1, 0);  /* DS read */ \
+        __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  /* 2 MFMA  */ \
+        __builtin_amdgcn_sched_group_barrier(0x020, 1, 0);  /* VMEM rd */ \
+        __builtin_amdgcn_sched_group_barrier(0x008, 2, 0);  /* 2 MFMA  */ \
+      }                                                                 \
+    }                                                                   \
     if (DISC == 1) __builtin_amdgcn_s_barrier();
   // NOTE: no barrier after the MFMA cluster — the next phase's reads may
   // run while the SIMD partner is still in this cluster (complementary
@@ -396,5 +410,6 @@ extern "C" void launch_gemm256b_disc(int disc, void* C, const void* A,
                        nullptr, nullptr, nullptr, M, N, K, 1.0f, 1e-6f);
   };
   if (disc == 1) go(gemm256b_kernel<0, 1>);
+  else if (disc == 3) go(gemm256b_kernel<0, 3>);
   else go(gemm256b_kernel<0, 2>);
 }

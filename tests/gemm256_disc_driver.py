@@ -39,11 +39,12 @@ def main():
         # numerics: all disciplines must agree exactly (same math order)
         d0 = ops.gemm256_nt(a[:512], b)
         same = all(torch.equal(d0, hip.gemm256_nt_disc(a[:512], b, d))
-                   for d in (1, 2))
+                   for d in (1, 2, 3))
         variants = {
             "disc0": lambda: ops.gemm256_nt(a, b),
             "disc1": lambda: hip.gemm256_nt_disc(a, b, 1),
             "disc2": lambda: hip.gemm256_nt_disc(a, b, 2),
+            "disc3": lambda: hip.gemm256_nt_disc(a, b, 3),
             "blaslt": lambda: torch.matmul(a, b.t()),
         }
         for fn in variants.values():  # warmup

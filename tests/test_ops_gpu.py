@@ -408,3 +408,50 @@ class TestRopeQkvDecode:
         mask = torch.ones(Smax, dtype=torch.bool)
         mask[pos] = False
         assert kc[:, :, mask].abs().max().item() == 0.0
+
+
+class TestGemm256W:
+    """32x32x16-MFMA GEMM variant (probe kernel, kept bound): numerics
+    for all three epilogues vs fp32 torch, incl. M-tail."""
+
+    @pytest.mark.parametrize("m", [300, 512])
+    def test_all_epilogues(self, m):
+        hip = ops._try_load()
+        n, k = 1536, 4096
+        a = (torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        b = (torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        r = (torch.randn(m, n, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        stat = ops.rowsumsq(a)
+        ref0 = torch.matmul(a.float(), b.float().t())
+
+        def rel(got, ref):
+            return ((got.float() - ref).abs().max() / ref.abs().max()).item()
+
+        assert rel(hip.gemm256_w(a, b, 0, None, None, 0, 0), ref0) < 0.02
+        scale = torch.rsqrt(stat.float() / k + 1e-5)[:, None]
+        ref0s = ref0 * scale
+        assert rel(hip.gemm256_w(a, b, 0, None, stat, 1.0 / k, 1e-5), ref0s) < 0.02
+        g, u = ref0s[:, 0::2], ref0s[:, 1::2]
+        ref1 = torch.nn.functional.silu(g) * u
+        assert rel(hip.gemm256_w(a, b, 1, None, stat, 1.0 / k, 1e-5), ref1) < 0.03
+        ref2 = ref0 + r.float()
+        assert rel(hip.gemm256_w(a, b, 2, r, None, 0, 0), ref2) < 0.02
+
+
+class TestAttnPrefillPipe:
+    """Pipelined prefill attention (probe kernel, kept bound): exact
+    parity with the shipped lockstep kernel on causal + edge shapes."""
+
+    @pytest.mark.parametrize("b,s,hq,hkv,causal",
+                             [(2, 333, 8, 2, True), (1, 512, 32, 8, False)])
+    def test_matches_lockstep(self, b, s, hq, hkv, causal):
+        hip = ops._try_load()
+        D = 128
+        q = torch.randn(b, s, hq, D, dtype=torch.bfloat16, device="cuda")
+        k = torch.randn(b, s, hkv, D, dtype=torch.bfloat16, device="cuda")
+        v = torch.randn(b, s, hkv, D, dtype=torch.bfloat16, device="cuda")
+        scale = 1.0 / math.sqrt(D)
+        lock = ops.attn_prefill(q, k, v, scale, causal)
+        pipe = hip.attn_prefill_pipe(q, k, v, scale, causal)
+        err = _mae(pipe, lock)
+        assert err < 0.005, f"pipe vs lockstep max err {err}"

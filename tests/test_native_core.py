@@ -270,6 +270,29 @@ spec:
         assert st["steps"]["watch"]["phase"] == "Succeeded"
         assert elapsed < 2.0, f"wait waited for the poll timer: {elapsed:.3f}s"
 
+    def test_wait_timeout_with_event_wakeup(self, rig):
+        """The event-driven re-check must not defeat wait timeouts: a
+        condition that never becomes true still times out (tag-6 timer)
+        even though other steps complete and tick the run."""
+        eng, nr = rig
+        _apply(
+            eng,
+            """
+kind: Story
+metadata: {name: n8c}
+spec:
+  steps:
+    - {name: busy, type: sleep, with: {duration: 20ms}}
+    - name: watch
+      type: wait
+      with: {until: "{{ false }}", pollInterval: 5s, timeout: 60ms, onTimeout: skip}
+""",
+        )
+        st = nr.run_story("default/n8c", {}, timeout=10)
+        assert st["phase"] == "Succeeded"
+        assert st["steps"]["watch"]["phase"] == "Skipped"
+        assert st["steps"]["busy"]["phase"] == "Succeeded"
+
     def test_gate_approved(self, rig):
         eng, nr = rig
         story = _apply(

@@ -14,6 +14,8 @@
 //    XOR swizzle lives on the per-lane SOURCE address and the matching
 //    read offset (guide rule 21)
 //  - XCD-aware bijective blockIdx swizzle (T1) for L2 locality
+#include <cstdlib>
+
 #include "common.h"
 
 #define GM_BM 128
@@ -278,7 +280,16 @@ static void gemv2_dispatch(void* C, const void* A, const void* W,
                            const void* resid, int M, int N, int K,
                            float stat_mul, float stat_eps, hipStream_t stream) {
   int blocks = (N + 3) / 4;
-  if (blocks > 2048) blocks = 2048;
+  // Persistent-grid cap: dispatching thousands of 256-thread WGs costs
+  // ~5-15 us of ramp on a kernel whose streaming work is itself ~10-40
+  // us; fewer WGs walking more rows each amortize it (the n loop is
+  // already persistent).  Overridable for probes via BOBRA_GEMV_CAP.
+  int cap = 2048;
+  if (const char* e = getenv("BOBRA_GEMV_CAP")) {
+    int v = atoi(e);
+    if (v > 0) cap = v;
+  }
+  if (blocks > cap) blocks = cap;
   dim3 grid(blocks), block(256);
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,

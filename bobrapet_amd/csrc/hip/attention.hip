@@ -49,11 +49,14 @@ __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
 
 template <int VARIANT>  // ablation bitmask: 1=stage 2=qk+softmax 4=pv (7=full)
 __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
-    unsigned short* __restrict__ out,      // [B,S,Hq,D]
-    const unsigned short* __restrict__ q,  // [B,S,Hq,D]
+    unsigned short* __restrict__ out,      // [B,S,Hq,D] (contiguous)
+    const unsigned short* __restrict__ q,  // [B,S,Hq,D] (seq stride q_sstride_in)
     const unsigned short* __restrict__ k,  // [B,S,Hkv,D]
     const unsigned short* __restrict__ v,  // [B,S,Hkv,D] (row stride may differ)
-    int B, int Hq, int Hkv, int S, long v_sstride, float scale, int causal,
+    int B, int Hq, int Hkv, int S, long v_sstride, long q_sstride_in,
+    const float* __restrict__ inv_freq,  // optional [64]: rope Q on load
+    int pos0,                            // rope position offset (seq shards)
+    float scale, int causal,
     float* __restrict__ stats) {  // optional [B,Hq,S,2] (m, l) exp2-domain
   // two K+V^T buffer pairs; pointers computed per use (an addrspace(3)
   // pointer array fails to compile as a static initializer)
@@ -79,10 +82,15 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
   const int hi = lane >> 5;    // half-wave
   const int l31 = lane & 31;
 
-  // BSHD strides: consecutive sequence positions are H*D elements apart
-  const long q_sstride = (long)Hq * D_HEAD;
+  // BSHD strides: consecutive sequence positions are H*D elements apart.
+  // q may be a strided view straight into the fused qkv projection
+  // (q_sstride_in), in which case inv_freq != nullptr applies rope to the
+  // Q rows ON LOAD (pairs (d, d+64) sit in the same lane: qf[s]/qf[s+4])
+  // — the rope kernel then only processes K and Q never round-trips HBM.
+  const long o_sstride = (long)Hq * D_HEAD;
   const long kv_sstride = (long)Hkv * D_HEAD;
-  const long q_base = (long)b * S * q_sstride + (long)hq * D_HEAD;
+  const long q_base = (long)b * S * q_sstride_in + (long)hq * D_HEAD;
+  const long o_base = (long)b * S * o_sstride + (long)hq * D_HEAD;
   const long kv_base = (long)b * S * kv_sstride + (long)hkv * D_HEAD;
   // v may be a strided view (e.g. a slice of the fused qkv projection)
   const long v_base = (long)b * S * v_sstride + (long)hkv * D_HEAD;
@@ -95,7 +103,7 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
   // ---- load Q fragments: B-operand layout, 8 slices of d (16 each) ----
   bf16x8 qf[8];
   {
-    const unsigned short* qrow = q + q_base + (long)my_q * q_sstride;
+    const unsigned short* qrow = q + q_base + (long)my_q * q_sstride_in;
     bool valid = my_q < S;
 #pragma unroll
     for (int s = 0; s < 8; ++s) {
@@ -104,6 +112,25 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
         qf[s] = *reinterpret_cast<const bf16x8*>(qrow + d0);
       } else {
         qf[s] = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+    }
+    if (inv_freq != nullptr && valid) {
+      // rope Q in-register: d = s*16 + hi*8 + j pairs with d+64 at s+4.
+      // ~64 sincos per lane ONCE per workgroup pass — amortized over the
+      // whole KV loop (the old rope kernel cost ~47 us/layer for Q)
+      const float pos = (float)(pos0 + my_q);
+#pragma unroll
+      for (int s = 0; s < 4; ++s) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = s * 16 + hi * 8 + j;
+          float c, sn;
+          __sincosf(pos * inv_freq[d], &sn, &c);
+          const float a = (float)qf[s][j];
+          const float b2 = (float)qf[s + 4][j];
+          qf[s][j] = (__bf16)(a * c - b2 * sn);
+          qf[s + 4][j] = (__bf16)(b2 * c + a * sn);
+        }
       }
     }
   }
@@ -337,7 +364,7 @@ __global__ __launch_bounds__(512, 2) void attn_prefill_kernel(
     int qrow = (r & 3) + 8 * (r >> 2) + 4 * hi;
     int qg = qbase + NWAVES * qrow + wid;
     if (qg >= S) continue;
-    unsigned short* orow = out + q_base + (long)qg * q_sstride;
+    unsigned short* orow = out + o_base + (long)qg * o_sstride;
 #pragma unroll
     for (int dt = 0; dt < 4; ++dt)
       orow[dt * 32 + l31] = f2bf(o_acc[dt][r] * inv_for_row[r]);
@@ -360,7 +387,28 @@ extern "C" void launch_attn_prefill(void* out, const void* q, const void* k,
   hipLaunchKernelGGL((attn_prefill_kernel<7>), grid, block, 0, stream,
                      (unsigned short*)out, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v, B, Hq,
-                     Hkv, S, v_sstride, scale, causal, (float*)stats);
+                     Hkv, S, v_sstride, (long)Hq * D_HEAD, (const float*)nullptr,
+                     0, scale, causal, (float*)stats);
+}
+
+// q is a strided view into the fused qkv projection (seq stride
+// q_sstride, head-contiguous); rope applied to Q rows on load using
+// inv_freq [64] at positions pos0 + row — the rope kernel then only
+// processes K and the Q rows never round-trip HBM
+extern "C" void launch_attn_prefill_qrope(void* out, const void* q,
+                                          const void* k, const void* v, int B,
+                                          int Hq, int Hkv, int S,
+                                          long v_sstride, long q_sstride,
+                                          const void* inv_freq, int pos0,
+                                          float scale, int causal,
+                                          hipStream_t stream) {
+  int nqblk = (S + WG_QROWS - 1) / WG_QROWS;
+  dim3 grid(B * Hq * nqblk), block(512);
+  hipLaunchKernelGGL((attn_prefill_kernel<7>), grid, block, 0, stream,
+                     (unsigned short*)out, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v, B, Hq,
+                     Hkv, S, v_sstride, q_sstride, (const float*)inv_freq,
+                     pos0, scale, causal, (float*)nullptr);
 }
 
 // ---------------------------------------------------------------------------
@@ -675,7 +723,9 @@ extern "C" void launch_attn_prefill_variant(int variant, void* out,
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)out,
                        (const unsigned short*)q, (const unsigned short*)k,
                        (const unsigned short*)v, B, Hq, Hkv, S,
-                       (long)Hkv * D_HEAD, scale, causal, (float*)nullptr);
+                       (long)Hkv * D_HEAD, (long)Hq * D_HEAD,
+                       (const float*)nullptr, 0, scale, causal,
+                       (float*)nullptr);
   };
   if (variant == 1) args(attn_prefill_kernel<1>);
   else if (variant == 3) args(attn_prefill_kernel<3>);

@@ -39,6 +39,9 @@ void launch_dbg_mfma(void*, const void*, const void*, hipStream_t);
 void launch_attn_prefill_pipe(void*, const void*, const void*, const void*,
                               int, int, int, int, long, float, int, void*,
                               hipStream_t);
+void launch_attn_prefill_qrope(void*, const void*, const void*, const void*,
+                               int, int, int, int, long, long, const void*,
+                               int, float, int, hipStream_t);
 void launch_attn_prefill_variant(int, void*, const void*, const void*,
                                  const void*, int, int, int, int, float, int,
                                  hipStream_t);
@@ -180,6 +183,53 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                       B, Hq, Hkv, S, (long)v.stride(1), (float)scale,
                       causal ? 1 : 0, nullptr, cur_stream());
   return out;
+}
+
+torch::Tensor attn_prefill_qrope(torch::Tensor q, torch::Tensor k,
+                                 torch::Tensor v, torch::Tensor inv_freq,
+                                 int64_t pos0, double scale, bool causal) {
+  // q is a STRIDED view into the fused qkv projection; rope is applied to
+  // the Q rows on load (inv_freq [D/2] f32) — no q copy, no q rope kernel
+  check_bf16(k, "k");
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == torch::kBFloat16, "q dtype");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128, "attn_prefill_qrope: D must be 128");
+  TORCH_CHECK(Hq % Hkv == 0, "attn_prefill_qrope: Hq % Hkv");
+  TORCH_CHECK(k.size(1) == S && v.size(1) == S, "attn_prefill_qrope: S mismatch");
+  TORCH_CHECK(q.stride(3) == 1 && q.stride(2) == D, "q head must be contiguous");
+  TORCH_CHECK(q.stride(0) == q.stride(1) * S, "q batch stride");
+  TORCH_CHECK(v.is_cuda() && v.scalar_type() == torch::kBFloat16, "v dtype");
+  TORCH_CHECK(v.stride(3) == 1 && v.stride(2) == D, "v head must be contiguous");
+  TORCH_CHECK(v.stride(0) == v.stride(1) * S, "v batch stride");
+  TORCH_CHECK(inv_freq.is_cuda() && inv_freq.scalar_type() == torch::kFloat32 &&
+                  inv_freq.numel() == D / 2 && inv_freq.is_contiguous(),
+              "inv_freq must be f32 [D/2] contiguous");
+  auto out = torch::empty({B, S, Hq, D}, k.options());
+  launch_attn_prefill_qrope(out.data_ptr(), q.data_ptr(), k.data_ptr(),
+                            v.data_ptr(), B, Hq, Hkv, S, (long)v.stride(1),
+                            (long)q.stride(1), inv_freq.data_ptr(), (int)pos0,
+                            (float)scale, causal ? 1 : 0, cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> rope_k_only(torch::Tensor qkv, int64_t Hq,
+                                       int64_t Hk, int64_t D,
+                                       torch::Tensor cos_t,
+                                       torch::Tensor sin_t) {
+  // rope ONLY the K heads out of the fused qkv rows (Q is roped on load
+  // inside attn_prefill_qrope); returns rotated contiguous k [T,Hk,D]
+  check_bf16(qkv, "qkv");
+  const long T = qkv.size(0);
+  TORCH_CHECK(qkv.is_contiguous(), "rope_k_only: qkv must be contiguous");
+  TORCH_CHECK(cos_t.size(0) == T && cos_t.size(1) == D / 2, "rope_k_only tables");
+  auto kout = torch::empty({T, Hk, D}, qkv.options());
+  const unsigned short* base =
+      (const unsigned short*)qkv.data_ptr() + (long)Hq * D;
+  launch_rope_qkv(nullptr, kout.data_ptr(), base, cos_t.data_ptr(),
+                  sin_t.data_ptr(), (int)T, 0, (int)Hk, (int)D,
+                  (long)qkv.size(1), cur_stream());
+  return {kout};
 }
 
 torch::Tensor attn_prefill_pipe(torch::Tensor q, torch::Tensor k,
@@ -571,6 +621,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "prefill attention + per-row (m,l) softmax stats");
   m.def("attn_prefill", &attn_prefill, "flash attention prefill (MFMA)");
   m.def("attn_prefill_pipe", &attn_prefill_pipe);
+  m.def("attn_prefill_qrope", &attn_prefill_qrope);
+  m.def("rope_k_only", &rope_k_only);
   m.def("attn_decode", &attn_decode, "decode attention w/ KV cache");
   m.def("rope_qkv_decode", &rope_qkv_decode,
         "fused decode rope + KV-cache append (graph-replayable)");

@@ -262,18 +262,39 @@ class LlamaModel:
         if fill_cache:
             self._alloc_cache(B, max(cfg.max_seq_len, S))
 
+        fuse_qrope = ring is None and x2.is_cuda
+        if fuse_qrope:
+            self._ensure_inv_freq()
+        nq = cfg.num_heads * cfg.head_dim
+        nkv = cfg.num_kv_heads * cfg.head_dim
         for li, lw in enumerate(self.layers):
             qkv = ops.gemm256_nt(x2, lw.w_qkv, stat, self._inv_h, cfg.rms_eps)
-            qh, kh, vh = ops.rope_qkv_split(
-                qkv.view(B, S, -1), B, S, cfg.num_heads, cfg.num_kv_heads,
-                cfg.head_dim, cos_f, sin_f,
-            )
+            if fuse_qrope:
+                # rope only K (Q is roped on load inside the attention
+                # kernel); q and v are STRIDED views into qkv — neither
+                # round-trips HBM through a rope/copy kernel
+                kh = ops.rope_k_from_qkv(
+                    qkv, cfg.num_heads, cfg.num_kv_heads, cfg.head_dim,
+                    cos_f, sin_f,
+                ).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+                q3 = qkv.view(B, S, -1)
+                qh = q3[..., :nq].unflatten(-1, (cfg.num_heads, cfg.head_dim))
+                vh = q3[..., nq + nkv:].unflatten(
+                    -1, (cfg.num_kv_heads, cfg.head_dim))
+            else:
+                qh, kh, vh = ops.rope_qkv_split(
+                    qkv.view(B, S, -1), B, S, cfg.num_heads, cfg.num_kv_heads,
+                    cfg.head_dim, cos_f, sin_f,
+                )
             if fill_cache:
                 kc, vc = self._kv_cache[li]
                 kc[:, :, :S] = kh.transpose(1, 2)
                 vc[:, :, :S] = vh.transpose(1, 2)
             if ring is not None:
                 attn = ring(qh, kh, vh, self.scale, causal=True)
+            elif fuse_qrope:
+                attn = ops.attn_prefill_qrope(
+                    qh, kh, vh, self._inv_freq, pos0, self.scale, causal=True)
             else:
                 attn = ops.attn_prefill(qh, kh, vh, self.scale, causal=True)
             a2 = attn.reshape(M, cfg.num_heads * cfg.head_dim)

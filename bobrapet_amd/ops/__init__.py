@@ -249,6 +249,38 @@ def rope_qkv_split(qkv, B, S, Hq, Hkv, D, cos_t, sin_t):
     return q, k, v
 
 
+def rope_k_from_qkv(qkv_flat, Hq, Hkv, D, cos_t, sin_t):
+    """Rope ONLY the K heads out of fused qkv rows [T, rowlen] → k
+    [T, Hkv, D] contiguous.  Pairs with attn_prefill_qrope (which ropes Q
+    on load): Q never round-trips HBM through a rope kernel."""
+    if qkv_flat.is_cuda:
+        (k,) = _require_ext().rope_k_only(qkv_flat.contiguous(), Hq, Hkv, D,
+                                          cos_t, sin_t)
+        return k
+    nq = Hq * D
+    k = qkv_flat[..., nq : nq + Hkv * D].reshape(-1, Hkv, D)
+    return rope_ref(k, cos_t, sin_t)
+
+
+def attn_prefill_qrope(q_view, k, v, inv_freq, pos0, scale, causal=True):
+    """Prefill attention with rope applied to Q ON LOAD from a strided
+    view into the fused qkv projection (csrc/hip/attention.hip
+    launch_attn_prefill_qrope).  k must already be roped
+    (rope_k_from_qkv); inv_freq is the f32 [D/2] frequency vector and
+    pos0 the sequence-shard offset."""
+    if q_view.is_cuda:
+        return _require_ext().attn_prefill_qrope(q_view, k, v, inv_freq,
+                                                 int(pos0), float(scale),
+                                                 bool(causal))
+    B, S, Hq, D = q_view.shape
+    pos = torch.arange(pos0, pos0 + S, dtype=torch.float32)
+    ang = pos[:, None] * inv_freq.float()[None, :]
+    qr = rope_ref(q_view.reshape(B * S, Hq, D), ang.cos().repeat(B, 1),
+                  ang.sin().repeat(B, 1)).view(B, S, Hq, D)
+    return attn_ref(qr.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                    scale, causal).transpose(1, 2)
+
+
 def silu_mul_fused(gate_up: torch.Tensor) -> torch.Tensor:
     """silu(g)*u from the fused [.., 2I] gate_up projection (no slicing
     copies on GPU)."""

@@ -455,3 +455,57 @@ class TestAttnPrefillPipe:
         pipe = hip.attn_prefill_pipe(q, k, v, scale, causal)
         err = _mae(pipe, lock)
         assert err < 0.005, f"pipe vs lockstep max err {err}"
+
+
+class TestAttnPrefillQRope:
+    """Fused Q-rope-on-load attention (strided q/v views into the fused
+    qkv projection) must match the separate rope_qkv_split + attn_prefill
+    path.  In-kernel __sincosf vs the host cos/sin tables differ at f32
+    ulp scale — tolerance covers that, not layout bugs."""
+
+    def test_matches_separate_rope(self):
+        B, S, Hq, Hkv, D = 2, 384, 8, 2, 128
+        theta = 500000.0
+        scale = 1.0 / math.sqrt(D)
+        rowlen = (Hq + 2 * Hkv) * D
+        qkv = (torch.randn(B * S, rowlen, dtype=torch.bfloat16, device="cuda")
+               * 0.5).contiguous()
+        pos = torch.arange(S, device="cuda")
+        cos_t, sin_t = ops.rope_tables(pos, D, theta)
+        cos_f, sin_f = cos_t.repeat(B, 1), sin_t.repeat(B, 1)
+        qh_o, kh_o, vh_o = ops.rope_qkv_split(
+            qkv.view(B, S, -1), B, S, Hq, Hkv, D, cos_f, sin_f)
+        old = ops.attn_prefill(qh_o, kh_o, vh_o, scale, causal=True)
+
+        inv_freq = 1.0 / (theta ** (torch.arange(0, D, 2, dtype=torch.float32,
+                                                 device="cuda") / D))
+        kh = ops.rope_k_from_qkv(qkv, Hq, Hkv, D, cos_f, sin_f).view(B, S, Hkv, D)
+        assert _mae(kh, kh_o) < 1e-6  # same kernel math, k slice only
+        q3 = qkv.view(B, S, -1)
+        qh = q3[..., : Hq * D].unflatten(-1, (Hq, D))
+        vh = q3[..., (Hq + Hkv) * D:].unflatten(-1, (Hkv, D))
+        new = ops.attn_prefill_qrope(qh, kh, vh, inv_freq, 0, scale)
+        err = _mae(new, old)
+        assert err < 0.01, f"fused qrope vs separate path max err {err}"
+
+    def test_pos0_offset(self):
+        # seq-shard offset: fused path at pos0=P equals tables built at P
+        B, S, Hq, Hkv, D, P = 1, 128, 4, 4, 128, 256
+        theta = 500000.0
+        scale = 1.0 / math.sqrt(D)
+        qkv = (torch.randn(B * S, (Hq + 2 * Hkv) * D, dtype=torch.bfloat16,
+                           device="cuda") * 0.5).contiguous()
+        pos = torch.arange(P, P + S, device="cuda")
+        cos_t, sin_t = ops.rope_tables(pos, D, theta)
+        qh_o, kh_o, vh_o = ops.rope_qkv_split(
+            qkv.view(B, S, -1), B, S, Hq, Hkv, D, cos_t, sin_t)
+        old = ops.attn_prefill(qh_o, kh_o, vh_o, scale, causal=True)
+        inv_freq = 1.0 / (theta ** (torch.arange(0, D, 2, dtype=torch.float32,
+                                                 device="cuda") / D))
+        kh = ops.rope_k_from_qkv(qkv, Hq, Hkv, D, cos_t, sin_t).view(B, S, Hkv, D)
+        q3 = qkv.view(B, S, -1)
+        qh = q3[..., : Hq * D].unflatten(-1, (Hq, D))
+        vh = q3[..., (Hq + Hkv) * D:].unflatten(-1, (Hkv, D))
+        new = ops.attn_prefill_qrope(qh, kh, vh, inv_freq, P, scale)
+        err = _mae(new, old)
+        assert err < 0.01, f"pos0 offset max err {err}"

@@ -145,11 +145,19 @@ class _Ingress:
         if sizes is None:
             self._push_one(header, payload)
             return 1
+        # validate the whole batch BEFORE applying any frame — a malformed
+        # sizes vector must not partially apply
+        if (not isinstance(sizes, list)
+                or any(not isinstance(sz, int) or sz < 4 for sz in sizes)
+                or sum(sizes) != len(payload)):
+            raise ValueError("batched frame: sizes do not tile the payload")
         off = 0
-        for sz in sizes:  # batched message: fan the frames back out
-            h, p = unpack_frame(payload[off : off + sz])
-            self._push_one(h, p)
+        frames = []
+        for sz in sizes:
+            frames.append(unpack_frame(payload[off : off + sz]))
             off += sz
+        for h, p in frames:  # batched message: fan the frames back out
+            self._push_one(h, p)
         return len(sizes)
 
     def push_packet(self, data: bytes, ctx) -> bytes:

@@ -42,13 +42,12 @@ void NativeEngine::loop() {
     if (events_.empty()) {
       if (!ntickets_.empty()) {
         // GPU work in flight: nap briefly, then re-poll
-        cv_.wait_for(g, std::chrono::microseconds(20));
+        timed_wait(g, 20e-6);
       } else if (timers_.empty()) {
         cv_.wait(g);
       } else {
         double dt = timers_.top().at - now();
-        if (dt > 0)
-          cv_.wait_for(g, std::chrono::duration<double>(dt));
+        if (dt > 0) timed_wait(g, dt);
       }
       continue;
     }

@@ -271,8 +271,26 @@ class NativeEngine {
       done_cv_.wait(g, pred);
       return true;
     }
+#ifdef BOBRA_TSAN_COMPAT
+    // see timed_wait(): clockwait is invisible to GCC-11 libtsan and
+    // corrupts the mutex model for every later report
+    double deadline = now() + timeout_s;
+    while (!pred()) {
+      double dt = deadline - now();
+      if (dt <= 0) return pred();
+      struct timespec ts;
+      clock_gettime(CLOCK_REALTIME, &ts);
+      long ns = ts.tv_nsec + (long)(dt * 1e9);
+      ts.tv_sec += ns / 1000000000L;
+      ts.tv_nsec = ns % 1000000000L;
+      pthread_cond_timedwait(done_cv_.native_handle(),
+                             g.mutex()->native_handle(), &ts);
+    }
+    return true;
+#else
     return done_cv_.wait_for(
         g, std::chrono::duration<double>(timeout_s), pred);
+#endif
   }
 
   // snapshot for the host language
@@ -406,6 +424,27 @@ class NativeEngine {
 
   void loop();
   void tick(Run& run);
+
+  // condition_variable::wait_for uses pthread_cond_clockwait (steady
+  // clock) on this glibc, which GCC-11's libtsan does NOT intercept —
+  // a TSan build then believes the mutex was never released and every
+  // later access reports as a race.  Under BOBRA_TSAN_COMPAT wait on
+  // the REALTIME clock through the intercepted pthread_cond_timedwait;
+  // the production build keeps the steady-clock wait (REALTIME is
+  // jump-sensitive, acceptable only for the sanitizer run).
+  void timed_wait(std::unique_lock<std::mutex>& g, double seconds) {
+#ifdef BOBRA_TSAN_COMPAT
+    struct timespec ts;
+    clock_gettime(CLOCK_REALTIME, &ts);
+    long ns = ts.tv_nsec + (long)(seconds * 1e9);
+    ts.tv_sec += ns / 1000000000L;
+    ts.tv_nsec = ns % 1000000000L;
+    pthread_cond_timedwait(cv_.native_handle(), g.mutex()->native_handle(),
+                           &ts);
+#else
+    cv_.wait_for(g, std::chrono::duration<double>(seconds));
+#endif
+  }
   void sync_primitives(Run& run, const Plan& plan);
   bool phase_pass(Run& run, const Plan& plan);
   bool steps_settled(Run& run, const Plan& plan, int begin, int end,

@@ -177,6 +177,8 @@ class StepExecutor:
                 sr.status.output = hit
                 sr.status.cache_hit = True
                 sr.status.finished_at = monotonic_now()
+                if hit not in (None, {}, []):
+                    sr.status.last_output_at = sr.status.finished_at
                 state.phase = Phase.SUCCEEDED
                 state.output = hit
                 state.finished_at = sr.status.finished_at
@@ -518,6 +520,8 @@ def finish_step_run(
             return False
         sr.status.phase = phase
         sr.status.output = output if output is not None else sr.status.output
+        if output not in (None, {}, []):
+            sr.status.last_output_at = monotonic_now()
         sr.status.error = error
         sr.status.exit_code = exit_code
         sr.status.exit_class = exit_class

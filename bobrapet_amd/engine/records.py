@@ -255,6 +255,10 @@ class StepRunStatus:
     effects: _t.List[EffectRecord] = field(default_factory=list)
     logs: _t.List[str] = field(default_factory=list)
     cache_hit: bool = False
+    # time of the most recent non-empty output write (cache hits included,
+    # EMPTY outputs excluded — reference: applyCacheHit / LastOutputAt,
+    # steprun_last_output_at_test.go behavior)
+    last_output_at: _t.Optional[float] = None
     started_at: _t.Optional[float] = None
     finished_at: _t.Optional[float] = None
     message: str = ""

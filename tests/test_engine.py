@@ -824,6 +824,11 @@ spec:
         assert r2.phase == Phase.SUCCEEDED
         assert r2.step_states["work"].output == r1.step_states["work"].output
         assert eng.metrics.counter_value("steprun_cache_lookups_total", result="hit") == 1
+        # cache hit sets lastOutputAt (reference: applyCacheHit /
+        # steprun_last_output_at_test.go); non-empty output required
+        sr2 = [s2 for s2 in eng.store.step_runs_of(r2.key) if s2.spec.step_name == "work"][0]
+        assert sr2.status.cache_hit
+        assert sr2.status.last_output_at is not None
 
     def test_effect_ledger_exactly_once(self, eng):
         from bobrapet_amd.engine.effects import EffectLedger

@@ -207,6 +207,33 @@ spec:
         with pytest.raises(ValueError):
             eng.submit_run("default/schema-story", {"count": 1})
 
+    def test_inputs_schema_defaults_do_not_mutate_submitted_dict(self, eng):
+        # resolving inputs (schema defaults injected) must not write back
+        # into the caller's payload (reference behavior:
+        # steprun_input_exposure_test.go — resolved bytes carry defaults,
+        # the persisted spec input stays as submitted)
+        _story(
+            eng,
+            """
+kind: Story
+metadata: {name: schema-story2}
+spec:
+  inputsSchema:
+    type: object
+    properties:
+      secret: {type: string}
+      mode: {type: string, default: safe}
+  steps:
+    - {name: a, ref: {name: echoer}, with: {m: "{{ inputs.mode }}"}}
+  output: {m: "{{ steps.a.output.m }}"}
+""",
+        )
+        submitted = {"secret": "value"}
+        run = eng.run_story("default/schema-story2", submitted, timeout=10)
+        assert run.output == {"m": "safe"}
+        assert submitted == {"secret": "value"}  # caller dict untouched
+        assert run.inputs == {"secret": "value", "mode": "safe"}
+
 
 class TestFailureMachinery:
     def test_fail_fast_skips_and_fails(self, eng):

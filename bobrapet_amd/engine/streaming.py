@@ -145,7 +145,7 @@ class _Stage(threading.Thread):
             self._loop()
         except Exception as exc:  # stage crash fails the pipeline
             self.error = f"{type(exc).__name__}: {exc}"
-            self.sr.on_stage_error(self.step.name, self.error)
+            self.sr.on_stage_error(self.step.name, self.error, fatal=True)
         finally:
             # downstream rings close only when EVERY lane of this step is
             # done (a shared ring closed early would cut sibling lanes off)
@@ -431,7 +431,10 @@ class StreamingRun:
         self.story = story
         self.topo = topology.analyze(story)
         self.canceled = False
+        self.degraded: _t.Optional[str] = None  # fatal stage crash message
         self._lock = threading.Lock()
+        self._finish_lock = threading.Lock()
+        self._finish_result: _t.Optional[StoryRun] = None
         self._leaf_packets = 0
         self._last_outputs: _t.List = []
         self._recorded: _t.List = []     # recording entries (bounded)
@@ -635,7 +638,18 @@ class StreamingRun:
         return n
 
     def finish(self, timeout: float = 60.0) -> StoryRun:
-        """Close the ingress, drain and finalize the run."""
+        """Close the ingress, drain and finalize the run.  Idempotent (a
+        topology termination may have finalized concurrently).  A failed
+        pipeline with compensations/finally hands the run to the batch DAG
+        machinery, which launches those steps and finalizes with the
+        proper phase (Compensated etc.)."""
+        with self._finish_lock:
+            if self._finish_result is not None:
+                return self._finish_result
+            self._finish_result = self._finish_inner(timeout)
+            return self._finish_result
+
+    def _finish_inner(self, timeout: float) -> StoryRun:
         for rings in self.ingress:
             for ring in rings:
                 ring.close()
@@ -670,6 +684,15 @@ class StreamingRun:
                 failed = True
             else:
                 st.phase = Phase.SUCCEEDED
+        if failed and (self.story.compensations or self.story.finally_):
+            # topology terminated with compensations declared: the batch
+            # DAG takes over (main steps are terminal; it launches the
+            # compensation/finally steps and finalizes the phase)
+            self.run.phase = Phase.RUNNING
+            self.run.output = {"packets": self._leaf_packets, "stages": len(agg)}
+            self.engine._streams.pop(self.run.key, None)
+            self.engine._post(("tick", self.run.key))
+            return self.engine.wait(self.run, timeout=timeout)
         self.run.phase = Phase.FAILED if failed else Phase.FINISHED
         self.run.output = {
             "packets": self._leaf_packets,
@@ -751,8 +774,26 @@ class StreamingRun:
                         self._last_outputs.append(_strip_tensors(packet))
         self.engine.metrics.inc("stream_packets_total", stage=stage)
 
-    def on_stage_error(self, stage: str, message: str) -> None:
+    def on_stage_error(self, stage: str, message: str, fatal: bool = False) -> None:
         self.engine.metrics.inc("stream_stage_errors_total", stage=stage)
+        if not fatal:
+            return
+        # a stage crash terminates the topology (reference: the DAG
+        # reconciler's Degraded(TopologyTerminated) handling — main steps
+        # fail and the compensation phase is entered); close the ingress
+        # and finalize from a separate thread (the dying stage's own
+        # thread cannot join itself)
+        with self._lock:
+            if self.degraded is not None:
+                return
+            self.degraded = f"{stage}: {message}"
+        for rings in self.ingress:
+            for ring in rings:
+                ring.close()
+        threading.Thread(
+            target=lambda: self.finish(timeout=10.0), daemon=True,
+            name=f"stream-terminate-{self.run.name}",
+        ).start()
 
     @property
     def leaf_packets(self) -> int:

@@ -833,3 +833,73 @@ spec:
 
         with _p.raises(ClientError):
             c.run("default", "no-such-run")
+
+
+class TestTopologyTermination:
+    """A fatal stage crash terminates the topology and enters the
+    compensation phase (reference behavior:
+    TestDAGReconciler_RealtimeTopologyTerminatedTriggersCompensation /
+    ...TriggersFinally in dag_test.go)."""
+
+    RES = """
+kind: EngramTemplate
+metadata: {name: crash-tpl}
+spec: {builtin: crash}
+---
+kind: Engram
+metadata: {name: crasher}
+spec: {templateRef: {name: crash-tpl}}
+---
+kind: Story
+metadata: {name: doomed}
+spec:
+  pattern: streaming
+  steps:
+    - {name: feed, ref: {name: transformer}}
+    - {name: sink, ref: {name: crasher}, needs: [feed]}
+  compensations:
+    - {name: rollback, ref: {name: echoer}, with: {v: "rolled-back"}}
+"""
+
+    def _register_crash(self):
+        from bobrapet_amd.engrams import registry
+        from bobrapet_amd.engrams.base import Engram
+
+        class Crash(Engram):
+            builtin = "crash"
+
+            def run(self, ctx):
+                raise RuntimeError("stage blew up")
+
+        registry.register("crash", Crash)
+
+    def test_crash_triggers_compensation(self, eng):
+        self._register_crash()
+        eng.apply_yaml(self.RES)
+        stream = eng.submit_stream("default/doomed")
+        stream.push({"items": [{"v": 1}]})
+        # the crash auto-terminates the topology (no finish() needed):
+        # the run must reach a terminal phase with the compensation ran
+        deadline = time.time() + 15
+        run = None
+        while time.time() < deadline:
+            run = eng.store.get_story_run(stream.run.key)
+            if run.is_terminal:
+                break
+            time.sleep(0.02)
+        assert run is not None and run.is_terminal, run.phase
+        assert run.phase == Phase.COMPENSATED, run.phase
+        assert run.step_states["sink"].phase == Phase.FAILED
+        comp = run.step_states.get("rollback")
+        assert comp is not None and comp.phase == Phase.SUCCEEDED
+        assert comp.output == {"v": "rolled-back"}
+
+    def test_finish_idempotent_after_termination(self, eng):
+        self._register_crash()
+        eng.apply_yaml(self.RES)
+        stream = eng.submit_stream("default/doomed")
+        stream.push({"items": [{"v": 1}]})
+        run = stream.finish(timeout=15)  # may race the auto-finalize
+        assert run.is_terminal
+        run2 = stream.finish(timeout=5)
+        assert run2.key == run.key and run2.is_terminal

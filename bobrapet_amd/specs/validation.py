@@ -153,6 +153,34 @@ def validate_story(story: T.Story) -> ValidationResult:
                         f"routing.maxDownstreams={cap}"
                     )
 
+    # context-variable discipline in `with` blocks (reference:
+    # story_webhook_test.go — "rejects packet context in batch with
+    # blocks", "rejects steps context in streaming with blocks",
+    # "rejects now() in streaming with blocks"; step `runtime` templates
+    # are the per-packet surface and stay exempt)
+    _BLOCK_RE = re.compile(r"\{\{(.*?)\}\}", re.S)
+    for s in story.all_steps():
+        if not s.with_:
+            continue
+        blocks = " ".join(_BLOCK_RE.findall(json.dumps(s.with_)))
+        if story.pattern == StoryPattern.STREAMING:
+            if re.search(r"\bsteps\.", blocks):
+                res.error(
+                    f"step {s.name!r}: `steps.` context is not available in "
+                    f"streaming `with` blocks (packets flow, steps do not finish)"
+                )
+            if "now(" in blocks:
+                res.error(
+                    f"step {s.name!r}: now() is non-deterministic per packet "
+                    f"and not allowed in streaming `with` blocks"
+                )
+        else:
+            if re.search(r"\bpacket\.", blocks):
+                res.error(
+                    f"step {s.name!r}: `packet.` context only exists in "
+                    f"streaming stories"
+                )
+
     # declared transports must exist in the story's transport list
     declared = {t.name for t in story.transports}
     for s in story.all_steps():

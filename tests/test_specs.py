@@ -360,3 +360,68 @@ spec:
         )
         res = validation.validate_story(story)
         assert not res.errors, res.errors
+
+
+class TestContextVariableDiscipline:
+    """with-block context rules per pattern (reference:
+    story_webhook_test.go packet/steps/now() context tests)."""
+
+    def _validate(self, yaml_text):
+        from bobrapet_amd.specs import load_yaml
+        from bobrapet_amd.specs.validation import validate_story
+
+        (story,) = load_yaml(yaml_text)
+        return validate_story(story)
+
+    def test_rejects_packet_context_in_batch(self):
+        res = self._validate("""
+kind: Story
+metadata: {name: b1}
+spec:
+  steps:
+    - {name: a, ref: {name: x}, with: {value: "{{ packet.id }}"}}
+""")
+        assert any("packet" in e for e in res.errors), res.errors
+
+    def test_rejects_steps_context_in_streaming_with(self):
+        res = self._validate("""
+kind: Story
+metadata: {name: s1}
+spec:
+  pattern: streaming
+  steps:
+    - {name: a, ref: {name: x}}
+    - {name: b, ref: {name: x}, needs: [a], with: {value: "{{ steps.a.output }}"}}
+""")
+        assert any("steps" in e for e in res.errors), res.errors
+
+    def test_rejects_now_in_streaming_with(self):
+        res = self._validate("""
+kind: Story
+metadata: {name: s2}
+spec:
+  pattern: streaming
+  steps:
+    - {name: a, ref: {name: x}, with: {t: "{{ now() }}"}}
+""")
+        assert any("now()" in e for e in res.errors), res.errors
+
+    def test_allows_steps_context_in_batch_and_runtime_packet(self):
+        res = self._validate("""
+kind: Story
+metadata: {name: ok1}
+spec:
+  steps:
+    - {name: a, ref: {name: x}, with: {v: 1}}
+    - {name: b, ref: {name: x}, needs: [a], with: {v: "{{ steps.a.output.v }}"}}
+""")
+        assert not any("context" in e or "packet" in e for e in res.errors), res.errors
+        res2 = self._validate("""
+kind: Story
+metadata: {name: ok2}
+spec:
+  pattern: streaming
+  steps:
+    - {name: a, ref: {name: x}, runtime: {item: "{{ packet.id }}"}}
+""")
+        assert not res2.errors, res2.errors

@@ -245,11 +245,12 @@ def _validate_step_list(
         if s.with_ is not None and _json_size(s.with_) > MAX_BLOCK_BYTES:
             res.error(f"step {s.name!r}: 'with' block exceeds {MAX_BLOCK_BYTES} bytes")
         if has_type:
-            _validate_primitive_with(res, s)
+            _validate_primitive_with(res, s, story)
     return names
 
 
-def _validate_primitive_with(res: ValidationResult, step: T.Step) -> None:
+def _validate_primitive_with(res: ValidationResult, step: T.Step,
+                             story: _t.Optional[T.Story] = None) -> None:
     """Shape checks for primitive `with` schemas (reference: dag.go:1549-1668,
     step_executor.go:1084-1107)."""
     w = step.with_ if isinstance(step.with_, dict) else {}
@@ -271,8 +272,28 @@ def _validate_primitive_with(res: ValidationResult, step: T.Step) -> None:
         ) not in (None, "success", "failure", "cancel"):
             res.error(f"step {step.name!r}: stop phase/mode invalid")
     elif st == StepType.EXECUTE_STORY:
-        if not (w.get("storyRef") or w.get("story")):
+        target = w.get("storyRef") or w.get("story")
+        if not target:
             res.error(f"step {step.name!r}: executeStory requires with.storyRef")
+        else:
+            # self-reference and cross-namespace rejection at apply time
+            # (reference: story_webhook_test.go "rejects executeStory steps
+            # that reference the same story" / "...another namespace")
+            t_ns = (w.get("namespace") or story.namespace) if story else None
+            t_name = target
+            if isinstance(target, dict):
+                t_ns = target.get("namespace") or t_ns
+                t_name = target.get("name")
+            if story is not None and t_name == story.name and t_ns == story.namespace:
+                res.error(
+                    f"step {step.name!r}: executeStory must not reference its "
+                    f"own story (unbounded recursion)"
+                )
+            if story is not None and t_ns != story.namespace:
+                res.error(
+                    f"step {step.name!r}: executeStory may not reference a "
+                    f"story in another namespace ({t_ns!r})"
+                )
     elif st == StepType.PARALLEL:
         branches = w.get("steps")
         if not isinstance(branches, list) or not branches:

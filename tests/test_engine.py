@@ -690,17 +690,55 @@ spec:
         assert run.phase == Phase.FAILED
 
     def test_recursion_depth_cap(self, eng):
+        # DIRECT self-reference is rejected at apply time (webhook parity);
+        # the runtime depth cap guards INDIRECT cycles (A -> B -> A)
         eng.apply_yaml(
             """
 kind: Story
 metadata: {name: recur}
 spec:
   steps:
-    - {name: again, type: executeStory, with: {storyRef: recur}}
+    - {name: hop, type: executeStory, with: {storyRef: recur2}}
+---
+kind: Story
+metadata: {name: recur2}
+spec:
+  steps:
+    - {name: back, type: executeStory, with: {storyRef: recur}}
 """
         )
         run = eng.run_story("default/recur", {}, timeout=20)
         assert run.phase == Phase.FAILED
+
+    def test_execute_story_self_reference_rejected_at_apply(self, eng):
+        from bobrapet_amd.specs.validation import SpecValidationError
+
+        with pytest.raises(SpecValidationError):
+            eng.apply_yaml(
+                """
+kind: Story
+metadata: {name: selfref}
+spec:
+  steps:
+    - {name: again, type: executeStory, with: {storyRef: selfref}}
+"""
+            )
+
+    def test_execute_story_cross_namespace_rejected_at_apply(self, eng):
+        from bobrapet_amd.specs.validation import SpecValidationError
+
+        with pytest.raises(SpecValidationError):
+            eng.apply_yaml(
+                """
+kind: Story
+metadata: {name: xns}
+spec:
+  steps:
+    - name: other
+      type: executeStory
+      with: {storyRef: {name: elsewhere, namespace: other-ns}}
+"""
+            )
 
 
 class TestControl:

@@ -193,6 +193,20 @@ def validate_story(story: T.Story) -> ValidationResult:
             if ref not in known and ref not in aliases:
                 res.warn(f"story output references unknown step {ref!r}")
 
+    # per-step template refs to unknown steps (reference: "warns on unknown
+    # step references in template expressions"); known refs become implicit
+    # deps at runtime, unknown ones can never resolve
+    for s in story.all_steps():
+        for blob in (s.with_, s.if_, getattr(s, "requires", None)):
+            if blob is None:
+                continue
+            for ref in tdeps.extract_referenced_steps(json.dumps(blob, default=str)):
+                if ref not in known and ref not in aliases:
+                    res.warn(
+                        f"step {s.name!r} references unknown step {ref!r} in a "
+                        f"template expression"
+                    )
+
     # schemas must be JSON-schema-shaped mappings (story_webhook.go:321-330)
     for label, schema in (("inputsSchema", story.inputs_schema), ("outputsSchema", story.outputs_schema)):
         if schema is not None and not isinstance(schema, dict):

@@ -425,3 +425,14 @@ spec:
     - {name: a, ref: {name: x}, runtime: {item: "{{ packet.id }}"}}
 """)
         assert not res2.errors, res2.errors
+
+    def test_warns_on_unknown_step_ref_in_with(self):
+        res = self._validate("""
+kind: Story
+metadata: {name: w1}
+spec:
+  steps:
+    - {name: a, ref: {name: x}, with: {v: "{{ steps.nosuch.output.v }}"}}
+""")
+        assert any("nosuch" in w for w in res.warnings), res.warnings
+        assert not res.errors

@@ -411,6 +411,23 @@ def validate_impulse(
         res.error("impulse.templateRef is required")
     if impulse.story_ref is None or not impulse.story_ref.name:
         res.error("impulse.storyRef is required")
+    # cross-namespace story references rejected (impulse_webhook_test.go)
+    if (impulse.story_ref is not None and impulse.story_ref.namespace
+            and impulse.story_ref.namespace != impulse.namespace):
+        res.error(
+            f"impulse.storyRef may not reference a story in another "
+            f"namespace ({impulse.story_ref.namespace!r})"
+        )
+    # throttle sanity (impulse_webhook_test.go: negative maxInFlight
+    # rejected; zero retry delays are fine elsewhere)
+    th = impulse.throttle
+    if th is not None:
+        if th.max_in_flight is not None and th.max_in_flight < 0:
+            res.error("impulse.throttle.maxInFlight must be >= 0")
+        if th.rate_per_second is not None and th.rate_per_second < 0:
+            res.error("impulse.throttle.ratePerSecond must be >= 0")
+        if th.burst is not None and th.burst < 0:
+            res.error("impulse.throttle.burst must be >= 0")
     if template is not None and template.config_schema is not None:
         from ..utils.jsonschema import validate_instance
 

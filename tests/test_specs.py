@@ -436,3 +436,33 @@ spec:
 """)
         assert any("nosuch" in w for w in res.warnings), res.warnings
         assert not res.errors
+
+
+class TestImpulseValidationGuards:
+    def _impulse(self, yaml_text):
+        from bobrapet_amd.specs import load_yaml
+        from bobrapet_amd.specs.validation import validate_impulse
+
+        (imp,) = load_yaml(yaml_text)
+        return validate_impulse(imp)
+
+    def test_rejects_cross_namespace_story_ref(self):
+        res = self._impulse("""
+kind: Impulse
+metadata: {name: imp1, namespace: default}
+spec:
+  templateRef: {name: t}
+  storyRef: {name: s, namespace: other}
+""")
+        assert any("namespace" in e for e in res.errors), res.errors
+
+    def test_rejects_negative_throttle(self):
+        res = self._impulse("""
+kind: Impulse
+metadata: {name: imp2}
+spec:
+  templateRef: {name: t}
+  storyRef: {name: s}
+  throttle: {maxInFlight: -1}
+""")
+        assert any("maxInFlight" in e for e in res.errors), res.errors

@@ -445,6 +445,20 @@ def validate_transport(transport: T.Transport) -> ValidationResult:
             f"unknown transport driver {transport.driver!r} "
             f"(known: {list(T.KNOWN_TRANSPORT_DRIVERS)})"
         )
+    # plaintext default security mode rejected (reference:
+    # transport_webhook_test.go RejectsPlaintextDefaultSecurityMode — the
+    # default wiring must never downgrade the wire to plaintext; an
+    # explicit per-binding override remains possible)
+    ds = transport.extra.get("defaultSettings")
+    if isinstance(ds, dict):
+        env = ds.get("env")
+        if isinstance(env, dict) and str(
+            env.get("BUBU_TRANSPORT_SECURITY_MODE", "")
+        ).lower() == "plaintext":
+            res.error(
+                "transport defaultSettings must not set "
+                "BUBU_TRANSPORT_SECURITY_MODE=plaintext"
+            )
     s = transport.streaming
     if s is not None:
         lane_names = [l.name for l in s.lanes]

@@ -466,3 +466,18 @@ spec:
   throttle: {maxInFlight: -1}
 """)
         assert any("maxInFlight" in e for e in res.errors), res.errors
+
+    def test_transport_rejects_plaintext_default_security(self):
+        from bobrapet_amd.specs import load_yaml
+        from bobrapet_amd.specs.validation import validate_transport
+
+        (tr,) = load_yaml("""
+kind: Transport
+metadata: {name: t1}
+spec:
+  driver: xgmi
+  defaultSettings:
+    env: {BUBU_TRANSPORT_SECURITY_MODE: plaintext}
+""")
+        res = validate_transport(tr)
+        assert any("plaintext" in e for e in res.errors), res.errors

@@ -163,10 +163,19 @@ class TriggerAdmission:
             eng.metrics.inc("storytriggers_total", decision=str(trigger.decision))
             return trigger
 
-    @staticmethod
-    def _run_matches(run: StoryRun, trigger: StoryTrigger) -> bool:
+    def _run_matches(self, run: StoryRun, trigger: StoryTrigger) -> bool:
         """(reference: storyRunMatchesTrigger 331-406): trigger-token match
-        plus input-hash immutability."""
+        plus input-hash immutability and impulse provenance."""
+        # same identity resubmitted with DIFFERENT inputs is a conflict —
+        # compare against the recorded trigger (raw-input hash; the run's
+        # own hash may include schema defaults)
+        prior = self.triggers.get(trigger.dedupe_token)
+        if prior is not None and prior.input_hash != trigger.input_hash:
+            return False
+        # impulse provenance must agree (reference: "does not reuse a
+        # StoryRun when impulse provenance differs")
+        if trigger.impulse and run.labels.get("impulse") not in (None, trigger.impulse):
+            return False
         if trigger.dedupe_token in run.trigger_tokens:
             return True
         return run.input_hash == trigger.input_hash

@@ -1156,3 +1156,74 @@ spec:
             assert st4["scanCapped"] is True
         finally:
             eng.stop()
+
+
+class TestTriggerConflicts:
+    """Trigger admission edge cases (reference:
+    storytrigger_controller_test.go: conflicting existing run → Rejected;
+    differing impulse provenance → no reuse; concurrent duplicates
+    collapse to one StoryRun)."""
+
+    STORY = """
+kind: Story
+metadata: {name: trig-tgt}
+spec:
+  steps:
+    - {name: a, ref: {name: echoer}, with: {v: "{{ inputs.x }}"}}
+  output: {v: "{{ steps.a.output.v }}"}
+"""
+
+    def test_conflicting_inputs_rejected(self, eng):
+        from bobrapet_amd.engine.triggers import StoryTrigger
+        from bobrapet_amd.enums import TriggerDecision
+
+        _story(eng, self.STORY)
+        t1 = eng.triggers.submit(StoryTrigger(
+            submission_id="s1", key="same-key", story_name="trig-tgt",
+            inputs={"x": 1}))
+        assert t1.decision == TriggerDecision.CREATED
+        eng.wait(t1.story_run_ref, timeout=10)
+        t2 = eng.triggers.submit(StoryTrigger(
+            submission_id="s2", key="same-key", story_name="trig-tgt",
+            inputs={"x": 2}))
+        assert t2.decision == TriggerDecision.REJECTED
+        assert "identity collision" in t2.message
+
+    def test_impulse_provenance_blocks_reuse(self, eng):
+        from bobrapet_amd.engine.triggers import StoryTrigger
+        from bobrapet_amd.enums import TriggerDecision
+
+        _story(eng, self.STORY)
+        t1 = eng.triggers.submit(StoryTrigger(
+            submission_id="p1", key="prov-key", story_name="trig-tgt",
+            inputs={"x": 1}, impulse="default/imp-a"))
+        assert t1.decision == TriggerDecision.CREATED
+        # same identity+inputs but a DIFFERENT impulse: not reused
+        t2 = eng.triggers.submit(StoryTrigger(
+            submission_id="p1", key="prov-key", story_name="trig-tgt",
+            inputs={"x": 1}, impulse="default/imp-b"))
+        assert t2.decision == TriggerDecision.REJECTED
+
+    def test_concurrent_duplicates_collapse(self, eng):
+        import concurrent.futures as cf
+
+        from bobrapet_amd.engine.triggers import StoryTrigger
+        from bobrapet_amd.enums import TriggerDecision
+
+        _story(eng, self.STORY)
+        before = len(eng.store.all_runs())
+
+        def fire(i):
+            return eng.triggers.submit(StoryTrigger(
+                submission_id="dup", key="dup-key", story_name="trig-tgt",
+                inputs={"x": 9}))
+
+        with cf.ThreadPoolExecutor(max_workers=8) as ex:
+            results = list(ex.map(fire, range(16)))
+        created = [r for r in results if r.decision == TriggerDecision.CREATED]
+        reused = [r for r in results if r.decision == TriggerDecision.REUSED]
+        assert len(created) == 1, [str(r.decision) for r in results]
+        assert len(created) + len(reused) == 16
+        assert len(eng.store.all_runs()) == before + 1
+        refs = {r.story_run_ref for r in results}
+        assert len(refs) == 1

@@ -111,6 +111,17 @@ class EffectLedger:
                     n += 1
             return n
 
+    def prune_prefix(self, prefix: str) -> int:
+        """Drop claims under a key prefix (claims are run-scoped:
+        "run/step/key").  Called when the owning StoryRun is deleted by
+        retention cleanup — the reference's orphan-claim GC (EffectClaim
+        owner references cascade when the StepRun disappears)."""
+        with self._lock:
+            doomed = [k for k in self._claims if k.startswith(prefix)]
+            for k in doomed:
+                del self._claims[k]
+            return len(doomed)
+
     def get(self, key: str) -> _t.Optional[EffectClaim]:
         with self._lock:
             return self._claims.get(key)

@@ -1040,4 +1040,9 @@ class RunEngine:
                 self.store.delete_steps_of(run.key)
             self.store.delete_story_run(run.key)
             self._run_done.pop(run.key, None)
+            # orphan effect claims: the run (and its StepRuns) are gone, so
+            # the run-scoped claims can never be referenced again
+            pruned = self.effects.prune_prefix(f"{run.name}/")
+            if pruned:
+                self.metrics.inc("resource_cleanup_total", kind="effectclaims")
             self.metrics.inc("resource_cleanup_total", kind="storyrun")

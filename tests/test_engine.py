@@ -1227,3 +1227,51 @@ spec:
         assert len(eng.store.all_runs()) == before + 1
         refs = {r.story_run_ref for r in results}
         assert len(refs) == 1
+
+
+def test_retention_prunes_orphan_effect_claims():
+    """Reference: effectclaim_controller_test.go — orphan claims whose
+    StepRun is gone are deleted.  Claims are run-scoped here, so retention
+    cleanup of the run prunes them."""
+    from bobrapet_amd.engine import EngineConfig, RunEngine
+
+    from bobrapet_amd.engrams import registry as ereg
+    from bobrapet_amd.engrams.base import Engram, EngramResult
+
+    class Effectful(Engram):
+        builtin = "effectful"
+
+        def run(self, ctx):
+            ctx.record_effect(ctx.config.get("effectKey", "k"), "test effect")
+            return EngramResult(output={"done": True})
+
+    ereg.register("effectful", Effectful)
+    eng = RunEngine(EngineConfig(
+        cpu_workers=1, child_ttl_seconds=0.05, storyrun_retention_seconds=0.1,
+    )).start()
+    try:
+        eng.apply_yaml("""
+kind: EngramTemplate
+metadata: {name: fx-tpl}
+spec: {builtin: effectful}
+---
+kind: Engram
+metadata: {name: fx}
+spec: {templateRef: {name: fx-tpl}}
+---
+kind: Story
+metadata: {name: fx-story}
+spec:
+  steps:
+    - {name: charge, ref: {name: fx}, with: {effectKey: pay-1}}
+""")
+        run = eng.run_story("default/fx-story", {}, timeout=10)
+        assert run.phase == Phase.SUCCEEDED
+        assert len(eng.effects) >= 1
+        deadline = time.time() + 10
+        while time.time() < deadline and eng.store.try_get_story_run(run.key) is not None:
+            time.sleep(0.02)
+        assert eng.store.try_get_story_run(run.key) is None
+        assert len(eng.effects) == 0, "claims must be pruned with the run"
+    finally:
+        eng.stop()

@@ -1242,7 +1242,9 @@ def test_retention_prunes_orphan_effect_claims():
         builtin = "effectful"
 
         def run(self, ctx):
-            ctx.record_effect(ctx.config.get("effectKey", "k"), "test effect")
+            key = ((ctx.config or {}).get("effectKey")
+                   or (ctx.input or {}).get("effectKey", "k"))
+            ctx.record_effect(key, "test effect")
             return EngramResult(output={"done": True})
 
     ereg.register("effectful", Effectful)

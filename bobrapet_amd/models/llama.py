@@ -340,25 +340,33 @@ class LlamaModel:
         cfg = self.cfg
         B = ids.shape[0]
         inv_freq = self._inv_freq
-        use_gemv = B <= 8 and self.device.type == "cuda"
+        on_cuda = self.device.type == "cuda"
+        use_gemv = B <= 8 and on_cuda
+        use_gemm = (not use_gemv) and on_cuda  # large-batch decode: the
+        # SAME fused-epilogue gemm256 chain the prefill layers use (weight
+        # reads stay single-pass; M=B rides the kernel's M-tail clamp) —
+        # no eager/at::native fallback on any CUDA decode path
         cos_t = sin_t = None
         pos_l = None
-        if not use_gemv:
-            # eager fallback path still builds host-side tables
+        if not on_cuda:
+            # CPU reference path builds host-side tables
             pos_l = pos_i32.to(torch.long)
             ang = pos_l.float().reshape(1, 1) * inv_freq[None, :]
             cos_t = torch.cos(ang).expand(B, -1).contiguous()
             sin_t = torch.sin(ang).expand(B, -1).contiguous()
 
         x2 = self.embed[ids].view(B, cfg.hidden_size)  # residual stream
+        stat = ops.rowsumsq(x2) if use_gemm else None
         L_dev = (pos_i32 + 1).contiguous()
         for li, lw in enumerate(self.layers):
             kc, vc = self._kv_cache[li]
-            if use_gemv or not x2.is_cuda:
-                qkv = ops.gemv_norm(x2, lw.w_qkv, self._inv_h, cfg.rms_eps)
-            else:
-                qkv = self._qkv_torch(x2, lw)
             if use_gemv:
+                qkv = ops.gemv_norm(x2, lw.w_qkv, self._inv_h, cfg.rms_eps)
+            elif use_gemm:
+                qkv = ops.gemm256_nt(x2, lw.w_qkv, stat, self._inv_h, cfg.rms_eps)
+            else:
+                qkv = ops.gemv_norm(x2, lw.w_qkv, self._inv_h, cfg.rms_eps)
+            if on_cuda:
                 # fused rope + cache append: one kernel instead of rope +
                 # two index_copys + two layout copies (x32 layers/step)
                 qf = ops.rope_qkv_decode(
@@ -375,33 +383,18 @@ class LlamaModel:
                 vc.index_copy_(2, pos_l, v.reshape(B, cfg.num_kv_heads, 1, cfg.head_dim).contiguous())
             attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
             a2 = attn.reshape(B, cfg.num_heads * cfg.head_dim)
-            if use_gemv or not x2.is_cuda:
+            if use_gemm:
+                x2, stat = ops.gemm256_resid(a2, lw.w_o, x2)
+                act = ops.gemm256_swiglu(x2, lw.w_gate_up, stat, self._inv_h, cfg.rms_eps)
+                x2, stat = ops.gemm256_resid(act, lw.w_down, x2)
+            else:
                 x2 = ops.gemv_resid(a2, lw.w_o, x2)
                 act = ops.gemv_swiglu_norm(x2, lw.w_gate_up, self._inv_h, cfg.rms_eps)
                 x2 = ops.gemv_resid(act, lw.w_down, x2)
-            else:
-                x2 = x2 + torch.matmul(a2, lw.w_o.t())
-                act = self._swiglu_torch(x2, lw)
-                x2 = x2 + torch.matmul(act, lw.w_down.t())
-        if use_gemv or not x2.is_cuda:
-            return ops.gemv_norm(x2, self.lm_head, self._inv_h, cfg.rms_eps)
-        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
-        return ((x2.float() * s) @ self.lm_head.float().t()).to(x2.dtype)
+        if use_gemm:
+            return ops.gemm256_nt(x2, self.lm_head, stat, self._inv_h, cfg.rms_eps)
+        return ops.gemv_norm(x2, self.lm_head, self._inv_h, cfg.rms_eps)
 
-    def _qkv_torch(self, x2, lw):
-        """Large-batch decode fallback (eager torch over folded weights)."""
-        cfg = self.cfg
-        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
-        return ((x2.float() * s) @ lw.w_qkv.float().t()).to(x2.dtype)
-
-    def _swiglu_torch(self, x2, lw):
-        cfg = self.cfg
-        s = torch.rsqrt(x2.float().pow(2).mean(-1, keepdim=True) + cfg.rms_eps)
-        gu = (x2.float() * s) @ lw.w_gate_up.float().t()
-        g, u = gu[:, 0::2], gu[:, 1::2]
-        return (g * torch.sigmoid(g) * u).to(x2.dtype)
-
-    @torch.no_grad()
     def decode_step_graphed(self, ids: torch.Tensor) -> torch.Tensor:
         """hipGraph-captured decode (guide: capture launch-bound inner loops
         in hipGraphs): one replay per token; position/length live in device

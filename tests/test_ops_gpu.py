@@ -324,6 +324,25 @@ class TestModel:
         scale = full.float().abs().max().item()
         assert err / max(scale, 1) < 0.1, (err, scale)
 
+    def test_large_batch_decode_matches_gemv_path(self):
+        # B > 8 routes decode through the fused gemm256 chain; it must
+        # agree with the B<=8 weight-streaming GEMV path on the same model
+        from bobrapet_amd.models.llama import LlamaModel
+
+        m = LlamaModel("llama-tiny", device="cuda")
+        B, S = 16, 64
+        ids = torch.randint(0, 1024, (B, S), device="cuda")
+        m.prefill(ids, fill_cache=True)
+        nxt = torch.randint(0, 1024, (B,), device="cuda")
+        big = m.decode_step(nxt)  # B=16 -> gemm256 chain
+        # same cache state, row 0 alone -> GEMV chain
+        m2 = LlamaModel("llama-tiny", device="cuda")
+        m2.prefill(ids[:1], fill_cache=True)
+        one = m2.decode_step(nxt[:1])
+        err = (big[:1].float() - one.float()).abs().max().item()
+        scale = one.float().abs().max().item()
+        assert err / max(scale, 1) < 0.05, (err, scale)
+
 
 @pytest.mark.gpu
 class TestAttnPrefillStats:

@@ -521,7 +521,7 @@ torch::Tensor gemv_norm(torch::Tensor a, torch::Tensor w, double mul,
   check_bf16(a, "a");
   check_bf16(w, "w");
   const int M = a.size(0), K = a.size(1), N = w.size(0);
-  TORCH_CHECK(M <= 32 && w.size(1) == K && K % 8 == 0, "gemv_norm: bad shape");
+  TORCH_CHECK(M <= 8 && w.size(1) == K && K % 8 == 0, "gemv_norm: bad shape");
   auto c = torch::empty({M, N}, a.options());
   launch_gemv2(1, c.data_ptr(), a.data_ptr(), w.data_ptr(), nullptr, M, N, K,
                (float)mul, (float)eps, cur_stream());
@@ -534,7 +534,7 @@ torch::Tensor gemv_resid(torch::Tensor a, torch::Tensor w,
   check_bf16(w, "w");
   check_bf16(resid, "resid");
   const int M = a.size(0), K = a.size(1), N = w.size(0);
-  TORCH_CHECK(M <= 32 && w.size(1) == K && K % 8 == 0, "gemv_resid: bad shape");
+  TORCH_CHECK(M <= 8 && w.size(1) == K && K % 8 == 0, "gemv_resid: bad shape");
   TORCH_CHECK(resid.numel() == (long)M * N, "gemv_resid: resid shape");
   auto c = torch::empty({M, N}, a.options());
   launch_gemv2(2, c.data_ptr(), a.data_ptr(), w.data_ptr(), resid.data_ptr(),
@@ -548,7 +548,7 @@ torch::Tensor gemv_swiglu_norm(torch::Tensor a, torch::Tensor w_i, double mul,
   check_bf16(a, "a");
   check_bf16(w_i, "w");
   const int M = a.size(0), K = a.size(1), N = w_i.size(0) / 2;
-  TORCH_CHECK(M <= 32 && w_i.size(1) == K && K % 8 == 0 && w_i.size(0) % 2 == 0,
+  TORCH_CHECK(M <= 8 && w_i.size(1) == K && K % 8 == 0 && w_i.size(0) % 2 == 0,
               "gemv_swiglu_norm: bad shape");
   auto c = torch::empty({M, N}, a.options());
   launch_gemv2(3, c.data_ptr(), a.data_ptr(), w_i.data_ptr(), nullptr, M, N, K,

@@ -205,15 +205,13 @@ __global__ __launch_bounds__(256) void gemv_bf16_nt_kernel(
 //         — gate_up intermediate and the silu_mul kernel both eliminated.
 // ---------------------------------------------------------------------------
 
-template <int M, int EPI>  // M = compile-time row BUCKET (>= Mr)
+template <int M, int EPI>
 __global__ __launch_bounds__(256) void gemv2_kernel(
-    unsigned short* __restrict__ C,        // [Mr][N]
-    const unsigned short* __restrict__ A,  // [Mr][K]
+    unsigned short* __restrict__ C,        // [M][N]
+    const unsigned short* __restrict__ A,  // [M][K]
     const unsigned short* __restrict__ W,  // [N][K] (EPI 3: [2N][K])
-    const unsigned short* __restrict__ resid,  // [Mr][N] (EPI 2)
-    int N, int K, float stat_mul, float stat_eps, int Mr) {
-  // rows m >= Mr read row Mr-1 (clamped) and never write — the bucket
-  // form serves any batch up to M with one weight stream (B<=32 decode)
+    const unsigned short* __restrict__ resid,  // [M][N] (EPI 2)
+    int N, int K, float stat_mul, float stat_eps) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
   const int waves_total = (gridDim.x * blockDim.x) >> 6;
@@ -240,8 +238,7 @@ __global__ __launch_bounds__(256) void gemv2_kernel(
       if (EPI == 3) uv = *reinterpret_cast<const ushort8v*>(wu + k0);
 #pragma unroll
       for (int m = 0; m < M; ++m) {
-        const int mr = m < Mr ? m : Mr - 1;
-        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)mr * K + k0);
+        ushort8v av = *reinterpret_cast<const ushort8v*>(A + (long)m * K + k0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const float a = bf2f(av[j]);
@@ -261,7 +258,7 @@ __global__ __launch_bounds__(256) void gemv2_kernel(
     for (int m = 0; m < M; ++m) {
       float g = wave_reduce_sum(accg[m]);
       float u = (EPI == 3) ? wave_reduce_sum(accu[m]) : 0.f;
-      if (lane == 0 && m < Mr) {
+      if (lane == 0) {
         float out;
         if (EPI == 3) {
           g *= scale[m];
@@ -297,10 +294,9 @@ static void gemv2_dispatch(void* C, const void* A, const void* W,
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, block, 0, stream, (unsigned short*)C,
                        (const unsigned short*)A, (const unsigned short*)W,
-                       (const unsigned short*)resid, N, K, stat_mul, stat_eps,
-                       M);
+                       (const unsigned short*)resid, N, K, stat_mul, stat_eps);
   };
-  switch (M) {  // exact up to 8, then buckets (clamped rows, guarded writes)
+  switch (M) {
     case 1: launch(gemv2_kernel<1, EPI>); break;
     case 2: launch(gemv2_kernel<2, EPI>); break;
     case 3: launch(gemv2_kernel<3, EPI>); break;
@@ -308,12 +304,7 @@ static void gemv2_dispatch(void* C, const void* A, const void* W,
     case 5: launch(gemv2_kernel<5, EPI>); break;
     case 6: launch(gemv2_kernel<6, EPI>); break;
     case 7: launch(gemv2_kernel<7, EPI>); break;
-    case 8: launch(gemv2_kernel<8, EPI>); break;
-    default:
-      if (M <= 16) launch(gemv2_kernel<16, EPI>);
-      else if (M <= 24) launch(gemv2_kernel<24, EPI>);
-      else launch(gemv2_kernel<32, EPI>);
-      break;
+    default: launch(gemv2_kernel<8, EPI>); break;
   }
 }
 

@@ -341,12 +341,11 @@ class LlamaModel:
         B = ids.shape[0]
         inv_freq = self._inv_freq
         on_cuda = self.device.type == "cuda"
-        # the weight-streaming GEMV serves up to B=32 (bucketed-M kernels:
-        # one weight stream regardless of batch); beyond that the fused
-        # gemm256 chain (the prefill layer code) amortizes its 256-row
-        # tiles — no eager/at::native fallback on any CUDA decode path
-        use_gemv = B <= 32 and on_cuda
-        use_gemm = (not use_gemv) and on_cuda
+        use_gemv = B <= 8 and on_cuda
+        use_gemm = (not use_gemv) and on_cuda  # large-batch decode: the
+        # SAME fused-epilogue gemm256 chain the prefill layers use (weight
+        # reads stay single-pass; M=B rides the kernel's M-tail clamp) —
+        # no eager/at::native fallback on any CUDA decode path
         cos_t = sin_t = None
         pos_l = None
         if not on_cuda:

@@ -165,10 +165,33 @@ class ResourceRegistry:
         if story is None:
             raise NotFound(f"story {story_key}")
         res = V.validate_story(story)
+        # soft admission: warn about engram refs that do not resolve yet —
+        # across MAIN, COMPENSATION, FINALLY and parallel branches
+        # (reference: TestValidateEngramReferencesIncludesCompensation /
+        # ...IncludesFinallyExecuteStory — the scan must not stop at main)
+        warnings = list(res.warnings)
+        def _scan(steps):
+            for st in steps:
+                if st.ref is not None and st.ref.name:
+                    ns = st.ref.namespace or story.namespace
+                    with self._lock:
+                        known = f"{ns}/{st.ref.name}" in self.engrams
+                    if not known:
+                        warnings.append(
+                            f"step {st.name!r} references unknown engram "
+                            f"{ns}/{st.ref.name}"
+                        )
+                w = st.with_ if isinstance(st.with_, dict) else {}
+                branches = w.get("steps")
+                if isinstance(branches, list):
+                    from ..specs import types as _T
+                    _scan([_T._step_from_dict(dict(b)) for b in branches
+                           if isinstance(b, dict)])
+        _scan(story.all_steps())
         return {
             "validationStatus": "valid" if res.ok else "invalid",
             "validationErrors": res.errors,
-            "validationWarnings": res.warnings,
+            "validationWarnings": warnings,
             "usageCount": self.story_usage_count(story_key),
             "stepsTotal": len(story.steps),
             "generation": story.generation,

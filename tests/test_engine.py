@@ -1277,3 +1277,44 @@ spec:
         assert len(eng.effects) == 0, "claims must be pruned with the run"
     finally:
         eng.stop()
+
+
+def test_story_status_warns_unknown_engrams_in_all_phases():
+    """Engram-reference scan covers main, compensation, finally AND
+    parallel branches (reference: ValidateEngramReferencesIncludes
+    Compensation/Finally)."""
+    from bobrapet_amd.engine import EngineConfig, RunEngine
+
+    eng = RunEngine(EngineConfig(cpu_workers=1)).start()
+    try:
+        eng.apply_yaml("""
+kind: EngramTemplate
+metadata: {name: e-tpl}
+spec: {builtin: echo}
+---
+kind: Engram
+metadata: {name: known}
+spec: {templateRef: {name: e-tpl}}
+---
+kind: Story
+metadata: {name: scan-story}
+spec:
+  steps:
+    - {name: a, ref: {name: known}, with: {v: 1}}
+    - name: fan
+      type: parallel
+      with:
+        steps:
+          - {name: b1, ref: {name: ghost-branch}, with: {v: 2}}
+  compensations:
+    - {name: undo, ref: {name: ghost-comp}, with: {v: 3}}
+  finally:
+    - {name: always, ref: {name: ghost-finally}, with: {v: 4}}
+""")
+        st = eng.registry.story_status("default/scan-story")
+        joined = " ".join(st["validationWarnings"])
+        for ghost in ("ghost-branch", "ghost-comp", "ghost-finally"):
+            assert ghost in joined, (ghost, st["validationWarnings"])
+        assert "engram default/known" not in joined
+    finally:
+        eng.stop()

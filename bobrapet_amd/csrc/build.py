@@ -22,7 +22,7 @@ BUILD_DIR = os.path.join(PKG_DIR, "csrc", "build")
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 
-KERNEL_SOURCES = ["elementwise.hip", "attention.hip", "gemm.hip", "gemm256.hip", "gemm256b.hip", "gemm256w.hip", "debug.hip"]
+KERNEL_SOURCES = ["elementwise.hip", "attention.hip", "gemm.hip", "gemm256.hip", "gemm256b.hip", "gemm256w.hip", "gemmsk.hip", "debug.hip"]
 BINDING_SOURCE = "bindings.cpp"
 
 

@@ -59,6 +59,10 @@ void launch_gemm256b_disc(int, void*, const void*, const void*, int, int,
 void launch_gemm256w(int, void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, float, float,
                      hipStream_t);
+void launch_gemmsk(int, void*, void*, const void*, const void*, const void*,
+                   const void*, void*, int, int, int, float, float,
+                   hipStream_t);
+long gemmsk_ws_elems(int, int);
 void launch_gemm256b(int, void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, float, float,
                      hipStream_t);
@@ -460,6 +464,36 @@ torch::Tensor gemm256_nt_disc(torch::Tensor a, torch::Tensor b,
   return c;
 }
 
+// skinny-M GEMM (decode batch 9..32): epi 0 rowscale, 1 swiglu, 2 resid
+std::vector<torch::Tensor> gemmsk(torch::Tensor a, torch::Tensor b,
+                                  int64_t epi,
+                                  c10::optional<torch::Tensor> resid,
+                                  c10::optional<torch::Tensor> stat,
+                                  double stat_mul, double stat_eps) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  const int M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(M <= 32, "gemmsk: M must be <= 32");
+  TORCH_CHECK(N % 64 == 0 && K % 64 == 0, "gemmsk: N%64, K%64 required");
+  TORCH_CHECK(a.is_contiguous() && b.is_contiguous(), "gemmsk: contiguous");
+  auto c = epi == 1 ? torch::empty({M, N / 2}, a.options())
+                    : torch::empty({M, N}, a.options());
+  torch::Tensor ws = torch::empty(
+      {gemmsk_ws_elems(N, K)}, a.options().dtype(torch::kFloat32));
+  torch::Tensor statout;
+  void* statout_p = nullptr;
+  if (epi == 2) {
+    statout = torch::empty({M}, a.options().dtype(torch::kFloat32));
+    statout_p = statout.data_ptr();
+  }
+  launch_gemmsk((int)epi, c.data_ptr(), ws.data_ptr(), a.data_ptr(),
+                b.data_ptr(), resid ? resid->data_ptr() : nullptr,
+                stat_ptr(stat, M), statout_p, M, N, K, (float)stat_mul,
+                (float)stat_eps, cur_stream());
+  if (epi == 2) return {c, statout};
+  return {c};
+}
+
 torch::Tensor gemm256_w(torch::Tensor a, torch::Tensor b, int64_t epi,
                         c10::optional<torch::Tensor> resid,
                         c10::optional<torch::Tensor> stat, double stat_mul,
@@ -634,6 +668,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nt", &gemm_nt, "bf16 MFMA GEMM: [M,K] @ [N,K]^T");
   m.def("gemm256_nt_disc", &gemm256_nt_disc);
   m.def("gemm256_w", &gemm256_w);
+  m.def("gemmsk", &gemmsk);
   m.def("gemm256_nt", &gemm256_nt,
         "256-tile bf16 MFMA GEMM, optional fused row-scale epilogue");
   m.def("gemm256_swiglu", &gemm256_swiglu,

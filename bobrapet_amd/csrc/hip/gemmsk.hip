@@ -97,7 +97,9 @@ __global__ __launch_bounds__(256, 2) void gemmsk_kernel(
   const unsigned short* brow = B + (long)(n0 + wid * 16 + l15) * K + k0;
 
   auto mfma_block = [&](int buf, int nks) {
-    // nks ks-blocks of 32 k each
+    // nks ks-blocks of 32 k each; unroll so 4 B loads stay in flight
+    // (runtime trip count otherwise serializes the 600-cycle loads)
+#pragma unroll 4
     for (int b = 0; b < nks; ++b) {
       const unsigned short* ab = a_lds[buf] + b * (32 * 32);
       // A frags: rows l15 (+16), k = lg*8..+7 within the ks-block
@@ -249,7 +251,11 @@ extern "C" void launch_gemmsk(int epi, void* C, void* ws, const void* A,
   if (epi != 1) {  // swiglu callers (gate/up) have huge N; keep KS=1 there
     while (KS * ntiles < 256 && (K / (KS * 2)) % 32 == 0 && K / (KS * 2) >= 64)
       KS *= 2;
-    while (K / KS > SK_KC_MAX && (K / (KS * 2)) % 32 == 0) KS *= 2;
+    // bound per-WG serial K only while the grid still has headroom
+    // (e.g. lm_head N=128256 already fills the chip at KS=1)
+    while (KS * ntiles < 1024 && K / KS > SK_KC_MAX &&
+           (K / (KS * 2)) % 32 == 0)
+      KS *= 2;
   }
   if (epi == 2 && KS == 1) KS = 2;  // resid epilogue lives in the reduce
   dim3 grid(ntiles, KS), block(256);
@@ -290,6 +296,8 @@ extern "C" long gemmsk_ws_elems(int N, int K) {
   int KS = 1;
   while (KS * ntiles < 256 && (K / (KS * 2)) % 32 == 0 && K / (KS * 2) >= 64)
     KS *= 2;
-  while (K / KS > SK_KC_MAX && (K / (KS * 2)) % 32 == 0) KS *= 2;
+  while (KS * ntiles < 1024 && K / KS > SK_KC_MAX && (K / (KS * 2)) % 32 == 0)
+    KS *= 2;
+  if (KS == 1) KS = 2;  // epi==2 may force a split; allocate for it
   return (long)KS * SK_M * N;
 }

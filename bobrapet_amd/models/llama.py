@@ -341,11 +341,13 @@ class LlamaModel:
         B = ids.shape[0]
         inv_freq = self._inv_freq
         on_cuda = self.device.type == "cuda"
+        # decode compute tiers (all hand-written kernels, no at::native):
+        #   B <= 8   weight-streaming fused GEMVs
+        #   B <= 32  skinny-M MFMA GEMM (gemmsk: N-tile x K-split grid)
+        #   B >  32  the prefill gemm256 chain (256-row tiles amortize)
         use_gemv = B <= 8 and on_cuda
-        use_gemm = (not use_gemv) and on_cuda  # large-batch decode: the
-        # SAME fused-epilogue gemm256 chain the prefill layers use (weight
-        # reads stay single-pass; M=B rides the kernel's M-tail clamp) —
-        # no eager/at::native fallback on any CUDA decode path
+        use_sk = (not use_gemv) and B <= 32 and on_cuda
+        use_gemm = (not use_gemv) and (not use_sk) and on_cuda
         cos_t = sin_t = None
         pos_l = None
         if not on_cuda:
@@ -356,12 +358,14 @@ class LlamaModel:
             sin_t = torch.sin(ang).expand(B, -1).contiguous()
 
         x2 = self.embed[ids].view(B, cfg.hidden_size)  # residual stream
-        stat = ops.rowsumsq(x2) if use_gemm else None
+        stat = ops.rowsumsq(x2) if (use_gemm or use_sk) else None
         L_dev = (pos_i32 + 1).contiguous()
         for li, lw in enumerate(self.layers):
             kc, vc = self._kv_cache[li]
             if use_gemv:
                 qkv = ops.gemv_norm(x2, lw.w_qkv, self._inv_h, cfg.rms_eps)
+            elif use_sk:
+                qkv = ops.gemmsk_nt(x2, lw.w_qkv, stat, self._inv_h, cfg.rms_eps)
             elif use_gemm:
                 qkv = ops.gemm256_nt(x2, lw.w_qkv, stat, self._inv_h, cfg.rms_eps)
             else:
@@ -383,7 +387,11 @@ class LlamaModel:
                 vc.index_copy_(2, pos_l, v.reshape(B, cfg.num_kv_heads, 1, cfg.head_dim).contiguous())
             attn = ops.attn_decode_t(qf, kc, vc, L_dev, self.scale)
             a2 = attn.reshape(B, cfg.num_heads * cfg.head_dim)
-            if use_gemm:
+            if use_sk:
+                x2, stat = ops.gemmsk_resid(a2, lw.w_o, x2)
+                act = ops.gemmsk_swiglu(x2, lw.w_gate_up, stat, self._inv_h, cfg.rms_eps)
+                x2, stat = ops.gemmsk_resid(act, lw.w_down, x2)
+            elif use_gemm:
                 x2, stat = ops.gemm256_resid(a2, lw.w_o, x2)
                 act = ops.gemm256_swiglu(x2, lw.w_gate_up, stat, self._inv_h, cfg.rms_eps)
                 x2, stat = ops.gemm256_resid(act, lw.w_down, x2)
@@ -391,6 +399,8 @@ class LlamaModel:
                 x2 = ops.gemv_resid(a2, lw.w_o, x2)
                 act = ops.gemv_swiglu_norm(x2, lw.w_gate_up, self._inv_h, cfg.rms_eps)
                 x2 = ops.gemv_resid(act, lw.w_down, x2)
+        if use_sk:
+            return ops.gemmsk_nt(x2, self.lm_head, stat, self._inv_h, cfg.rms_eps)
         if use_gemm:
             return ops.gemm256_nt(x2, self.lm_head, stat, self._inv_h, cfg.rms_eps)
         return ops.gemv_norm(x2, self.lm_head, self._inv_h, cfg.rms_eps)

@@ -249,6 +249,36 @@ def rope_qkv_split(qkv, B, S, Hq, Hkv, D, cos_t, sin_t):
     return q, k, v
 
 
+def gemmsk_nt(a, b, stat, mul: float, eps: float):
+    """Skinny-M (<=32) GEMM with the rowscale epilogue — the decode
+    batch-9..32 projection path (csrc/hip/gemmsk.hip)."""
+    if a.is_cuda:
+        (c,) = _require_ext().gemmsk(a, b, 0, None, stat, mul, eps)
+        return c
+    s = torch.rsqrt(stat.float() * mul + eps)
+    return ((a.float() @ b.float().t()) * s[:, None]).to(a.dtype)
+
+
+def gemmsk_swiglu(a, b_interleaved, stat, mul: float, eps: float):
+    if a.is_cuda:
+        (c,) = _require_ext().gemmsk(a, b_interleaved, 1, None, stat, mul, eps)
+        return c
+    s = torch.rsqrt(stat.float() * mul + eps)
+    c = (a.float() @ b_interleaved.float().t()) * s[:, None]
+    g, u = c[:, 0::2], c[:, 1::2]
+    return (torch.nn.functional.silu(g) * u).to(a.dtype)
+
+
+def gemmsk_resid(a, b, resid):
+    """out = a @ b.T + resid; also returns the row sumsq of out (the
+    next projection's norm statistic)."""
+    if a.is_cuda:
+        c, stat = _require_ext().gemmsk(a, b, 2, resid, None, 0.0, 0.0)
+        return c, stat
+    c = (a.float() @ b.float().t() + resid.float())
+    return c.to(a.dtype), c.pow(2).sum(-1)
+
+
 def rope_k_from_qkv(qkv_flat, Hq, Hkv, D, cos_t, sin_t):
     """Rope ONLY the K heads out of fused qkv rows [T, rowlen] → k
     [T, Hkv, D] contiguous.  Pairs with attn_prefill_qrope (which ropes Q

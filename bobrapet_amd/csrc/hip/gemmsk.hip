@@ -203,21 +203,27 @@ __global__ __launch_bounds__(256, 2) void gemmsk_kernel(
 
 // reduce ws[KS][M][N] -> bf16 with epilogue.
 // EPI 0: rowscale (stat_in); EPI 2: + resid, row-sumsq -> stat_out.
+// Grid (Mr rows x N-chunks of 2048) so the reduce uses the whole chip;
+// the EPI2 row statistic accumulates via one atomicAdd per (row, chunk)
+// block — the same discipline as gemm256's resid epilogue (stat_out is
+// zeroed by the launcher).
+#define SK_RC 2048
 template <int EPI>
 __global__ __launch_bounds__(256) void gemmsk_reduce_kernel(
     unsigned short* __restrict__ C, const float* __restrict__ ws,
     const unsigned short* __restrict__ resid,
     const float* __restrict__ stat_in, float* __restrict__ stat_out,
     int Mr, int N, int KS, float stat_mul, float stat_eps) {
-  // block = one row, 256 threads stride N
   const int row = blockIdx.x;
+  const int c0 = blockIdx.y * SK_RC;
+  const int cend = min(c0 + SK_RC, N);
   if (row >= Mr) return;
   __shared__ float ssq_lds[256];
   const float sc = (EPI == 0 && stat_in != nullptr)
                        ? rsqrtf(stat_in[row] * stat_mul + stat_eps)
                        : 1.0f;
   float ss = 0.0f;
-  for (int n = threadIdx.x; n < N; n += 256) {
+  for (int n = c0 + threadIdx.x; n < cend; n += 256) {
     float v = 0.0f;
     for (int k = 0; k < KS; ++k) v += ws[((long)k * SK_M + row) * N + n];
     if (EPI == 2) {
@@ -235,7 +241,7 @@ __global__ __launch_bounds__(256) void gemmsk_reduce_kernel(
       if (threadIdx.x < off) ssq_lds[threadIdx.x] += ssq_lds[threadIdx.x + off];
       __syncthreads();
     }
-    if (threadIdx.x == 0) stat_out[row] = ssq_lds[0];
+    if (threadIdx.x == 0) atomicAdd(stat_out + row, ssq_lds[0]);
   }
 }
 
@@ -277,7 +283,9 @@ extern "C" void launch_gemmsk(int epi, void* C, void* ws, const void* A,
                      (unsigned short*)nullptr, (float*)ws,
                      (const unsigned short*)A, (const unsigned short*)B,
                      (const float*)nullptr, M, N, K, KS, stat_mul, stat_eps);
-  dim3 g2(SK_M), b2(256);
+  dim3 g2(SK_M, (N + SK_RC - 1) / SK_RC), b2(256);
+  if (epi == 2 && stat_out != nullptr)
+    (void)hipMemsetAsync(stat_out, 0, (size_t)M * sizeof(float), stream);
   if (epi == 2)
     hipLaunchKernelGGL((gemmsk_reduce_kernel<2>), g2, b2, 0, stream,
                        (unsigned short*)C, (const float*)ws,

@@ -96,9 +96,11 @@ __global__ __launch_bounds__(256, 2) void gemmsk_kernel(
   // at this WG's K-chunk
   const unsigned short* brow = B + (long)(n0 + wid * 16 + l15) * K + k0;
 
+  int bofs = 0;  // k progress of the B stream (elements)
   auto mfma_block = [&](int buf, int nks) {
-    // nks ks-blocks of 32 k each; unroll so 4 B loads stay in flight
-    // (runtime trip count otherwise serializes the 600-cycle loads)
+    // nks ks-blocks of 32 k each; unroll so 4 B loads stay in flight —
+    // indexed addressing (brow + b*32), NOT a serial pointer bump, so
+    // the unrolled iterations' load addresses are independent
 #pragma unroll 4
     for (int b = 0; b < nks; ++b) {
       const unsigned short* ab = a_lds[buf] + b * (32 * 32);
@@ -108,11 +110,12 @@ __global__ __launch_bounds__(256, 2) void gemmsk_kernel(
       const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
           ab + (16 + l15) * 32 + (((lg * 16) ^ sk_swz(16 + l15)) >> 1));
       // B frag straight from HBM (each weight row read once)
-      const bf16x8 bf = *reinterpret_cast<const bf16x8*>(brow + lg * 8);
+      const bf16x8 bf =
+          *reinterpret_cast<const bf16x8*>(brow + bofs + b * 32 + lg * 8);
       acc[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bf, acc[0], 0, 0, 0);
       acc[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bf, acc[1], 0, 0, 0);
-      brow += 32;
     }
+    bofs += nks * 32;
   };
 
   if (nkb > 0) stage_full(0, 0);

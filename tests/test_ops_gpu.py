@@ -528,3 +528,32 @@ class TestAttnPrefillQRope:
         new = ops.attn_prefill_qrope(qh, kh, vh, inv_freq, P, scale)
         err = _mae(new, old)
         assert err < 0.01, f"pos0 offset max err {err}"
+
+
+class TestGemmSk:
+    """Skinny-M GEMM (decode batch 9-32 tier): all epilogues vs fp32
+    torch at several M, including the K-split reduce path."""
+
+    @pytest.mark.parametrize("m", [9, 16, 32])
+    def test_all_epilogues(self, m):
+        hip = ops._try_load()
+        k, n = 4096, 1536
+        a = (torch.randn(m, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        b = (torch.randn(n, k, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        r = (torch.randn(m, n, dtype=torch.bfloat16, device="cuda") * 0.3).contiguous()
+        stat = ops.rowsumsq(a)
+        ref0 = torch.matmul(a.float(), b.float().t())
+        scale = torch.rsqrt(stat.float() / k + 1e-5)[:, None]
+
+        def rel(got, ref):
+            return ((got.float() - ref).abs().max() / ref.abs().max()).item()
+
+        assert rel(ops.gemmsk_nt(a, b, stat, 1.0 / k, 1e-5), ref0 * scale) < 0.02
+        g, u = (ref0 * scale)[:, 0::2], (ref0 * scale)[:, 1::2]
+        ref1 = torch.nn.functional.silu(g) * u
+        assert rel(ops.gemmsk_swiglu(a, b, stat, 1.0 / k, 1e-5), ref1) < 0.03
+        c2, s2 = ops.gemmsk_resid(a, b, r)
+        ref2 = ref0 + r.float()
+        assert rel(c2, ref2) < 0.02
+        assert rel(s2, ref2.pow(2).sum(-1)) < 1e-4
+        del hip

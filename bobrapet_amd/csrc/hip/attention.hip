@@ -17,6 +17,8 @@
 //   A[i][k]: i = lane&31, k = 8*(lane>>5) + j   (j in 0..7)
 //   B[k][j]: j = lane&31, k = 8*(lane>>5) + jj
 //   C[i][j]: j = lane&31, i = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+#include <cstdlib>
+
 #include "common.h"
 
 #define D_HEAD 128
@@ -1034,7 +1036,13 @@ extern "C" void launch_attn_decode(void* out, void* workspace, const void* q,
   // 16-256 CUs (37.6 vs 22.3 us at L=256; 787 vs 45.5 at L=8192; decode
   // e2e 6.53 vs 3.96 ms/tok).  The two-pass "fixed cost" IS the
   // parallelism floor, not kernel overhead — docs/PERF.md late round 2.)
-  if (workspace != nullptr && G <= DEC_GMAX && B * Hq < 512) {
+  // routing override for A/B probes: BOBRA_DEC_ATTN=chunk|single
+  int route = 0;  // 0 auto, 1 force chunk, 2 force single
+  if (const char* e = getenv("BOBRA_DEC_ATTN"))
+    route = e[0] == 'c' ? 1 : e[0] == 's' ? 2 : 0;
+  const bool want_chunk =
+      route == 1 || (route == 0 && B * Hq < 512);
+  if (workspace != nullptr && G <= DEC_GMAX && want_chunk) {
     const int nchunk = (Smax + DEC_CHUNK - 1) / DEC_CHUNK;
     const int total = B * Hkv * nchunk;
     dim3 g1(total < 2048 ? total : 2048), b1(256);

@@ -942,6 +942,157 @@ __global__ __launch_bounds__(64, 8) void attn_decode_combine_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// FUSED small-batch decode attention: ONE kernel per step instead of the
+// two-pass chunk+combine.  One workgroup per (b,hkv); the 4 waves stride
+// 64-row KV sub-blocks with a per-wave ONLINE softmax merge (rescale o by
+// exp(m_old - m_new) when a later sub-block raises the max), then one
+// cross-wave LDS merge writes the normalized output directly — no
+// workspace round-trip, no combine launch.  Routed at B*Hkv <= 16 (the
+// b1/b2 serving case, where the two-pass form spent ~23 us/layer of
+// fixed cost); larger batches keep the chunked form (more WGs = more
+// CUs at work).  L may be a device scalar (hipGraph-replayable); the
+// sub-block loop handles any L without re-capture.
+// ---------------------------------------------------------------------------
+template <bool L_FROM_DEV>
+__global__ __launch_bounds__(256, 4) void attn_decode_fused_kernel(
+    unsigned short* __restrict__ out,       // [B][Hq][D]
+    const unsigned short* __restrict__ q,   // [B][Hq][D]
+    const unsigned short* __restrict__ kc,  // [B][Hkv][Smax][D]
+    const unsigned short* __restrict__ vc,
+    int B, int Hq, int Hkv, int Smax, int L,
+    const int* __restrict__ L_dev, float scale) {
+  if constexpr (L_FROM_DEV) L = *L_dev;
+  const int G = Hq / Hkv;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  const int hkv = blockIdx.x % Hkv;
+  const int b = blockIdx.x / Hkv;
+
+  __shared__ float q_lds[DEC_GMAX][D_HEAD];
+  __shared__ float p_lds[4][DEC_GMAX][WAVE];
+  __shared__ float sm[DEC_GMAX][4], sl[DEC_GMAX][4], so[DEC_GMAX][4][D_HEAD];
+
+  for (int i = tid; i < G * D_HEAD; i += blockDim.x) {
+    int g = i / D_HEAD, d = i % D_HEAD;
+    q_lds[g][d] = bf2f(q[((long)b * Hq + hkv * G + g) * D_HEAD + d]);
+  }
+  __syncthreads();
+  const long base = (((long)b * Hkv + hkv) * Smax) * D_HEAD;
+
+  float m_w[DEC_GMAX], l_w[DEC_GMAX], o0[DEC_GMAX], o1[DEC_GMAX];
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    m_w[g] = -1e30f;
+    l_w[g] = 0.f;
+    o0[g] = o1[g] = 0.f;
+  }
+
+  for (int r0 = wid * WAVE; r0 < L; r0 += 4 * WAVE) {
+    const int r = r0 + lane;
+    const bool valid = r < L;
+    const int r_safe = valid ? r : L - 1;
+    // phase A: lane-per-row scores
+    float sacc[DEC_GMAX];
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) sacc[g] = 0.f;
+    const unsigned short* krow = kc + base + (long)r_safe * D_HEAD;
+#pragma unroll
+    for (int d0 = 0; d0 < D_HEAD / 8; ++d0) {
+      ushort8v k8 = *reinterpret_cast<const ushort8v*>(krow + d0 * 8);
+      float kf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) kf[j] = bf2f(k8[j]);
+#pragma unroll
+      for (int g = 0; g < DEC_GMAX; ++g) {
+        if (g >= G) break;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) sacc[g] += kf[j] * q_lds[g][d0 * 8 + j];
+      }
+    }
+    // online merge of this sub-block into the wave's running (m, l, o)
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) {
+      if (g >= G) break;
+      float sv = valid ? sacc[g] * scale : -1e30f;
+      float m_t = wave_reduce_max(sv);
+      float m_new = fmaxf(m_w[g], m_t);
+      float alpha = (m_w[g] <= -1e30f) ? 0.f : __expf(m_w[g] - m_new);
+      if (m_new <= -1e30f) alpha = 1.f;
+      float pv = (sv <= -1e30f) ? 0.f : __expf(sv - m_new);
+      p_lds[wid][g][lane] = pv;
+      l_w[g] = l_w[g] * alpha + wave_reduce_sum(pv);
+      o0[g] *= alpha;
+      o1[g] *= alpha;
+      m_w[g] = m_new;
+    }
+    // phase B: broadcast-PV accumulation, 8-row batches (see the chunk
+    // kernel note on load-latency serialization)
+    const int rows = min(WAVE, L - r0);
+    const unsigned short* vrow0 = vc + base + (long)r0 * D_HEAD + lane * 2;
+    int rr = 0;
+    for (; rr + 8 <= rows; rr += 8) {
+      ushort2 v8[8];
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        v8[u] = *reinterpret_cast<const ushort2*>(vrow0 + (long)(rr + u) * D_HEAD);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        float vx = bf2f(v8[u].x), vy = bf2f(v8[u].y);
+#pragma unroll
+        for (int g = 0; g < DEC_GMAX; ++g) {
+          if (g >= G) break;
+          float pv = p_lds[wid][g][rr + u];
+          o0[g] += pv * vx;
+          o1[g] += pv * vy;
+        }
+      }
+    }
+    for (; rr < rows; ++rr) {
+      ushort2 v2 = *reinterpret_cast<const ushort2*>(vrow0 + (long)rr * D_HEAD);
+      float vx = bf2f(v2.x), vy = bf2f(v2.y);
+#pragma unroll
+      for (int g = 0; g < DEC_GMAX; ++g) {
+        if (g >= G) break;
+        float pv = p_lds[wid][g][rr];
+        o0[g] += pv * vx;
+        o1[g] += pv * vy;
+      }
+    }
+  }
+
+  // cross-wave merge (waves covered disjoint sub-blocks) + direct output
+#pragma unroll
+  for (int g = 0; g < DEC_GMAX; ++g) {
+    if (g >= G) break;
+    sm[g][wid] = m_w[g];
+    sl[g][wid] = l_w[g];
+    so[g][wid][lane * 2] = o0[g];
+    so[g][wid][lane * 2 + 1] = o1[g];
+  }
+  __syncthreads();
+  if (wid == 0) {
+#pragma unroll
+    for (int g = 0; g < DEC_GMAX; ++g) {
+      if (g >= G) break;
+      float m_g = fmaxf(fmaxf(sm[g][0], sm[g][1]), fmaxf(sm[g][2], sm[g][3]));
+      float l_g = 0.f, a0 = 0.f, a1 = 0.f;
+#pragma unroll
+      for (int w = 0; w < 4; ++w) {
+        float aw = (sm[g][w] <= -1e30f) ? 0.f : __expf(sm[g][w] - m_g);
+        l_g += sl[g][w] * aw;
+        a0 += so[g][w][lane * 2] * aw;
+        a1 += so[g][w][lane * 2 + 1] * aw;
+      }
+      float inv = (l_g > 0.f) ? 1.f / l_g : 0.f;
+      unsigned short* orow = out + ((long)b * Hq + hkv * G + g) * D_HEAD;
+      orow[lane * 2] = f2bf(a0 * inv);
+      orow[lane * 2 + 1] = f2bf(a1 * inv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Decode attention (single new token, GQA, KV cache).
 //   q: [B,Hq,D] bf16; kc/vc: [B,Hkv,Smax,D] bf16; out: [B,Hq,D]
 // One workgroup (4 waves) per (b,hq); waves stride the kv length; memory-
@@ -1040,6 +1191,21 @@ extern "C" void launch_attn_decode(void* out, void* workspace, const void* q,
   int route = 0;  // 0 auto, 1 force chunk, 2 force single
   if (const char* e = getenv("BOBRA_DEC_ATTN"))
     route = e[0] == 'c' ? 1 : e[0] == 's' ? 2 : 0;
+  if (route == 0 && G <= DEC_GMAX && B * Hkv >= 64) {
+    // large-batch: one WG per (b,hkv) — full grid, KV read once
+    dim3 g0(B * Hkv), b0(256);
+    if (L_dev != nullptr)
+      hipLaunchKernelGGL((attn_decode_fused_kernel<true>), g0, b0, 0, stream,
+                         (unsigned short*)out, (const unsigned short*)q,
+                         (const unsigned short*)kc, (const unsigned short*)vc,
+                         B, Hq, Hkv, Smax, 0, (const int*)L_dev, scale);
+    else
+      hipLaunchKernelGGL((attn_decode_fused_kernel<false>), g0, b0, 0, stream,
+                         (unsigned short*)out, (const unsigned short*)q,
+                         (const unsigned short*)kc, (const unsigned short*)vc,
+                         B, Hq, Hkv, Smax, L, (const int*)nullptr, scale);
+    return;
+  }
   const bool want_chunk =
       route == 1 || (route == 0 && B * Hq < 512);
   if (workspace != nullptr && G <= DEC_GMAX && want_chunk) {

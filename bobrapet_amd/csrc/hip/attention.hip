@@ -1187,12 +1187,15 @@ extern "C" void launch_attn_decode(void* out, void* workspace, const void* q,
   // 16-256 CUs (37.6 vs 22.3 us at L=256; 787 vs 45.5 at L=8192; decode
   // e2e 6.53 vs 3.96 ms/tok).  The two-pass "fixed cost" IS the
   // parallelism floor, not kernel overhead — docs/PERF.md late round 2.)
-  // routing override for A/B probes: BOBRA_DEC_ATTN=chunk|single
-  int route = 0;  // 0 auto, 1 force chunk, 2 force single
+  // routing override for A/B probes: BOBRA_DEC_ATTN=chunk|single|fused.
+  // The fused kernel is NEVER auto-routed: even at B*Hkv >= 64 (full
+  // grid, KV read once) its per-WG serial sub-block walk loses to the
+  // massively-parallel chunk/single grids (b8 L1024: 117 vs 61 us;
+  // decode b8 9.6 -> 10.5 ms when it was routed) — kept as a probe.
+  int route = 0;  // 0 auto, 1 force chunk, 2 force single, 3 force fused
   if (const char* e = getenv("BOBRA_DEC_ATTN"))
-    route = e[0] == 'c' ? 1 : e[0] == 's' ? 2 : 0;
-  if (route == 0 && G <= DEC_GMAX && B * Hkv >= 64) {
-    // large-batch: one WG per (b,hkv) — full grid, KV read once
+    route = e[0] == 'c' ? 1 : e[0] == 's' ? 2 : e[0] == 'f' ? 3 : 0;
+  if (route == 3 && G <= DEC_GMAX) {
     dim3 g0(B * Hkv), b0(256);
     if (L_dev != nullptr)
       hipLaunchKernelGGL((attn_decode_fused_kernel<true>), g0, b0, 0, stream,

@@ -476,6 +476,9 @@ std::vector<torch::Tensor> gemmsk(torch::Tensor a, torch::Tensor b,
   TORCH_CHECK(M <= 32, "gemmsk: M must be <= 32");
   TORCH_CHECK(N % 64 == 0 && K % 64 == 0, "gemmsk: N%64, K%64 required");
   TORCH_CHECK(a.is_contiguous() && b.is_contiguous(), "gemmsk: contiguous");
+  TORCH_CHECK(epi != 2 || (resid && resid->numel() == (long)M * N &&
+                           resid->is_contiguous()),
+              "gemmsk: resid must be contiguous [M, N]");
   auto c = epi == 1 ? torch::empty({M, N / 2}, a.options())
                     : torch::empty({M, N}, a.options());
   torch::Tensor ws = torch::empty(

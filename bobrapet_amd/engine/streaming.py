@@ -58,6 +58,20 @@ class TransportBinding:
     phase: str = "Ready"
     generation: int = 0  # bumped on live cutover (reference: connector gen)
     heartbeat: float = field(default_factory=monotonic_now)
+    # per-lane negotiated codec (reference: DeriveNegotiatedCapabilities —
+    # defaults to the FIRST offered codec per modality, preserves an
+    # already-negotiated value across re-derivation)
+    negotiated: _t.Dict[str, str] = field(default_factory=dict)
+
+
+def derive_negotiated(binding: "TransportBinding",
+                      lanes: _t.Sequence[str]) -> None:
+    """Default each lane's negotiated codec to the binding's first
+    offered codec; existing negotiations are PRESERVED (reference:
+    pkg/transport/capabilities.go DeriveNegotiatedCapabilities)."""
+    first = binding.codecs[0] if binding.codecs else "json"
+    for lane in lanes or ("data",):
+        binding.negotiated.setdefault(lane, first)
 
 
 @dataclass
@@ -520,6 +534,9 @@ class StreamingRun:
                 upstream=[e.src for e in self.topo.upstream_of(name) if e.src],
                 downstream=[e.dst for e in self.topo.downstream_of(name)],
             )
+        lane_names = [l.name for l in (self.settings.lanes or [])] or ["data"]
+        for b in self.bindings.values():
+            derive_negotiated(b, lane_names)
         engine.metrics.set_gauge("transport_bindings_total", len(self.bindings))
         engine.metrics.set_gauge("transport_bindings_ready", len(self.bindings))
         run.phase = Phase.RUNNING

@@ -903,3 +903,18 @@ spec:
         assert run.is_terminal
         run2 = stream.finish(timeout=5)
         assert run2.key == run.key and run2.is_terminal
+
+
+def test_negotiated_capabilities_default_and_preserve():
+    """reference: pkg/transport/capabilities_test.go — defaults to the
+    first offered codec per lane; re-derivation preserves existing."""
+    from bobrapet_amd.engine.streaming import TransportBinding, derive_negotiated
+
+    b = TransportBinding(name="b", story_run="r", step="s",
+                         codecs=["tensor", "json"])
+    derive_negotiated(b, ["media", "data"])
+    assert b.negotiated == {"media": "tensor", "data": "tensor"}
+    b.negotiated["media"] = "json"  # operator override / prior negotiation
+    derive_negotiated(b, ["media", "data", "control"])
+    assert b.negotiated["media"] == "json"        # preserved
+    assert b.negotiated["control"] == "tensor"    # new lane defaulted
